@@ -1,0 +1,179 @@
+"""Training driver (SURVEY.md §3.2 — the hot loop — plus §3.3/§3.4).
+
+Epoch loop with: bf16 H2D input path, fused-kernel forward/backward via the
+op layer, bucketed RCCL gradient all-reduce overlapped with backward,
+optimizer + LR schedule + EMA per step, per-interval metric reduction and
+JSONL logging, periodic + best checkpoints, exact resume.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Optional
+
+import torch
+
+from ..models import build_model
+from ..parallel import DataParallelEngine, distributed as udist
+from ..utils import checkpoint as ckpt
+from ..utils import dataflow
+from ..utils.common import AverageMeter, JsonlWriter, accuracy, set_random_seed, setup_logger
+from ..utils.config import AttrDict
+from ..utils.optim import (CrossEntropyLabelSmooth, ExponentialMovingAverage,
+                           LRScheduler, get_optimizer)
+
+
+class Trainer:
+    def __init__(self, cfg: AttrDict) -> None:
+        self.cfg = cfg
+        self.rank = udist.init_dist(cfg.dist.backend, cfg.dist.timeout_sec)
+        self.world = udist.get_world_size()
+        self.device = (torch.device("cuda", udist.get_local_rank())
+                       if torch.cuda.is_available() else torch.device("cpu"))
+        if self.device.type == "cuda":
+            torch.cuda.set_device(self.device)
+        set_random_seed(cfg.train.seed + self.rank, cfg.train.deterministic)
+        self.logger = setup_logger(cfg.log.dir, self.rank)
+        self.jsonl = JsonlWriter(os.path.join(cfg.log.dir, "metrics.jsonl"),
+                                 cfg.log.jsonl and self.rank == 0)
+        self.dtype = (torch.bfloat16
+                      if (cfg.train.dtype == "bf16" and self.device.type == "cuda")
+                      else torch.float32)
+
+        model = build_model(cfg.model.name, **{
+            k: v for k, v in cfg.model.items() if k != "name"})
+        model = model.to(self.device)
+        if cfg.train.channels_last and self.device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+        self.engine = DataParallelEngine(
+            model, bucket_mb=cfg.dist.bucket_mb, overlap=cfg.dist.overlap)
+        self.model = model
+
+        self.train_loader, self.val_loader = dataflow.get_data_loaders(
+            cfg.data, seed=cfg.train.seed)
+        self.steps_per_epoch = len(self.train_loader)
+        self.criterion = CrossEntropyLabelSmooth(cfg.optim.label_smoothing)
+        self.optimizer = get_optimizer(model, cfg.optim)
+        self.lr_sched = LRScheduler(self.optimizer, cfg.optim,
+                                    self.steps_per_epoch, cfg.train.epochs)
+        self.ema = (ExponentialMovingAverage(model, cfg.ema.decay)
+                    if cfg.ema.enabled else None)
+        self.start_epoch = 0
+        self.global_step = 0
+        self.best_top1 = 0.0
+
+        if cfg.train.pretrained:
+            ckpt.load_pretrained(cfg.train.pretrained, model)
+            self.logger.info("loaded pretrained %s", cfg.train.pretrained)
+        if cfg.train.resume and os.path.exists(cfg.train.resume):
+            state = ckpt.load_checkpoint(
+                cfg.train.resume, model, self.optimizer, self.ema,
+                self.lr_sched, map_location=str(self.device))
+            self.start_epoch = state.get("epoch", 0)
+            self.global_step = state.get("global_step", 0)
+            self.best_top1 = state.get("meta", {}).get("best_top1", 0.0)
+            self.logger.info("resumed from %s at epoch %d",
+                             cfg.train.resume, self.start_epoch)
+
+    # ------------------------------------------------------------------
+
+    def _to_device(self, x: torch.Tensor, y: torch.Tensor):
+        x = x.to(self.device, non_blocking=True)
+        if self.dtype is torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        if self.cfg.train.channels_last and self.device.type == "cuda":
+            x = x.contiguous(memory_format=torch.channels_last)
+        return x, y.to(self.device, non_blocking=True)
+
+    def train_one_epoch(self, epoch: int) -> None:
+        cfg = self.cfg
+        self.engine.train()
+        sampler = getattr(self.train_loader, "sampler", None)
+        if hasattr(sampler, "set_epoch"):
+            sampler.set_epoch(epoch)
+        meters = {k: AverageMeter(k) for k in ("loss", "top1", "top5", "img_s")}
+        t_last = time.time()
+        for it, (x, y) in enumerate(self.train_loader):
+            x, y = self._to_device(x, y)
+            lr = self.lr_sched.step(self.global_step)
+            self.engine.zero_grad()
+            logits = self.engine(x)
+            loss = self.criterion(logits, y)
+            loss.backward()
+            self.engine.finish()
+            self.optimizer.step()
+            if self.ema is not None:
+                self.ema.update()
+            self.global_step += 1
+
+            with torch.no_grad():
+                top1, top5 = accuracy(logits.detach(), y, (1, 5))
+            bs = x.shape[0]
+            now = time.time()
+            meters["loss"].update(loss.item(), bs)
+            meters["top1"].update(top1.item(), bs)
+            meters["top5"].update(top5.item(), bs)
+            meters["img_s"].update(bs * self.world / max(now - t_last, 1e-9))
+            t_last = now
+            if it % cfg.train.log_interval == 0 and self.rank == 0:
+                self.logger.info(
+                    "epoch %d it %d/%d loss %.4f top1 %.2f lr %.4g img/s %.0f",
+                    epoch, it, self.steps_per_epoch, meters["loss"].avg,
+                    meters["top1"].avg, lr, meters["img_s"].val)
+                self.jsonl.write(kind="train", epoch=epoch, step=self.global_step,
+                                 loss=meters["loss"].avg, top1=meters["top1"].avg,
+                                 lr=lr, img_s=meters["img_s"].val)
+
+    @torch.no_grad()
+    def validate(self, epoch: int, use_ema: Optional[bool] = None) -> float:
+        cfg = self.cfg
+        use_ema = (cfg.val.use_ema if use_ema is None else use_ema) and self.ema
+        if use_ema:
+            self.ema.swap()
+        self.engine.eval()
+        counters = torch.zeros(3, device=self.device)  # n, top1-correct, top5-correct
+        for x, y in self.val_loader:
+            x, y = self._to_device(x, y)
+            logits = self.model(x)
+            top1, top5 = accuracy(logits, y, (1, 5))
+            bs = x.shape[0]
+            counters += torch.tensor(
+                [bs, top1.item() * bs / 100, top5.item() * bs / 100],
+                device=self.device)
+        counters = udist.dist_all_reduce_tensor(counters, op="sum")
+        top1 = 100.0 * counters[1].item() / max(counters[0].item(), 1)
+        top5 = 100.0 * counters[2].item() / max(counters[0].item(), 1)
+        if use_ema:
+            self.ema.swap()
+        if self.rank == 0:
+            tag = "ema" if use_ema else "raw"
+            self.logger.info("VAL epoch %d (%s): top1 %.3f top5 %.3f",
+                             epoch, tag, top1, top5)
+            self.jsonl.write(kind="val", epoch=epoch, top1=top1, top5=top5,
+                             ema=bool(use_ema))
+        return top1
+
+    def save(self, epoch: int, top1: float) -> None:
+        if self.rank != 0:
+            return
+        meta = {"best_top1": self.best_top1, "config": self.cfg.to_dict()}
+        path = os.path.join(self.cfg.log.dir, "checkpoint.pt")
+        ckpt.save_checkpoint(path, self.model, self.optimizer, self.ema,
+                             self.lr_sched, epoch + 1, self.global_step, meta)
+        if top1 > self.best_top1:
+            self.best_top1 = top1
+            ckpt.save_checkpoint(os.path.join(self.cfg.log.dir, "best.pt"),
+                                 self.model, epoch=epoch + 1,
+                                 global_step=self.global_step, meta=meta)
+
+    def fit(self) -> None:
+        cfg = self.cfg
+        for epoch in range(self.start_epoch, cfg.train.epochs):
+            self.train_one_epoch(epoch)
+            top1 = 0.0
+            if (epoch + 1) % cfg.val.interval_epochs == 0:
+                top1 = self.validate(epoch)
+            if (epoch + 1) % cfg.train.save_interval_epochs == 0:
+                self.save(epoch, top1)
+        self.jsonl.close()

@@ -1,0 +1,369 @@
+"""GPU op path: custom autograd.Functions over the CDNA4 HIP kernels.
+
+Layout contract with the extension (`ops/csrc`): activations are NCHW
+logical / ``torch.channels_last`` memory (== NHWC rows, C fastest), compute
+dtype bf16, accumulation and statistics fp32.  Weight tensors are re-packed
+per call into the kernel-friendly layouts:
+
+- pointwise 1x1 [Cout, Cin, 1, 1]   -> GEMM B operand [Cin, Cout] bf16
+- depthwise  [C, 1, k, k]           -> [k*k, C] bf16 (C fastest)
+- stem dense [Cout, 3, 3, 3]        -> [3, 3, 3, Cout] bf16 (Cout fastest)
+
+Gradients w.r.t. parameters are produced in fp32 (master-weight dtype);
+gradients w.r.t. activations in bf16.  All reductions (BN stats, wgrad)
+are two-stage slab reductions — deterministic, no atomics.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from .backend import ext
+
+BF16 = torch.bfloat16
+CL = torch.channels_last
+
+# activation codes shared with csrc/common.h
+ACT_IDENTITY = 0
+ACT_RELU = 1
+ACT_RELU6 = 2
+ACT_HSWISH = 3
+ACT_HSIGMOID = 4
+
+
+def _nhwc(x: torch.Tensor) -> torch.Tensor:
+    """bf16 + channels_last view of a 4D activation tensor."""
+    if x.dim() == 2:
+        x = x.view(x.shape[0], x.shape[1], 1, 1)
+    return x.to(BF16).contiguous(memory_format=CL)
+
+
+def _empty_like_nhwc(x: torch.Tensor, c: Optional[int] = None,
+                     h: Optional[int] = None, w: Optional[int] = None) -> torch.Tensor:
+    n = x.shape[0]
+    c = c if c is not None else x.shape[1]
+    h = h if h is not None else x.shape[2]
+    w = w if w is not None else x.shape[3]
+    return torch.empty((n, c, h, w), dtype=BF16, device=x.device,
+                       memory_format=CL)
+
+
+def _conv_out_hw(h: int, w: int, k: int, s: int, p: int) -> Tuple[int, int]:
+    return (h + 2 * p - k) // s + 1, (w + 2 * p - k) // s + 1
+
+
+def _conv_forward(x: torch.Tensor, w_packed: torch.Tensor, kind: str,
+                  k: int, s: int, p: int) -> torch.Tensor:
+    """Raw conv z = conv(x) with packed bf16 weights; returns bf16 NHWC."""
+    e = ext()
+    n, cin, h, wdt = x.shape
+    if kind == "pw":
+        cout = w_packed.shape[1]
+        z = _empty_like_nhwc(x, c=cout)
+        e.pw_conv(x, w_packed, z)
+    elif kind == "dw":
+        ho, wo = _conv_out_hw(h, wdt, k, s, p)
+        z = _empty_like_nhwc(x, h=ho, w=wo)
+        e.dw_conv_fwd(x, w_packed, z, k, s, p)
+    elif kind == "stem":
+        cout = w_packed.shape[3]
+        ho, wo = _conv_out_hw(h, wdt, k, s, p)
+        z = _empty_like_nhwc(x, c=cout, h=ho, w=wo)
+        e.stem_conv_fwd(x, w_packed, z, k, s, p)
+    else:
+        raise ValueError(kind)
+    return z
+
+
+class _ConvBNActFn(torch.autograd.Function):
+    """Fused conv → BN(train|eval) → act [+ residual] on the HIP engine."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bn_w, bn_b, residual,
+                running_mean, running_var, kind, k, s, p,
+                act, training, momentum, eps):
+        e = ext()
+        x = _nhwc(x)
+        cout = weight.shape[0]
+        cin = x.shape[1]
+        # pack weights into kernel layout (bf16)
+        if kind == "pw":
+            w_packed = weight.view(cout, cin).t().to(BF16).contiguous()
+        elif kind == "dw":
+            w_packed = weight.view(cout, k * k).t().to(BF16).contiguous()
+        elif kind == "stem":
+            w_packed = weight.permute(2, 3, 1, 0).to(BF16).contiguous()
+        else:
+            raise ValueError(kind)
+
+        z = _conv_forward(x, w_packed, kind, k, s, p)
+
+        mean = torch.empty(cout, dtype=torch.float32, device=x.device)
+        invstd = torch.empty_like(mean)
+        if training:
+            e.bn_stats(z, mean, invstd, running_mean, running_var,
+                       momentum, eps)
+        else:
+            mean.copy_(running_mean)
+            invstd.copy_((running_var.float() + eps).rsqrt())
+
+        y = torch.empty_like(z)
+        res = residual
+        if res is not None:
+            res = _nhwc(res)
+        e.bn_act_fwd(z, mean, invstd, bn_w, bn_b, act,
+                     y, res)
+
+        ctx.save_for_backward(x, z, mean, invstd, bn_w, bn_b, weight)
+        ctx.kind, ctx.k, ctx.s, ctx.p = kind, k, s, p
+        ctx.act, ctx.training, ctx.eps = act, training, eps
+        ctx.has_residual = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        e = ext()
+        x, z, mean, invstd, bn_w, bn_b, weight = ctx.saved_tensors
+        kind, k, s, p = ctx.kind, ctx.k, ctx.s, ctx.p
+        dy = dy.to(BF16).contiguous(memory_format=CL)
+        cout = weight.shape[0]
+        cin = x.shape[1]
+
+        dgamma = torch.empty(cout, dtype=torch.float32, device=x.device)
+        dbeta = torch.empty_like(dgamma)
+        dz = torch.empty_like(z)
+        e.bn_act_bwd(dy, z, mean, invstd, bn_w, bn_b, ctx.act,
+                     bool(ctx.training), dgamma, dbeta, dz)
+
+        dx = None
+        if ctx.needs_input_grad[0]:
+            if kind == "pw":
+                w_t = weight.view(cout, cin).to(BF16).contiguous()  # [Cout,Cin]
+                dx = _empty_like_nhwc(x)
+                e.pw_conv(dz, w_t, dx)
+            elif kind == "dw":
+                w_packed = weight.view(cout, k * k).t().to(BF16).contiguous()
+                dx = _empty_like_nhwc(x)
+                e.dw_conv_dgrad(dz, w_packed, dx, k, s, p)
+            else:
+                raise NotImplementedError(
+                    "stem conv input gradient is not implemented "
+                    "(the stem is the first layer; its input needs no grad)")
+
+        dweight = None
+        if ctx.needs_input_grad[1]:
+            if kind == "pw":
+                dwp = torch.empty(cin, cout, dtype=torch.float32,
+                                  device=x.device)
+                e.pw_conv_wgrad(x, dz, dwp)
+                dweight = dwp.t().reshape(cout, cin, 1, 1)
+            elif kind == "dw":
+                dwp = torch.empty(k * k, cout, dtype=torch.float32,
+                                  device=x.device)
+                e.dw_conv_wgrad(x, dz, dwp, k, s, p)
+                dweight = dwp.t().reshape(cout, 1, k, k)
+            else:  # stem
+                dwp = torch.empty(k, k, cin, cout, dtype=torch.float32,
+                                  device=x.device)
+                e.stem_conv_wgrad(x, dz, dwp, k, s, p)
+                dweight = dwp.permute(3, 2, 0, 1).contiguous()
+
+        dres = dy if ctx.has_residual else None
+        return (dx, dweight, dgamma, dbeta, dres) + (None,) * 10
+
+
+def conv_bn_act(x, weight, bn_weight, bn_bias, running_mean, running_var, *,
+                stride, padding, groups, act, training, momentum, eps,
+                residual=None, conv_bias=None):
+    cout, cin_g, kh, kw = weight.shape
+    if bn_weight is None or conv_bias is not None:
+        raise NotImplementedError(
+            "HIP conv_bn_act requires BN and no conv bias; the zoo's only "
+            "bias convs are post-pool 1x1s, expressed as linear()")
+    if kh != kw:
+        raise ValueError("square kernels only")
+    if groups == 1 and kh == 1:
+        kind = "pw"
+    elif groups == cout and cin_g == 1:
+        kind = "dw"
+    elif groups == 1 and cin_g == 3 and kh == 3:
+        kind = "stem"
+    else:
+        raise NotImplementedError(
+            f"conv shape not in the zoo: k={kh} groups={groups} cin={cin_g}")
+    return _ConvBNActFn.apply(
+        x, weight, bn_weight, bn_bias, residual, running_mean, running_var,
+        kind, kh, stride, padding, act, training, momentum, eps)
+
+
+class _LinearActFn(torch.autograd.Function):
+    """y = act(x @ W^T + b) via the pointwise MFMA GEMM (M = batch)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, act):
+        e = ext()
+        x2 = x.to(BF16).contiguous()
+        if x2.dim() != 2:
+            x2 = x2.view(x2.shape[0], -1)
+        nout, nin = weight.shape
+        wt = weight.t().to(BF16).contiguous()  # [in, out]
+        z = torch.empty(x2.shape[0], nout, dtype=BF16, device=x.device)
+        e.pw_conv(x2.view(x2.shape[0], nin, 1, 1), wt,
+                  z.view(z.shape[0], nout, 1, 1))
+        y = torch.empty_like(z)
+        e.bias_act_fwd(z, bias.float() if bias is not None else None, act, y)
+        ctx.save_for_backward(x2, z, weight,
+                              bias if bias is not None else torch.empty(0))
+        ctx.act = act
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        e = ext()
+        x2, z, weight, bias = ctx.saved_tensors
+        dy = dy.to(BF16).contiguous()
+        nout, nin = weight.shape
+        dz = torch.empty_like(dy)
+        dbias = (torch.empty(nout, dtype=torch.float32, device=dy.device)
+                 if ctx.has_bias else None)
+        e.bias_act_bwd(dy, z, bias.float() if ctx.has_bias else None,
+                       ctx.act, dz, dbias)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            w = weight.to(BF16).contiguous()  # [out, in]
+            dx = torch.empty_like(x2)
+            e.pw_conv(dz.view(-1, nout, 1, 1), w, dx.view(-1, nin, 1, 1))
+        dweight = None
+        if ctx.needs_input_grad[1]:
+            dwp = torch.empty(nin, nout, dtype=torch.float32, device=dy.device)
+            e.pw_conv_wgrad(x2.view(-1, nin, 1, 1), dz.view(-1, nout, 1, 1), dwp)
+            dweight = dwp.t().contiguous()
+        return dx, dweight, dbias, None
+
+
+def linear(x, weight, bias=None, act=ACT_IDENTITY):
+    return _LinearActFn.apply(x, weight, bias, act)
+
+
+class _GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        e = ext()
+        x = _nhwc(x)
+        out = torch.empty(x.shape[0], x.shape[1], dtype=BF16, device=x.device)
+        e.avgpool_fwd(x, out)
+        ctx.in_shape = x.shape
+        ctx.dev = x.device
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        e = ext()
+        n, c, h, w = ctx.in_shape
+        dy = dy.to(BF16).contiguous()
+        dx = torch.empty((n, c, h, w), dtype=BF16, device=ctx.dev,
+                         memory_format=CL)
+        e.avgpool_bwd(dy, dx)
+        return dx
+
+
+def global_avg_pool(x):
+    return _GlobalAvgPoolFn.apply(x)
+
+
+class _SqueezeExciteFn(torch.autograd.Function):
+    """SE block (K7): pool → FC+ReLU → FC+hsigmoid → channel scale.
+
+    The two FCs run on the pointwise GEMM at M = N (batch); pool/scale are
+    dedicated reduction/broadcast kernels.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        e = ext()
+        x = _nhwc(x)
+        n, c, h, wdt = x.shape
+        r = w1.shape[0]
+        pooled = torch.empty(n, c, dtype=BF16, device=x.device)
+        e.avgpool_fwd(x, pooled)
+        w1t = w1.view(r, c).t().to(BF16).contiguous()
+        hz = torch.empty(n, r, dtype=BF16, device=x.device)
+        e.pw_conv(pooled.view(n, c, 1, 1), w1t, hz.view(n, r, 1, 1))
+        ha = torch.empty_like(hz)
+        e.bias_act_fwd(hz, b1.float(), ACT_RELU, ha)
+        w2t = w2.view(c, r).t().to(BF16).contiguous()
+        sz = torch.empty(n, c, dtype=BF16, device=x.device)
+        e.pw_conv(ha.view(n, r, 1, 1), w2t, sz.view(n, c, 1, 1))
+        s = torch.empty_like(sz)
+        e.bias_act_fwd(sz, b2.float(), ACT_HSIGMOID, s)
+        y = torch.empty_like(x)
+        e.rowscale_fwd(x, s, y)
+        ctx.save_for_backward(x, pooled, hz, ha, sz, s, w1, b1, w2, b2)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        e = ext()
+        x, pooled, hz, ha, sz, s, w1, b1, w2, b2 = ctx.saved_tensors
+        n, c, h, wdt = x.shape
+        r = w1.shape[0]
+        dy = dy.to(BF16).contiguous(memory_format=CL)
+        # d(x*s): dx_part = dy*s (broadcast), ds = sum_hw dy*x
+        dx = torch.empty_like(x)
+        ds = torch.empty(n, c, dtype=BF16, device=x.device)
+        e.rowscale_bwd(dy, x, s, dx, ds)
+        # hsigmoid gate backward
+        dsz = torch.empty_like(ds)
+        db2 = torch.empty(c, dtype=torch.float32, device=x.device)
+        e.bias_act_bwd(ds, sz, b2.float(), ACT_HSIGMOID, dsz, db2)
+        # fc2 backward
+        w2m = w2.view(c, r).to(BF16).contiguous()
+        dha = torch.empty_like(ha)
+        e.pw_conv(dsz.view(n, c, 1, 1), w2m, dha.view(n, r, 1, 1))
+        dw2p = torch.empty(r, c, dtype=torch.float32, device=x.device)
+        e.pw_conv_wgrad(ha.view(n, r, 1, 1), dsz.view(n, c, 1, 1), dw2p)
+        dw2 = dw2p.t().reshape_as(w2)
+        # relu backward
+        dhz = torch.empty_like(dha)
+        db1 = torch.empty(r, dtype=torch.float32, device=x.device)
+        e.bias_act_bwd(dha, hz, b1.float(), ACT_RELU, dhz, db1)
+        # fc1 backward
+        w1m = w1.view(r, c).to(BF16).contiguous()
+        dpooled = torch.empty_like(pooled)
+        e.pw_conv(dhz.view(n, r, 1, 1), w1m, dpooled.view(n, c, 1, 1))
+        dw1p = torch.empty(c, r, dtype=torch.float32, device=x.device)
+        e.pw_conv_wgrad(pooled.view(n, c, 1, 1), dhz.view(n, r, 1, 1), dw1p)
+        dw1 = dw1p.t().reshape_as(w1)
+        # pool backward folds into dx: dx += broadcast(dpooled)/(H*W)
+        e.avgpool_bwd_add(dpooled, dx)
+        return dx, dw1, db1, dw2, db2
+
+
+def squeeze_excite(x, w1, b1, w2, b2):
+    return _SqueezeExciteFn.apply(x, w1, b1, w2, b2)
+
+
+class _LabelSmoothCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target, smoothing):
+        e = ext()
+        logits = logits.to(BF16).contiguous()
+        loss = torch.empty((), dtype=torch.float32, device=logits.device)
+        dlogits = torch.empty(logits.shape, dtype=torch.float32,
+                              device=logits.device)
+        e.label_smooth_ce(logits, target.to(torch.int64).contiguous(),
+                          float(smoothing), loss, dlogits)
+        ctx.save_for_backward(dlogits)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        (dlogits,) = ctx.saved_tensors
+        return dlogits * dloss, None, None
+
+
+def label_smooth_ce(logits, target, smoothing):
+    return _LabelSmoothCEFn.apply(logits, target, smoothing)

@@ -1,0 +1,148 @@
+"""Data-parallel engine: bucketed gradient all-reduce over RCCL/xGMI.
+
+Replaces the reference's thin ``AllReduceDistributedDataParallel``
+ with an MI355X-first design:
+
+- Parameters are bucketed in reverse registration order (the order their
+  grads are produced during backward) into flat fp32 buffers; ``p.grad``
+  is a *view* into its bucket, so autograd accumulates directly into
+  contiguous memory and each all-reduce is one flat message — no
+  gather/scatter copies.
+- Default bucket count ≈ the 7 xGMI point-to-point links per GPU: the node
+  is fully connected (no NVSwitch), so several in-flight ring all-reduces
+  spread across links; with overlap on, each bucket's reduce launches as
+  soon as backward finishes its last parameter and runs on RCCL's comm
+  stream under the remaining backward.
+- ``finish()`` waits on all handles and averages.  The same code runs the
+  CPU fake-world tests over gloo (world_size 2–8, no GPUs).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from .distributed import broadcast_module, get_world_size
+
+
+class _Bucket:
+    __slots__ = ("params", "flat", "pending", "handle", "ready_event")
+
+    def __init__(self, params: List[torch.nn.Parameter], device, dtype) -> None:
+        self.params = params
+        total = sum(p.numel() for p in params)
+        self.flat = torch.zeros(total, device=device, dtype=dtype)
+        off = 0
+        for p in params:
+            p.grad = self.flat[off:off + p.numel()].view_as(p)
+            off += p.numel()
+        self.pending = 0
+        self.handle = None
+
+
+class DataParallelEngine(nn.Module):
+    """Wrap a module for single-node data parallelism.
+
+    Usage::
+
+        engine = DataParallelEngine(model, bucket_mb=4, overlap=True)
+        engine.zero_grad()
+        loss = crit(engine(x), y)
+        loss.backward()          # buckets all-reduce as they fill
+        engine.finish()          # wait + average
+        optimizer.step()
+    """
+
+    def __init__(self, module: nn.Module, bucket_mb: float = 4.0,
+                 overlap: bool = True, grad_dtype: torch.dtype = torch.float32,
+                 broadcast_params: bool = True) -> None:
+        super().__init__()
+        self.module = module
+        self.world_size = get_world_size()
+        self.overlap = overlap and self.world_size > 1
+        self.grad_dtype = grad_dtype
+        if broadcast_params and self.world_size > 1:
+            broadcast_module(module)
+        params = [p for p in module.parameters() if p.requires_grad]
+        device = params[0].device if params else torch.device("cpu")
+        # reverse registration order ≈ grad production order during backward
+        params = list(reversed(params))
+        bucket_elems = max(int(bucket_mb * 1024 * 1024 / 4), 1)
+        self.buckets: List[_Bucket] = []
+        cur: List[nn.Parameter] = []
+        cur_n = 0
+        for p in params:
+            cur.append(p)
+            cur_n += p.numel()
+            if cur_n >= bucket_elems:
+                self.buckets.append(_Bucket(cur, device, grad_dtype))
+                cur, cur_n = [], 0
+        if cur:
+            self.buckets.append(_Bucket(cur, device, grad_dtype))
+        self._param_bucket: Dict[nn.Parameter, _Bucket] = {}
+        for b in self.buckets:
+            for p in b.params:
+                self._param_bucket[p] = b
+        self._hooks = []
+        if self.overlap:
+            for p in params:
+                h = p.register_post_accumulate_grad_hook(self._grad_hook)
+                self._hooks.append(h)
+        self._launched: List[_Bucket] = []
+
+    # -- training-step API -------------------------------------------------
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def zero_grad(self, set_to_none: bool = False) -> None:  # noqa: ARG002
+        """Zero the flat bucket buffers (grads are views; never set to None)."""
+        for b in self.buckets:
+            b.flat.zero_()
+            b.pending = len(b.params)
+            b.handle = None
+        self._launched.clear()
+
+    def _grad_hook(self, p: nn.Parameter) -> None:
+        b = self._param_bucket[p]
+        b.pending -= 1
+        if b.pending == 0 and self.world_size > 1:
+            b.handle = dist.all_reduce(b.flat, async_op=True)
+            self._launched.append(b)
+
+    def finish(self) -> None:
+        """Wait for in-flight reduces (or run them now) and average."""
+        if self.world_size <= 1:
+            return
+        if self.overlap:
+            for b in self.buckets:
+                if b.handle is None:
+                    # a bucket whose params got no grads this step: reduce
+                    # anyway so ranks stay collectively consistent
+                    b.handle = dist.all_reduce(b.flat, async_op=True)
+            for b in self.buckets:
+                b.handle.wait()
+                b.handle = None
+        else:
+            for b in self.buckets:
+                dist.all_reduce(b.flat)
+        inv = 1.0 / self.world_size
+        torch._foreach_mul_([b.flat for b in self.buckets], inv)
+
+    # -- passthrough -------------------------------------------------------
+
+    def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, *args, **kwargs):
+        return self.module.load_state_dict(*args, **kwargs)
+
+    def train(self, mode: bool = True):
+        self.module.train(mode)
+        return self
+
+    def eval(self):
+        return self.train(False)

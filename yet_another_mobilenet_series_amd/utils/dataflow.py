@@ -1,0 +1,160 @@
+"""Data pipeline (SURVEY.md §2.2 "Data pipeline", §3.5).
+
+Loaders:
+- ``synthetic``: random images + fixed-seed labels, device-shape identical
+  to ImageNet 224² — what BASELINE.json's benchmark configs use (no
+  network ⇒ no real dataset on these boxes).
+- ``imagefolder``: ImageNet-style ``root/<split>/<class>/*.jpg`` directory
+  via PIL (torchvision is not installed in this image; the transform set —
+  RandomResizedCrop(224) + hflip for train, Resize(256)+CenterCrop(224)
+  for val — is implemented here directly).
+
+All loaders return standard ``torch.utils.data.DataLoader`` objects with a
+``DistributedSampler`` when the process group is initialised.
+"""
+
+from __future__ import annotations
+
+import os
+import random
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader, Dataset, DistributedSampler
+
+try:
+    from PIL import Image
+    HAS_PIL = True
+except ImportError:  # pragma: no cover
+    HAS_PIL = False
+
+IMAGENET_MEAN = (0.485, 0.456, 0.406)
+IMAGENET_STD = (0.229, 0.224, 0.225)
+
+
+class SyntheticDataset(Dataset):
+    """Random images, deterministic per-index labels."""
+
+    def __init__(self, size: int = 12800, image_size: int = 224,
+                 num_classes: int = 1000, seed: int = 0) -> None:
+        self.size = size
+        self.image_size = image_size
+        self.num_classes = num_classes
+        self.seed = seed
+
+    def __len__(self) -> int:
+        return self.size
+
+    def __getitem__(self, idx: int) -> Tuple[torch.Tensor, int]:
+        g = torch.Generator().manual_seed(self.seed * 1000003 + idx)
+        img = torch.randn(3, self.image_size, self.image_size, generator=g)
+        label = int(torch.randint(0, self.num_classes, (1,), generator=g))
+        return img, label
+
+
+def _pil_to_tensor(img) -> torch.Tensor:
+    arr = np.asarray(img, dtype=np.uint8)
+    if arr.ndim == 2:
+        arr = np.stack([arr] * 3, axis=-1)
+    t = torch.from_numpy(arr).permute(2, 0, 1).float().div_(255.0)
+    mean = torch.tensor(IMAGENET_MEAN).view(3, 1, 1)
+    std = torch.tensor(IMAGENET_STD).view(3, 1, 1)
+    return (t - mean) / std
+
+
+class ImageFolderDataset(Dataset):
+    """Minimal ImageFolder with Inception-style train augmentation."""
+
+    EXT = (".jpg", ".jpeg", ".png", ".bmp", ".webp")
+
+    def __init__(self, root: str, train: bool, image_size: int = 224) -> None:
+        if not HAS_PIL:
+            raise RuntimeError("PIL is required for ImageFolderDataset")
+        self.root = root
+        self.train = train
+        self.image_size = image_size
+        classes = sorted(d for d in os.listdir(root)
+                         if os.path.isdir(os.path.join(root, d)))
+        self.class_to_idx = {c: i for i, c in enumerate(classes)}
+        self.samples: List[Tuple[str, int]] = []
+        for c in classes:
+            cdir = os.path.join(root, c)
+            for fn in sorted(os.listdir(cdir)):
+                if fn.lower().endswith(self.EXT):
+                    self.samples.append((os.path.join(cdir, fn),
+                                         self.class_to_idx[c]))
+        if not self.samples:
+            raise RuntimeError(f"no images found under {root}")
+
+    def __len__(self) -> int:
+        return len(self.samples)
+
+    def _random_resized_crop(self, img):
+        w, h = img.size
+        area = w * h
+        for _ in range(10):
+            target_area = random.uniform(0.08, 1.0) * area
+            ar = random.uniform(3 / 4, 4 / 3)
+            cw = int(round((target_area * ar) ** 0.5))
+            ch = int(round((target_area / ar) ** 0.5))
+            if cw <= w and ch <= h:
+                x0 = random.randint(0, w - cw)
+                y0 = random.randint(0, h - ch)
+                img = img.crop((x0, y0, x0 + cw, y0 + ch))
+                return img.resize((self.image_size, self.image_size),
+                                  Image.BILINEAR)
+        # fallback: center crop
+        return self._center_crop(img)
+
+    def _center_crop(self, img):
+        scale = 256 / min(img.size)
+        img = img.resize((max(1, round(img.size[0] * scale)),
+                          max(1, round(img.size[1] * scale))), Image.BILINEAR)
+        w, h = img.size
+        s = self.image_size
+        x0, y0 = (w - s) // 2, (h - s) // 2
+        return img.crop((x0, y0, x0 + s, y0 + s))
+
+    def __getitem__(self, idx: int) -> Tuple[torch.Tensor, int]:
+        path, label = self.samples[idx]
+        img = Image.open(path).convert("RGB")
+        if self.train:
+            img = self._random_resized_crop(img)
+            if random.random() < 0.5:
+                img = img.transpose(Image.FLIP_LEFT_RIGHT)
+        else:
+            img = self._center_crop(img)
+        return _pil_to_tensor(img), label
+
+
+def build_dataset(cfg, train: bool, seed: int = 0) -> Dataset:
+    if cfg.dataset == "synthetic":
+        size = cfg.synthetic_size if train else max(cfg.synthetic_size // 10, 64)
+        return SyntheticDataset(size=size, image_size=cfg.image_size,
+                                seed=seed + (0 if train else 1))
+    if cfg.dataset in ("imagefolder", "imagenet"):
+        split = "train" if train else "val"
+        return ImageFolderDataset(os.path.join(cfg.root, split), train,
+                                  cfg.image_size)
+    raise ValueError(f"unknown dataset {cfg.dataset!r}")
+
+
+def get_data_loader(cfg, train: bool, batch_size: Optional[int] = None,
+                    seed: int = 0, distributed: Optional[bool] = None) -> DataLoader:
+    import torch.distributed as dist
+    ds = build_dataset(cfg, train, seed)
+    if distributed is None:
+        distributed = dist.is_initialized()
+    sampler = (DistributedSampler(ds, shuffle=train, seed=seed)
+               if distributed else None)
+    bs = batch_size or cfg.batch_size_per_gpu
+    return DataLoader(
+        ds, batch_size=bs, shuffle=(train and sampler is None),
+        sampler=sampler, num_workers=cfg.workers, pin_memory=torch.cuda.is_available(),
+        drop_last=train, persistent_workers=cfg.workers > 0)
+
+
+def get_data_loaders(cfg, seed: int = 0):
+    return (get_data_loader(cfg, True, seed=seed),
+            get_data_loader(cfg, False, seed=seed))

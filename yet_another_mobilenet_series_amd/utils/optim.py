@@ -1,0 +1,218 @@
+"""Optimization layer (SURVEY.md §2.2 "Optimizers/LR", "Losses", "EMA").
+
+- SGD (momentum, nesterov) and TF-semantics RMSprop (epsilon inside the
+  sqrt, momentum on the update) — the MobileNetV3 paper recipe.
+- Param-group splitter exempting BN affine params and biases from weight
+  decay.
+- Warmup + {cosine, exponential-decay, multistep} LR schedules (per-step
+  granularity).
+- CrossEntropyLabelSmooth (functional lives in ops; module wrapper here).
+- ExponentialMovingAverage over params + BN running stats, decay ~0.9999.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, Iterable, List, Optional
+
+import torch
+import torch.nn as nn
+
+from ..ops import functional as OF
+
+
+class CrossEntropyLabelSmooth(nn.Module):
+    def __init__(self, smoothing: float = 0.1) -> None:
+        super().__init__()
+        self.smoothing = smoothing
+
+    def forward(self, logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+        return OF.label_smooth_ce(logits, target, self.smoothing)
+
+
+class RMSpropTF(torch.optim.Optimizer):
+    """RMSprop with TensorFlow semantics.
+
+    Differences from ``torch.optim.RMSprop`` (they matter for the
+    MobileNetV3 recipe):
+    - epsilon is added INSIDE the sqrt: ``update = g / sqrt(ms + eps)``;
+    - the square-accumulator is initialised to 1 (TF's ``ms`` init), not 0;
+    - momentum buffers accumulate the LR-scaled update
+      (``mom = m*mom + lr*g/sqrt(ms+eps)``; param -= mom), like TF's
+      RMSPropOptimizer;
+    - weight decay is applied to the gradient (L2, not decoupled).
+    """
+
+    def __init__(self, params, lr: float = 0.01, alpha: float = 0.9,
+                 eps: float = 1e-3, momentum: float = 0.9,
+                 weight_decay: float = 0.0) -> None:
+        defaults = dict(lr=lr, alpha=alpha, eps=eps, momentum=momentum,
+                        weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            lr = group["lr"]
+            alpha = group["alpha"]
+            eps = group["eps"]
+            momentum = group["momentum"]
+            wd = group["weight_decay"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad
+                state = self.state[p]
+                if not state:
+                    state["square_avg"] = torch.ones_like(p)
+                    if momentum > 0:
+                        state["momentum_buffer"] = torch.zeros_like(p)
+                if wd != 0:
+                    g = g.add(p, alpha=wd)
+                sq = state["square_avg"]
+                sq.mul_(alpha).addcmul_(g, g, value=1 - alpha)
+                upd = g / (sq + eps).sqrt()
+                if momentum > 0:
+                    buf = state["momentum_buffer"]
+                    buf.mul_(momentum).add_(upd, alpha=lr)
+                    p.add_(buf, alpha=-1)
+                else:
+                    p.add_(upd, alpha=-lr)
+        return loss
+
+
+def split_param_groups(model: nn.Module, weight_decay: float,
+                       no_wd_on_bn_bias: bool = True) -> List[Dict]:
+    """Two param groups: weights with WD, {BN affine, biases} without.
+
+    Matches the reference recipe ("param groups exempting
+    BN/bias from weight decay").
+    """
+    if not no_wd_on_bn_bias:
+        return [{"params": list(model.parameters()), "weight_decay": weight_decay}]
+    decay, no_decay = [], []
+    for name, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        if p.ndim <= 1 or name.endswith(".bias"):
+            no_decay.append(p)
+        else:
+            decay.append(p)
+    return [{"params": decay, "weight_decay": weight_decay},
+            {"params": no_decay, "weight_decay": 0.0}]
+
+
+def get_optimizer(model: nn.Module, cfg) -> torch.optim.Optimizer:
+    groups = split_param_groups(model, cfg.weight_decay, cfg.no_wd_on_bn_bias)
+    if cfg.name == "sgd":
+        return torch.optim.SGD(groups, lr=cfg.lr, momentum=cfg.momentum,
+                               nesterov=cfg.nesterov)
+    if cfg.name in ("rmsprop", "rmsprop_tf"):
+        return RMSpropTF(groups, lr=cfg.lr, alpha=cfg.rmsprop_alpha,
+                         eps=cfg.rmsprop_eps, momentum=cfg.momentum)
+    raise ValueError(f"unknown optimizer {cfg.name!r}")
+
+
+class LRScheduler:
+    """Per-step LR schedule with linear warmup, then cosine / exponential /
+    multistep decay.  Stateless in the optimizer: ``step(global_step)`` sets
+    every group's lr (so resume only needs the step counter)."""
+
+    def __init__(self, optimizer: torch.optim.Optimizer, cfg,
+                 steps_per_epoch: int, epochs: int) -> None:
+        self.optimizer = optimizer
+        self.base_lrs = [g["lr"] for g in optimizer.param_groups]
+        self.schedule = cfg.lr_schedule
+        self.warmup_steps = int(cfg.warmup_epochs * steps_per_epoch)
+        self.warmup_lr = cfg.warmup_lr
+        self.total_steps = int(epochs * steps_per_epoch)
+        self.steps_per_epoch = steps_per_epoch
+        self.exp_decay_rate = cfg.exp_decay_rate
+        self.exp_decay_steps = max(int(cfg.exp_decay_epochs * steps_per_epoch), 1)
+        self.milestones = [int(m * steps_per_epoch)
+                           for m in cfg.multistep_milestones]
+        self.multistep_gamma = cfg.multistep_gamma
+        self.last_step = -1
+
+    def lr_at(self, step: int, base_lr: float) -> float:
+        if self.warmup_steps > 0 and step < self.warmup_steps:
+            frac = step / self.warmup_steps
+            return self.warmup_lr + (base_lr - self.warmup_lr) * frac
+        s = step - self.warmup_steps
+        total = max(self.total_steps - self.warmup_steps, 1)
+        if self.schedule == "cosine":
+            return 0.5 * base_lr * (1 + math.cos(math.pi * min(s / total, 1.0)))
+        if self.schedule in ("exp", "exponential"):
+            return base_lr * (self.exp_decay_rate ** (s // self.exp_decay_steps))
+        if self.schedule == "multistep":
+            k = sum(1 for m in self.milestones if step >= m)
+            return base_lr * (self.multistep_gamma ** k)
+        if self.schedule in ("const", "none"):
+            return base_lr
+        raise ValueError(f"unknown lr schedule {self.schedule!r}")
+
+    def step(self, global_step: Optional[int] = None) -> float:
+        if global_step is None:
+            global_step = self.last_step + 1
+        self.last_step = global_step
+        lr = 0.0
+        for group, base in zip(self.optimizer.param_groups, self.base_lrs):
+            lr = self.lr_at(global_step, base)
+            group["lr"] = lr
+        return lr
+
+    def state_dict(self) -> Dict:
+        return {"last_step": self.last_step}
+
+    def load_state_dict(self, state: Dict) -> None:
+        self.last_step = state["last_step"]
+        self.step(self.last_step)
+
+
+class ExponentialMovingAverage:
+    """EMA of all parameters and float buffers (BN running stats), decay
+    ≈0.9999 (SURVEY.md §2.2 "EMA").  ``swap()`` exchanges the shadow with
+    the live weights for evaluation / checkpointing; call again to restore.
+    """
+
+    def __init__(self, model: nn.Module, decay: float = 0.9999) -> None:
+        self.decay = decay
+        self.model = model
+        self.shadow: Dict[str, torch.Tensor] = {
+            k: v.detach().clone().float()
+            for k, v in self._named_tensors(model)}
+
+    @staticmethod
+    def _named_tensors(model: nn.Module):
+        for k, v in model.state_dict().items():
+            if v.dtype.is_floating_point:
+                yield k, v
+
+    @torch.no_grad()
+    def update(self, model: Optional[nn.Module] = None) -> None:
+        model = model or self.model
+        d = self.decay
+        for k, v in self._named_tensors(model):
+            s = self.shadow[k]
+            s.mul_(d).add_(v.float(), alpha=1 - d)
+
+    @torch.no_grad()
+    def swap(self, model: Optional[nn.Module] = None) -> None:
+        """Swap shadow <-> live tensors (in place, dtype-preserving)."""
+        model = model or self.model
+        for k, v in self._named_tensors(model):
+            s = self.shadow[k]
+            tmp = v.detach().clone()
+            v.copy_(s.to(v.dtype))
+            s.copy_(tmp.float())
+
+    def state_dict(self) -> Dict[str, torch.Tensor]:
+        return {"decay": self.decay, "shadow": self.shadow}
+
+    def load_state_dict(self, state: Dict) -> None:
+        self.decay = state["decay"]
+        self.shadow = {k: v.clone() for k, v in state["shadow"].items()}
