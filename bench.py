@@ -1,0 +1,152 @@
+#!/usr/bin/env python3
+"""Benchmark: the BASELINE.json headline metric — images/sec (whole node),
+MobileNetV3-Large 1.0x 224 bf16 training, synthetic data, random-init
+weights, N GPUs of one node (DP over RCCL/xGMI).
+
+Single GPU:   python bench.py --steps 30 --warmup 10
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from yet_another_mobilenet_series_amd import models  # noqa: E402
+from yet_another_mobilenet_series_amd.ops import functional as OF  # noqa: E402
+from yet_another_mobilenet_series_amd.ops.backend import (  # noqa: E402
+    load_extension, set_engine_mode)
+from yet_another_mobilenet_series_amd.parallel import (  # noqa: E402
+    DataParallelEngine, distributed as udist)
+from yet_another_mobilenet_series_amd.utils.optim import (  # noqa: E402
+    RMSpropTF, split_param_groups)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="mobilenet_v3_large")
+    p.add_argument("--batch", type=int, default=256,
+                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--engine", default="hip", choices=["hip", "reference"],
+                   help="reference = eager fp32->bf16 torch ops (A/B baseline)")
+    p.add_argument("--graph", action="store_true", default=False,
+                   help="capture the step in a hipGraph (torch.cuda.CUDAGraph)")
+    p.add_argument("--optimizer", default="sgd", choices=["sgd", "rmsprop"])
+    p.add_argument("--bucket-mb", type=float, default=4.0)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    rank = udist.init_dist("nccl" if torch.cuda.is_available() else "gloo")
+    world = udist.get_world_size()
+    dev = (torch.device("cuda", udist.get_local_rank())
+           if torch.cuda.is_available() else torch.device("cpu"))
+    on_gpu = dev.type == "cuda"
+    if on_gpu:
+        torch.cuda.set_device(dev)
+        load_extension(required=args.engine == "hip")
+    set_engine_mode(args.engine)
+
+    torch.manual_seed(1234 + rank)
+    model = models.build_model(args.model, dropout=0.0).to(dev)
+    if on_gpu:
+        model = model.to(memory_format=torch.channels_last)
+    engine = DataParallelEngine(model, bucket_mb=args.bucket_mb, overlap=True)
+    if args.optimizer == "sgd":
+        opt = torch.optim.SGD(split_param_groups(model, 4e-5), lr=0.4,
+                              momentum=0.9, nesterov=True)
+    else:
+        opt = RMSpropTF(split_param_groups(model, 1e-5), lr=0.064,
+                        alpha=0.9, eps=1e-3, momentum=0.9)
+    model.train()
+
+    x = torch.randn(args.batch, 3, 224, 224, device=dev)
+    if on_gpu:
+        x = x.bfloat16().contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (args.batch,), device=dev)
+
+    def step():
+        engine.zero_grad()
+        loss = OF.label_smooth_ce(engine(x), y, 0.1)
+        loss.backward()
+        engine.finish()
+        opt.step()
+        return loss
+
+    graphed = None
+    if args.graph and on_gpu:
+        # warm up allocator + RCCL before capture
+        for _ in range(3):
+            step()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            static_loss = step()
+        graphed = (g, static_loss)
+
+    def run_one():
+        if graphed is not None:
+            graphed[0].replay()
+            return graphed[1]
+        return step()
+
+    for _ in range(max(args.warmup, 1)):
+        run_one()
+
+    udist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    loss = None
+    for _ in range(args.steps):
+        loss = run_one()
+    udist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+
+    # max over ranks = whole-job wall time
+    t = torch.tensor([dt], device=dev if on_gpu else "cpu")
+    if world > 1:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    dt = t.item()
+    ms_per_step = dt / args.steps * 1000.0
+    imgs_per_sec = args.batch * world * args.steps / dt
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node) MobileNetV3-Large 224 bf16 train",
+            "value": round(imgs_per_sec, 1),
+            "unit": "img/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "loss": round(float(loss), 4) if loss is not None else None,
+            "config": {"model": args.model, "global_batch": args.batch * world,
+                       "image_size": 224, "parallelism": f"dp{world}",
+                       "engine": args.engine, "graph": graphed is not None,
+                       "optimizer": args.optimizer},
+        }))
+    udist.destroy()
+
+
+if __name__ == "__main__":
+    main()

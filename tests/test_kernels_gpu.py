@@ -1,0 +1,330 @@
+"""HIP kernel numerics tests (MI355X, @gpu): each csrc kernel vs a plain
+PyTorch fp32 reference computed FROM THE SAME bf16-rounded inputs, so the
+only difference is accumulation order — tolerances are tight.
+
+Shapes parametrize over the zoo's distinct conv-shape classes
+(SURVEY.md Appendix A): skinny-K expands, wide-K projects, SE FCs, 112²
+wide-spatial depthwise through 7² channel-rich depthwise.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+BF16 = torch.bfloat16
+CL = torch.channels_last
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from yet_another_mobilenet_series_amd.ops.backend import ext as get_ext
+    return get_ext()
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+def randn_nhwc(n, c, h, w, seed=0, scale=1.0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    t = torch.randn(n, c, h, w, generator=g) * scale
+    return t.to(dev()).to(BF16).contiguous(memory_format=CL)
+
+
+def assert_close(got, want, rtol=1e-2, atol=None, msg=""):
+    got = got.float().cpu()
+    want = want.float().cpu()
+    if atol is None:
+        atol = 1e-2 * want.abs().mean().clamp(min=1e-3).item()
+    torch.testing.assert_close(got, want, rtol=rtol, atol=atol, msg=msg)
+
+
+# ---------------------------------------------------------------- pw GEMM
+
+@pytest.mark.parametrize("m,k,n", [
+    (96, 16, 64), (256, 672, 168), (1000, 72, 24), (12544, 160, 960),
+    (4096, 960, 160), (256, 1280, 1000), (98, 8, 16),
+])
+def test_pw_conv_matches_fp32(ext, m, k, n):
+    x = torch.randn(m, k, device=dev()).to(BF16)
+    w = torch.randn(n, k, device=dev()).to(BF16) * 0.1
+    y = torch.empty(m, n, device=dev(), dtype=BF16)
+    ext.pw_conv(x.view(m, k, 1, 1), w, y.view(m, n, 1, 1))
+    want = (x.float() @ w.float().t()).to(BF16)
+    assert_close(y, want)
+
+
+def test_pw_conv_identity_transpose_detect(ext):
+    """A = I with ASYMMETRIC B catches swapped output layouts."""
+    k = 64
+    x = torch.eye(k, device=dev()).to(BF16)
+    w = torch.arange(k * k, device=dev()).float().view(k, k)
+    w = (w / (k * k)).to(BF16)  # asymmetric
+    y = torch.empty(k, k, device=dev(), dtype=BF16)
+    ext.pw_conv(x.view(k, k, 1, 1), w, y.view(k, k, 1, 1))
+    assert torch.equal(y.float().cpu(), w.float().t().cpu())
+
+
+@pytest.mark.parametrize("m,n1,n2", [(512, 64, 32), (12544, 16, 96),
+                                     (50176, 64, 24), (256, 960, 240)])
+def test_pw_wgrad_matches_fp32(ext, m, n1, n2):
+    a = torch.randn(m, n1, device=dev()).to(BF16)
+    b = torch.randn(m, n2, device=dev()).to(BF16)
+    dw = torch.empty(n1, n2, device=dev(), dtype=torch.float32)
+    ext.pw_conv_wgrad(a.view(m, n1, 1, 1), b.view(m, n2, 1, 1), dw)
+    want = a.float().t() @ b.float()
+    assert_close(dw, want, rtol=1e-2, atol=1e-2 * max(m, 1) ** 0.5 * 0.05)
+
+
+def test_pw_wgrad_deterministic(ext):
+    m, n1, n2 = 30000, 48, 32
+    a = torch.randn(m, n1, device=dev()).to(BF16)
+    b = torch.randn(m, n2, device=dev()).to(BF16)
+    outs = []
+    for _ in range(2):
+        dw = torch.empty(n1, n2, device=dev(), dtype=torch.float32)
+        ext.pw_conv_wgrad(a.view(m, n1, 1, 1), b.view(m, n2, 1, 1), dw)
+        outs.append(dw.cpu())
+    assert torch.equal(outs[0], outs[1])
+
+
+# ------------------------------------------------------------- depthwise
+
+DW_SHAPES = [  # (n, c, h, k, s) — zoo extremes from SURVEY.md §2.5
+    (2, 16, 112, 3, 1), (2, 64, 112, 3, 2), (2, 72, 56, 5, 2),
+    (2, 120, 28, 5, 1), (2, 240, 28, 3, 2), (2, 672, 14, 5, 2),
+    (2, 960, 8, 5, 1), (3, 8, 9, 3, 1),
+]
+
+
+@pytest.mark.parametrize("n,c,h,k,s", DW_SHAPES)
+def test_dw_fwd(ext, n, c, h, k, s):
+    p = k // 2
+    x = randn_nhwc(n, c, h, h, seed=1)
+    w = (torch.randn(k * k, c, device=dev()) * 0.3).to(BF16)
+    ho = (h + 2 * p - k) // s + 1
+    y = torch.empty(n, c, ho, ho, device=dev(), dtype=BF16,
+                    memory_format=CL)
+    ext.dw_conv_fwd(x, w, y, k, s, p)
+    wref = w.float().t().reshape(c, 1, k, k)
+    want = F.conv2d(x.float(), wref, stride=s, padding=p, groups=c)
+    assert_close(y, want)
+
+
+@pytest.mark.parametrize("n,c,h,k,s", DW_SHAPES)
+def test_dw_dgrad(ext, n, c, h, k, s):
+    p = k // 2
+    ho = (h + 2 * p - k) // s + 1
+    dy = randn_nhwc(n, c, ho, ho, seed=2)
+    w = (torch.randn(k * k, c, device=dev()) * 0.3).to(BF16)
+    dx = torch.empty(n, c, h, h, device=dev(), dtype=BF16, memory_format=CL)
+    ext.dw_conv_dgrad(dy, w, dx, k, s, p)
+    wref = w.float().t().reshape(c, 1, k, k)
+    want = torch.nn.grad.conv2d_input(
+        (n, c, h, h), wref, dy.float(), stride=s, padding=p, groups=c)
+    assert_close(dx, want)
+
+
+@pytest.mark.parametrize("n,c,h,k,s", DW_SHAPES)
+def test_dw_wgrad(ext, n, c, h, k, s):
+    p = k // 2
+    ho = (h + 2 * p - k) // s + 1
+    x = randn_nhwc(n, c, h, h, seed=3)
+    dy = randn_nhwc(n, c, ho, ho, seed=4)
+    dw = torch.empty(k * k, c, device=dev(), dtype=torch.float32)
+    ext.dw_conv_wgrad(x, dy, dw, k, s, p)
+    want = torch.nn.grad.conv2d_weight(
+        x.float(), (c, 1, k, k), dy.float(), stride=s, padding=p, groups=c)
+    want = want.reshape(c, k * k).t()
+    assert_close(dw, want, atol=0.05 * (n * ho * ho) ** 0.5 * 0.1)
+
+
+# ------------------------------------------------------------------ stem
+
+@pytest.mark.parametrize("cout", [16, 24, 32])
+def test_stem_fwd(ext, cout):
+    n, h = 2, 32
+    x = randn_nhwc(n, 3, h, h, seed=5)
+    w = (torch.randn(3, 3, 3, cout, device=dev()) * 0.2).to(BF16)
+    ho = h // 2
+    y = torch.empty(n, cout, ho, ho, device=dev(), dtype=BF16,
+                    memory_format=CL)
+    ext.stem_conv_fwd(x, w, y, 3, 2, 1)
+    wref = w.float().permute(3, 2, 0, 1)  # [cout, cin, kh, kw]
+    want = F.conv2d(x.float(), wref, stride=2, padding=1)
+    assert_close(y, want)
+
+
+def test_stem_wgrad(ext):
+    n, h, cout = 2, 32, 32
+    x = randn_nhwc(n, 3, h, h, seed=6)
+    dy = randn_nhwc(n, cout, h // 2, h // 2, seed=7)
+    dw = torch.empty(27, cout, device=dev(), dtype=torch.float32)
+    ext.stem_conv_wgrad(x, dy, dw, 3, 2, 1)
+    want = torch.nn.grad.conv2d_weight(
+        x.float(), (cout, 3, 3, 3), dy.float(), stride=2, padding=1)
+    want = want.permute(2, 3, 1, 0).reshape(27, cout)
+    assert_close(dw, want, atol=0.05 * (n * (h // 2) ** 2) ** 0.5 * 0.1)
+
+
+# ---------------------------------------------------------------- BN+act
+
+def _bn_setup(n=4, c=72, h=14, seed=8):
+    z = randn_nhwc(n, c, h, h, seed=seed, scale=2.0)
+    gamma = (torch.rand(c, device=dev()) + 0.5).float()
+    beta = (torch.randn(c, device=dev()) * 0.2).float()
+    return z, gamma, beta
+
+
+def test_bn_stats(ext):
+    z, _, _ = _bn_setup()
+    c = z.shape[1]
+    mean = torch.empty(c, device=dev(), dtype=torch.float32)
+    invstd = torch.empty_like(mean)
+    rm = torch.zeros(c, device=dev())
+    rv = torch.ones(c, device=dev())
+    ext.bn_stats(z, mean, invstd, rm, rv, 0.1, 1e-5)
+    zf = z.float()
+    want_mean = zf.mean(dim=(0, 2, 3))
+    want_var = zf.var(dim=(0, 2, 3), unbiased=False)
+    assert_close(mean, want_mean, rtol=1e-3, atol=1e-3)
+    assert_close(invstd, (want_var + 1e-5).rsqrt(), rtol=1e-3, atol=1e-3)
+    m = z.shape[0] * z.shape[2] * z.shape[3]
+    assert_close(rm, 0.1 * want_mean, rtol=1e-3, atol=1e-4)
+    assert_close(rv, 0.9 + 0.1 * want_var * m / (m - 1), rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize("act", [0, 1, 2, 3])
+@pytest.mark.parametrize("residual", [False, True])
+def test_bn_act_fwd(ext, act, residual):
+    z, gamma, beta = _bn_setup()
+    zf = z.float()
+    mean = zf.mean(dim=(0, 2, 3))
+    invstd = (zf.var(dim=(0, 2, 3), unbiased=False) + 1e-5).rsqrt()
+    res = randn_nhwc(*z.shape, seed=9) if residual else None
+    y = torch.empty_like(z)
+    ext.bn_act_fwd(z, mean, invstd, gamma, beta, act, y, res)
+    v = (zf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
+    v = gamma.view(1, -1, 1, 1) * v + beta.view(1, -1, 1, 1)
+    want = [lambda t: t, F.relu, F.relu6, F.hardswish][act](v)
+    if residual:
+        want = want + res.float()
+    assert_close(y, want, rtol=1e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("act", [0, 2, 3])
+@pytest.mark.parametrize("training", [True, False])
+def test_bn_act_bwd(ext, act, training):
+    torch.manual_seed(0)
+    z, gamma, beta = _bn_setup(c=64)
+    n, c, h, w = z.shape
+    m = n * h * w
+    dy = randn_nhwc(n, c, h, w, seed=10)
+    zf = z.float()
+    mean = zf.mean(dim=(0, 2, 3))
+    invstd = (zf.var(dim=(0, 2, 3), unbiased=False) + 1e-5).rsqrt()
+    dgamma = torch.empty(c, device=dev(), dtype=torch.float32)
+    dbeta = torch.empty_like(dgamma)
+    dz = torch.empty_like(z)
+    ext.bn_act_bwd(dy, z, mean, invstd, gamma, beta, act, training,
+                   dgamma, dbeta, dz)
+    # autograd reference on the same bf16-rounded values
+    zr = zf.clone().requires_grad_(True)
+    mu = mean.view(1, -1, 1, 1)
+    iv = invstd.view(1, -1, 1, 1)
+    if training:
+        mu_r = zr.mean(dim=(0, 2, 3), keepdim=True)
+        var_r = zr.var(dim=(0, 2, 3), unbiased=False, keepdim=True)
+        zhat = (zr - mu_r) * (var_r + 1e-5).rsqrt()
+    else:
+        zhat = (zr - mu) * iv
+    gr = gamma.clone().requires_grad_(True)
+    br = beta.clone().requires_grad_(True)
+    v = gr.view(1, -1, 1, 1) * zhat + br.view(1, -1, 1, 1)
+    out = [lambda t: t, F.relu, F.relu6, F.hardswish][act](v)
+    out.backward(dy.float())
+    assert_close(dgamma, gr.grad, rtol=2e-2, atol=0.05 * m ** 0.5 * 0.02)
+    assert_close(dbeta, br.grad, rtol=2e-2, atol=0.05 * m ** 0.5 * 0.02)
+    assert_close(dz, zr.grad, rtol=5e-2, atol=5e-3)
+
+
+# ----------------------------------------------------------------- misc
+
+@pytest.mark.parametrize("act", [1, 4])
+def test_bias_act_fwd_bwd(ext, act):
+    m, c = 37, 64
+    z = torch.randn(m, c, device=dev()).to(BF16)
+    bias = torch.randn(c, device=dev()).float()
+    y = torch.empty_like(z)
+    ext.bias_act_fwd(z, bias, act, y)
+    v = z.float() + bias
+    want = F.relu(v) if act == 1 else F.hardsigmoid(v)
+    assert_close(y, want, rtol=1e-2, atol=1e-2)
+
+    dy = torch.randn(m, c, device=dev()).to(BF16)
+    dz = torch.empty_like(dy)
+    dbias = torch.empty(c, device=dev(), dtype=torch.float32)
+    ext.bias_act_bwd(dy, z, bias, act, dz, dbias)
+    vr = z.float().clone().requires_grad_(True)
+    br = bias.clone().requires_grad_(True)
+    out = F.relu(vr + br) if act == 1 else F.hardsigmoid(vr + br)
+    out.backward(dy.float())
+    assert_close(dz, vr.grad, rtol=1e-2, atol=1e-3)
+    assert_close(dbias, br.grad, rtol=1e-2, atol=1e-2)
+
+
+def test_rowscale_fwd_bwd(ext):
+    n, c, h = 3, 48, 14
+    x = randn_nhwc(n, c, h, h, seed=11)
+    s = torch.rand(n, c, device=dev()).to(BF16)
+    y = torch.empty_like(x)
+    ext.rowscale_fwd(x, s, y)
+    want = x.float() * s.float().view(n, c, 1, 1)
+    assert_close(y, want)
+
+    dy = randn_nhwc(n, c, h, h, seed=12)
+    dx = torch.empty_like(x)
+    ds = torch.empty(n, c, device=dev(), dtype=BF16)
+    ext.rowscale_bwd(dy, x, s, dx, ds)
+    assert_close(dx, dy.float() * s.float().view(n, c, 1, 1))
+    assert_close(ds, (dy.float() * x.float()).sum(dim=(2, 3)),
+                 rtol=2e-2, atol=0.1)
+
+
+def test_avgpool_fwd_bwd(ext):
+    n, c, h = 3, 96, 7
+    x = randn_nhwc(n, c, h, h, seed=13)
+    out = torch.empty(n, c, device=dev(), dtype=BF16)
+    ext.avgpool_fwd(x, out)
+    assert_close(out, x.float().mean(dim=(2, 3)), rtol=1e-2, atol=1e-2)
+
+    dp = torch.randn(n, c, device=dev()).to(BF16)
+    dx = torch.empty_like(x)
+    ext.avgpool_bwd(dp, dx)
+    want = (dp.float() / (h * h)).view(n, c, 1, 1).expand(n, c, h, h)
+    assert_close(dx, want)
+    dx2 = dx.clone()
+    ext.avgpool_bwd_add(dp, dx2)
+    assert_close(dx2, 2 * want, rtol=1e-2, atol=1e-3)
+
+
+def test_label_smooth_ce(ext):
+    n, c = 16, 1000
+    logits = (torch.randn(n, c, device=dev()) * 3).to(BF16)
+    target = torch.randint(0, c, (n,), device=dev())
+    loss = torch.empty((), device=dev(), dtype=torch.float32)
+    dlogits = torch.empty(n, c, device=dev(), dtype=torch.float32)
+    ext.label_smooth_ce(logits, target, 0.1, loss, dlogits)
+    lf = logits.float()
+    logp = F.log_softmax(lf, dim=1)
+    nll = -logp.gather(1, target.view(-1, 1)).squeeze(1)
+    smooth = -logp.mean(dim=1)
+    want = (0.9 * nll + 0.1 * smooth).mean()
+    assert_close(loss, want, rtol=1e-3, atol=1e-3)
+    lr = lf.clone().requires_grad_(True)
+    logp = F.log_softmax(lr, dim=1)
+    ((0.9 * (-logp.gather(1, target.view(-1, 1)).squeeze(1))
+      + 0.1 * (-logp.mean(dim=1))).mean()).backward()
+    assert_close(dlogits, lr.grad, rtol=1e-3, atol=1e-5)

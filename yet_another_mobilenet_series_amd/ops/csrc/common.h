@@ -1,0 +1,98 @@
+// Common device helpers for the gfx950 (CDNA4) kernel engine.
+//
+// Conventions (see ops/hip_ops.py for the host-side contract):
+//  - activations: NHWC rows, C fastest, bf16 storage, fp32 math
+//  - per-channel statistics / parameter tensors: fp32
+//  - wave size 64 (CDNA), block sizes multiples of 64
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define DEV_INLINE __device__ __forceinline__
+
+typedef unsigned short ushort_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;   // MFMA A/B frag (4 VGPRs)
+typedef __attribute__((ext_vector_type(4))) short short4_t; // 8-byte vector load
+typedef __attribute__((ext_vector_type(8))) short short8_t; // 16-byte vector load
+typedef __attribute__((ext_vector_type(4))) float float4_t;
+
+// ---- bf16 <-> f32 (bf16 = upper 16 bits of f32; RNE on the way down) ----
+DEV_INLINE float bf2f(ushort_t u) {
+  union { uint32_t i; float f; } v;
+  v.i = uint32_t(u) << 16;
+  return v.f;
+}
+
+DEV_INLINE ushort_t f2bf(float f) {
+  union { float f; uint32_t i; } v;
+  v.f = f;
+  uint32_t i = v.i;
+  // round-to-nearest-even; NaN quietens
+  if ((i & 0x7fffffffu) > 0x7f800000u) return ushort_t((i >> 16) | 0x40);
+  i += 0x7fffu + ((i >> 16) & 1u);
+  return ushort_t(i >> 16);
+}
+
+// ---- activations (codes shared with ops/hip_ops.py) ----
+#define ACT_IDENTITY 0
+#define ACT_RELU 1
+#define ACT_RELU6 2
+#define ACT_HSWISH 3
+#define ACT_HSIGMOID 4
+
+template <int ACT>
+DEV_INLINE float act_fwd(float v) {
+  if (ACT == ACT_RELU) return fmaxf(v, 0.f);
+  if (ACT == ACT_RELU6) return fminf(fmaxf(v, 0.f), 6.f);
+  if (ACT == ACT_HSWISH) return v * fminf(fmaxf(v + 3.f, 0.f), 6.f) * (1.f / 6.f);
+  if (ACT == ACT_HSIGMOID) return fminf(fmaxf(v + 3.f, 0.f), 6.f) * (1.f / 6.f);
+  return v;
+}
+
+// derivative w.r.t. the pre-activation value v
+template <int ACT>
+DEV_INLINE float act_bwd(float v) {
+  if (ACT == ACT_RELU) return v > 0.f ? 1.f : 0.f;
+  if (ACT == ACT_RELU6) return (v > 0.f && v < 6.f) ? 1.f : 0.f;
+  if (ACT == ACT_HSWISH) {
+    if (v <= -3.f) return 0.f;
+    if (v >= 3.f) return 1.f;
+    return (2.f * v + 3.f) * (1.f / 6.f);
+  }
+  if (ACT == ACT_HSIGMOID) return (v > -3.f && v < 3.f) ? (1.f / 6.f) : 0.f;
+  return 1.f;
+}
+
+// dispatch an `auto fn = [&](auto act_tag){...}` over a runtime act code
+#define ACT_SWITCH(ACT_CODE, ...)                                      \
+  switch (ACT_CODE) {                                                  \
+    case ACT_IDENTITY: { constexpr int kAct = ACT_IDENTITY; __VA_ARGS__; break; } \
+    case ACT_RELU:     { constexpr int kAct = ACT_RELU;     __VA_ARGS__; break; } \
+    case ACT_RELU6:    { constexpr int kAct = ACT_RELU6;    __VA_ARGS__; break; } \
+    case ACT_HSWISH:   { constexpr int kAct = ACT_HSWISH;   __VA_ARGS__; break; } \
+    case ACT_HSIGMOID: { constexpr int kAct = ACT_HSIGMOID; __VA_ARGS__; break; } \
+  }
+
+DEV_INLINE int64_t idx4(int n, int h, int w, int c, int H, int W, int C) {
+  return ((int64_t(n) * H + h) * W + w) * C + c;
+}
+
+constexpr int ceil_div(int64_t a, int64_t b) { return int((a + b - 1) / b); }
+
+// Wave-level f32 sum over all 64 lanes (butterfly shuffles).
+DEV_INLINE float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+#define HIP_KERNEL_ASSERT_OK(expr)                                   \
+  do {                                                               \
+    hipError_t _e = (expr);                                          \
+    if (_e != hipSuccess) {                                          \
+      printf("HIP error %s at %s:%d\n", hipGetErrorString(_e),       \
+             __FILE__, __LINE__);                                    \
+    }                                                                \
+  } while (0)

@@ -1,0 +1,338 @@
+// Depthwise conv (K3/K4) and stem dense conv (K1) for gfx950, NHWC bf16.
+//
+// NHWC is the natural depthwise layout on CDNA4: channels are the fastest
+// axis, so a wave's 64 lanes reading one spatial position across channels
+// is one coalesced transaction, and the per-channel filter taps broadcast
+// from L1.  Forward/dgrad are flat vec8-per-thread elementwise kernels
+// (memory-bound; the k*k window reuse is served by L1/L2); wgrad reduces
+// deterministically via per-(chunk,mlane) fp32 slabs + slab_reduce.
+
+#include <algorithm>
+
+#include "common.h"
+#include "engine.h"
+
+namespace yams {
+
+void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
+                 hipStream_t s);
+
+namespace {
+
+constexpr int MAX_BLOCKS = 4096;
+
+template <int K>
+__global__ __launch_bounds__(256) void dw_fwd_kernel(
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
+    ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
+    int stride, int pad, int64_t total8) {
+  const int C8 = C >> 3;
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c8 = int(idx % C8);
+    int64_t rest = idx / C8;
+    const int wo = int(rest % Wo);
+    rest /= Wo;
+    const int ho = int(rest % Ho);
+    const int n = int(rest / Ho);
+    const int c = c8 * 8;
+    float acc[8] = {};
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int hi = ho * stride + i - pad;
+      if (hi < 0 || hi >= H) continue;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        const int wi = wo * stride + j - pad;
+        if (wi < 0 || wi >= W) continue;
+        short8_t xv = *reinterpret_cast<const short8_t*>(
+            &x[idx4(n, hi, wi, c, H, W, C)]);
+        short8_t wv = *reinterpret_cast<const short8_t*>(
+            &w[(i * K + j) * C + c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[e] = fmaf(bf2f(ushort_t(xv[e])), bf2f(ushort_t(wv[e])), acc[e]);
+      }
+    }
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
+    *reinterpret_cast<short8_t*>(&y[idx4(n, ho, wo, c, Ho, Wo, C)]) = out;
+  }
+}
+
+template <int K>
+__global__ __launch_bounds__(256) void dw_dgrad_kernel(
+    const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
+    ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
+    int stride, int pad, int64_t total8) {
+  const int C8 = C >> 3;
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c8 = int(idx % C8);
+    int64_t rest = idx / C8;
+    const int wi = int(rest % W);
+    rest /= W;
+    const int hi = int(rest % H);
+    const int n = int(rest / H);
+    const int c = c8 * 8;
+    float acc[8] = {};
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int hnum = hi + pad - i;
+      if (hnum < 0 || hnum % stride) continue;
+      const int ho = hnum / stride;
+      if (ho >= Ho) continue;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        const int wnum = wi + pad - j;
+        if (wnum < 0 || wnum % stride) continue;
+        const int wo = wnum / stride;
+        if (wo >= Wo) continue;
+        short8_t gv = *reinterpret_cast<const short8_t*>(
+            &dy[idx4(n, ho, wo, c, Ho, Wo, C)]);
+        short8_t wv = *reinterpret_cast<const short8_t*>(
+            &w[(i * K + j) * C + c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[e] = fmaf(bf2f(ushort_t(gv[e])), bf2f(ushort_t(wv[e])), acc[e]);
+      }
+    }
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
+    *reinterpret_cast<short8_t*>(&dx[idx4(n, hi, wi, c, H, W, C)]) = out;
+  }
+}
+
+// wgrad: each thread owns one channel lane and k*k fp32 accumulators over a
+// chunk of output positions; each (chunk, mlane) writes its own slab row.
+template <int K>
+__global__ __launch_bounds__(256) void dw_wgrad_kernel(
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+    float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
+    int stride, int pad, int chunks) {
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  const int c0 = (blockIdx.x % c_tiles) * CT;
+  const int chunk = blockIdx.x / c_tiles;
+  const int c = c0 + int(threadIdx.x) % CT;
+  const int mlane = int(threadIdx.x) / CT;
+  if (c >= C) return;
+
+  const int64_t P = int64_t(N) * Ho * Wo;
+  const int64_t per = (P + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi_ = std::min<int64_t>(lo + per, P);
+
+  float acc[K * K] = {};
+  for (int64_t p = lo + mlane; p < hi_; p += ML) {
+    const int wo = int(p % Wo);
+    int64_t rest = p / Wo;
+    const int ho = int(rest % Ho);
+    const int n = int(rest / Ho);
+    const float g = bf2f(dy[idx4(n, ho, wo, c, Ho, Wo, C)]);
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      const int hi2 = ho * stride + i - pad;
+      if (hi2 < 0 || hi2 >= H) continue;
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        const int wi = wo * stride + j - pad;
+        if (wi < 0 || wi >= W) continue;
+        acc[i * K + j] =
+            fmaf(g, bf2f(x[idx4(n, hi2, wi, c, H, W, C)]), acc[i * K + j]);
+      }
+    }
+  }
+  float* slab = ws + int64_t(chunk * ML + mlane) * K * K * C;
+#pragma unroll
+  for (int t = 0; t < K * K; ++t) slab[t * C + c] = acc[t];
+}
+
+// stem forward: flat vec8 over output channels; 27 taps, x broadcast from L1
+__global__ __launch_bounds__(256) void stem_fwd_kernel(
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
+    ushort_t* __restrict__ y, int N, int H, int W, int Cout, int Ho, int Wo,
+    int k, int stride, int pad, int64_t total8) {
+  const int C8 = Cout >> 3;
+  const int Cin = 3;
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c8 = int(idx % C8);
+    int64_t rest = idx / C8;
+    const int wo = int(rest % Wo);
+    rest /= Wo;
+    const int ho = int(rest % Ho);
+    const int n = int(rest / Ho);
+    const int co = c8 * 8;
+    float acc[8] = {};
+    for (int i = 0; i < k; ++i) {
+      const int hi = ho * stride + i - pad;
+      if (hi < 0 || hi >= H) continue;
+      for (int j = 0; j < k; ++j) {
+        const int wi = wo * stride + j - pad;
+        if (wi < 0 || wi >= W) continue;
+#pragma unroll
+        for (int ci = 0; ci < 3; ++ci) {
+          const float xv = bf2f(x[idx4(n, hi, wi, ci, H, W, Cin)]);
+          short8_t wv = *reinterpret_cast<const short8_t*>(
+              &w[((i * k + j) * Cin + ci) * Cout + co]);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc[e] = fmaf(xv, bf2f(ushort_t(wv[e])), acc[e]);
+        }
+      }
+    }
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
+    *reinterpret_cast<short8_t*>(&y[idx4(n, ho, wo, co, Ho, Wo, Cout)]) = out;
+  }
+}
+
+// stem wgrad: thread = cout lane, k*k*3 fp32 accumulators
+__global__ __launch_bounds__(256) void stem_wgrad_kernel(
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+    float* __restrict__ ws, int N, int H, int W, int Cout, int Ho, int Wo,
+    int k, int stride, int pad, int chunks) {
+  const int Cin = 3;
+  const int CT = Cout < 256 ? Cout : 256;
+  const int ML = 256 / CT;
+  const int co = int(threadIdx.x) % CT;
+  const int mlane = int(threadIdx.x) / CT;
+  const int chunk = blockIdx.x;
+  if (co >= Cout) return;
+
+  const int64_t P = int64_t(N) * Ho * Wo;
+  const int64_t per = (P + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi_ = std::min<int64_t>(lo + per, P);
+
+  float acc[27] = {};  // k=3 fixed for the zoo stem
+  for (int64_t p = lo + mlane; p < hi_; p += ML) {
+    const int wo = int(p % Wo);
+    int64_t rest = p / Wo;
+    const int ho = int(rest % Ho);
+    const int n = int(rest / Ho);
+    const float g = bf2f(dy[idx4(n, ho, wo, co, Ho, Wo, Cout)]);
+    for (int i = 0; i < k; ++i) {
+      const int hi2 = ho * stride + i - pad;
+      if (hi2 < 0 || hi2 >= H) continue;
+      for (int j = 0; j < k; ++j) {
+        const int wi = wo * stride + j - pad;
+        if (wi < 0 || wi >= W) continue;
+#pragma unroll
+        for (int ci = 0; ci < 3; ++ci)
+          acc[(i * k + j) * Cin + ci] =
+              fmaf(g, bf2f(x[idx4(n, hi2, wi, ci, H, W, Cin)]),
+                   acc[(i * k + j) * Cin + ci]);
+      }
+    }
+  }
+  float* slab = ws + int64_t(chunk * ML + mlane) * (k * k * Cin) * Cout;
+  for (int t = 0; t < k * k * Cin; ++t) slab[t * Cout + co] = acc[t];
+}
+
+int grid_for(int64_t total) {
+  return int(std::min<int64_t>((total + 255) / 256, MAX_BLOCKS));
+}
+
+}  // namespace
+
+void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
+                 int C, int Ho, int Wo, int k, int stride, int pad,
+                 hipStream_t s) {
+  const int64_t total8 = int64_t(N) * Ho * Wo * (C >> 3);
+  auto* xp = static_cast<const ushort_t*>(x);
+  auto* wp = static_cast<const ushort_t*>(w);
+  auto* yp = static_cast<ushort_t*>(y);
+  if (k == 3)
+    dw_fwd_kernel<3><<<grid_for(total8), 256, 0, s>>>(
+        xp, wp, yp, N, H, W, C, Ho, Wo, stride, pad, total8);
+  else
+    dw_fwd_kernel<5><<<grid_for(total8), 256, 0, s>>>(
+        xp, wp, yp, N, H, W, C, Ho, Wo, stride, pad, total8);
+}
+
+void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
+                   int W, int C, int Ho, int Wo, int k, int stride, int pad,
+                   hipStream_t s) {
+  const int64_t total8 = int64_t(N) * H * W * (C >> 3);
+  auto* gp = static_cast<const ushort_t*>(dy);
+  auto* wp = static_cast<const ushort_t*>(w);
+  auto* xp = static_cast<ushort_t*>(dx);
+  if (k == 3)
+    dw_dgrad_kernel<3><<<grid_for(total8), 256, 0, s>>>(
+        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
+  else
+    dw_dgrad_kernel<5><<<grid_for(total8), 256, 0, s>>>(
+        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
+}
+
+int dw_wgrad_chunks(int N, int Ho, int Wo) {
+  const int64_t P = int64_t(N) * Ho * Wo;
+  int c = int(std::min<int64_t>((P + 511) / 512, 256));
+  return c < 1 ? 1 : c;
+}
+
+int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k) {
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  return int64_t(dw_wgrad_chunks(N, Ho, Wo)) * ML * k * k * C;
+}
+
+void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
+                   int N, int H, int W, int C, int Ho, int Wo, int k,
+                   int stride, int pad, hipStream_t s) {
+  const int chunks = dw_wgrad_chunks(N, Ho, Wo);
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  auto* xp = static_cast<const ushort_t*>(x);
+  auto* gp = static_cast<const ushort_t*>(dy);
+  if (k == 3)
+    dw_wgrad_kernel<3><<<dim3(c_tiles * chunks), 256, 0, s>>>(
+        xp, gp, ws, N, H, W, C, Ho, Wo, stride, pad, chunks);
+  else
+    dw_wgrad_kernel<5><<<dim3(c_tiles * chunks), 256, 0, s>>>(
+        xp, gp, ws, N, H, W, C, Ho, Wo, stride, pad, chunks);
+  slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks * ML, s);
+}
+
+void stem_conv_fwd(const void* x, const void* w, void* y, int N, int H,
+                   int W, int Cout, int Ho, int Wo, int k, int stride,
+                   int pad, hipStream_t s) {
+  const int64_t total8 = int64_t(N) * Ho * Wo * (Cout >> 3);
+  stem_fwd_kernel<<<grid_for(total8), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
+      static_cast<ushort_t*>(y), N, H, W, Cout, Ho, Wo, k, stride, pad,
+      total8);
+}
+
+int stem_wgrad_chunks(int N, int Ho, int Wo) {
+  const int64_t P = int64_t(N) * Ho * Wo;
+  int c = int(std::min<int64_t>((P + 511) / 512, 512));
+  return c < 1 ? 1 : c;
+}
+
+int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k) {
+  const int CT = Cout < 256 ? Cout : 256;
+  const int ML = 256 / CT;
+  return int64_t(stem_wgrad_chunks(N, Ho, Wo)) * ML * k * k * 3 * Cout;
+}
+
+void stem_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
+                     int N, int H, int W, int Cout, int Ho, int Wo, int k,
+                     int stride, int pad, hipStream_t s) {
+  const int chunks = stem_wgrad_chunks(N, Ho, Wo);
+  const int CT = Cout < 256 ? Cout : 256;
+  const int ML = 256 / CT;
+  stem_wgrad_kernel<<<dim3(chunks), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(dy), ws,
+      N, H, W, Cout, Ho, Wo, k, stride, pad, chunks);
+  slab_reduce(ws, dwgt, int64_t(k) * k * 3 * Cout, chunks * ML, s);
+}
+
+}  // namespace yams

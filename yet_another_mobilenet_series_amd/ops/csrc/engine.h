@@ -1,0 +1,108 @@
+// Kernel launcher declarations: the C ABI between bindings.cpp (compiled
+// against torch headers) and the gfx950 kernels in *.hip.
+//
+// Pointer conventions: `bf16*` passed as const void*/void*; fp32 as float*.
+// All launches are asynchronous on the given hipStream_t.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace yams {
+
+// --- GEMM family (pointwise conv K2, FC K10) ----------------------------
+// y[M,N] = x[M,K] @ w[N,K]^T   (both operands K-major; w is the natural
+// [Cout, Cin] torch layout for forward, its transpose for dgrad)
+void pw_conv(const void* x, const void* w, void* y,
+             int64_t M, int K, int N, hipStream_t s);
+
+// dw[N1,N2] = a[M,N1]^T @ b[M,N2], fp32 out, deterministic split-M slabs
+// workspace: caller-allocated fp32 of size splits(M)*N1*N2 (see
+// pw_wgrad_workspace_elems)
+void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
+                   int64_t M, int N1, int N2, hipStream_t s);
+int pw_wgrad_splits(int64_t M);
+int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2);
+
+// out[i] = sum_s ws[s*elems + i] (deterministic slab combine)
+void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
+                 hipStream_t s);
+
+// --- depthwise conv (K3/K4) --------------------------------------------
+// w layout [k*k, C] (C fastest)
+void dw_conv_fwd(const void* x, const void* w, void* y,
+                 int N, int H, int W, int C, int Ho, int Wo,
+                 int k, int stride, int pad, hipStream_t s);
+void dw_conv_dgrad(const void* dy, const void* w, void* dx,
+                   int N, int H, int W, int C, int Ho, int Wo,
+                   int k, int stride, int pad, hipStream_t s);
+// dwgt fp32 [k*k, C]; ws fp32 [chunks, k*k, C]
+void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
+                   int N, int H, int W, int C, int Ho, int Wo,
+                   int k, int stride, int pad, hipStream_t s);
+int dw_wgrad_chunks(int N, int Ho, int Wo);
+int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k);
+
+// --- stem dense conv (K1): cin=3, k=3, s=2; w layout [3,3,3,Cout] -------
+void stem_conv_fwd(const void* x, const void* w, void* y,
+                   int N, int H, int W, int Cout, int Ho, int Wo,
+                   int k, int stride, int pad, hipStream_t s);
+void stem_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
+                     int N, int H, int W, int Cout, int Ho, int Wo,
+                     int k, int stride, int pad, hipStream_t s);
+int stem_wgrad_chunks(int N, int Ho, int Wo);
+int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k);
+
+// --- BatchNorm (K5) -----------------------------------------------------
+// training stats over M=N*H*W rows of z[M,C]; ws fp32 [2, chunks, C]
+void bn_stats(const void* z, float* mean, float* invstd,
+              float* running_mean, float* running_var, float* ws,
+              int64_t M, int C, float momentum, float eps, hipStream_t s);
+int bn_stats_chunks(int64_t M);
+int64_t bn_stats_ws_elems(int64_t M, int C);
+
+// y = act(gamma*(z-mean)*invstd + beta) [+ residual]
+void bn_act_fwd(const void* z, const float* mean, const float* invstd,
+                const float* gamma, const float* beta, int act,
+                void* y, const void* residual, int64_t M, int C,
+                hipStream_t s);
+
+// backward: dgamma/dbeta + dz; ws fp32 [2, chunks, C]
+void bn_act_bwd(const void* dy, const void* z, const float* mean,
+                const float* invstd, const float* gamma, const float* beta,
+                int act, bool training, float* dgamma, float* dbeta,
+                void* dz, float* ws, int64_t M, int C, hipStream_t s);
+
+// --- small fused elementwise/reduction kernels --------------------------
+// y = act(z + bias) on [M, C] (bias nullable)
+void bias_act_fwd(const void* z, const float* bias, int act, void* y,
+                  int64_t M, int C, hipStream_t s);
+// dz = dy * act'(z+bias); dbias = sum_m dz (nullable); ws [chunks, C]
+void bias_act_bwd(const void* dy, const void* z, const float* bias, int act,
+                  void* dz, float* dbias, float* ws, int64_t M, int C,
+                  hipStream_t s);
+int bias_bwd_chunks(int64_t M);
+int64_t bias_bwd_ws_elems(int64_t M, int C);
+
+// y[n,hw,c] = x[n,hw,c] * scale[n,c]
+void rowscale_fwd(const void* x, const void* scale, void* y,
+                  int N, int HW, int C, hipStream_t s);
+// dx = dy*scale ; dscale[n,c] = sum_hw dy*x
+void rowscale_bwd(const void* dy, const void* x, const void* scale,
+                  void* dx, void* dscale, int N, int HW, int C,
+                  hipStream_t s);
+
+// out[n,c] = mean_hw x[n,hw,c]
+void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
+                 hipStream_t s);
+// dx[n,hw,c] = dpooled[n,c]/HW   (accumulate=true adds into dx)
+void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,
+                 bool accumulate, hipStream_t s);
+
+// --- loss (K11): label-smoothed softmax CE, mean over batch -------------
+// loss: fp32 scalar; dlogits fp32 [N, C]; rowloss ws fp32 [N]
+void label_smooth_ce(const void* logits, const int64_t* target,
+                     float smoothing, float* loss, float* dlogits,
+                     float* rowloss, int N, int C, hipStream_t s);
+
+}  // namespace yams

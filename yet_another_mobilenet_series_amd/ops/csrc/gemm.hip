@@ -1,0 +1,267 @@
+// MFMA bf16 GEMM family for gfx950: the pointwise-conv (K2) / FC (K10)
+// compute path, plus the deterministic split-M wgrad GEMM.
+//
+// pw_conv:  y[M,N] = x[M,K] @ w[N,K]^T
+//   - implicit GEMM over NHWC rows: M = N*H*W, K = Cin, N = Cout
+//   - both operands K-major, so A-frag and B-frag LDS reads are contiguous
+//     16-byte ds_read_b128 (no transpose staging)
+//   - v_mfma_f32_16x16x32_bf16, fp32 accumulate; masked edges so any
+//     (M, K, N) with K % 8 == 0 works
+//
+// pw_conv_wgrad: dw[N1,N2] = a[M,N1]^T @ b[M,N2]
+//   - M (= batch*spatial) is the reduction axis; fixed split-M partial
+//     slabs + a separate reduce kernel: deterministic, no atomics
+//     (SURVEY.md §5.2).
+
+#include <algorithm>
+
+#include "common.h"
+#include "engine.h"
+
+namespace yams {
+
+// ---------------------------------------------------------------------------
+// pw_conv: 128x64 tile, BK=32, 4 waves as 2(M)x2(N), 64x32 per wave.
+// ---------------------------------------------------------------------------
+namespace {
+
+constexpr int BM = 128, BN = 64, BK = 32;
+constexpr int LDA = BK + 8;  // +16B pad: conflict-free b128 column reads
+constexpr int FRAG_M = 4, FRAG_N = 2;  // 16x16 frags per wave
+
+__global__ __launch_bounds__(256) void pw_conv_kernel(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
+    ushort_t* __restrict__ C, int64_t M, int K, int N) {
+  __shared__ ushort_t Alds[BM * LDA];
+  __shared__ ushort_t Blds[BN * LDA];
+
+  const int tid = threadIdx.x;
+  const int64_t bm = int64_t(blockIdx.x) * BM;
+  const int bn = blockIdx.y * BN;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wrow0 = (wave >> 1) * 64, wcol0 = (wave & 1) * 32;
+  const int fr = lane & 15, fk = (lane >> 4) * 8;
+
+  f32x4 acc[FRAG_M][FRAG_N];
+#pragma unroll
+  for (int i = 0; i < FRAG_M; ++i)
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int a_row = tid >> 1, a_seg = (tid & 1) * 16;   // 2 thr/row x 128
+  const int b_row = tid >> 2, b_seg = (tid & 3) * 8;    // 4 thr/row x 64
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // stage A tile [BM][BK]
+    {
+      const int64_t gm = bm + a_row;
+      const int gk = k0 + a_seg;
+      ushort_t* dst = &Alds[a_row * LDA + a_seg];
+      if (gm < M && gk + 16 <= K) {
+        const ushort_t* src = &A[gm * K + gk];
+        *reinterpret_cast<short8_t*>(dst) =
+            *reinterpret_cast<const short8_t*>(src);
+        *reinterpret_cast<short8_t*>(dst + 8) =
+            *reinterpret_cast<const short8_t*>(src + 8);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          dst[j] = (gm < M && gk + j < K) ? A[gm * K + gk + j] : ushort_t(0);
+      }
+    }
+    // stage B tile [BN][BK] from Bt[N][K]
+    {
+      const int gn = bn + b_row;
+      const int gk = k0 + b_seg;
+      ushort_t* dst = &Blds[b_row * LDA + b_seg];
+      if (gn < N && gk + 8 <= K) {
+        *reinterpret_cast<short4_t*>(dst) =
+            *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gk]);
+        *reinterpret_cast<short4_t*>(dst + 4) =
+            *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gk + 4]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dst[j] = (gn < N && gk + j < K)
+                       ? Bt[int64_t(gn) * K + gk + j] : ushort_t(0);
+      }
+    }
+    __syncthreads();
+
+    bf16x8 a_frag[FRAG_M], b_frag[FRAG_N];
+#pragma unroll
+    for (int i = 0; i < FRAG_M; ++i)
+      a_frag[i] = *reinterpret_cast<const bf16x8*>(
+          &Alds[(wrow0 + i * 16 + fr) * LDA + fk]);
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j)
+      b_frag[j] = *reinterpret_cast<const bf16x8*>(
+          &Blds[(wcol0 + j * 16 + fr) * LDA + fk]);
+#pragma unroll
+    for (int i = 0; i < FRAG_M; ++i)
+#pragma unroll
+      for (int j = 0; j < FRAG_N; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // epilogue: C/D frag layout col = lane&15, row = (lane>>4)*4 + r
+  const int col_l = fr, row_l4 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FRAG_M; ++i) {
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j) {
+      const int col = bn + wcol0 + j * 16 + col_l;
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
+        if (row < M) C[row * N + col] = f2bf(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void pw_conv(const void* x, const void* w, void* y,
+             int64_t M, int K, int N, hipStream_t s) {
+  dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
+  pw_conv_kernel<<<grid, 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
+      static_cast<ushort_t*>(y), M, K, N);
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dw[N1,N2] = a[M,N1]^T @ b[M,N2] via split-M fp32 slabs.
+// 64x64 tile per block, 16x16 threads x (4x4 per thread), VALU fp32 FMA
+// (the reduction axis streams through once — memory-bound, MFMA not needed).
+// ---------------------------------------------------------------------------
+namespace {
+
+constexpr int WT = 64;        // output tile side
+constexpr int MT = 32;        // m rows per LDS pass
+constexpr int WLD = WT + 4;   // pad 8B
+
+__global__ __launch_bounds__(256) void atb_kernel(
+    const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
+    float* __restrict__ ws, int64_t M, int N1, int N2, int splits) {
+  __shared__ ushort_t Al[MT * WLD];
+  __shared__ ushort_t Bl[MT * WLD];
+
+  const int tid = threadIdx.x;
+  const int t1 = (tid & 15) * 4;   // rows of dw (N1 axis)
+  const int t2 = (tid >> 4) * 4;   // cols of dw (N2 axis)
+  const int n1_tiles = ceil_div(N1, WT);
+  const int tile = blockIdx.x % (n1_tiles * ceil_div(N2, WT));
+  const int split = blockIdx.x / (n1_tiles * ceil_div(N2, WT));
+  const int n1_0 = (tile % n1_tiles) * WT;
+  const int n2_0 = (tile / n1_tiles) * WT;
+
+  const int64_t m_per = (M + splits - 1) / splits;
+  const int64_t m_lo = split * m_per;
+  const int64_t m_hi = m_lo + m_per < M ? m_lo + m_per : M;
+
+  float acc[4][4] = {};
+  const int s_row = tid >> 3, s_seg = (tid & 7) * 8;  // 8 thr/row x 32 rows
+
+  for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
+    const int64_t gm = m0 + s_row;
+    {
+      ushort_t* dst = &Al[s_row * WLD + s_seg];
+      const int gn = n1_0 + s_seg;
+      if (gm < m_hi && gn + 8 <= N1) {
+        *reinterpret_cast<short8_t*>(dst) =
+            *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dst[j] = (gm < m_hi && gn + j < N1) ? A[gm * N1 + gn + j]
+                                              : ushort_t(0);
+      }
+    }
+    {
+      ushort_t* dst = &Bl[s_row * WLD + s_seg];
+      const int gn = n2_0 + s_seg;
+      if (gm < m_hi && gn + 8 <= N2) {
+        *reinterpret_cast<short8_t*>(dst) =
+            *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dst[j] = (gm < m_hi && gn + j < N2) ? B[gm * N2 + gn + j]
+                                              : ushort_t(0);
+      }
+    }
+    __syncthreads();
+
+    const int mm_end = (m_hi - m0) < MT ? int(m_hi - m0) : MT;
+    for (int mm = 0; mm < mm_end; ++mm) {
+      short4_t av = *reinterpret_cast<const short4_t*>(&Al[mm * WLD + t1]);
+      short4_t bv = *reinterpret_cast<const short4_t*>(&Bl[mm * WLD + t2]);
+      float a0 = bf2f(ushort_t(av.x)), a1 = bf2f(ushort_t(av.y)),
+            a2 = bf2f(ushort_t(av.z)), a3 = bf2f(ushort_t(av.w));
+      float b0 = bf2f(ushort_t(bv.x)), b1 = bf2f(ushort_t(bv.y)),
+            b2 = bf2f(ushort_t(bv.z)), b3 = bf2f(ushort_t(bv.w));
+      float a[4] = {a0, a1, a2, a3}, b[4] = {b0, b1, b2, b3};
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
+    }
+    __syncthreads();
+  }
+
+  float* out = ws + int64_t(split) * N1 * N2;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int n1 = n1_0 + t1 + i;
+    if (n1 >= N1) continue;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int n2 = n2_0 + t2 + j;
+      if (n2 < N2) out[int64_t(n1) * N2 + n2] = acc[i][j];
+    }
+  }
+}
+
+__global__ void slab_reduce_kernel(const float* __restrict__ ws,
+                                   float* __restrict__ out, int64_t elems,
+                                   int splits) {
+  for (int64_t i = int64_t(blockIdx.x) * blockDim.x + threadIdx.x; i < elems;
+       i += int64_t(gridDim.x) * blockDim.x) {
+    float v = 0.f;
+    for (int sp = 0; sp < splits; ++sp) v += ws[int64_t(sp) * elems + i];
+    out[i] = v;
+  }
+}
+
+}  // namespace
+
+void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
+                 hipStream_t s) {
+  const int blocks = int(std::min<int64_t>((elems + 255) / 256, 2048));
+  slab_reduce_kernel<<<dim3(blocks), 256, 0, s>>>(ws, out, elems, splits);
+}
+
+int pw_wgrad_splits(int64_t M) {
+  int s = int((M + 2047) / 2048);
+  return s < 1 ? 1 : (s > 64 ? 64 : s);
+}
+
+int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2) {
+  return int64_t(pw_wgrad_splits(M)) * N1 * N2;
+}
+
+void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
+                   int64_t M, int N1, int N2, hipStream_t s) {
+  const int splits = pw_wgrad_splits(M);
+  const int tiles = ceil_div(N1, WT) * ceil_div(N2, WT);
+  atb_kernel<<<dim3(tiles * splits), 256, 0, s>>>(
+      static_cast<const ushort_t*>(a), static_cast<const ushort_t*>(b), ws,
+      M, N1, N2, splits);
+  slab_reduce(ws, dw, int64_t(N1) * N2, splits, s);
+}
+
+}  // namespace yams

@@ -1,0 +1,330 @@
+// Small fused kernels: bias+activation (head/SE FCs), SE channel scale
+// (K7 pieces), global average pool (K9), label-smoothed softmax CE (K11).
+// All NHWC/bf16 with fp32 math; reductions are deterministic (fixed-order
+// LDS/slab combines, no atomics).
+
+#include <algorithm>
+
+#include "common.h"
+#include "engine.h"
+
+namespace yams {
+
+namespace {
+
+int elem_grid(int64_t total) {
+  return int(std::min<int64_t>((total + 255) / 256, 4096));
+}
+
+// --- bias + act on [M, C] (vec8) ---------------------------------------
+template <int ACT>
+__global__ __launch_bounds__(256) void bias_act_fwd_kernel(
+    const ushort_t* __restrict__ z, const float* __restrict__ bias,
+    ushort_t* __restrict__ y, int64_t total8, int C8) {
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c = int(idx % C8) * 8;
+    short8_t zv = *reinterpret_cast<const short8_t*>(&z[idx * 8]);
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = bf2f(ushort_t(zv[e]));
+      if (bias != nullptr) v += bias[c + e];
+      out[e] = short(f2bf(act_fwd<ACT>(v)));
+    }
+    *reinterpret_cast<short8_t*>(&y[idx * 8]) = out;
+  }
+}
+
+// --- bias_act backward: dz everywhere + per-channel dbias slabs ---------
+template <int ACT>
+__global__ __launch_bounds__(256) void bias_act_bwd_kernel(
+    const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
+    const float* __restrict__ bias, ushort_t* __restrict__ dz,
+    float* __restrict__ ws, int64_t M, int C, int chunks) {
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
+  const int chunk = blockIdx.x / c_tiles;
+  const int mlane = int(threadIdx.x) / CT;
+  if (c >= C) return;
+  const float b = bias != nullptr ? bias[c] : 0.f;
+  const int64_t per = (M + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi = std::min<int64_t>(lo + per, M);
+  float s = 0.f;
+  for (int64_t m = lo + mlane; m < hi; m += ML) {
+    const float v0 = bf2f(z[m * C + c]) + b;
+    const float d = bf2f(dy[m * C + c]) * act_bwd<ACT>(v0);
+    dz[m * C + c] = f2bf(d);
+    s += d;
+  }
+  if (ws != nullptr) ws[int64_t(chunk * ML + mlane) * C + c] = s;
+}
+
+// --- SE channel scale: y[n,hw,c] = x * s[n,c] ---------------------------
+__global__ __launch_bounds__(256) void rowscale_fwd_kernel(
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ scale,
+    ushort_t* __restrict__ y, int64_t total8, int HW, int C8) {
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c8 = int(idx % C8);
+    const int n = int(idx / (int64_t(C8) * HW));
+    short8_t xv = *reinterpret_cast<const short8_t*>(&x[idx * 8]);
+    short8_t sv = *reinterpret_cast<const short8_t*>(
+        &scale[(int64_t(n) * C8 + c8) * 8]);
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = short(f2bf(bf2f(ushort_t(xv[e])) * bf2f(ushort_t(sv[e]))));
+    *reinterpret_cast<short8_t*>(&y[idx * 8]) = out;
+  }
+}
+
+// dx = dy * s (broadcast over hw); one block per image: ds[n,c] = sum_hw dy*x
+__global__ __launch_bounds__(256) void rowscale_bwd_kernel(
+    const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
+    const ushort_t* __restrict__ scale, ushort_t* __restrict__ dx,
+    ushort_t* __restrict__ dscale, int HW, int C) {
+  __shared__ float red[256];
+  const int n = blockIdx.x;
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  const int64_t base = int64_t(n) * HW * C;
+  for (int ct = 0; ct < c_tiles; ++ct) {
+    const int c = ct * CT + int(threadIdx.x) % CT;
+    const int mlane = int(threadIdx.x) / CT;
+    float acc = 0.f;
+    if (c < C) {
+      const float sv = bf2f(scale[int64_t(n) * C + c]);
+      for (int hw = mlane; hw < HW; hw += ML) {
+        const float g = bf2f(dy[base + int64_t(hw) * C + c]);
+        acc = fmaf(g, bf2f(x[base + int64_t(hw) * C + c]), acc);
+        dx[base + int64_t(hw) * C + c] = f2bf(g * sv);
+      }
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    if (threadIdx.x < CT && c < C) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+      dscale[int64_t(n) * C + c] = f2bf(t);
+    }
+    __syncthreads();
+  }
+}
+
+// --- global average pool ------------------------------------------------
+template <bool SUM_ONLY>
+__global__ __launch_bounds__(256) void avgpool_fwd_kernel(
+    const ushort_t* __restrict__ x, ushort_t* __restrict__ out, int HW,
+    int C) {
+  __shared__ float red[256];
+  const int n = blockIdx.x;
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  const int64_t base = int64_t(n) * HW * C;
+  for (int ct = 0; ct < c_tiles; ++ct) {
+    const int c = ct * CT + int(threadIdx.x) % CT;
+    const int mlane = int(threadIdx.x) / CT;
+    float acc = 0.f;
+    if (c < C)
+      for (int hw = mlane; hw < HW; hw += ML)
+        acc += bf2f(x[base + int64_t(hw) * C + c]);
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    if (threadIdx.x < CT && c < C) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+      out[int64_t(n) * C + c] = f2bf(t / HW);
+    }
+    __syncthreads();
+  }
+}
+
+template <bool ACCUM>
+__global__ __launch_bounds__(256) void avgpool_bwd_kernel(
+    const ushort_t* __restrict__ dpooled, ushort_t* __restrict__ dx,
+    int64_t total8, int HW, int C8, float inv_hw) {
+  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
+    const int c8 = int(idx % C8);
+    const int n = int(idx / (int64_t(C8) * HW));
+    short8_t gv = *reinterpret_cast<const short8_t*>(
+        &dpooled[(int64_t(n) * C8 + c8) * 8]);
+    short8_t out;
+    if (ACCUM) {
+      short8_t cur = *reinterpret_cast<const short8_t*>(&dx[idx * 8]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        out[e] = short(f2bf(bf2f(ushort_t(cur[e])) +
+                            bf2f(ushort_t(gv[e])) * inv_hw));
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        out[e] = short(f2bf(bf2f(ushort_t(gv[e])) * inv_hw));
+    }
+    *reinterpret_cast<short8_t*>(&dx[idx * 8]) = out;
+  }
+}
+
+// --- label-smoothed softmax CE: one block per row -----------------------
+__global__ __launch_bounds__(256) void ce_row_kernel(
+    const ushort_t* __restrict__ logits, const int64_t* __restrict__ target,
+    float smoothing, float* __restrict__ dlogits,
+    float* __restrict__ rowloss, int N, int C) {
+  __shared__ float red[256];
+  const int n = blockIdx.x;
+  const int tid = threadIdx.x;
+  const ushort_t* row = logits + int64_t(n) * C;
+  // 1) max
+  float m = -3.4e38f;
+  for (int c = tid; c < C; c += 256) m = fmaxf(m, bf2f(row[c]));
+  red[tid] = m;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (tid < off) red[tid] = fmaxf(red[tid], red[tid + off]);
+    __syncthreads();
+  }
+  m = red[0];
+  __syncthreads();
+  // 2) sum exp and sum logits
+  float se = 0.f, sl = 0.f;
+  for (int c = tid; c < C; c += 256) {
+    const float v = bf2f(row[c]);
+    se += __expf(v - m);
+    sl += v;
+  }
+  red[tid] = se;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (tid < off) red[tid] += red[tid + off];
+    __syncthreads();
+  }
+  se = red[0];
+  __syncthreads();
+  red[tid] = sl;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (tid < off) red[tid] += red[tid + off];
+    __syncthreads();
+  }
+  sl = red[0];
+  __syncthreads();
+  const float log_z = m + __logf(se);
+  const int64_t y = target[n];
+  const float inv_n = 1.f / float(N);
+  // 3) dlogits = (softmax - ((1-s)*onehot + s/C)) / N ; rowloss
+  for (int c = tid; c < C; c += 256) {
+    const float p = __expf(bf2f(row[c]) - m) / se;
+    float t = smoothing / C;
+    if (c == int(y)) t += 1.f - smoothing;
+    dlogits[int64_t(n) * C + c] = (p - t) * inv_n;
+  }
+  if (tid == 0) {
+    const float xy = bf2f(row[y]);
+    rowloss[n] = log_z - (1.f - smoothing) * xy - (smoothing / C) * sl;
+  }
+}
+
+__global__ void ce_final_kernel(const float* __restrict__ rowloss,
+                                float* __restrict__ loss, int N) {
+  __shared__ float red[256];
+  float s = 0.f;
+  for (int n = threadIdx.x; n < N; n += 256) s += rowloss[n];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) *loss = red[0] / N;
+}
+
+}  // namespace
+
+void bias_act_fwd(const void* z, const float* bias, int act, void* y,
+                  int64_t M, int C, hipStream_t s) {
+  const int64_t total8 = M * (C >> 3);
+  ACT_SWITCH(act, {
+    bias_act_fwd_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
+        static_cast<const ushort_t*>(z), bias, static_cast<ushort_t*>(y),
+        total8, C >> 3);
+  });
+}
+
+int bias_bwd_chunks(int64_t M) {
+  int c = int(std::min<int64_t>((M + 255) / 256, 64));
+  return c < 1 ? 1 : c;
+}
+
+int64_t bias_bwd_ws_elems(int64_t M, int C) {
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  return int64_t(bias_bwd_chunks(M)) * ML * C;
+}
+
+void bias_act_bwd(const void* dy, const void* z, const float* bias, int act,
+                  void* dz, float* dbias, float* ws, int64_t M, int C,
+                  hipStream_t s) {
+  const int chunks = bias_bwd_chunks(M);
+  const int CT = C < 256 ? C : 256;
+  const int ML = 256 / CT;
+  const int c_tiles = (C + CT - 1) / CT;
+  ACT_SWITCH(act, {
+    bias_act_bwd_kernel<kAct><<<dim3(c_tiles * chunks), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(z),
+        bias, static_cast<ushort_t*>(dz), dbias != nullptr ? ws : nullptr,
+        M, C, chunks);
+  });
+  if (dbias != nullptr) slab_reduce(ws, dbias, C, chunks * ML, s);
+}
+
+void rowscale_fwd(const void* x, const void* scale, void* y, int N, int HW,
+                  int C, hipStream_t s) {
+  const int64_t total8 = int64_t(N) * HW * (C >> 3);
+  rowscale_fwd_kernel<<<elem_grid(total8), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(scale),
+      static_cast<ushort_t*>(y), total8, HW, C >> 3);
+}
+
+void rowscale_bwd(const void* dy, const void* x, const void* scale, void* dx,
+                  void* dscale, int N, int HW, int C, hipStream_t s) {
+  rowscale_bwd_kernel<<<dim3(N), 256, 0, s>>>(
+      static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
+      static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
+      static_cast<ushort_t*>(dscale), HW, C);
+}
+
+void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
+                 hipStream_t s) {
+  avgpool_fwd_kernel<false><<<dim3(N), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C);
+}
+
+void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,
+                 bool accumulate, hipStream_t s) {
+  const int64_t total8 = int64_t(N) * HW * (C >> 3);
+  if (accumulate)
+    avgpool_bwd_kernel<true><<<elem_grid(total8), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dpooled), static_cast<ushort_t*>(dx),
+        total8, HW, C >> 3, 1.f / HW);
+  else
+    avgpool_bwd_kernel<false><<<elem_grid(total8), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dpooled), static_cast<ushort_t*>(dx),
+        total8, HW, C >> 3, 1.f / HW);
+}
+
+void label_smooth_ce(const void* logits, const int64_t* target,
+                     float smoothing, float* loss, float* dlogits,
+                     float* rowloss, int N, int C, hipStream_t s) {
+  ce_row_kernel<<<dim3(N), 256, 0, s>>>(
+      static_cast<const ushort_t*>(logits), target, smoothing, dlogits,
+      rowloss, N, C);
+  ce_final_kernel<<<dim3(1), 256, 0, s>>>(rowloss, loss, N);
+}
+
+}  // namespace yams

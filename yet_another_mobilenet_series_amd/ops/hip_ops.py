@@ -90,7 +90,7 @@ class _ConvBNActFn(torch.autograd.Function):
         cin = x.shape[1]
         # pack weights into kernel layout (bf16)
         if kind == "pw":
-            w_packed = weight.view(cout, cin).t().to(BF16).contiguous()
+            w_packed = weight.view(cout, cin).to(BF16).contiguous()
         elif kind == "dw":
             w_packed = weight.view(cout, k * k).t().to(BF16).contiguous()
         elif kind == "stem":
@@ -140,7 +140,7 @@ class _ConvBNActFn(torch.autograd.Function):
         dx = None
         if ctx.needs_input_grad[0]:
             if kind == "pw":
-                w_t = weight.view(cout, cin).to(BF16).contiguous()  # [Cout,Cin]
+                w_t = weight.view(cout, cin).t().to(BF16).contiguous()  # [Cin,Cout]
                 dx = _empty_like_nhwc(x)
                 e.pw_conv(dz, w_t, dx)
             elif kind == "dw":
@@ -155,20 +155,22 @@ class _ConvBNActFn(torch.autograd.Function):
         dweight = None
         if ctx.needs_input_grad[1]:
             if kind == "pw":
-                dwp = torch.empty(cin, cout, dtype=torch.float32,
+                dwp = torch.empty(cout, cin, dtype=torch.float32,
                                   device=x.device)
-                e.pw_conv_wgrad(x, dz, dwp)
-                dweight = dwp.t().reshape(cout, cin, 1, 1)
+                e.pw_conv_wgrad(dz, x, dwp)
+                dweight = dwp.reshape(cout, cin, 1, 1)
             elif kind == "dw":
                 dwp = torch.empty(k * k, cout, dtype=torch.float32,
                                   device=x.device)
                 e.dw_conv_wgrad(x, dz, dwp, k, s, p)
                 dweight = dwp.t().reshape(cout, 1, k, k)
             else:  # stem
-                dwp = torch.empty(k, k, cin, cout, dtype=torch.float32,
+                dwp = torch.empty(k * k * cin, cout, dtype=torch.float32,
                                   device=x.device)
                 e.stem_conv_wgrad(x, dz, dwp, k, s, p)
-                dweight = dwp.permute(3, 2, 0, 1).contiguous()
+                # [k*k*3, Cout] -> [Cout, Cin, k, k]
+                dweight = (dwp.view(k, k, cin, cout)
+                           .permute(3, 2, 0, 1).contiguous())
 
         dres = dy if ctx.has_residual else None
         return (dx, dweight, dgamma, dbeta, dres) + (None,) * 10
@@ -208,7 +210,7 @@ class _LinearActFn(torch.autograd.Function):
         if x2.dim() != 2:
             x2 = x2.view(x2.shape[0], -1)
         nout, nin = weight.shape
-        wt = weight.t().to(BF16).contiguous()  # [in, out]
+        wt = weight.to(BF16).contiguous()  # [out, in], K-major
         z = torch.empty(x2.shape[0], nout, dtype=BF16, device=x.device)
         e.pw_conv(x2.view(x2.shape[0], nin, 1, 1), wt,
                   z.view(z.shape[0], nout, 1, 1))
@@ -233,14 +235,14 @@ class _LinearActFn(torch.autograd.Function):
                        ctx.act, dz, dbias)
         dx = None
         if ctx.needs_input_grad[0]:
-            w = weight.to(BF16).contiguous()  # [out, in]
+            w = weight.t().to(BF16).contiguous()  # [in, out]
             dx = torch.empty_like(x2)
             e.pw_conv(dz.view(-1, nout, 1, 1), w, dx.view(-1, nin, 1, 1))
         dweight = None
         if ctx.needs_input_grad[1]:
-            dwp = torch.empty(nin, nout, dtype=torch.float32, device=dy.device)
-            e.pw_conv_wgrad(x2.view(-1, nin, 1, 1), dz.view(-1, nout, 1, 1), dwp)
-            dweight = dwp.t().contiguous()
+            dwp = torch.empty(nout, nin, dtype=torch.float32, device=dy.device)
+            e.pw_conv_wgrad(dz.view(-1, nout, 1, 1), x2.view(-1, nin, 1, 1), dwp)
+            dweight = dwp
         return dx, dweight, dbias, None
 
 
@@ -289,14 +291,14 @@ class _SqueezeExciteFn(torch.autograd.Function):
         r = w1.shape[0]
         pooled = torch.empty(n, c, dtype=BF16, device=x.device)
         e.avgpool_fwd(x, pooled)
-        w1t = w1.view(r, c).t().to(BF16).contiguous()
+        w1m = w1.view(r, c).to(BF16).contiguous()
         hz = torch.empty(n, r, dtype=BF16, device=x.device)
-        e.pw_conv(pooled.view(n, c, 1, 1), w1t, hz.view(n, r, 1, 1))
+        e.pw_conv(pooled.view(n, c, 1, 1), w1m, hz.view(n, r, 1, 1))
         ha = torch.empty_like(hz)
         e.bias_act_fwd(hz, b1.float(), ACT_RELU, ha)
-        w2t = w2.view(c, r).t().to(BF16).contiguous()
+        w2m = w2.view(c, r).to(BF16).contiguous()
         sz = torch.empty(n, c, dtype=BF16, device=x.device)
-        e.pw_conv(ha.view(n, r, 1, 1), w2t, sz.view(n, c, 1, 1))
+        e.pw_conv(ha.view(n, r, 1, 1), w2m, sz.view(n, c, 1, 1))
         s = torch.empty_like(sz)
         e.bias_act_fwd(sz, b2.float(), ACT_HSIGMOID, s)
         y = torch.empty_like(x)
@@ -320,23 +322,23 @@ class _SqueezeExciteFn(torch.autograd.Function):
         db2 = torch.empty(c, dtype=torch.float32, device=x.device)
         e.bias_act_bwd(ds, sz, b2.float(), ACT_HSIGMOID, dsz, db2)
         # fc2 backward
-        w2m = w2.view(c, r).to(BF16).contiguous()
+        w2t = w2.view(c, r).t().to(BF16).contiguous()  # [r, c]
         dha = torch.empty_like(ha)
-        e.pw_conv(dsz.view(n, c, 1, 1), w2m, dha.view(n, r, 1, 1))
-        dw2p = torch.empty(r, c, dtype=torch.float32, device=x.device)
-        e.pw_conv_wgrad(ha.view(n, r, 1, 1), dsz.view(n, c, 1, 1), dw2p)
-        dw2 = dw2p.t().reshape_as(w2)
+        e.pw_conv(dsz.view(n, c, 1, 1), w2t, dha.view(n, r, 1, 1))
+        dw2p = torch.empty(c, r, dtype=torch.float32, device=x.device)
+        e.pw_conv_wgrad(dsz.view(n, c, 1, 1), ha.view(n, r, 1, 1), dw2p)
+        dw2 = dw2p.reshape_as(w2)
         # relu backward
         dhz = torch.empty_like(dha)
         db1 = torch.empty(r, dtype=torch.float32, device=x.device)
         e.bias_act_bwd(dha, hz, b1.float(), ACT_RELU, dhz, db1)
         # fc1 backward
-        w1m = w1.view(r, c).to(BF16).contiguous()
+        w1t = w1.view(r, c).t().to(BF16).contiguous()  # [c, r]
         dpooled = torch.empty_like(pooled)
-        e.pw_conv(dhz.view(n, r, 1, 1), w1m, dpooled.view(n, c, 1, 1))
-        dw1p = torch.empty(c, r, dtype=torch.float32, device=x.device)
-        e.pw_conv_wgrad(pooled.view(n, c, 1, 1), dhz.view(n, r, 1, 1), dw1p)
-        dw1 = dw1p.t().reshape_as(w1)
+        e.pw_conv(dhz.view(n, r, 1, 1), w1t, dpooled.view(n, c, 1, 1))
+        dw1p = torch.empty(r, c, dtype=torch.float32, device=x.device)
+        e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, c, 1, 1), dw1p)
+        dw1 = dw1p.reshape_as(w1)
         # pool backward folds into dx: dx += broadcast(dpooled)/(H*W)
         e.avgpool_bwd_add(dpooled, dx)
         return dx, dw1, db1, dw2, db2
@@ -357,12 +359,13 @@ class _LabelSmoothCEFn(torch.autograd.Function):
         e.label_smooth_ce(logits, target.to(torch.int64).contiguous(),
                           float(smoothing), loss, dlogits)
         ctx.save_for_backward(dlogits)
+        ctx.logits_dtype = logits.dtype
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
         (dlogits,) = ctx.saved_tensors
-        return dlogits * dloss, None, None
+        return (dlogits * dloss).to(ctx.logits_dtype), None, None
 
 
 def label_smooth_ce(logits, target, smoothing):
