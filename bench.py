@@ -77,13 +77,25 @@ def main():
         x = x.bfloat16().contiguous(memory_format=torch.channels_last)
     y = torch.randint(0, 1000, (args.batch,), device=dev)
 
+    import contextlib
+
+    def _fwd_ctx():
+        # reference engine on GPU: eager torch ops under bf16 autocast
+        # (the honest MIOpen-based baseline the HIP engine must beat)
+        if args.engine == "reference" and on_gpu:
+            return torch.autocast("cuda", torch.bfloat16)
+        return contextlib.nullcontext()
+
     def step():
         engine.zero_grad()
-        loss = OF.label_smooth_ce(engine(x), y, 0.1)
+        with _fwd_ctx():
+            logits = engine(x)
+        loss = OF.label_smooth_ce(logits.float() if args.engine == "reference"
+                                  else logits, y, 0.1)
         loss.backward()
         engine.finish()
         opt.step()
-        return loss
+        return loss.detach()
 
     graphed = None
     if args.graph and on_gpu:

@@ -32,7 +32,7 @@ __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
   const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C) return;
+  if (c >= C || mlane >= ML) return;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
@@ -113,7 +113,7 @@ __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
   const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C) return;
+  if (c >= C || mlane >= ML) return;
   const float a = gamma[c] * invstd[c];
   const float b = beta[c] - mean[c] * a;
   const int64_t per = (M + chunks - 1) / chunks;

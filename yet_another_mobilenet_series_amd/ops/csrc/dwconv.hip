@@ -119,7 +119,7 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int chunk = blockIdx.x / c_tiles;
   const int c = c0 + int(threadIdx.x) % CT;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C) return;
+  if (c >= C || mlane >= ML) return;
 
   const int64_t P = int64_t(N) * Ho * Wo;
   const int64_t per = (P + chunks - 1) / chunks;
@@ -203,7 +203,7 @@ __global__ __launch_bounds__(256) void stem_wgrad_kernel(
   const int co = int(threadIdx.x) % CT;
   const int mlane = int(threadIdx.x) / CT;
   const int chunk = blockIdx.x;
-  if (co >= Cout) return;
+  if (co >= Cout || mlane >= ML) return;
 
   const int64_t P = int64_t(N) * Ho * Wo;
   const int64_t per = (P + chunks - 1) / chunks;

@@ -48,7 +48,7 @@ __global__ __launch_bounds__(256) void bias_act_bwd_kernel(
   const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C) return;
+  if (c >= C || mlane >= ML) return;
   const float b = bias != nullptr ? bias[c] : 0.f;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
@@ -97,7 +97,7 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
     const int c = ct * CT + int(threadIdx.x) % CT;
     const int mlane = int(threadIdx.x) / CT;
     float acc = 0.f;
-    if (c < C) {
+    if (c < C && mlane < ML) {
       const float sv = bf2f(scale[int64_t(n) * C + c]);
       for (int hw = mlane; hw < HW; hw += ML) {
         const float g = bf2f(dy[base + int64_t(hw) * C + c]);
@@ -131,7 +131,7 @@ __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
     const int c = ct * CT + int(threadIdx.x) % CT;
     const int mlane = int(threadIdx.x) / CT;
     float acc = 0.f;
-    if (c < C)
+    if (c < C && mlane < ML)
       for (int hw = mlane; hw < HW; hw += ML)
         acc += bf2f(x[base + int64_t(hw) * C + c]);
     red[threadIdx.x] = acc;

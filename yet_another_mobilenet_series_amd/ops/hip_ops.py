@@ -60,7 +60,7 @@ def _conv_forward(x: torch.Tensor, w_packed: torch.Tensor, kind: str,
     e = ext()
     n, cin, h, wdt = x.shape
     if kind == "pw":
-        cout = w_packed.shape[1]
+        cout = w_packed.shape[0]
         z = _empty_like_nhwc(x, c=cout)
         e.pw_conv(x, w_packed, z)
     elif kind == "dw":
