@@ -179,13 +179,14 @@ def _bn_setup(n=4, c=72, h=14, seed=8):
 
 
 def test_bn_stats(ext):
-    z, _, _ = _bn_setup()
+    z, gamma, beta = _bn_setup()
     c = z.shape[1]
     mean = torch.empty(c, device=dev(), dtype=torch.float32)
     invstd = torch.empty_like(mean)
+    ab = torch.empty(2 * c, device=dev(), dtype=torch.float32)
     rm = torch.zeros(c, device=dev())
     rv = torch.ones(c, device=dev())
-    ext.bn_stats(z, mean, invstd, rm, rv, 0.1, 1e-5)
+    ext.bn_stats(z, mean, invstd, ab, gamma, beta, rm, rv, 0.1, 1e-5)
     zf = z.float()
     want_mean = zf.mean(dim=(0, 2, 3))
     want_var = zf.var(dim=(0, 2, 3), unbiased=False)
@@ -194,6 +195,9 @@ def test_bn_stats(ext):
     m = z.shape[0] * z.shape[2] * z.shape[3]
     assert_close(rm, 0.1 * want_mean, rtol=1e-3, atol=1e-4)
     assert_close(rv, 0.9 + 0.1 * want_var * m / (m - 1), rtol=1e-3, atol=1e-3)
+    a = gamma * invstd
+    assert_close(ab[:c], a, rtol=1e-4, atol=1e-5)
+    assert_close(ab[c:], beta - mean * a, rtol=1e-4, atol=1e-4)
 
 
 @pytest.mark.parametrize("act", [0, 1, 2, 3])
@@ -205,7 +209,9 @@ def test_bn_act_fwd(ext, act, residual):
     invstd = (zf.var(dim=(0, 2, 3), unbiased=False) + 1e-5).rsqrt()
     res = randn_nhwc(*z.shape, seed=9) if residual else None
     y = torch.empty_like(z)
-    ext.bn_act_fwd(z, mean, invstd, gamma, beta, act, y, res)
+    a = gamma * invstd
+    ab = torch.cat([a, beta - mean * a]).contiguous()
+    ext.bn_act_fwd(z, ab, act, y, res)
     v = (zf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
     v = gamma.view(1, -1, 1, 1) * v + beta.view(1, -1, 1, 1)
     want = [lambda t: t, F.relu, F.relu6, F.hardswish][act](v)

@@ -161,12 +161,14 @@ void stem_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
 
 // ---- BatchNorm ---------------------------------------------------------
 
-void bn_stats(const Tensor& z, Tensor& mean, Tensor& invstd,
+void bn_stats(const Tensor& z, Tensor& mean, Tensor& invstd, Tensor& ab,
+              const Tensor& gamma, const Tensor& beta,
               c10::optional<Tensor> running_mean,
               c10::optional<Tensor> running_var, double momentum,
               double eps) {
   auto zr = rows_of(z, "z");
   auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+  TORCH_CHECK(ab.numel() == 2 * zr.C, "ab must be [2C]");
   float* rm = nullptr;
   float* rv = nullptr;
   if (running_mean.has_value()) {
@@ -174,19 +176,18 @@ void bn_stats(const Tensor& z, Tensor& mean, Tensor& invstd,
     rv = running_var->data_ptr<float>();
   }
   yams::bn_stats(zr.ptr, mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                 rm, rv, ws.data_ptr<float>(), zr.M, zr.C, float(momentum),
-                 float(eps), cur_stream());
+                 ab.data_ptr<float>(), fp32_req(gamma, "gamma"),
+                 fp32_req(beta, "beta"), rm, rv, ws.data_ptr<float>(), zr.M,
+                 zr.C, float(momentum), float(eps), cur_stream());
 }
 
-void bn_act_fwd(const Tensor& z, const Tensor& mean, const Tensor& invstd,
-                const Tensor& gamma, const Tensor& beta, int64_t act,
-                Tensor& y, c10::optional<Tensor> residual) {
+void bn_act_fwd(const Tensor& z, const Tensor& ab, int64_t act, Tensor& y,
+                c10::optional<Tensor> residual) {
   auto zr = rows_of(z, "z");
   const void* res = nullptr;
   if (residual.has_value()) res = rows_of(*residual, "residual").ptr;
-  yams::bn_act_fwd(zr.ptr, fp32_req(mean, "mean"), fp32_req(invstd, "invstd"),
-                   fp32_req(gamma, "gamma"), fp32_req(beta, "beta"),
-                   int(act), y.data_ptr(), res, zr.M, zr.C, cur_stream());
+  yams::bn_act_fwd(zr.ptr, fp32_req(ab, "ab"), int(act), y.data_ptr(), res,
+                   zr.M, zr.C, cur_stream());
 }
 
 void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
@@ -197,12 +198,13 @@ void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
   auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+  auto coef = f32_ws(z, 4 * zr.C);
   yams::bn_act_bwd(gr.ptr, zr.ptr, fp32_req(mean, "mean"),
                    fp32_req(invstd, "invstd"), fp32_req(gamma, "gamma"),
                    fp32_req(beta, "beta"), int(act), training,
                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                   dz.data_ptr(), ws.data_ptr<float>(), zr.M, zr.C,
-                   cur_stream());
+                   dz.data_ptr(), ws.data_ptr<float>(),
+                   coef.data_ptr<float>(), zr.M, zr.C, cur_stream());
 }
 
 // ---- misc --------------------------------------------------------------

@@ -1,17 +1,24 @@
 // BatchNorm (K5) + fused activation (K6) + residual (K8) for gfx950.
 //
-// Training statistics are fp32 two-stage slab reductions over the bf16
-// conv output z[M, C] (M = N*H*W rows): per-(chunk,mlane) partial
-// sum/sumsq slabs, then a finalize kernel producing mean/invstd and the
-// running-stat update (PyTorch convention: running_var gets the UNBIASED
-// variance).  Deterministic — no atomics.
+// Design (2nd iteration, sized from rocprofv3 stats on MI355X):
+//  - partial kernels fill the chip: grid = c_tiles x chunks with chunks
+//    scaled so the launch has ~4096 workgroups (256 CUs x 8 XCDs need
+//    well over 256 blocks); per-block mlane partials fold through LDS so
+//    the slab count stays = chunks (keeps the combine cheap)
+//  - finalize kernels are split-parallel: 256 threads = 16 channels x 16
+//    slab lanes, LDS tree over lanes (the v1 one-thread-per-channel
+//    serial loop over 4k slabs was 150 us per call)
+//  - normalize/act kernels consume precomputed per-channel {a, b}
+//    (a = gamma*invstd, b = beta - mean*a) so the hot loop loads 2
+//    fp32 arrays, not 4-6
 //
-// Backward identity used (training mode):
-//   v0   = gamma*zhat + beta,  zhat = (z-mean)*invstd
-//   da   = dy * act'(v0)
-//   dgamma = sum da*zhat ; dbeta = sum da
-//   dz   = gamma*invstd * (da - dbeta/M - zhat*dgamma/M)
-// Eval mode drops the two mean-correction terms.
+// Backward identity (training):
+//   v0 = a*z + b ; da = dy * act'(v0)
+//   dgamma = sum(da*zhat) ; dbeta = sum(da)
+//   dz = a*da + K0 - K2*z,  K2 = a*invstd*dgamma/M,
+//                           K0 = -a*dbeta/M + mean*K2
+// Eval mode: dz = a*da.
+// All reductions deterministic (fixed-order slabs, no atomics).
 
 #include <algorithm>
 
@@ -22,49 +29,98 @@ namespace yams {
 
 namespace {
 
-// --- stats partial: thread = (channel lane, m lane) ---------------------
+constexpr int TARGET_BLOCKS = 4096;
+
+struct RedGeom {
+  int CT, ML, c_tiles, chunks;
+};
+
+RedGeom red_geom(int64_t M, int C) {
+  RedGeom g;
+  g.CT = C < 256 ? C : 256;
+  g.ML = 256 / g.CT;
+  g.c_tiles = (C + g.CT - 1) / g.CT;
+  int64_t max_chunks = (M + 63) / 64;
+  int64_t want = TARGET_BLOCKS / g.c_tiles;
+  g.chunks = int(std::min<int64_t>(std::max<int64_t>(want, 1), max_chunks));
+  return g;
+}
+
+// --- stats partial: per-(chunk) slab after in-block mlane fold ----------
 __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
     const ushort_t* __restrict__ z, float* __restrict__ ws, int64_t M, int C,
     int chunks) {
+  __shared__ float red[256];
   const int CT = C < 256 ? C : 256;
   const int ML = 256 / CT;
   const int c_tiles = (C + CT - 1) / CT;
   const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C || mlane >= ML) return;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
   float s1 = 0.f, s2 = 0.f;
-  for (int64_t m = lo + mlane; m < hi; m += ML) {
-    const float v = bf2f(z[m * C + c]);
-    s1 += v;
-    s2 = fmaf(v, v, s2);
+  if (c < C && mlane < ML) {
+    for (int64_t m = lo + mlane; m < hi; m += ML) {
+      const float v = bf2f(z[m * C + c]);
+      s1 += v;
+      s2 = fmaf(v, v, s2);
+    }
   }
-  const int64_t splits = int64_t(chunks) * ML;
-  const int64_t slot = int64_t(chunk) * ML + mlane;
-  ws[slot * C + c] = s1;
-  ws[(splits + slot) * C + c] = s2;
+  red[threadIdx.x] = s1;
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT) && c < C) {
+    float t = 0.f;
+    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+    ws[int64_t(chunk) * C + c] = t;
+  }
+  __syncthreads();
+  red[threadIdx.x] = s2;
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT) && c < C) {
+    float t = 0.f;
+    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+    ws[(int64_t(chunks) + chunk) * C + c] = t;
+  }
 }
 
-__global__ void bn_stats_finalize_kernel(
+// --- split-parallel finalize: 16 channels x 16 slab lanes per block -----
+__global__ __launch_bounds__(256) void bn_stats_finalize_kernel(
     const float* __restrict__ ws, float* __restrict__ mean,
-    float* __restrict__ invstd, float* __restrict__ running_mean,
-    float* __restrict__ running_var, int64_t M, int C, int splits,
-    float momentum, float eps) {
-  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-       c += gridDim.x * blockDim.x) {
-    float s1 = 0.f, s2 = 0.f;
-    for (int sp = 0; sp < splits; ++sp) {
+    float* __restrict__ invstd, float* __restrict__ ab,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    int64_t M, int C, int splits, float momentum, float eps) {
+  __shared__ float red[256];
+  const int cl = int(threadIdx.x) % 16;          // channel lane
+  const int sl = int(threadIdx.x) / 16;          // split lane
+  const int c = blockIdx.x * 16 + cl;
+  float s1 = 0.f, s2 = 0.f;
+  if (c < C) {
+    for (int sp = sl; sp < splits; sp += 16) {
       s1 += ws[int64_t(sp) * C + c];
       s2 += ws[int64_t(splits + sp) * C + c];
     }
+  }
+  red[threadIdx.x] = s1;
+  __syncthreads();
+  if (sl == 0 && c < C)
+    for (int l = 1; l < 16; ++l) s1 += red[cl + l * 16];
+  __syncthreads();
+  red[threadIdx.x] = s2;
+  __syncthreads();
+  if (sl == 0 && c < C) {
+    for (int l = 1; l < 16; ++l) s2 += red[cl + l * 16];
     const float mu = s1 / M;
     float var = s2 / M - mu * mu;
     var = fmaxf(var, 0.f);
+    const float inv = rsqrtf(var + eps);
     mean[c] = mu;
-    invstd[c] = rsqrtf(var + eps);
+    invstd[c] = inv;
+    const float a = gamma[c] * inv;
+    ab[c] = a;
+    ab[C + c] = beta[c] - mu * a;
     if (running_mean != nullptr) {
       const float unbiased = M > 1 ? var * (float(M) / float(M - 1)) : var;
       running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
@@ -73,25 +129,28 @@ __global__ void bn_stats_finalize_kernel(
   }
 }
 
-// --- forward normalize+act(+residual): flat vec8 ------------------------
+// --- forward normalize+act(+residual): flat vec8, 2 coef arrays ---------
 template <int ACT, bool RES>
 __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
-    const ushort_t* __restrict__ z, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ gamma,
-    const float* __restrict__ beta, ushort_t* __restrict__ y,
-    const ushort_t* __restrict__ res, int64_t total8, int C8) {
+    const ushort_t* __restrict__ z, const float* __restrict__ ab,
+    ushort_t* __restrict__ y, const ushort_t* __restrict__ res,
+    int64_t total8, int C8, int C) {
   for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
        idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
     const int c = int(idx % C8) * 8;
     const int64_t off = idx * 8;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
+    float4_t a0 = *reinterpret_cast<const float4_t*>(&ab[c]);
+    float4_t a1 = *reinterpret_cast<const float4_t*>(&ab[c + 4]);
+    float4_t b0 = *reinterpret_cast<const float4_t*>(&ab[C + c]);
+    float4_t b1 = *reinterpret_cast<const float4_t*>(&ab[C + c + 4]);
     short8_t rv;
     if (RES) rv = *reinterpret_cast<const short8_t*>(&res[off]);
     short8_t out;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const float a = gamma[c + e] * invstd[c + e];
-      const float b = beta[c + e] - mean[c + e] * a;
+      const float a = e < 4 ? a0[e] : a1[e - 4];
+      const float b = e < 4 ? b0[e] : b1[e - 4];
       float v = act_fwd<ACT>(fmaf(bf2f(ushort_t(zv[e])), a, b));
       if (RES) v += bf2f(ushort_t(rv[e]));
       out[e] = short(f2bf(v));
@@ -100,83 +159,127 @@ __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
   }
 }
 
-// --- backward reduce: per-channel sum(da), sum(da*z) --------------------
+// --- backward reduce: per-chunk sum(da), sum(da*z) slabs ----------------
 template <int ACT>
 __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ ws, int64_t M, int C, int chunks) {
+  __shared__ float red[256];
   const int CT = C < 256 ? C : 256;
   const int ML = 256 / CT;
   const int c_tiles = (C + CT - 1) / CT;
   const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C || mlane >= ML) return;
-  const float a = gamma[c] * invstd[c];
-  const float b = beta[c] - mean[c] * a;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
   float s1 = 0.f, s2 = 0.f;
-  for (int64_t m = lo + mlane; m < hi; m += ML) {
-    const float zv = bf2f(z[m * C + c]);
-    const float da = bf2f(dy[m * C + c]) * act_bwd<ACT>(fmaf(zv, a, b));
-    s1 += da;
-    s2 = fmaf(da, zv, s2);
+  if (c < C && mlane < ML) {
+    const float a = gamma[c] * invstd[c];
+    const float b = beta[c] - mean[c] * a;
+    for (int64_t m = lo + mlane; m < hi; m += ML) {
+      const float zv = bf2f(z[m * C + c]);
+      const float da = bf2f(dy[m * C + c]) * act_bwd<ACT>(fmaf(zv, a, b));
+      s1 += da;
+      s2 = fmaf(da, zv, s2);
+    }
   }
-  const int64_t splits = int64_t(chunks) * ML;
-  const int64_t slot = int64_t(chunk) * ML + mlane;
-  ws[slot * C + c] = s1;
-  ws[(splits + slot) * C + c] = s2;
+  red[threadIdx.x] = s1;
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT) && c < C) {
+    float t = 0.f;
+    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+    ws[int64_t(chunk) * C + c] = t;
+  }
+  __syncthreads();
+  red[threadIdx.x] = s2;
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT) && c < C) {
+    float t = 0.f;
+    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
+    ws[(int64_t(chunks) + chunk) * C + c] = t;
+  }
 }
 
-__global__ void bn_bwd_finalize_kernel(
+// finalize -> dgamma/dbeta + apply coefficients {a, b, K0, K2}
+template <bool TRAIN>
+__global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     const float* __restrict__ ws, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, int C, int splits) {
-  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
-       c += gridDim.x * blockDim.x) {
-    float s1 = 0.f, s2 = 0.f;
-    for (int sp = 0; sp < splits; ++sp) {
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, float* __restrict__ coef, int64_t M, int C,
+    int splits) {
+  __shared__ float red[256];
+  const int cl = int(threadIdx.x) % 16;
+  const int sl = int(threadIdx.x) / 16;
+  const int c = blockIdx.x * 16 + cl;
+  float s1 = 0.f, s2 = 0.f;
+  if (c < C) {
+    for (int sp = sl; sp < splits; sp += 16) {
       s1 += ws[int64_t(sp) * C + c];
       s2 += ws[int64_t(splits + sp) * C + c];
     }
+  }
+  red[threadIdx.x] = s1;
+  __syncthreads();
+  if (sl == 0 && c < C)
+    for (int l = 1; l < 16; ++l) s1 += red[cl + l * 16];
+  __syncthreads();
+  red[threadIdx.x] = s2;
+  __syncthreads();
+  if (sl == 0 && c < C) {
+    for (int l = 1; l < 16; ++l) s2 += red[cl + l * 16];
+    const float mu = mean[c], inv = invstd[c];
+    const float dg = (s2 - mu * s1) * inv;  // sum(da*zhat)
+    dgamma[c] = dg;
     dbeta[c] = s1;
-    dgamma[c] = (s2 - mean[c] * s1) * invstd[c];  // = sum(da * zhat)
+    const float a = gamma[c] * inv;
+    coef[c] = a;
+    coef[C + c] = beta[c] - mu * a;
+    if (TRAIN) {
+      const float inv_m = 1.f / float(M);
+      const float k2 = a * inv * dg * inv_m;
+      coef[2 * C + c] = -a * s1 * inv_m + mu * k2;  // K0
+      coef[3 * C + c] = k2;                         // K2
+    } else {
+      coef[2 * C + c] = 0.f;
+      coef[3 * C + c] = 0.f;
+    }
   }
 }
 
-template <int ACT, bool TRAIN>
+template <int ACT>
 __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ gamma, const float* __restrict__ beta,
-    const float* __restrict__ dgamma, const float* __restrict__ dbeta,
-    ushort_t* __restrict__ dz, int64_t total8, int C8, float inv_m) {
+    const float* __restrict__ coef, ushort_t* __restrict__ dz,
+    int64_t total8, int C8, int C) {
   for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
        idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
     const int c = int(idx % C8) * 8;
     const int64_t off = idx * 8;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
     short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
+    float4_t a0 = *reinterpret_cast<const float4_t*>(&coef[c]);
+    float4_t a1 = *reinterpret_cast<const float4_t*>(&coef[c + 4]);
+    float4_t b0 = *reinterpret_cast<const float4_t*>(&coef[C + c]);
+    float4_t b1 = *reinterpret_cast<const float4_t*>(&coef[C + c + 4]);
+    float4_t k00 = *reinterpret_cast<const float4_t*>(&coef[2 * C + c]);
+    float4_t k01 = *reinterpret_cast<const float4_t*>(&coef[2 * C + c + 4]);
+    float4_t k20 = *reinterpret_cast<const float4_t*>(&coef[3 * C + c]);
+    float4_t k21 = *reinterpret_cast<const float4_t*>(&coef[3 * C + c + 4]);
     short8_t out;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const float inv = invstd[c + e];
-      const float a = gamma[c + e] * inv;
-      const float b = beta[c + e] - mean[c + e] * a;
+      const float a = e < 4 ? a0[e] : a1[e - 4];
+      const float b = e < 4 ? b0[e] : b1[e - 4];
+      const float k0 = e < 4 ? k00[e] : k01[e - 4];
+      const float k2 = e < 4 ? k20[e] : k21[e - 4];
       const float zf = bf2f(ushort_t(zv[e]));
       const float da = bf2f(ushort_t(gv[e])) * act_bwd<ACT>(fmaf(zf, a, b));
-      float d;
-      if (TRAIN) {
-        const float zhat = (zf - mean[c + e]) * inv;
-        d = a * (da - dbeta[c + e] * inv_m - zhat * dgamma[c + e] * inv_m);
-      } else {
-        d = a * da;
-      }
-      out[e] = short(f2bf(d));
+      out[e] = short(f2bf(fmaf(a, da, k0) - k2 * zf));
     }
     *reinterpret_cast<short8_t*>(&dz[off]) = out;
   }
@@ -188,75 +291,63 @@ int elem_grid(int64_t total8) {
 
 }  // namespace
 
-int bn_stats_chunks(int64_t M) {
-  int c = int(std::min<int64_t>((M + 511) / 512, 256));
-  return c < 1 ? 1 : c;
-}
+int bn_stats_chunks(int64_t M) { return red_geom(M, 256).chunks; }
 
 int64_t bn_stats_ws_elems(int64_t M, int C) {
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  return 2LL * bn_stats_chunks(M) * ML * C;
+  return 2LL * red_geom(M, C).chunks * C;
 }
 
-void bn_stats(const void* z, float* mean, float* invstd, float* running_mean,
+void bn_stats(const void* z, float* mean, float* invstd, float* ab,
+              const float* gamma, const float* beta, float* running_mean,
               float* running_var, float* ws, int64_t M, int C,
               float momentum, float eps, hipStream_t s) {
-  const int chunks = bn_stats_chunks(M);
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
-  bn_stats_partial_kernel<<<dim3(c_tiles * chunks), 256, 0, s>>>(
-      static_cast<const ushort_t*>(z), ws, M, C, chunks);
-  bn_stats_finalize_kernel<<<dim3(ceil_div(C, 256)), 256, 0, s>>>(
-      ws, mean, invstd, running_mean, running_var, M, C, chunks * ML,
-      momentum, eps);
+  const RedGeom g = red_geom(M, C);
+  bn_stats_partial_kernel<<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
+      static_cast<const ushort_t*>(z), ws, M, C, g.chunks);
+  bn_stats_finalize_kernel<<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+      ws, mean, invstd, ab, gamma, beta, running_mean, running_var, M, C,
+      g.chunks, momentum, eps);
 }
 
-void bn_act_fwd(const void* z, const float* mean, const float* invstd,
-                const float* gamma, const float* beta, int act, void* y,
+void bn_act_fwd(const void* z, const float* ab, int act, void* y,
                 const void* residual, int64_t M, int C, hipStream_t s) {
   const int64_t total8 = M * (C >> 3);
-  const int C8 = C >> 3;
   auto* zp = static_cast<const ushort_t*>(z);
   auto* yp = static_cast<ushort_t*>(y);
   auto* rp = static_cast<const ushort_t*>(residual);
   ACT_SWITCH(act, {
     if (rp != nullptr)
       bn_act_fwd_kernel<kAct, true><<<elem_grid(total8), 256, 0, s>>>(
-          zp, mean, invstd, gamma, beta, yp, rp, total8, C8);
+          zp, ab, yp, rp, total8, C >> 3, C);
     else
       bn_act_fwd_kernel<kAct, false><<<elem_grid(total8), 256, 0, s>>>(
-          zp, mean, invstd, gamma, beta, yp, nullptr, total8, C8);
+          zp, ab, yp, nullptr, total8, C >> 3, C);
   });
 }
 
 void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
-                void* dz, float* ws, int64_t M, int C, hipStream_t s) {
-  const int chunks = bn_stats_chunks(M);
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
+                void* dz, float* ws, float* coef, int64_t M, int C,
+                hipStream_t s) {
+  const RedGeom g = red_geom(M, C);
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* zp = static_cast<const ushort_t*>(z);
   auto* dzp = static_cast<ushort_t*>(dz);
   const int64_t total8 = M * (C >> 3);
-  const float inv_m = 1.f / float(M);
   ACT_SWITCH(act, {
-    bn_bwd_partial_kernel<kAct><<<dim3(c_tiles * chunks), 256, 0, s>>>(
-        gp, zp, mean, invstd, gamma, beta, ws, M, C, chunks);
-    bn_bwd_finalize_kernel<<<dim3(ceil_div(C, 256)), 256, 0, s>>>(
-        ws, mean, invstd, dgamma, dbeta, C, chunks * ML);
+    bn_bwd_partial_kernel<kAct><<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
+        gp, zp, mean, invstd, gamma, beta, ws, M, C, g.chunks);
     if (training)
-      bn_bwd_apply_kernel<kAct, true><<<elem_grid(total8), 256, 0, s>>>(
-          gp, zp, mean, invstd, gamma, beta, dgamma, dbeta, dzp, total8,
-          C >> 3, inv_m);
+      bn_bwd_finalize_kernel<true><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+          ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
+          g.chunks);
     else
-      bn_bwd_apply_kernel<kAct, false><<<elem_grid(total8), 256, 0, s>>>(
-          gp, zp, mean, invstd, gamma, beta, dgamma, dbeta, dzp, total8,
-          C >> 3, inv_m);
+      bn_bwd_finalize_kernel<false><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+          ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
+          g.chunks);
+    bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
+        gp, zp, coef, dzp, total8, C >> 3, C);
   });
 }
 

@@ -105,50 +105,76 @@ __global__ __launch_bounds__(256) void dw_dgrad_kernel(
   }
 }
 
-// wgrad: each thread owns one channel lane and k*k fp32 accumulators over a
-// chunk of output positions; each (chunk, mlane) writes its own slab row.
-template <int K>
+// wgrad: thread = channel lane; 4 consecutive wo outputs share one x row
+// window in registers (kills the k-fold x re-read of v1); per-chunk fp32
+// slab after an in-block LDS fold over the m lanes.
+template <int K, int S>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
-    int stride, int pad, int chunks) {
+    int pad, int chunks) {
+  __shared__ float red[256];
+  constexpr int XW = 3 * S + K;  // x window for 4 outputs
   const int CT = C < 256 ? C : 256;
   const int ML = 256 / CT;
   const int c_tiles = (C + CT - 1) / CT;
-  const int c0 = (blockIdx.x % c_tiles) * CT;
+  const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
   const int chunk = blockIdx.x / c_tiles;
-  const int c = c0 + int(threadIdx.x) % CT;
   const int mlane = int(threadIdx.x) / CT;
-  if (c >= C || mlane >= ML) return;
 
-  const int64_t P = int64_t(N) * Ho * Wo;
+  const int Wo4 = (Wo + 3) / 4;
+  const int64_t P = int64_t(N) * Ho * Wo4;
   const int64_t per = (P + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float acc[K * K] = {};
-  for (int64_t p = lo + mlane; p < hi_; p += ML) {
-    const int wo = int(p % Wo);
-    int64_t rest = p / Wo;
-    const int ho = int(rest % Ho);
-    const int n = int(rest / Ho);
-    const float g = bf2f(dy[idx4(n, ho, wo, c, Ho, Wo, C)]);
+  if (c < C && mlane < ML) {
+    for (int64_t p = lo + mlane; p < hi_; p += ML) {
+      const int w4 = int(p % Wo4);
+      int64_t rest = p / Wo4;
+      const int ho = int(rest % Ho);
+      const int n = int(rest / Ho);
+      const int wo0 = w4 * 4;
+      float g[4];
 #pragma unroll
-    for (int i = 0; i < K; ++i) {
-      const int hi2 = ho * stride + i - pad;
-      if (hi2 < 0 || hi2 >= H) continue;
+      for (int q = 0; q < 4; ++q)
+        g[q] = (wo0 + q < Wo)
+                   ? bf2f(dy[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) : 0.f;
 #pragma unroll
-      for (int j = 0; j < K; ++j) {
-        const int wi = wo * stride + j - pad;
-        if (wi < 0 || wi >= W) continue;
-        acc[i * K + j] =
-            fmaf(g, bf2f(x[idx4(n, hi2, wi, c, H, W, C)]), acc[i * K + j]);
+      for (int i = 0; i < K; ++i) {
+        const int hi2 = ho * S + i - pad;
+        if (hi2 < 0 || hi2 >= H) continue;
+        float xr[XW];
+        const int wbase = wo0 * S - pad;
+#pragma unroll
+        for (int t = 0; t < XW; ++t) {
+          const int wi = wbase + t;
+          xr[t] = (wi >= 0 && wi < W)
+                      ? bf2f(x[idx4(n, hi2, wi, c, H, W, C)]) : 0.f;
+        }
+#pragma unroll
+        for (int j = 0; j < K; ++j) {
+          float sum = 0.f;
+#pragma unroll
+          for (int q = 0; q < 4; ++q) sum = fmaf(g[q], xr[q * S + j], sum);
+          acc[i * K + j] += sum;
+        }
       }
     }
   }
-  float* slab = ws + int64_t(chunk * ML + mlane) * K * K * C;
+  float* slab = ws + int64_t(chunk) * K * K * C;
 #pragma unroll
-  for (int t = 0; t < K * K; ++t) slab[t * C + c] = acc[t];
+  for (int t = 0; t < K * K; ++t) {
+    red[threadIdx.x] = acc[t];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT) && c < C) {
+      float v = 0.f;
+      for (int l = 0; l < ML; ++l) v += red[threadIdx.x + l * CT];
+      slab[t * C + c] = v;
+    }
+    __syncthreads();
+  }
 }
 
 // stem forward: flat vec8 over output channels; 27 taps, x broadcast from L1
@@ -192,18 +218,18 @@ __global__ __launch_bounds__(256) void stem_fwd_kernel(
   }
 }
 
-// stem wgrad: thread = cout lane, k*k*3 fp32 accumulators
+// stem wgrad: thread = cout lane, 27 fp32 accumulators, LDS mlane fold
 __global__ __launch_bounds__(256) void stem_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int Cout, int Ho, int Wo,
     int k, int stride, int pad, int chunks) {
+  __shared__ float red[256];
   const int Cin = 3;
   const int CT = Cout < 256 ? Cout : 256;
   const int ML = 256 / CT;
   const int co = int(threadIdx.x) % CT;
   const int mlane = int(threadIdx.x) / CT;
   const int chunk = blockIdx.x;
-  if (co >= Cout || mlane >= ML) return;
 
   const int64_t P = int64_t(N) * Ho * Wo;
   const int64_t per = (P + chunks - 1) / chunks;
@@ -211,28 +237,39 @@ __global__ __launch_bounds__(256) void stem_wgrad_kernel(
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float acc[27] = {};  // k=3 fixed for the zoo stem
-  for (int64_t p = lo + mlane; p < hi_; p += ML) {
-    const int wo = int(p % Wo);
-    int64_t rest = p / Wo;
-    const int ho = int(rest % Ho);
-    const int n = int(rest / Ho);
-    const float g = bf2f(dy[idx4(n, ho, wo, co, Ho, Wo, Cout)]);
-    for (int i = 0; i < k; ++i) {
-      const int hi2 = ho * stride + i - pad;
-      if (hi2 < 0 || hi2 >= H) continue;
-      for (int j = 0; j < k; ++j) {
-        const int wi = wo * stride + j - pad;
-        if (wi < 0 || wi >= W) continue;
+  if (co < Cout && mlane < ML) {
+    for (int64_t p = lo + mlane; p < hi_; p += ML) {
+      const int wo = int(p % Wo);
+      int64_t rest = p / Wo;
+      const int ho = int(rest % Ho);
+      const int n = int(rest / Ho);
+      const float g = bf2f(dy[idx4(n, ho, wo, co, Ho, Wo, Cout)]);
+      for (int i = 0; i < k; ++i) {
+        const int hi2 = ho * stride + i - pad;
+        if (hi2 < 0 || hi2 >= H) continue;
+        for (int j = 0; j < k; ++j) {
+          const int wi = wo * stride + j - pad;
+          if (wi < 0 || wi >= W) continue;
 #pragma unroll
-        for (int ci = 0; ci < 3; ++ci)
-          acc[(i * k + j) * Cin + ci] =
-              fmaf(g, bf2f(x[idx4(n, hi2, wi, ci, H, W, Cin)]),
-                   acc[(i * k + j) * Cin + ci]);
+          for (int ci = 0; ci < 3; ++ci)
+            acc[(i * k + j) * Cin + ci] =
+                fmaf(g, bf2f(x[idx4(n, hi2, wi, ci, H, W, Cin)]),
+                     acc[(i * k + j) * Cin + ci]);
+        }
       }
     }
   }
-  float* slab = ws + int64_t(chunk * ML + mlane) * (k * k * Cin) * Cout;
-  for (int t = 0; t < k * k * Cin; ++t) slab[t * Cout + co] = acc[t];
+  float* slab = ws + int64_t(chunk) * (k * k * Cin) * Cout;
+  for (int t = 0; t < k * k * Cin; ++t) {
+    red[threadIdx.x] = acc[t];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT) && co < Cout) {
+      float v = 0.f;
+      for (int l = 0; l < ML; ++l) v += red[threadIdx.x + l * CT];
+      slab[t * Cout + co] = v;
+    }
+    __syncthreads();
+  }
 }
 
 int grid_for(int64_t total) {
@@ -272,33 +309,39 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
 }
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
-  const int64_t P = int64_t(N) * Ho * Wo;
-  int c = int(std::min<int64_t>((P + 511) / 512, 256));
+  const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
+  int c = int(std::min<int64_t>((P + 15) / 16, 4096));
   return c < 1 ? 1 : c;
 }
 
 int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k) {
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  return int64_t(dw_wgrad_chunks(N, Ho, Wo)) * ML * k * k * C;
+  const int c_tiles = (C + 255) / 256;
+  return int64_t((dw_wgrad_chunks(N, Ho, Wo) + c_tiles - 1) / c_tiles)
+         * k * k * C;
 }
 
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo, int k,
                    int stride, int pad, hipStream_t s) {
-  const int chunks = dw_wgrad_chunks(N, Ho, Wo);
   const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
   const int c_tiles = (C + CT - 1) / CT;
+  const int chunks = (dw_wgrad_chunks(N, Ho, Wo) + c_tiles - 1) / c_tiles;
   auto* xp = static_cast<const ushort_t*>(x);
   auto* gp = static_cast<const ushort_t*>(dy);
-  if (k == 3)
-    dw_wgrad_kernel<3><<<dim3(c_tiles * chunks), 256, 0, s>>>(
-        xp, gp, ws, N, H, W, C, Ho, Wo, stride, pad, chunks);
+  dim3 grid(c_tiles * chunks);
+  if (k == 3 && stride == 1)
+    dw_wgrad_kernel<3, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
+                                               Wo, pad, chunks);
+  else if (k == 3 && stride == 2)
+    dw_wgrad_kernel<3, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
+                                               Wo, pad, chunks);
+  else if (k == 5 && stride == 1)
+    dw_wgrad_kernel<5, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
+                                               Wo, pad, chunks);
   else
-    dw_wgrad_kernel<5><<<dim3(c_tiles * chunks), 256, 0, s>>>(
-        xp, gp, ws, N, H, W, C, Ho, Wo, stride, pad, chunks);
-  slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks * ML, s);
+    dw_wgrad_kernel<5, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
+                                               Wo, pad, chunks);
+  slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }
 
 void stem_conv_fwd(const void* x, const void* w, void* y, int N, int H,
@@ -313,26 +356,22 @@ void stem_conv_fwd(const void* x, const void* w, void* y, int N, int H,
 
 int stem_wgrad_chunks(int N, int Ho, int Wo) {
   const int64_t P = int64_t(N) * Ho * Wo;
-  int c = int(std::min<int64_t>((P + 511) / 512, 512));
+  int c = int(std::min<int64_t>((P + 63) / 64, 4096));
   return c < 1 ? 1 : c;
 }
 
 int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k) {
-  const int CT = Cout < 256 ? Cout : 256;
-  const int ML = 256 / CT;
-  return int64_t(stem_wgrad_chunks(N, Ho, Wo)) * ML * k * k * 3 * Cout;
+  return int64_t(stem_wgrad_chunks(N, Ho, Wo)) * k * k * 3 * Cout;
 }
 
 void stem_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                      int N, int H, int W, int Cout, int Ho, int Wo, int k,
                      int stride, int pad, hipStream_t s) {
   const int chunks = stem_wgrad_chunks(N, Ho, Wo);
-  const int CT = Cout < 256 ? Cout : 256;
-  const int ML = 256 / CT;
   stem_wgrad_kernel<<<dim3(chunks), 256, 0, s>>>(
       static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(dy), ws,
       N, H, W, Cout, Ho, Wo, k, stride, pad, chunks);
-  slab_reduce(ws, dwgt, int64_t(k) * k * 3 * Cout, chunks * ML, s);
+  slab_reduce(ws, dwgt, int64_t(k) * k * 3 * Cout, chunks, s);
 }
 
 }  // namespace yams

@@ -21,7 +21,7 @@ void pw_conv(const void* x, const void* w, void* y,
 // pw_wgrad_workspace_elems)
 void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
                    int64_t M, int N1, int N2, hipStream_t s);
-int pw_wgrad_splits(int64_t M);
+int pw_wgrad_splits(int64_t M, int N1, int N2);
 int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2);
 
 // out[i] = sum_s ws[s*elems + i] (deterministic slab combine)
@@ -54,24 +54,26 @@ int stem_wgrad_chunks(int N, int Ho, int Wo);
 int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k);
 
 // --- BatchNorm (K5) -----------------------------------------------------
-// training stats over M=N*H*W rows of z[M,C]; ws fp32 [2, chunks, C]
-void bn_stats(const void* z, float* mean, float* invstd,
-              float* running_mean, float* running_var, float* ws,
-              int64_t M, int C, float momentum, float eps, hipStream_t s);
+// training stats over M=N*H*W rows of z[M,C]; ws fp32 [2, chunks, C].
+// Also emits ab[2C] = {a = gamma*invstd, b = beta - mean*a} for the
+// normalize kernels.
+void bn_stats(const void* z, float* mean, float* invstd, float* ab,
+              const float* gamma, const float* beta, float* running_mean,
+              float* running_var, float* ws, int64_t M, int C,
+              float momentum, float eps, hipStream_t s);
 int bn_stats_chunks(int64_t M);
 int64_t bn_stats_ws_elems(int64_t M, int C);
 
-// y = act(gamma*(z-mean)*invstd + beta) [+ residual]
-void bn_act_fwd(const void* z, const float* mean, const float* invstd,
-                const float* gamma, const float* beta, int act,
-                void* y, const void* residual, int64_t M, int C,
-                hipStream_t s);
+// y = act(a*z + b) [+ residual], ab as produced by bn_stats
+void bn_act_fwd(const void* z, const float* ab, int act, void* y,
+                const void* residual, int64_t M, int C, hipStream_t s);
 
-// backward: dgamma/dbeta + dz; ws fp32 [2, chunks, C]
+// backward: dgamma/dbeta + dz; ws fp32 [2, chunks, C]; coef fp32 [4C]
 void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
-                void* dz, float* ws, int64_t M, int C, hipStream_t s);
+                void* dz, float* ws, float* coef, int64_t M, int C,
+                hipStream_t s);
 
 // --- small fused elementwise/reduction kernels --------------------------
 // y = act(z + bias) on [M, C] (bias nullable)

@@ -226,14 +226,26 @@ __global__ __launch_bounds__(256) void atb_kernel(
   }
 }
 
-__global__ void slab_reduce_kernel(const float* __restrict__ ws,
-                                   float* __restrict__ out, int64_t elems,
-                                   int splits) {
-  for (int64_t i = int64_t(blockIdx.x) * blockDim.x + threadIdx.x; i < elems;
-       i += int64_t(gridDim.x) * blockDim.x) {
+// split-parallel: 256 threads = 16 element lanes x 16 split lanes
+__global__ __launch_bounds__(256) void slab_reduce_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, int64_t elems,
+    int splits) {
+  __shared__ float red[256];
+  const int el = int(threadIdx.x) % 16;
+  const int sl = int(threadIdx.x) / 16;
+  for (int64_t base = int64_t(blockIdx.x) * 16; base < elems;
+       base += int64_t(gridDim.x) * 16) {
+    const int64_t i = base + el;
     float v = 0.f;
-    for (int sp = 0; sp < splits; ++sp) v += ws[int64_t(sp) * elems + i];
-    out[i] = v;
+    if (i < elems)
+      for (int sp = sl; sp < splits; sp += 16) v += ws[int64_t(sp) * elems + i];
+    red[threadIdx.x] = v;
+    __syncthreads();
+    if (sl == 0 && i < elems) {
+      for (int l = 1; l < 16; ++l) v += red[el + l * 16];
+      out[i] = v;
+    }
+    __syncthreads();
   }
 }
 
@@ -241,22 +253,26 @@ __global__ void slab_reduce_kernel(const float* __restrict__ ws,
 
 void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s) {
-  const int blocks = int(std::min<int64_t>((elems + 255) / 256, 2048));
+  const int blocks = int(std::min<int64_t>((elems + 15) / 16, 8192));
   slab_reduce_kernel<<<dim3(blocks), 256, 0, s>>>(ws, out, elems, splits);
 }
 
-int pw_wgrad_splits(int64_t M) {
-  int s = int((M + 2047) / 2048);
-  return s < 1 ? 1 : (s > 64 ? 64 : s);
+int pw_wgrad_splits(int64_t M, int N1, int N2) {
+  const int tiles = ceil_div(N1, WT) * ceil_div(N2, WT);
+  const int64_t max_s = (M + 127) / 128;
+  int64_t s = 2048 / tiles;
+  if (s < 1) s = 1;
+  if (s > max_s) s = max_s;
+  return int(s < 1 ? 1 : s);
 }
 
 int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2) {
-  return int64_t(pw_wgrad_splits(M)) * N1 * N2;
+  return int64_t(pw_wgrad_splits(M, N1, N2)) * N1 * N2;
 }
 
 void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
                    int64_t M, int N1, int N2, hipStream_t s) {
-  const int splits = pw_wgrad_splits(M);
+  const int splits = pw_wgrad_splits(M, N1, N2);
   const int tiles = ceil_div(N1, WT) * ceil_div(N2, WT);
   atb_kernel<<<dim3(tiles * splits), 256, 0, s>>>(
       static_cast<const ushort_t*>(a), static_cast<const ushort_t*>(b), ws,

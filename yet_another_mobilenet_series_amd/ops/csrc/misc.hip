@@ -82,66 +82,91 @@ __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
   }
 }
 
-// dx = dy * s (broadcast over hw); one block per image: ds[n,c] = sum_hw dy*x
+// dx = dy * s (broadcast over hw); grid (n x c8 tiles); vec8 channels,
+// hw lanes folded through LDS: ds[n,c] = sum_hw dy*x
 __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ scale, ushort_t* __restrict__ dx,
-    ushort_t* __restrict__ dscale, int HW, int C) {
-  __shared__ float red[256];
-  const int n = blockIdx.x;
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
+    ushort_t* __restrict__ dscale, int HW, int C, int c8_tiles) {
+  __shared__ float red[256 * 8];
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int n = blockIdx.x / c8_tiles;
+  const int c8 = (blockIdx.x % c8_tiles) * CT8 + int(threadIdx.x) % CT8;
+  const int hwlane = int(threadIdx.x) / CT8;
   const int64_t base = int64_t(n) * HW * C;
-  for (int ct = 0; ct < c_tiles; ++ct) {
-    const int c = ct * CT + int(threadIdx.x) % CT;
-    const int mlane = int(threadIdx.x) / CT;
-    float acc = 0.f;
-    if (c < C && mlane < ML) {
-      const float sv = bf2f(scale[int64_t(n) * C + c]);
-      for (int hw = mlane; hw < HW; hw += ML) {
-        const float g = bf2f(dy[base + int64_t(hw) * C + c]);
-        acc = fmaf(g, bf2f(x[base + int64_t(hw) * C + c]), acc);
-        dx[base + int64_t(hw) * C + c] = f2bf(g * sv);
+  const int c = c8 * 8;
+  float acc[8] = {};
+  float sv[8];
+  if (c8 < C8 && hwlane < ML) {
+    short8_t s8 = *reinterpret_cast<const short8_t*>(
+        &scale[int64_t(n) * C + c]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) sv[e] = bf2f(ushort_t(s8[e]));
+    for (int hw = hwlane; hw < HW; hw += ML) {
+      const int64_t off = base + int64_t(hw) * C + c;
+      short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
+      short8_t x8 = *reinterpret_cast<const short8_t*>(&x[off]);
+      short8_t o8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float g = bf2f(ushort_t(g8[e]));
+        acc[e] = fmaf(g, bf2f(ushort_t(x8[e])), acc[e]);
+        o8[e] = short(f2bf(g * sv[e]));
       }
+      *reinterpret_cast<short8_t*>(&dx[off]) = o8;
     }
-    red[threadIdx.x] = acc;
-    __syncthreads();
-    if (threadIdx.x < CT && c < C) {
-      float t = 0.f;
-      for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-      dscale[int64_t(n) * C + c] = f2bf(t);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = acc[e];
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+    short8_t out;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = 0.f;
+      for (int l = 0; l < ML; ++l) v += red[(threadIdx.x + l * CT8) * 8 + e];
+      out[e] = short(f2bf(v));
     }
-    __syncthreads();
+    *reinterpret_cast<short8_t*>(&dscale[int64_t(n) * C + c]) = out;
   }
 }
 
 // --- global average pool ------------------------------------------------
-template <bool SUM_ONLY>
 __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
     const ushort_t* __restrict__ x, ushort_t* __restrict__ out, int HW,
-    int C) {
-  __shared__ float red[256];
-  const int n = blockIdx.x;
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
+    int C, int c8_tiles) {
+  __shared__ float red[256 * 8];
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int n = blockIdx.x / c8_tiles;
+  const int c8 = (blockIdx.x % c8_tiles) * CT8 + int(threadIdx.x) % CT8;
+  const int hwlane = int(threadIdx.x) / CT8;
   const int64_t base = int64_t(n) * HW * C;
-  for (int ct = 0; ct < c_tiles; ++ct) {
-    const int c = ct * CT + int(threadIdx.x) % CT;
-    const int mlane = int(threadIdx.x) / CT;
-    float acc = 0.f;
-    if (c < C && mlane < ML)
-      for (int hw = mlane; hw < HW; hw += ML)
-        acc += bf2f(x[base + int64_t(hw) * C + c]);
-    red[threadIdx.x] = acc;
-    __syncthreads();
-    if (threadIdx.x < CT && c < C) {
-      float t = 0.f;
-      for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-      out[int64_t(n) * C + c] = f2bf(t / HW);
+  const int c = c8 * 8;
+  float acc[8] = {};
+  if (c8 < C8 && hwlane < ML) {
+    for (int hw = hwlane; hw < HW; hw += ML) {
+      short8_t x8 = *reinterpret_cast<const short8_t*>(
+          &x[base + int64_t(hw) * C + c]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += bf2f(ushort_t(x8[e]));
     }
-    __syncthreads();
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = acc[e];
+  __syncthreads();
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+    short8_t o8;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = 0.f;
+      for (int l = 0; l < ML; ++l) v += red[(threadIdx.x + l * CT8) * 8 + e];
+      o8[e] = short(f2bf(v / HW));
+    }
+    *reinterpret_cast<short8_t*>(&out[int64_t(n) * C + c]) = o8;
   }
 }
 
@@ -293,16 +318,23 @@ void rowscale_fwd(const void* x, const void* scale, void* y, int N, int HW,
 
 void rowscale_bwd(const void* dy, const void* x, const void* scale, void* dx,
                   void* dscale, int N, int HW, int C, hipStream_t s) {
-  rowscale_bwd_kernel<<<dim3(N), 256, 0, s>>>(
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c8_tiles = (C8 + CT8 - 1) / CT8;
+  rowscale_bwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
       static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
       static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
-      static_cast<ushort_t*>(dscale), HW, C);
+      static_cast<ushort_t*>(dscale), HW, C, c8_tiles);
 }
 
 void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
                  hipStream_t s) {
-  avgpool_fwd_kernel<false><<<dim3(N), 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C);
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c8_tiles = (C8 + CT8 - 1) / CT8;
+  avgpool_fwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C,
+      c8_tiles);
 }
 
 void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,

@@ -102,19 +102,22 @@ class _ConvBNActFn(torch.autograd.Function):
 
         mean = torch.empty(cout, dtype=torch.float32, device=x.device)
         invstd = torch.empty_like(mean)
+        ab = torch.empty(2 * cout, dtype=torch.float32, device=x.device)
         if training:
-            e.bn_stats(z, mean, invstd, running_mean, running_var,
-                       momentum, eps)
+            e.bn_stats(z, mean, invstd, ab, bn_w, bn_b,
+                       running_mean, running_var, momentum, eps)
         else:
             mean.copy_(running_mean)
             invstd.copy_((running_var.float() + eps).rsqrt())
+            a = bn_w.float() * invstd
+            ab[:cout] = a
+            ab[cout:] = bn_b.float() - mean * a
 
         y = torch.empty_like(z)
         res = residual
         if res is not None:
             res = _nhwc(res)
-        e.bn_act_fwd(z, mean, invstd, bn_w, bn_b, act,
-                     y, res)
+        e.bn_act_fwd(z, ab, act, y, res)
 
         ctx.save_for_backward(x, z, mean, invstd, bn_w, bn_b, weight)
         ctx.kind, ctx.k, ctx.s, ctx.p = kind, k, s, p
