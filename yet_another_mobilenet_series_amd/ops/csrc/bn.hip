@@ -29,7 +29,7 @@ namespace yams {
 
 namespace {
 
-constexpr int TARGET_BLOCKS = 4096;
+constexpr int TARGET_BLOCKS = 1536;
 
 struct RedGeom {
   int CT, ML, c_tiles, chunks;

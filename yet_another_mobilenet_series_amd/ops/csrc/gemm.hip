@@ -140,8 +140,8 @@ void pw_conv(const void* x, const void* w, void* y,
 // ---------------------------------------------------------------------------
 namespace {
 
-constexpr int WT = 64;        // output tile side
-constexpr int MT = 32;        // m rows per LDS pass
+constexpr int WT = 128;       // output tile side
+constexpr int MT = 16;        // m rows per LDS pass
 constexpr int WLD = WT + 4;   // pad 8B
 
 __global__ __launch_bounds__(256) void atb_kernel(
@@ -151,8 +151,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
   __shared__ ushort_t Bl[MT * WLD];
 
   const int tid = threadIdx.x;
-  const int t1 = (tid & 15) * 4;   // rows of dw (N1 axis)
-  const int t2 = (tid >> 4) * 4;   // cols of dw (N2 axis)
+  const int t1 = (tid & 15) * 8;   // rows of dw (N1 axis)
+  const int t2 = (tid >> 4) * 8;   // cols of dw (N2 axis)
   const int n1_tiles = ceil_div(N1, WT);
   const int tile = blockIdx.x % (n1_tiles * ceil_div(N2, WT));
   const int split = blockIdx.x / (n1_tiles * ceil_div(N2, WT));
@@ -163,8 +163,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
   const int64_t m_lo = split * m_per;
   const int64_t m_hi = m_lo + m_per < M ? m_lo + m_per : M;
 
-  float acc[4][4] = {};
-  const int s_row = tid >> 3, s_seg = (tid & 7) * 8;  // 8 thr/row x 32 rows
+  float acc[8][8] = {};
+  const int s_row = tid >> 4, s_seg = (tid & 15) * 8;  // 16 thr/row x 16 rows
 
   for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
     const int64_t gm = m0 + s_row;
@@ -198,51 +198,54 @@ __global__ __launch_bounds__(256) void atb_kernel(
 
     const int mm_end = (m_hi - m0) < MT ? int(m_hi - m0) : MT;
     for (int mm = 0; mm < mm_end; ++mm) {
-      short4_t av = *reinterpret_cast<const short4_t*>(&Al[mm * WLD + t1]);
-      short4_t bv = *reinterpret_cast<const short4_t*>(&Bl[mm * WLD + t2]);
-      float a0 = bf2f(ushort_t(av.x)), a1 = bf2f(ushort_t(av.y)),
-            a2 = bf2f(ushort_t(av.z)), a3 = bf2f(ushort_t(av.w));
-      float b0 = bf2f(ushort_t(bv.x)), b1 = bf2f(ushort_t(bv.y)),
-            b2 = bf2f(ushort_t(bv.z)), b3 = bf2f(ushort_t(bv.w));
-      float a[4] = {a0, a1, a2, a3}, b[4] = {b0, b1, b2, b3};
+      short8_t av = *reinterpret_cast<const short8_t*>(&Al[mm * WLD + t1]);
+      short8_t bv = *reinterpret_cast<const short8_t*>(&Bl[mm * WLD + t2]);
+      float a[8], b[8];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int e = 0; e < 8; ++e) {
+        a[e] = bf2f(ushort_t(av[e]));
+        b[e] = bf2f(ushort_t(bv[e]));
+      }
 #pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
     }
     __syncthreads();
   }
 
   float* out = ws + int64_t(split) * N1 * N2;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < 8; ++i) {
     const int n1 = n1_0 + t1 + i;
     if (n1 >= N1) continue;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < 8; ++j) {
       const int n2 = n2_0 + t2 + j;
       if (n2 < N2) out[int64_t(n1) * N2 + n2] = acc[i][j];
     }
   }
 }
 
-// split-parallel: 256 threads = 16 element lanes x 16 split lanes
+// split-parallel: 256 threads = EL element lanes x SL split lanes
+template <int EL>
 __global__ __launch_bounds__(256) void slab_reduce_kernel(
     const float* __restrict__ ws, float* __restrict__ out, int64_t elems,
     int splits) {
+  constexpr int SL = 256 / EL;
   __shared__ float red[256];
-  const int el = int(threadIdx.x) % 16;
-  const int sl = int(threadIdx.x) / 16;
-  for (int64_t base = int64_t(blockIdx.x) * 16; base < elems;
-       base += int64_t(gridDim.x) * 16) {
+  const int el = int(threadIdx.x) % EL;
+  const int sl = int(threadIdx.x) / EL;
+  for (int64_t base = int64_t(blockIdx.x) * EL; base < elems;
+       base += int64_t(gridDim.x) * EL) {
     const int64_t i = base + el;
     float v = 0.f;
     if (i < elems)
-      for (int sp = sl; sp < splits; sp += 16) v += ws[int64_t(sp) * elems + i];
+      for (int sp = sl; sp < splits; sp += SL) v += ws[int64_t(sp) * elems + i];
     red[threadIdx.x] = v;
     __syncthreads();
     if (sl == 0 && i < elems) {
-      for (int l = 1; l < 16; ++l) v += red[el + l * 16];
+      for (int l = 1; l < SL; ++l) v += red[el + l * EL];
       out[i] = v;
     }
     __syncthreads();
@@ -253,8 +256,15 @@ __global__ __launch_bounds__(256) void slab_reduce_kernel(
 
 void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s) {
-  const int blocks = int(std::min<int64_t>((elems + 15) / 16, 8192));
-  slab_reduce_kernel<<<dim3(blocks), 256, 0, s>>>(ws, out, elems, splits);
+  if (splits >= 256) {
+    const int blocks = int(std::min<int64_t>((elems + 7) / 8, 8192));
+    slab_reduce_kernel<8><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
+                                                       splits);
+  } else {
+    const int blocks = int(std::min<int64_t>((elems + 15) / 16, 8192));
+    slab_reduce_kernel<16><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
+                                                        splits);
+  }
 }
 
 int pw_wgrad_splits(int64_t M, int N1, int N2) {

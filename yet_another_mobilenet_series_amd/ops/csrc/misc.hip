@@ -282,7 +282,7 @@ void bias_act_fwd(const void* z, const float* bias, int act, void* y,
 }
 
 int bias_bwd_chunks(int64_t M) {
-  int c = int(std::min<int64_t>((M + 255) / 256, 64));
+  int c = int(std::min<int64_t>((M + 3) / 4, 64));
   return c < 1 ? 1 : c;
 }
 
