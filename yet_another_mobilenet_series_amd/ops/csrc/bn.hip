@@ -37,51 +37,66 @@ struct RedGeom {
 
 RedGeom red_geom(int64_t M, int C) {
   RedGeom g;
-  g.CT = C < 256 ? C : 256;
+  const int C8 = C >> 3;
+  g.CT = C8 < 256 ? C8 : 256;
   g.ML = 256 / g.CT;
-  g.c_tiles = (C + g.CT - 1) / g.CT;
+  g.c_tiles = (C8 + g.CT - 1) / g.CT;
   int64_t max_chunks = (M + 63) / 64;
   int64_t want = TARGET_BLOCKS / g.c_tiles;
   g.chunks = int(std::min<int64_t>(std::max<int64_t>(want, 1), max_chunks));
   return g;
 }
 
-// --- stats partial: per-(chunk) slab after in-block mlane fold ----------
+// --- stats partial: vec8 channels, per-(chunk) slab after mlane fold ---
 __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
     const ushort_t* __restrict__ z, float* __restrict__ ws, int64_t M, int C,
     int chunks) {
-  __shared__ float red[256];
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
-  const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
+  __shared__ float red[256 * 8];
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
   const int chunk = blockIdx.x / c_tiles;
-  const int mlane = int(threadIdx.x) / CT;
+  const int mlane = int(threadIdx.x) / CT8;
+  const int c = c8 * 8;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
-  float s1 = 0.f, s2 = 0.f;
-  if (c < C && mlane < ML) {
+  float s1[8] = {}, s2[8] = {};
+  if (c8 < C8 && mlane < ML) {
     for (int64_t m = lo + mlane; m < hi; m += ML) {
-      const float v = bf2f(z[m * C + c]);
-      s1 += v;
-      s2 = fmaf(v, v, s2);
+      short8_t v8 = *reinterpret_cast<const short8_t*>(&z[m * C + c]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float v = bf2f(ushort_t(v8[e]));
+        s1[e] += v;
+        s2[e] = fmaf(v, v, s2[e]);
+      }
     }
   }
-  red[threadIdx.x] = s1;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s1[e];
   __syncthreads();
-  if (threadIdx.x < unsigned(CT) && c < C) {
-    float t = 0.f;
-    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-    ws[int64_t(chunk) * C + c] = t;
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[(threadIdx.x + l * CT8) * 8 + e];
+      ws[int64_t(chunk) * C + c + e] = t;
+    }
   }
   __syncthreads();
-  red[threadIdx.x] = s2;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s2[e];
   __syncthreads();
-  if (threadIdx.x < unsigned(CT) && c < C) {
-    float t = 0.f;
-    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-    ws[(int64_t(chunks) + chunk) * C + c] = t;
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[(threadIdx.x + l * CT8) * 8 + e];
+      ws[(int64_t(chunks) + chunk) * C + c + e] = t;
+    }
   }
 }
 
@@ -159,48 +174,68 @@ __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
   }
 }
 
-// --- backward reduce: per-chunk sum(da), sum(da*z) slabs ----------------
+// --- backward reduce: vec8 channels, per-chunk sum(da), sum(da*z) -------
 template <int ACT>
 __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ ws, int64_t M, int C, int chunks) {
-  __shared__ float red[256];
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
-  const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
+  __shared__ float red[256 * 8];
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
   const int chunk = blockIdx.x / c_tiles;
-  const int mlane = int(threadIdx.x) / CT;
+  const int mlane = int(threadIdx.x) / CT8;
+  const int c = c8 * 8;
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
-  float s1 = 0.f, s2 = 0.f;
-  if (c < C && mlane < ML) {
-    const float a = gamma[c] * invstd[c];
-    const float b = beta[c] - mean[c] * a;
+  float s1[8] = {}, s2[8] = {};
+  if (c8 < C8 && mlane < ML) {
+    float a[8], b[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      a[e] = gamma[c + e] * invstd[c + e];
+      b[e] = beta[c + e] - mean[c + e] * a[e];
+    }
     for (int64_t m = lo + mlane; m < hi; m += ML) {
-      const float zv = bf2f(z[m * C + c]);
-      const float da = bf2f(dy[m * C + c]) * act_bwd<ACT>(fmaf(zv, a, b));
-      s1 += da;
-      s2 = fmaf(da, zv, s2);
+      short8_t z8 = *reinterpret_cast<const short8_t*>(&z[m * C + c]);
+      short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[m * C + c]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float zv = bf2f(ushort_t(z8[e]));
+        const float da =
+            bf2f(ushort_t(g8[e])) * act_bwd<ACT>(fmaf(zv, a[e], b[e]));
+        s1[e] += da;
+        s2[e] = fmaf(da, zv, s2[e]);
+      }
     }
   }
-  red[threadIdx.x] = s1;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s1[e];
   __syncthreads();
-  if (threadIdx.x < unsigned(CT) && c < C) {
-    float t = 0.f;
-    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-    ws[int64_t(chunk) * C + c] = t;
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[(threadIdx.x + l * CT8) * 8 + e];
+      ws[int64_t(chunk) * C + c + e] = t;
+    }
   }
   __syncthreads();
-  red[threadIdx.x] = s2;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s2[e];
   __syncthreads();
-  if (threadIdx.x < unsigned(CT) && c < C) {
-    float t = 0.f;
-    for (int l = 0; l < ML; ++l) t += red[threadIdx.x + l * CT];
-    ws[(int64_t(chunks) + chunk) * C + c] = t;
+  if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float t = 0.f;
+      for (int l = 0; l < ML; ++l) t += red[(threadIdx.x + l * CT8) * 8 + e];
+      ws[(int64_t(chunks) + chunk) * C + c + e] = t;
+    }
   }
 }
 

@@ -33,6 +33,30 @@ ACT_HSWISH = 3
 ACT_HSIGMOID = 4
 
 
+
+def _packed(param: torch.Tensor, key: str, fn):
+    """Cache a packed/cast weight on the tensor, keyed by data version.
+
+    The pack ops re-run whenever the optimizer bumps the param version,
+    and ALWAYS during CUDA-graph capture (so replay re-reads the updated
+    fp32 master through the baked cast kernels)."""
+    if torch.cuda.is_current_stream_capturing():
+        return fn(param)
+    cache = getattr(param, "_yams_pack", None)
+    if cache is None:
+        cache = {}
+        try:
+            param._yams_pack = cache
+        except AttributeError:
+            return fn(param)
+    ver = param._version
+    hit = cache.get(key)
+    if hit is not None and hit[0] == ver:
+        return hit[1]
+    t = fn(param)
+    cache[key] = (ver, t)
+    return t
+
 def _nhwc(x: torch.Tensor) -> torch.Tensor:
     """bf16 + channels_last view of a 4D activation tensor."""
     if x.dim() == 2:
@@ -90,11 +114,14 @@ class _ConvBNActFn(torch.autograd.Function):
         cin = x.shape[1]
         # pack weights into kernel layout (bf16)
         if kind == "pw":
-            w_packed = weight.view(cout, cin).to(BF16).contiguous()
+            w_packed = _packed(weight, "pw", lambda w: w.view(
+                cout, cin).to(BF16).contiguous())
         elif kind == "dw":
-            w_packed = weight.view(cout, k * k).t().to(BF16).contiguous()
+            w_packed = _packed(weight, "dw", lambda w: w.view(
+                cout, k * k).t().to(BF16).contiguous())
         elif kind == "stem":
-            w_packed = weight.permute(2, 3, 1, 0).to(BF16).contiguous()
+            w_packed = _packed(weight, "stem", lambda w: w.permute(
+                2, 3, 1, 0).to(BF16).contiguous())
         else:
             raise ValueError(kind)
 
@@ -143,11 +170,13 @@ class _ConvBNActFn(torch.autograd.Function):
         dx = None
         if ctx.needs_input_grad[0]:
             if kind == "pw":
-                w_t = weight.view(cout, cin).t().to(BF16).contiguous()  # [Cin,Cout]
+                w_t = _packed(weight, "pw_t", lambda w: w.view(
+                    cout, cin).t().to(BF16).contiguous())  # [Cin,Cout]
                 dx = _empty_like_nhwc(x)
                 e.pw_conv(dz, w_t, dx)
             elif kind == "dw":
-                w_packed = weight.view(cout, k * k).t().to(BF16).contiguous()
+                w_packed = _packed(weight, "dw", lambda w: w.view(
+                    cout, k * k).t().to(BF16).contiguous())
                 dx = _empty_like_nhwc(x)
                 e.dw_conv_dgrad(dz, w_packed, dx, k, s, p)
             else:
@@ -213,7 +242,7 @@ class _LinearActFn(torch.autograd.Function):
         if x2.dim() != 2:
             x2 = x2.view(x2.shape[0], -1)
         nout, nin = weight.shape
-        wt = weight.to(BF16).contiguous()  # [out, in], K-major
+        wt = _packed(weight, "lin", lambda w: w.to(BF16).contiguous())
         z = torch.empty(x2.shape[0], nout, dtype=BF16, device=x.device)
         e.pw_conv(x2.view(x2.shape[0], nin, 1, 1), wt,
                   z.view(z.shape[0], nout, 1, 1))
@@ -238,7 +267,8 @@ class _LinearActFn(torch.autograd.Function):
                        ctx.act, dz, dbias)
         dx = None
         if ctx.needs_input_grad[0]:
-            w = weight.t().to(BF16).contiguous()  # [in, out]
+            w = _packed(weight, "lin_t",
+                        lambda wt_: wt_.t().to(BF16).contiguous())
             dx = torch.empty_like(x2)
             e.pw_conv(dz.view(-1, nout, 1, 1), w, dx.view(-1, nin, 1, 1))
         dweight = None
@@ -294,12 +324,12 @@ class _SqueezeExciteFn(torch.autograd.Function):
         r = w1.shape[0]
         pooled = torch.empty(n, c, dtype=BF16, device=x.device)
         e.avgpool_fwd(x, pooled)
-        w1m = w1.view(r, c).to(BF16).contiguous()
+        w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous())
         hz = torch.empty(n, r, dtype=BF16, device=x.device)
         e.pw_conv(pooled.view(n, c, 1, 1), w1m, hz.view(n, r, 1, 1))
         ha = torch.empty_like(hz)
         e.bias_act_fwd(hz, b1.float(), ACT_RELU, ha)
-        w2m = w2.view(c, r).to(BF16).contiguous()
+        w2m = _packed(w2, "se", lambda w: w.view(c, r).to(BF16).contiguous())
         sz = torch.empty(n, c, dtype=BF16, device=x.device)
         e.pw_conv(ha.view(n, r, 1, 1), w2m, sz.view(n, c, 1, 1))
         s = torch.empty_like(sz)
@@ -325,7 +355,8 @@ class _SqueezeExciteFn(torch.autograd.Function):
         db2 = torch.empty(c, dtype=torch.float32, device=x.device)
         e.bias_act_bwd(ds, sz, b2.float(), ACT_HSIGMOID, dsz, db2)
         # fc2 backward
-        w2t = w2.view(c, r).t().to(BF16).contiguous()  # [r, c]
+        w2t = _packed(w2, "se_t",
+                      lambda w: w.view(c, r).t().to(BF16).contiguous())
         dha = torch.empty_like(ha)
         e.pw_conv(dsz.view(n, c, 1, 1), w2t, dha.view(n, r, 1, 1))
         dw2p = torch.empty(c, r, dtype=torch.float32, device=x.device)
@@ -336,7 +367,8 @@ class _SqueezeExciteFn(torch.autograd.Function):
         db1 = torch.empty(r, dtype=torch.float32, device=x.device)
         e.bias_act_bwd(dha, hz, b1.float(), ACT_RELU, dhz, db1)
         # fc1 backward
-        w1t = w1.view(r, c).t().to(BF16).contiguous()  # [c, r]
+        w1t = _packed(w1, "se_t",
+                      lambda w: w.view(r, c).t().to(BF16).contiguous())
         dpooled = torch.empty_like(pooled)
         e.pw_conv(dhz.view(n, r, 1, 1), w1t, dpooled.view(n, c, 1, 1))
         dw1p = torch.empty(r, c, dtype=torch.float32, device=x.device)
