@@ -105,22 +105,24 @@ __global__ __launch_bounds__(256) void dw_dgrad_kernel(
   }
 }
 
-// wgrad: thread = channel lane; 4 consecutive wo outputs share one x row
-// window in registers (kills the k-fold x re-read of v1); per-chunk fp32
-// slab after an in-block LDS fold over the m lanes.
+// wgrad: thread = 4-channel vector lane; 4 consecutive wo outputs share
+// one x row window in registers; per-chunk fp32 slab after an LDS fold
+// over the m lanes.
 template <int K, int S>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
     int pad, int chunks) {
-  __shared__ float red[256];
+  __shared__ float red[256 * 4];
   constexpr int XW = 3 * S + K;  // x window for 4 outputs
-  const int CT = C < 256 ? C : 256;
-  const int ML = 256 / CT;
-  const int c_tiles = (C + CT - 1) / CT;
-  const int c = (blockIdx.x % c_tiles) * CT + int(threadIdx.x) % CT;
+  const int C4 = C >> 2;
+  const int CT4 = C4 < 256 ? C4 : 256;
+  const int ML = 256 / CT4;
+  const int c_tiles = (C4 + CT4 - 1) / CT4;
+  const int c4 = (blockIdx.x % c_tiles) * CT4 + int(threadIdx.x) % CT4;
   const int chunk = blockIdx.x / c_tiles;
-  const int mlane = int(threadIdx.x) / CT;
+  const int mlane = int(threadIdx.x) / CT4;
+  const int c = c4 * 4;
 
   const int Wo4 = (Wo + 3) / 4;
   const int64_t P = int64_t(N) * Ho * Wo4;
@@ -128,50 +130,72 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int64_t lo = chunk * per;
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
-  float acc[K * K] = {};
-  if (c < C && mlane < ML) {
+  float acc[K * K][4] = {};
+  if (c4 < C4 && mlane < ML) {
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
       const int w4 = int(p % Wo4);
       int64_t rest = p / Wo4;
       const int ho = int(rest % Ho);
       const int n = int(rest / Ho);
       const int wo0 = w4 * 4;
-      float g[4];
+      float g[4][4];
 #pragma unroll
-      for (int q = 0; q < 4; ++q)
-        g[q] = (wo0 + q < Wo)
-                   ? bf2f(dy[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) : 0.f;
+      for (int q = 0; q < 4; ++q) {
+        if (wo0 + q < Wo) {
+          short4_t g4 = *reinterpret_cast<const short4_t*>(
+              &dy[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]);
+#pragma unroll
+          for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
+        } else {
+#pragma unroll
+          for (int e = 0; e < 4; ++e) g[q][e] = 0.f;
+        }
+      }
 #pragma unroll
       for (int i = 0; i < K; ++i) {
         const int hi2 = ho * S + i - pad;
         if (hi2 < 0 || hi2 >= H) continue;
-        float xr[XW];
+        float xr[XW][4];
         const int wbase = wo0 * S - pad;
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
-          xr[t] = (wi >= 0 && wi < W)
-                      ? bf2f(x[idx4(n, hi2, wi, c, H, W, C)]) : 0.f;
+          if (wi >= 0 && wi < W) {
+            short4_t x4 = *reinterpret_cast<const short4_t*>(
+                &x[idx4(n, hi2, wi, c, H, W, C)]);
+#pragma unroll
+            for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
+          } else {
+#pragma unroll
+            for (int e = 0; e < 4; ++e) xr[t][e] = 0.f;
+          }
         }
 #pragma unroll
-        for (int j = 0; j < K; ++j) {
-          float sum = 0.f;
+        for (int j = 0; j < K; ++j)
 #pragma unroll
-          for (int q = 0; q < 4; ++q) sum = fmaf(g[q], xr[q * S + j], sum);
-          acc[i * K + j] += sum;
-        }
+          for (int e = 0; e < 4; ++e) {
+            float sum = 0.f;
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+              sum = fmaf(g[q][e], xr[q * S + j][e], sum);
+            acc[i * K + j][e] += sum;
+          }
       }
     }
   }
   float* slab = ws + int64_t(chunk) * K * K * C;
 #pragma unroll
   for (int t = 0; t < K * K; ++t) {
-    red[threadIdx.x] = acc[t];
+#pragma unroll
+    for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = acc[t][e];
     __syncthreads();
-    if (threadIdx.x < unsigned(CT) && c < C) {
-      float v = 0.f;
-      for (int l = 0; l < ML; ++l) v += red[threadIdx.x + l * CT];
-      slab[t * C + c] = v;
+    if (threadIdx.x < unsigned(CT4) && c4 < C4) {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        float v = 0.f;
+        for (int l = 0; l < ML; ++l) v += red[(threadIdx.x + l * CT4) * 4 + e];
+        slab[t * C + c + e] = v;
+      }
     }
     __syncthreads();
   }
@@ -315,7 +339,7 @@ int dw_wgrad_chunks(int N, int Ho, int Wo) {
 }
 
 int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k) {
-  const int c_tiles = (C + 255) / 256;
+  const int c_tiles = ((C >> 2) + 255) / 256;
   return int64_t((dw_wgrad_chunks(N, Ho, Wo) + c_tiles - 1) / c_tiles)
          * k * k * C;
 }
@@ -323,8 +347,9 @@ int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k) {
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo, int k,
                    int stride, int pad, hipStream_t s) {
-  const int CT = C < 256 ? C : 256;
-  const int c_tiles = (C + CT - 1) / CT;
+  const int C4 = C >> 2;
+  const int CT4 = C4 < 256 ? C4 : 256;
+  const int c_tiles = (C4 + CT4 - 1) / CT4;
   const int chunks = (dw_wgrad_chunks(N, Ho, Wo) + c_tiles - 1) / c_tiles;
   auto* xp = static_cast<const ushort_t*>(x);
   auto* gp = static_cast<const ushort_t*>(dy);
