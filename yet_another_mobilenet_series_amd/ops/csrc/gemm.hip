@@ -140,91 +140,120 @@ void pw_conv(const void* x, const void* w, void* y,
 // ---------------------------------------------------------------------------
 namespace {
 
-constexpr int WT = 128;       // output tile side
-constexpr int MT = 16;        // m rows per LDS pass
-constexpr int WLD = WT + 4;   // pad 8B
+constexpr int MT = 16;  // m rows per LDS pass
 
+// Adaptive output tile (W1 x W2 in {32,64,128}): small-channel conv wgrads
+// (e.g. dw 16->96 expands) waste up to 10x compute in a fixed 128x128
+// tile.  Thread layout 16x16, each thread V1 x V2 = (W1/16) x (W2/16)
+// fp32 accumulators; staging rows of max(W1,W2).
+template <int W1, int W2>
 __global__ __launch_bounds__(256) void atb_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     float* __restrict__ ws, int64_t M, int N1, int N2, int splits) {
-  __shared__ ushort_t Al[MT * WLD];
-  __shared__ ushort_t Bl[MT * WLD];
+  constexpr int V1 = W1 / 16, V2 = W2 / 16;
+  constexpr int LD1 = W1 + 4, LD2 = W2 + 4;
+  __shared__ ushort_t Al[MT * LD1];
+  __shared__ ushort_t Bl[MT * LD2];
 
   const int tid = threadIdx.x;
-  const int t1 = (tid & 15) * 8;   // rows of dw (N1 axis)
-  const int t2 = (tid >> 4) * 8;   // cols of dw (N2 axis)
-  const int n1_tiles = ceil_div(N1, WT);
-  const int tile = blockIdx.x % (n1_tiles * ceil_div(N2, WT));
-  const int split = blockIdx.x / (n1_tiles * ceil_div(N2, WT));
-  const int n1_0 = (tile % n1_tiles) * WT;
-  const int n2_0 = (tile / n1_tiles) * WT;
+  const int t1 = (tid & 15) * V1;
+  const int t2 = (tid >> 4) * V2;
+  const int n1_tiles = ceil_div(N1, W1);
+  const int n_tiles = n1_tiles * ceil_div(N2, W2);
+  const int tile = blockIdx.x % n_tiles;
+  const int split = blockIdx.x / n_tiles;
+  const int n1_0 = (tile % n1_tiles) * W1;
+  const int n2_0 = (tile / n1_tiles) * W2;
 
   const int64_t m_per = (M + splits - 1) / splits;
   const int64_t m_lo = split * m_per;
   const int64_t m_hi = m_lo + m_per < M ? m_lo + m_per : M;
 
-  float acc[8][8] = {};
-  const int s_row = tid >> 4, s_seg = (tid & 15) * 8;  // 16 thr/row x 16 rows
+  float acc[V1][V2] = {};
+  // staging: each thread loads  W/32 bf16x? ...  rows = 256/(W/8) per pass
+  constexpr int TPR1 = W1 / 8;   // threads per row (8 elems each)
+  constexpr int TPR2 = W2 / 8;
 
   for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
-    const int64_t gm = m0 + s_row;
     {
-      ushort_t* dst = &Al[s_row * WLD + s_seg];
-      const int gn = n1_0 + s_seg;
-      if (gm < m_hi && gn + 8 <= N1) {
-        *reinterpret_cast<short8_t*>(dst) =
-            *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
-      } else {
+      const int s_row = tid / TPR1, s_seg = (tid % TPR1) * 8;
+      for (int r = s_row; r < MT; r += 256 / TPR1) {
+        const int64_t gm = m0 + r;
+        ushort_t* dst = &Al[r * LD1 + s_seg];
+        const int gn = n1_0 + s_seg;
+        if (gm < m_hi && gn + 8 <= N1) {
+          *reinterpret_cast<short8_t*>(dst) =
+              *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          dst[j] = (gm < m_hi && gn + j < N1) ? A[gm * N1 + gn + j]
-                                              : ushort_t(0);
+          for (int j = 0; j < 8; ++j)
+            dst[j] = (gm < m_hi && gn + j < N1) ? A[gm * N1 + gn + j]
+                                                : ushort_t(0);
+        }
       }
     }
     {
-      ushort_t* dst = &Bl[s_row * WLD + s_seg];
-      const int gn = n2_0 + s_seg;
-      if (gm < m_hi && gn + 8 <= N2) {
-        *reinterpret_cast<short8_t*>(dst) =
-            *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
-      } else {
+      const int s_row = tid / TPR2, s_seg = (tid % TPR2) * 8;
+      for (int r = s_row; r < MT; r += 256 / TPR2) {
+        const int64_t gm = m0 + r;
+        ushort_t* dst = &Bl[r * LD2 + s_seg];
+        const int gn = n2_0 + s_seg;
+        if (gm < m_hi && gn + 8 <= N2) {
+          *reinterpret_cast<short8_t*>(dst) =
+              *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          dst[j] = (gm < m_hi && gn + j < N2) ? B[gm * N2 + gn + j]
-                                              : ushort_t(0);
+          for (int j = 0; j < 8; ++j)
+            dst[j] = (gm < m_hi && gn + j < N2) ? B[gm * N2 + gn + j]
+                                                : ushort_t(0);
+        }
       }
     }
     __syncthreads();
 
     const int mm_end = (m_hi - m0) < MT ? int(m_hi - m0) : MT;
     for (int mm = 0; mm < mm_end; ++mm) {
-      short8_t av = *reinterpret_cast<const short8_t*>(&Al[mm * WLD + t1]);
-      short8_t bv = *reinterpret_cast<const short8_t*>(&Bl[mm * WLD + t2]);
-      float a[8], b[8];
+      float a[V1], b[V2];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        a[e] = bf2f(ushort_t(av[e]));
-        b[e] = bf2f(ushort_t(bv[e]));
-      }
+      for (int e = 0; e < V1; ++e) a[e] = bf2f(Al[mm * LD1 + t1 + e]);
 #pragma unroll
-      for (int i = 0; i < 8; ++i)
+      for (int e = 0; e < V2; ++e) b[e] = bf2f(Bl[mm * LD2 + t2 + e]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
+      for (int i = 0; i < V1; ++i)
+#pragma unroll
+        for (int j = 0; j < V2; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
     }
     __syncthreads();
   }
 
   float* out = ws + int64_t(split) * N1 * N2;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < V1; ++i) {
     const int n1 = n1_0 + t1 + i;
     if (n1 >= N1) continue;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
+    for (int j = 0; j < V2; ++j) {
       const int n2 = n2_0 + t2 + j;
       if (n2 < N2) out[int64_t(n1) * N2 + n2] = acc[i][j];
     }
   }
+}
+
+int tile_of(int n) { return n <= 32 ? 32 : (n <= 64 ? 64 : 128); }
+
+void atb_launch(const ushort_t* a, const ushort_t* b, float* ws, int64_t M,
+                int N1, int N2, int splits, int blocks, hipStream_t s) {
+  const int w1 = tile_of(N1), w2 = tile_of(N2);
+  dim3 g(blocks);
+#define ATB_CASE(X, Y)                                                   \
+  if (w1 == X && w2 == Y) {                                              \
+    atb_kernel<X, Y><<<g, 256, 0, s>>>(a, b, ws, M, N1, N2, splits);     \
+    return;                                                              \
+  }
+  ATB_CASE(32, 32) ATB_CASE(32, 64) ATB_CASE(32, 128)
+  ATB_CASE(64, 32) ATB_CASE(64, 64) ATB_CASE(64, 128)
+  ATB_CASE(128, 32) ATB_CASE(128, 64) ATB_CASE(128, 128)
+#undef ATB_CASE
 }
 
 // split-parallel: 256 threads = EL element lanes x SL split lanes
@@ -268,7 +297,7 @@ void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
 }
 
 int pw_wgrad_splits(int64_t M, int N1, int N2) {
-  const int tiles = ceil_div(N1, WT) * ceil_div(N2, WT);
+  const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   const int64_t max_s = (M + 127) / 128;
   int64_t s = 2048 / tiles;
   if (s < 1) s = 1;
@@ -283,10 +312,10 @@ int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2) {
 void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
                    int64_t M, int N1, int N2, hipStream_t s) {
   const int splits = pw_wgrad_splits(M, N1, N2);
-  const int tiles = ceil_div(N1, WT) * ceil_div(N2, WT);
-  atb_kernel<<<dim3(tiles * splits), 256, 0, s>>>(
-      static_cast<const ushort_t*>(a), static_cast<const ushort_t*>(b), ws,
-      M, N1, N2, splits);
+  const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
+  atb_launch(static_cast<const ushort_t*>(a),
+             static_cast<const ushort_t*>(b), ws, M, N1, N2, splits,
+             tiles * splits, s);
   slab_reduce(ws, dw, int64_t(N1) * N2, splits, s);
 }
 

@@ -66,7 +66,18 @@ class DataParallelEngine(nn.Module):
         self.grad_dtype = grad_dtype
         if broadcast_params and self.world_size > 1:
             broadcast_module(module)
+        # single-process: skip flat buckets entirely — preset grad views
+        # force an extra AccumulateGrad add per param per step (~200 tiny
+        # kernels); plain assigned grads are free
+        self._bucketed = self.world_size > 1
         params = [p for p in module.parameters() if p.requires_grad]
+        if not self._bucketed:
+            self.buckets = []
+            self._param_bucket = {}
+            self._hooks = []
+            self._launched = []
+            self._params = params
+            return
         device = params[0].device if params else torch.device("cpu")
         # reverse registration order ≈ grad production order during backward
         params = list(reversed(params))
@@ -99,7 +110,12 @@ class DataParallelEngine(nn.Module):
         return self.module(*args, **kwargs)
 
     def zero_grad(self, set_to_none: bool = False) -> None:  # noqa: ARG002
-        """Zero the flat bucket buffers (grads are views; never set to None)."""
+        """Zero the flat bucket buffers (grads are views; never set to None).
+        Single-process mode: plain set-to-none."""
+        if not self._bucketed:
+            for p in self._params:
+                p.grad = None
+            return
         for b in self.buckets:
             b.flat.zero_()
             b.pending = len(b.params)
