@@ -82,7 +82,7 @@ def test_dp_matches_single_process(tmp_path, overlap):
 
 def test_bucket_layout_world1():
     model = _make_model()
-    eng = DataParallelEngine(model, bucket_mb=0.0001)
+    eng = DataParallelEngine(model, bucket_mb=0.0001, force_bucketed=True)
     total = sum(p.numel() for p in model.parameters())
     assert sum(b.flat.numel() for b in eng.buckets) == total
     assert len(eng.buckets) >= 2  # tiny bucket_mb forces multiple buckets
@@ -99,7 +99,7 @@ def test_bucket_layout_world1():
 
 def test_grad_order_reverse_registration():
     model = _make_model()
-    eng = DataParallelEngine(model, bucket_mb=1e-9)
+    eng = DataParallelEngine(model, bucket_mb=1e-9, force_bucketed=True)
     first_bucket_params = set(id(p) for p in eng.buckets[0].params)
     last_param = list(model.parameters())[-1]
     assert id(last_param) in first_bucket_params

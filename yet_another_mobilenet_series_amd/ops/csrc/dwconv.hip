@@ -334,7 +334,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
   const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
-  int c = int(std::min<int64_t>((P + 15) / 16, 4096));
+  int c = int(std::min<int64_t>((P + 3) / 4, 4096));
   return c < 1 ? 1 : c;
 }
 

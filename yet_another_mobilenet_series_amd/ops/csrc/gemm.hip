@@ -140,7 +140,7 @@ void pw_conv(const void* x, const void* w, void* y,
 // ---------------------------------------------------------------------------
 namespace {
 
-constexpr int MT = 16;  // m rows per LDS pass
+constexpr int MT = 32;  // m rows per LDS pass
 
 // Adaptive output tile (W1 x W2 in {32,64,128}): small-channel conv wgrads
 // (e.g. dw 16->96 expands) waste up to 10x compute in a fixed 128x128
@@ -214,10 +214,32 @@ __global__ __launch_bounds__(256) void atb_kernel(
     const int mm_end = (m_hi - m0) < MT ? int(m_hi - m0) : MT;
     for (int mm = 0; mm < mm_end; ++mm) {
       float a[V1], b[V2];
+      {  // V-wide vector loads (V*2 bytes, 8B-aligned for V>=4)
+        if (V1 >= 4) {
 #pragma unroll
-      for (int e = 0; e < V1; ++e) a[e] = bf2f(Al[mm * LD1 + t1 + e]);
+          for (int e = 0; e < V1; e += 4) {
+            short4_t v = *reinterpret_cast<const short4_t*>(
+                &Al[mm * LD1 + t1 + e]);
 #pragma unroll
-      for (int e = 0; e < V2; ++e) b[e] = bf2f(Bl[mm * LD2 + t2 + e]);
+            for (int q = 0; q < 4; ++q) a[e + q] = bf2f(ushort_t(v[q]));
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < V1; ++e) a[e] = bf2f(Al[mm * LD1 + t1 + e]);
+        }
+        if (V2 >= 4) {
+#pragma unroll
+          for (int e = 0; e < V2; e += 4) {
+            short4_t v = *reinterpret_cast<const short4_t*>(
+                &Bl[mm * LD2 + t2 + e]);
+#pragma unroll
+            for (int q = 0; q < 4; ++q) b[e + q] = bf2f(ushort_t(v[q]));
+          }
+        } else {
+#pragma unroll
+          for (int e = 0; e < V2; ++e) b[e] = bf2f(Bl[mm * LD2 + t2 + e]);
+        }
+      }
 #pragma unroll
       for (int i = 0; i < V1; ++i)
 #pragma unroll
@@ -299,7 +321,7 @@ void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
 int pw_wgrad_splits(int64_t M, int N1, int N2) {
   const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   const int64_t max_s = (M + 127) / 128;
-  int64_t s = 2048 / tiles;
+  int64_t s = 1024 / tiles;
   if (s < 1) s = 1;
   if (s > max_s) s = max_s;
   return int(s < 1 ? 1 : s);

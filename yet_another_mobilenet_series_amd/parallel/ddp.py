@@ -58,7 +58,8 @@ class DataParallelEngine(nn.Module):
 
     def __init__(self, module: nn.Module, bucket_mb: float = 4.0,
                  overlap: bool = True, grad_dtype: torch.dtype = torch.float32,
-                 broadcast_params: bool = True) -> None:
+                 broadcast_params: bool = True,
+                 force_bucketed: bool = False) -> None:
         super().__init__()
         self.module = module
         self.world_size = get_world_size()
@@ -69,7 +70,7 @@ class DataParallelEngine(nn.Module):
         # single-process: skip flat buckets entirely — preset grad views
         # force an extra AccumulateGrad add per param per step (~200 tiny
         # kernels); plain assigned grads are free
-        self._bucketed = self.world_size > 1
+        self._bucketed = self.world_size > 1 or force_bucketed
         params = [p for p in module.parameters() if p.requires_grad]
         if not self._bucketed:
             self.buckets = []
