@@ -21,48 +21,76 @@ namespace {
 
 constexpr int MAX_BLOCKS = 4096;
 
-template <int K>
+// forward: 4 consecutive wo outputs per thread share the x row window in
+// registers (k*k -> ~k*(k+3)/4 x loads per output).  FLIP reuses the same
+// kernel as the stride-1 dgrad (correlation with the flipped filter).
+template <int K, int S, bool FLIP>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
-    int stride, int pad, int64_t total8) {
+    int pad, int64_t total8) {
+  constexpr int XW = 3 * S + K;
   const int C8 = C >> 3;
+  const int Wo4 = (Wo + 3) / 4;
   for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
        idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
     const int c8 = int(idx % C8);
     int64_t rest = idx / C8;
-    const int wo = int(rest % Wo);
-    rest /= Wo;
+    const int w4 = int(rest % Wo4);
+    rest /= Wo4;
     const int ho = int(rest % Ho);
     const int n = int(rest / Ho);
     const int c = c8 * 8;
-    float acc[8] = {};
+    const int wo0 = w4 * 4;
+    float acc[4][8] = {};
 #pragma unroll
     for (int i = 0; i < K; ++i) {
-      const int hi = ho * stride + i - pad;
+      const int hi = ho * S + i - pad;
       if (hi < 0 || hi >= H) continue;
+      float xr[XW][8];
+      const int wbase = wo0 * S - pad;
+#pragma unroll
+      for (int t = 0; t < XW; ++t) {
+        const int wi = wbase + t;
+        if (wi >= 0 && wi < W) {
+          short8_t v = *reinterpret_cast<const short8_t*>(
+              &x[idx4(n, hi, wi, c, H, W, C)]);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) xr[t][e] = bf2f(ushort_t(v[e]));
+        } else {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) xr[t][e] = 0.f;
+        }
+      }
 #pragma unroll
       for (int j = 0; j < K; ++j) {
-        const int wi = wo * stride + j - pad;
-        if (wi < 0 || wi >= W) continue;
-        short8_t xv = *reinterpret_cast<const short8_t*>(
-            &x[idx4(n, hi, wi, c, H, W, C)]);
-        short8_t wv = *reinterpret_cast<const short8_t*>(
-            &w[(i * K + j) * C + c]);
+        const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+        short8_t wv = *reinterpret_cast<const short8_t*>(&w[tap * C + c]);
+        float wf[8];
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          acc[e] = fmaf(bf2f(ushort_t(xv[e])), bf2f(ushort_t(wv[e])), acc[e]);
+        for (int e = 0; e < 8; ++e) wf[e] = bf2f(ushort_t(wv[e]));
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc[q][e] = fmaf(xr[q * S + j][e], wf[e], acc[q][e]);
       }
     }
-    short8_t out;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
-    *reinterpret_cast<short8_t*>(&y[idx4(n, ho, wo, c, Ho, Wo, C)]) = out;
+    for (int q = 0; q < 4; ++q) {
+      if (wo0 + q >= Wo) break;
+      short8_t out;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[q][e]));
+      *reinterpret_cast<short8_t*>(
+          &y[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) = out;
+    }
   }
 }
 
+// dgrad for stride 2 (stride 1 routes through dw_fwd_kernel<K,1,FLIP=true>)
 template <int K>
-__global__ __launch_bounds__(256) void dw_dgrad_kernel(
+__global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
     int stride, int pad, int64_t total8) {
@@ -305,30 +333,49 @@ int grid_for(int64_t total) {
 void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
                  int C, int Ho, int Wo, int k, int stride, int pad,
                  hipStream_t s) {
-  const int64_t total8 = int64_t(N) * Ho * Wo * (C >> 3);
+  const int64_t total8 = int64_t(N) * Ho * ((Wo + 3) / 4) * (C >> 3);
   auto* xp = static_cast<const ushort_t*>(x);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
-  if (k == 3)
-    dw_fwd_kernel<3><<<grid_for(total8), 256, 0, s>>>(
-        xp, wp, yp, N, H, W, C, Ho, Wo, stride, pad, total8);
+  dim3 g(grid_for(total8));
+  if (k == 3 && stride == 1)
+    dw_fwd_kernel<3, 1, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
+                                                 Wo, pad, total8);
+  else if (k == 3 && stride == 2)
+    dw_fwd_kernel<3, 2, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
+                                                 Wo, pad, total8);
+  else if (k == 5 && stride == 1)
+    dw_fwd_kernel<5, 1, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
+                                                 Wo, pad, total8);
   else
-    dw_fwd_kernel<5><<<grid_for(total8), 256, 0, s>>>(
-        xp, wp, yp, N, H, W, C, Ho, Wo, stride, pad, total8);
+    dw_fwd_kernel<5, 2, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
+                                                 Wo, pad, total8);
 }
 
 void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
                    int W, int C, int Ho, int Wo, int k, int stride, int pad,
                    hipStream_t s) {
-  const int64_t total8 = int64_t(N) * H * W * (C >> 3);
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* xp = static_cast<ushort_t*>(dx);
+  if (stride == 1) {
+    // dgrad == correlation of dy with the flipped filter at same pad
+    const int64_t total8 = int64_t(N) * H * ((W + 3) / 4) * (C >> 3);
+    dim3 g(grid_for(total8));
+    if (k == 3)
+      dw_fwd_kernel<3, 1, true><<<g, 256, 0, s>>>(gp, wp, xp, N, Ho, Wo, C,
+                                                  H, W, pad, total8);
+    else
+      dw_fwd_kernel<5, 1, true><<<g, 256, 0, s>>>(gp, wp, xp, N, Ho, Wo, C,
+                                                  H, W, pad, total8);
+    return;
+  }
+  const int64_t total8 = int64_t(N) * H * W * (C >> 3);
   if (k == 3)
-    dw_dgrad_kernel<3><<<grid_for(total8), 256, 0, s>>>(
+    dw_dgrad_s2_kernel<3><<<grid_for(total8), 256, 0, s>>>(
         gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
   else
-    dw_dgrad_kernel<5><<<grid_for(total8), 256, 0, s>>>(
+    dw_dgrad_s2_kernel<5><<<grid_for(total8), 256, 0, s>>>(
         gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
 }
 
