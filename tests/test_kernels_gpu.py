@@ -143,30 +143,37 @@ def test_dw_wgrad(ext, n, c, h, k, s):
 
 # ------------------------------------------------------------------ stem
 
-@pytest.mark.parametrize("cout", [16, 24, 32])
-def test_stem_fwd(ext, cout):
+def test_im2row_matches_unfold(ext):
     n, h = 2, 32
     x = randn_nhwc(n, 3, h, h, seed=5)
-    w = (torch.randn(3, 3, 3, cout, device=dev()) * 0.2).to(BF16)
     ho = h // 2
-    y = torch.empty(n, cout, ho, ho, device=dev(), dtype=BF16,
-                    memory_format=CL)
-    ext.stem_conv_fwd(x, w, y, 3, 2, 1)
-    wref = w.float().permute(3, 2, 0, 1)  # [cout, cin, kh, kw]
-    want = F.conv2d(x.float(), wref, stride=2, padding=1)
-    assert_close(y, want)
+    kp = 32  # 27 rounded to /8
+    patches = torch.empty(n * ho * ho, kp, device=dev(), dtype=BF16)
+    ext.im2row(x, patches, 3, 2, 1)
+    want = torch.nn.functional.unfold(x.float(), 3, padding=1, stride=2)
+    # unfold: [n, cin*k*k, L] with cin-major taps; ours is tap-major, cin fastest
+    want = want.view(n, 3, 9, -1).permute(0, 3, 2, 1).reshape(n * ho * ho, 27)
+    assert torch.equal(patches[:, :27].float().cpu(), want.cpu())
+    assert patches[:, 27:].abs().sum().item() == 0
 
 
-def test_stem_wgrad(ext):
+def test_stem_conv_via_gemm(ext):
+    """stem conv == im2row + GEMM (checked against F.conv2d)."""
+    import torch.nn.functional as F
     n, h, cout = 2, 32, 32
     x = randn_nhwc(n, 3, h, h, seed=6)
-    dy = randn_nhwc(n, cout, h // 2, h // 2, seed=7)
-    dw = torch.empty(27, cout, device=dev(), dtype=torch.float32)
-    ext.stem_conv_wgrad(x, dy, dw, 3, 2, 1)
-    want = torch.nn.grad.conv2d_weight(
-        x.float(), (cout, 3, 3, 3), dy.float(), stride=2, padding=1)
-    want = want.permute(2, 3, 1, 0).reshape(27, cout)
-    assert_close(dw, want, atol=0.05 * (n * (h // 2) ** 2) ** 0.5 * 0.1)
+    wref = (torch.randn(cout, 3, 3, 3, device=dev()) * 0.2).to(BF16)
+    ho = h // 2
+    m = n * ho * ho
+    patches = torch.empty(m, 32, device=dev(), dtype=BF16)
+    ext.im2row(x, patches, 3, 2, 1)
+    wp = torch.zeros(cout, 32, device=dev(), dtype=BF16)
+    wp[:, :27] = wref.permute(0, 2, 3, 1).reshape(cout, 27)
+    y = torch.empty(m, cout, device=dev(), dtype=BF16)
+    ext.pw_conv(patches.view(m, 32, 1, 1), wp, y.view(m, cout, 1, 1))
+    want = F.conv2d(x.float(), wref.float(), stride=2, padding=1)
+    want = want.permute(0, 2, 3, 1).reshape(m, cout)
+    assert_close(y, want)
 
 
 # ---------------------------------------------------------------- BN+act

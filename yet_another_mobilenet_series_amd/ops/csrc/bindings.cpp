@@ -131,32 +131,22 @@ void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
                       int(stride), int(pad), cur_stream());
 }
 
-// ---- stem --------------------------------------------------------------
+// ---- stem: im2row (conv itself runs on the GEMM kernels) ---------------
 
-void stem_conv_fwd(const Tensor& x, const Tensor& w, Tensor& y, int64_t k,
-                   int64_t stride, int64_t pad) {
+void im2row(const Tensor& x, Tensor& patches, int64_t k, int64_t stride,
+            int64_t pad) {
   auto xr = rows_of(x, "x");
-  auto yr = rows_of(y, "y");
-  check_w(w);
-  TORCH_CHECK(xr.C == 3, "stem expects Cin=3");
-  yams::stem_conv_fwd(xr.ptr, w.data_ptr(), y.data_ptr(), int(x.size(0)),
-                      int(x.size(2)), int(x.size(3)), yr.C, int(y.size(2)),
-                      int(y.size(3)), int(k), int(stride), int(pad),
-                      cur_stream());
-}
-
-void stem_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
-                     int64_t k, int64_t stride, int64_t pad) {
-  auto xr = rows_of(x, "x");
-  auto gr = rows_of(dy, "dy");
-  TORCH_CHECK(dwgt.scalar_type() == torch::kFloat32 && dwgt.is_contiguous(),
-              "stem wgrad out must be fp32");
-  const int N = int(x.size(0)), H = int(x.size(2)), W = int(x.size(3));
-  const int Ho = int(dy.size(2)), Wo = int(dy.size(3));
-  auto ws = f32_ws(x, yams::stem_wgrad_ws_elems(N, Ho, Wo, gr.C, int(k)));
-  yams::stem_conv_wgrad(xr.ptr, gr.ptr, dwgt.data_ptr<float>(),
-                        ws.data_ptr<float>(), N, H, W, gr.C, Ho, Wo, int(k),
-                        int(stride), int(pad), cur_stream());
+  TORCH_CHECK(xr.C == 3, "im2row expects Cin=3");
+  TORCH_CHECK(patches.dim() == 2 && patches.is_contiguous() &&
+                  patches.scalar_type() == torch::kBFloat16,
+              "patches must be bf16 [M, KP]");
+  const int KP = int(patches.size(1));
+  const int Ho = int((x.size(2) + 2 * pad - k) / stride + 1);
+  const int Wo = int((x.size(3) + 2 * pad - k) / stride + 1);
+  TORCH_CHECK(patches.size(0) == x.size(0) * Ho * Wo, "patches row mismatch");
+  yams::im2row(xr.ptr, patches.data_ptr(), int(x.size(0)), int(x.size(2)),
+               int(x.size(3)), Ho, Wo, int(k), int(stride), int(pad), KP,
+               cur_stream());
 }
 
 // ---- BatchNorm ---------------------------------------------------------
@@ -288,8 +278,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dw_conv_fwd", &dw_conv_fwd);
   m.def("dw_conv_dgrad", &dw_conv_dgrad);
   m.def("dw_conv_wgrad", &dw_conv_wgrad);
-  m.def("stem_conv_fwd", &stem_conv_fwd);
-  m.def("stem_conv_wgrad", &stem_conv_wgrad);
+  m.def("im2row", &im2row);
   m.def("bn_stats", &bn_stats);
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("bn_act_bwd", &bn_act_bwd);

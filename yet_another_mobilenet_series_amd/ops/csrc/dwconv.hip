@@ -229,98 +229,34 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   }
 }
 
-// stem forward: flat vec8 over output channels; 27 taps, x broadcast from L1
-__global__ __launch_bounds__(256) void stem_fwd_kernel(
-    const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
-    ushort_t* __restrict__ y, int N, int H, int W, int Cout, int Ho, int Wo,
-    int k, int stride, int pad, int64_t total8) {
-  const int C8 = Cout >> 3;
+// im2row for the stem (K1): x[N,H,W,3] -> patches[M, KP] with
+// KP = round_up(k*k*3, 8) zero-padded columns, row m = (n,ho,wo).
+// The stem then runs on the MFMA GEMM: y = patches @ w[KP,Cout]^T, and its
+// wgrad on the split-M atb GEMM — no bespoke dense-conv kernels.
+__global__ __launch_bounds__(256) void im2row_kernel(
+    const ushort_t* __restrict__ x, ushort_t* __restrict__ patches,
+    int N, int H, int W, int Ho, int Wo, int k, int stride, int pad,
+    int KP, int64_t total) {
   const int Cin = 3;
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c8 = int(idx % C8);
-    int64_t rest = idx / C8;
-    const int wo = int(rest % Wo);
-    rest /= Wo;
+  for (int64_t m = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       m < total; m += int64_t(gridDim.x) * blockDim.x) {
+    const int wo = int(m % Wo);
+    int64_t rest = m / Wo;
     const int ho = int(rest % Ho);
     const int n = int(rest / Ho);
-    const int co = c8 * 8;
-    float acc[8] = {};
+    ushort_t* dst = &patches[m * KP];
+    int t = 0;
     for (int i = 0; i < k; ++i) {
       const int hi = ho * stride + i - pad;
-      if (hi < 0 || hi >= H) continue;
       for (int j = 0; j < k; ++j) {
         const int wi = wo * stride + j - pad;
-        if (wi < 0 || wi >= W) continue;
+        const bool ok = hi >= 0 && hi < H && wi >= 0 && wi < W;
 #pragma unroll
-        for (int ci = 0; ci < 3; ++ci) {
-          const float xv = bf2f(x[idx4(n, hi, wi, ci, H, W, Cin)]);
-          short8_t wv = *reinterpret_cast<const short8_t*>(
-              &w[((i * k + j) * Cin + ci) * Cout + co]);
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(xv, bf2f(ushort_t(wv[e])), acc[e]);
-        }
+        for (int ci = 0; ci < 3; ++ci)
+          dst[t++] = ok ? x[idx4(n, hi, wi, ci, H, W, Cin)] : ushort_t(0);
       }
     }
-    short8_t out;
-#pragma unroll
-    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
-    *reinterpret_cast<short8_t*>(&y[idx4(n, ho, wo, co, Ho, Wo, Cout)]) = out;
-  }
-}
-
-// stem wgrad: thread = cout lane, 27 fp32 accumulators, LDS mlane fold
-__global__ __launch_bounds__(256) void stem_wgrad_kernel(
-    const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-    float* __restrict__ ws, int N, int H, int W, int Cout, int Ho, int Wo,
-    int k, int stride, int pad, int chunks) {
-  __shared__ float red[256];
-  const int Cin = 3;
-  const int CT = Cout < 256 ? Cout : 256;
-  const int ML = 256 / CT;
-  const int co = int(threadIdx.x) % CT;
-  const int mlane = int(threadIdx.x) / CT;
-  const int chunk = blockIdx.x;
-
-  const int64_t P = int64_t(N) * Ho * Wo;
-  const int64_t per = (P + chunks - 1) / chunks;
-  const int64_t lo = chunk * per;
-  const int64_t hi_ = std::min<int64_t>(lo + per, P);
-
-  float acc[27] = {};  // k=3 fixed for the zoo stem
-  if (co < Cout && mlane < ML) {
-    for (int64_t p = lo + mlane; p < hi_; p += ML) {
-      const int wo = int(p % Wo);
-      int64_t rest = p / Wo;
-      const int ho = int(rest % Ho);
-      const int n = int(rest / Ho);
-      const float g = bf2f(dy[idx4(n, ho, wo, co, Ho, Wo, Cout)]);
-      for (int i = 0; i < k; ++i) {
-        const int hi2 = ho * stride + i - pad;
-        if (hi2 < 0 || hi2 >= H) continue;
-        for (int j = 0; j < k; ++j) {
-          const int wi = wo * stride + j - pad;
-          if (wi < 0 || wi >= W) continue;
-#pragma unroll
-          for (int ci = 0; ci < 3; ++ci)
-            acc[(i * k + j) * Cin + ci] =
-                fmaf(g, bf2f(x[idx4(n, hi2, wi, ci, H, W, Cin)]),
-                     acc[(i * k + j) * Cin + ci]);
-        }
-      }
-    }
-  }
-  float* slab = ws + int64_t(chunk) * (k * k * Cin) * Cout;
-  for (int t = 0; t < k * k * Cin; ++t) {
-    red[threadIdx.x] = acc[t];
-    __syncthreads();
-    if (threadIdx.x < unsigned(CT) && co < Cout) {
-      float v = 0.f;
-      for (int l = 0; l < ML; ++l) v += red[threadIdx.x + l * CT];
-      slab[t * Cout + co] = v;
-    }
-    __syncthreads();
+    for (; t < KP; ++t) dst[t] = ushort_t(0);
   }
 }
 
@@ -416,34 +352,12 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }
 
-void stem_conv_fwd(const void* x, const void* w, void* y, int N, int H,
-                   int W, int Cout, int Ho, int Wo, int k, int stride,
-                   int pad, hipStream_t s) {
-  const int64_t total8 = int64_t(N) * Ho * Wo * (Cout >> 3);
-  stem_fwd_kernel<<<grid_for(total8), 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
-      static_cast<ushort_t*>(y), N, H, W, Cout, Ho, Wo, k, stride, pad,
-      total8);
-}
-
-int stem_wgrad_chunks(int N, int Ho, int Wo) {
-  const int64_t P = int64_t(N) * Ho * Wo;
-  int c = int(std::min<int64_t>((P + 63) / 64, 4096));
-  return c < 1 ? 1 : c;
-}
-
-int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k) {
-  return int64_t(stem_wgrad_chunks(N, Ho, Wo)) * k * k * 3 * Cout;
-}
-
-void stem_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
-                     int N, int H, int W, int Cout, int Ho, int Wo, int k,
-                     int stride, int pad, hipStream_t s) {
-  const int chunks = stem_wgrad_chunks(N, Ho, Wo);
-  stem_wgrad_kernel<<<dim3(chunks), 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(dy), ws,
-      N, H, W, Cout, Ho, Wo, k, stride, pad, chunks);
-  slab_reduce(ws, dwgt, int64_t(k) * k * 3 * Cout, chunks, s);
+void im2row(const void* x, void* patches, int N, int H, int W, int Ho,
+            int Wo, int k, int stride, int pad, int KP, hipStream_t s) {
+  const int64_t total = int64_t(N) * Ho * Wo;
+  im2row_kernel<<<grid_for((total + 3) / 4), 256, 0, s>>>(
+      static_cast<const ushort_t*>(x), static_cast<ushort_t*>(patches),
+      N, H, W, Ho, Wo, k, stride, pad, KP, total);
 }
 
 }  // namespace yams

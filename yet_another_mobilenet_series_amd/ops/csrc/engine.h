@@ -43,15 +43,9 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
 int dw_wgrad_chunks(int N, int Ho, int Wo);
 int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k);
 
-// --- stem dense conv (K1): cin=3, k=3, s=2; w layout [3,3,3,Cout] -------
-void stem_conv_fwd(const void* x, const void* w, void* y,
-                   int N, int H, int W, int Cout, int Ho, int Wo,
-                   int k, int stride, int pad, hipStream_t s);
-void stem_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
-                     int N, int H, int W, int Cout, int Ho, int Wo,
-                     int k, int stride, int pad, hipStream_t s);
-int stem_wgrad_chunks(int N, int Ho, int Wo);
-int64_t stem_wgrad_ws_elems(int N, int Ho, int Wo, int Cout, int k);
+// --- stem (K1) runs as im2row + the GEMM family ------------------------
+void im2row(const void* x, void* patches, int N, int H, int W, int Ho,
+            int Wo, int k, int stride, int pad, int KP, hipStream_t s);
 
 // --- BatchNorm (K5) -----------------------------------------------------
 // training stats over M=N*H*W rows of z[M,C]; ws fp32 [2, chunks, C].

@@ -120,12 +120,31 @@ class _ConvBNActFn(torch.autograd.Function):
             w_packed = _packed(weight, "dw", lambda w: w.view(
                 cout, k * k).t().to(BF16).contiguous())
         elif kind == "stem":
-            w_packed = _packed(weight, "stem", lambda w: w.permute(
-                2, 3, 1, 0).to(BF16).contiguous())
+            # [Cout, 3, k, k] -> [Cout, k*k*3 padded to /8] matching im2row
+            kp = (k * k * cin + 7) // 8 * 8
+            def _pack_stem(w):
+                flat = w.permute(0, 2, 3, 1).reshape(cout, k * k * cin)
+                out = torch.zeros(cout, kp, dtype=BF16, device=w.device)
+                out[:, :k * k * cin] = flat.to(BF16)
+                return out
+            w_packed = _packed(weight, "stem", _pack_stem)
         else:
             raise ValueError(kind)
 
-        z = _conv_forward(x, w_packed, kind, k, s, p)
+        if kind == "stem":
+            n, _, h, wdt = x.shape
+            ho, wo = _conv_out_hw(h, wdt, k, s, p)
+            m = n * ho * wo
+            patches = torch.empty(m, w_packed.shape[1], dtype=BF16,
+                                  device=x.device)
+            e.im2row(x, patches, k, s, p)
+            z = _empty_like_nhwc(x, c=cout, h=ho, w=wo)
+            e.pw_conv(patches.view(m, -1, 1, 1), w_packed,
+                      z.permute(0, 2, 3, 1).reshape(m, cout)
+                      .view(m, cout, 1, 1))
+            ctx.stem_patches = patches
+        else:
+            z = _conv_forward(x, w_packed, kind, k, s, p)
 
         mean = torch.empty(cout, dtype=torch.float32, device=x.device)
         invstd = torch.empty_like(mean)
@@ -196,13 +215,19 @@ class _ConvBNActFn(torch.autograd.Function):
                                   device=x.device)
                 e.dw_conv_wgrad(x, dz, dwp, k, s, p)
                 dweight = dwp.t().reshape(cout, 1, k, k)
-            else:  # stem
-                dwp = torch.empty(k * k * cin, cout, dtype=torch.float32,
+            else:  # stem: wgrad on the split-M GEMM over saved patches
+                patches = ctx.stem_patches
+                m = patches.shape[0]
+                kp = patches.shape[1]
+                dwp = torch.empty(cout, kp, dtype=torch.float32,
                                   device=x.device)
-                e.stem_conv_wgrad(x, dz, dwp, k, s, p)
-                # [k*k*3, Cout] -> [Cout, Cin, k, k]
-                dweight = (dwp.view(k, k, cin, cout)
-                           .permute(3, 2, 0, 1).contiguous())
+                e.pw_conv_wgrad(
+                    dz.permute(0, 2, 3, 1).reshape(m, cout)
+                    .view(m, cout, 1, 1),
+                    patches.view(m, kp, 1, 1), dwp)
+                dweight = (dwp[:, :k * k * cin]
+                           .view(cout, k, k, cin)
+                           .permute(0, 3, 1, 2).contiguous())
 
         dres = dy if ctx.has_residual else None
         return (dx, dweight, dgamma, dbeta, dres) + (None,) * 10
