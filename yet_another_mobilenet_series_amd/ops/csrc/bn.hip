@@ -29,7 +29,7 @@ namespace yams {
 
 namespace {
 
-constexpr int TARGET_BLOCKS = 512;
+constexpr int TARGET_BLOCKS = 1024;
 
 struct RedGeom {
   int CT, ML, c_tiles, chunks;

@@ -244,7 +244,7 @@ __global__ __launch_bounds__(256) void im2row_kernel(
     int64_t rest = m / Wo;
     const int ho = int(rest % Ho);
     const int n = int(rest / Ho);
-    ushort_t* dst = &patches[m * KP];
+    ushort_t buf[32];
     int t = 0;
     for (int i = 0; i < k; ++i) {
       const int hi = ho * stride + i - pad;
@@ -253,10 +253,17 @@ __global__ __launch_bounds__(256) void im2row_kernel(
         const bool ok = hi >= 0 && hi < H && wi >= 0 && wi < W;
 #pragma unroll
         for (int ci = 0; ci < 3; ++ci)
-          dst[t++] = ok ? x[idx4(n, hi, wi, ci, H, W, Cin)] : ushort_t(0);
+          buf[t++] = ok ? x[idx4(n, hi, wi, ci, H, W, Cin)] : ushort_t(0);
       }
     }
-    for (; t < KP; ++t) dst[t] = ushort_t(0);
+    for (; t < KP; ++t) buf[t] = ushort_t(0);
+    ushort_t* dst = &patches[m * KP];
+#pragma unroll
+    for (int v = 0; v < 32; v += 8) {
+      if (v >= KP) break;
+      *reinterpret_cast<short8_t*>(&dst[v]) =
+          *reinterpret_cast<const short8_t*>(&buf[v]);
+    }
   }
 }
 
