@@ -1,0 +1,108 @@
+"""Fused inverted-residual block parity (@gpu): the one-Function fused path
+(BN stats in conv epilogues, BN+act as load-ops, dz as BNBWD load-ops) vs
+the plain fp32 torch composition from the same bf16-rounded inputs."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from yet_another_mobilenet_series_amd.models.mobilenet_base import InvertedResidual
+from yet_another_mobilenet_series_amd.ops.backend import set_engine_mode
+
+pytestmark = pytest.mark.gpu
+
+
+def _rel(a, b):
+    return ((a.float() - b.float()).norm() /
+            b.float().norm().clamp(min=1e-6)).item()
+
+
+CFGS = [
+    # cin, cout, k, s, expand, se_rd, act, use_res
+    dict(cin=24, cout=24, kernel_size=3, stride=1, expand_ch=72,
+         se_reduced_ch=0, act="relu6"),                      # V2-style residual
+    dict(cin=40, cout=40, kernel_size=5, stride=1, expand_ch=120,
+         se_reduced_ch=32, act="relu"),                      # V3 SE block
+    dict(cin=80, cout=112, kernel_size=3, stride=1, expand_ch=480,
+         se_reduced_ch=120, act="hswish"),                   # V3 SE hswish
+    dict(cin=16, cout=16, kernel_size=3, stride=2, expand_ch=16,
+         se_reduced_ch=8, act="relu"),                       # no-expand + SE
+    dict(cin=112, cout=160, kernel_size=5, stride=2, expand_ch=672,
+         se_reduced_ch=168, act="hswish"),                   # k5 s2 SE
+]
+
+
+@pytest.mark.parametrize("cfg", CFGS)
+@pytest.mark.parametrize("training", [True, False])
+def test_fused_block_parity(cfg, training):
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    block = InvertedResidual(**cfg).to(dev).to(
+        memory_format=torch.channels_last)
+    block.train(training)
+    x = (torch.randn(4, cfg["cin"], 28, 28, device=dev).bfloat16()
+         .contiguous(memory_format=torch.channels_last))
+
+    state0 = {k: v.clone() for k, v in block.state_dict().items()}
+
+    # fused HIP path
+    xh = x.clone().requires_grad_(True)
+    y = block(xh)
+    dy = (torch.randn_like(y.float()).bfloat16()
+          .contiguous(memory_format=torch.channels_last))
+    if training:
+        y.backward(dy)
+    g_hip = {n: (p.grad.float().clone() if p.grad is not None else None)
+             for n, p in block.named_parameters()}
+    run_hip = {k: v.clone() for k, v in block.state_dict().items()
+               if "running" in k}
+
+    # fp32 reference composition from the same bf16 values
+    block.load_state_dict(state0)
+    for p in block.parameters():
+        p.grad = None
+    set_engine_mode("reference")
+    try:
+        xr = x.float().clone().requires_grad_(True)
+        yr = block(xr)
+        if training:
+            yr.backward(dy.float())
+    finally:
+        set_engine_mode("hip")
+
+    assert _rel(y.detach(), yr.detach()) < 0.03, "forward mismatch"
+    if training:
+        assert _rel(xh.grad, xr.grad) < 0.08
+        for n, p in block.named_parameters():
+            assert g_hip[n] is not None, f"missing fused grad {n}"
+            e = _rel(g_hip[n], p.grad)
+            # bn biases of linear-BN project layers can be cancellation-noisy
+            denom_small = p.grad.float().norm().item() < 1e-3 * p.numel() ** 0.5
+            assert e < 0.1 or denom_small, (n, e)
+        # running stats updated identically
+        for k, v in block.state_dict().items():
+            if "running" in k:
+                assert _rel(run_hip[k], v) < 0.02, k
+
+
+def test_fused_block_determinism():
+    dev = torch.device("cuda:0")
+
+    def run():
+        torch.manual_seed(3)
+        block = InvertedResidual(cin=40, cout=40, kernel_size=5, stride=1,
+                                 expand_ch=120, se_reduced_ch=32,
+                                 act="hswish").to(dev).to(
+            memory_format=torch.channels_last)
+        block.train()
+        x = (torch.randn(4, 40, 28, 28, device=dev).bfloat16()
+             .contiguous(memory_format=torch.channels_last))
+        x.requires_grad_(True)
+        y = block(x)
+        y.sum().backward()
+        return (y.detach().float().cpu(), x.grad.float().cpu(),
+                block.dw.conv.weight.grad.cpu())
+
+    a, b = run(), run()
+    for t1, t2 in zip(a, b):
+        assert torch.equal(t1, t2)

@@ -91,6 +91,10 @@ class InvertedResidual(nn.Module):
         self.project = ConvBNAct(expand_ch, cout, 1, act="identity", **bn_kw)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.backend import use_hip
+        if use_hip(x):
+            from ..ops.fused_block import fused_inverted_residual
+            return fused_inverted_residual(x, self)
         y = self.expand(x) if self.expand is not None else x
         y = self.dw(y)
         if self.se is not None:

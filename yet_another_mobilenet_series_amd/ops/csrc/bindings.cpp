@@ -65,9 +65,36 @@ Tensor f32_ws(const Tensor& like, int64_t elems) {
   return torch::empty({elems}, like.options().dtype(torch::kFloat32));
 }
 
+yams::LopDesc make_lop(int64_t kind, int64_t act,
+                       const c10::optional<Tensor>& coef,
+                       const c10::optional<Tensor>& z, int C) {
+  yams::LopDesc d;
+  d.kind = int(kind);
+  d.act = int(act);
+  d.C = C;
+  if (kind != 0) {
+    TORCH_CHECK(coef.has_value(), "load-op requires coef tensor");
+    TORCH_CHECK(coef->scalar_type() == torch::kFloat32 &&
+                    coef->is_contiguous(),
+                "lop coef must be fp32 contiguous");
+    TORCH_CHECK(coef->numel() >= (kind == 2 ? 4 : 2) * int64_t(C),
+                "lop coef too small");
+    d.coef = coef->data_ptr<float>();
+  }
+  if (kind == 2) {
+    TORCH_CHECK(z.has_value(), "bnbwd load-op requires z tensor");
+    TORCH_CHECK(z->scalar_type() == torch::kBFloat16, "lop z must be bf16");
+    d.z = z->data_ptr();
+  }
+  return d;
+}
+
 // ---- GEMM family -------------------------------------------------------
 
-void pw_conv(const Tensor& x, const Tensor& w, Tensor& y) {
+void pw_conv(const Tensor& x, const Tensor& w, Tensor& y,
+             int64_t lop_kind, int64_t lop_act,
+             c10::optional<Tensor> lop_coef, c10::optional<Tensor> lop_z,
+             c10::optional<Tensor> stats_ws) {
   auto xr = rows_of(x, "x");
   auto yr = rows_of(y, "y");
   check_w(w);
@@ -75,11 +102,26 @@ void pw_conv(const Tensor& x, const Tensor& w, Tensor& y) {
               "pw_conv weight shape mismatch");
   TORCH_CHECK(xr.M == yr.M, "pw_conv row mismatch");
   TORCH_CHECK(xr.C % 8 == 0, "pw_conv K must be a multiple of 8");
-  yams::pw_conv(xr.ptr, w.data_ptr(), y.data_ptr(), xr.M, xr.C, yr.C,
-                cur_stream());
+  auto lop = make_lop(lop_kind, lop_act, lop_coef, lop_z, xr.C);
+  float* sw = nullptr;
+  if (stats_ws.has_value()) {
+    TORCH_CHECK(stats_ws->numel() >=
+                    2 * int64_t(yams::pw_stats_mblocks(xr.M)) * yr.C,
+                "stats_ws too small");
+    sw = stats_ws->data_ptr<float>();
+  }
+  yams::pw_conv(xr.ptr, w.data_ptr(), y.data_ptr(), xr.M, xr.C, yr.C, lop,
+                sw, cur_stream());
 }
 
-void pw_conv_wgrad(const Tensor& a, const Tensor& b, Tensor& dw) {
+int64_t pw_stats_mblocks(int64_t M) { return yams::pw_stats_mblocks(M); }
+
+void pw_conv_wgrad(const Tensor& a, const Tensor& b, Tensor& dw,
+                   int64_t lopa_kind, int64_t lopa_act,
+                   c10::optional<Tensor> lopa_coef,
+                   c10::optional<Tensor> lopa_z, int64_t lopb_kind,
+                   int64_t lopb_act, c10::optional<Tensor> lopb_coef,
+                   c10::optional<Tensor> lopb_z) {
   auto ar = rows_of(a, "a");
   auto br = rows_of(b, "b");
   TORCH_CHECK(ar.M == br.M, "wgrad row mismatch");
@@ -88,25 +130,39 @@ void pw_conv_wgrad(const Tensor& a, const Tensor& b, Tensor& dw) {
               "wgrad out must be fp32 [N1,N2]");
   auto ws = f32_ws(a, yams::pw_wgrad_ws_elems(ar.M, ar.C, br.C));
   yams::pw_conv_wgrad(ar.ptr, br.ptr, dw.data_ptr<float>(),
-                      ws.data_ptr<float>(), ar.M, ar.C, br.C, cur_stream());
+                      ws.data_ptr<float>(), ar.M, ar.C, br.C,
+                      make_lop(lopa_kind, lopa_act, lopa_coef, lopa_z, ar.C),
+                      make_lop(lopb_kind, lopb_act, lopb_coef, lopb_z, br.C),
+                      cur_stream());
 }
 
 // ---- depthwise ---------------------------------------------------------
 
 void dw_conv_fwd(const Tensor& x, const Tensor& w, Tensor& y, int64_t k,
-                 int64_t stride, int64_t pad) {
+                 int64_t stride, int64_t pad, int64_t lop_kind,
+                 int64_t lop_act, c10::optional<Tensor> lop_coef,
+                 c10::optional<Tensor> lop_z,
+                 c10::optional<Tensor> stats_ws) {
   auto xr = rows_of(x, "x");
   auto yr = rows_of(y, "y");
   check_w(w);
   TORCH_CHECK(xr.C == yr.C && xr.C % 8 == 0, "dw channel mismatch");
+  float* sw = stats_ws.has_value() ? stats_ws->data_ptr<float>() : nullptr;
   yams::dw_conv_fwd(xr.ptr, w.data_ptr(), y.data_ptr(), int(x.size(0)),
                     int(x.size(2)), int(x.size(3)), xr.C, int(y.size(2)),
                     int(y.size(3)), int(k), int(stride), int(pad),
+                    make_lop(lop_kind, lop_act, lop_coef, lop_z, xr.C), sw,
                     cur_stream());
 }
 
+int64_t dw_stats_chunks(int64_t N, int64_t Ho, int64_t Wo, int64_t C) {
+  return yams::dw_stats_chunks(int(N), int(Ho), int(Wo), int(C));
+}
+
 void dw_conv_dgrad(const Tensor& dy, const Tensor& w, Tensor& dx, int64_t k,
-                   int64_t stride, int64_t pad) {
+                   int64_t stride, int64_t pad, int64_t lop_kind,
+                   int64_t lop_act, c10::optional<Tensor> lop_coef,
+                   c10::optional<Tensor> lop_z) {
   auto gr = rows_of(dy, "dy");
   auto xr = rows_of(dx, "dx");
   check_w(w);
@@ -114,11 +170,18 @@ void dw_conv_dgrad(const Tensor& dy, const Tensor& w, Tensor& dx, int64_t k,
   yams::dw_conv_dgrad(gr.ptr, w.data_ptr(), dx.data_ptr(), int(dx.size(0)),
                       int(dx.size(2)), int(dx.size(3)), xr.C,
                       int(dy.size(2)), int(dy.size(3)), int(k), int(stride),
-                      int(pad), cur_stream());
+                      int(pad),
+                      make_lop(lop_kind, lop_act, lop_coef, lop_z, gr.C),
+                      cur_stream());
 }
 
 void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
-                   int64_t k, int64_t stride, int64_t pad) {
+                   int64_t k, int64_t stride, int64_t pad,
+                   int64_t lopx_kind, int64_t lopx_act,
+                   c10::optional<Tensor> lopx_coef,
+                   c10::optional<Tensor> lopx_z, int64_t lopdy_kind,
+                   int64_t lopdy_act, c10::optional<Tensor> lopdy_coef,
+                   c10::optional<Tensor> lopdy_z) {
   auto xr = rows_of(x, "x");
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(dwgt.scalar_type() == torch::kFloat32 && dwgt.is_contiguous(),
@@ -126,9 +189,12 @@ void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
   const int N = int(x.size(0)), H = int(x.size(2)), W = int(x.size(3));
   const int Ho = int(dy.size(2)), Wo = int(dy.size(3));
   auto ws = f32_ws(x, yams::dw_wgrad_ws_elems(N, Ho, Wo, xr.C, int(k)));
-  yams::dw_conv_wgrad(xr.ptr, gr.ptr, dwgt.data_ptr<float>(),
-                      ws.data_ptr<float>(), N, H, W, xr.C, Ho, Wo, int(k),
-                      int(stride), int(pad), cur_stream());
+  yams::dw_conv_wgrad(
+      xr.ptr, gr.ptr, dwgt.data_ptr<float>(), ws.data_ptr<float>(), N, H, W,
+      xr.C, Ho, Wo, int(k), int(stride), int(pad),
+      make_lop(lopx_kind, lopx_act, lopx_coef, lopx_z, xr.C),
+      make_lop(lopdy_kind, lopdy_act, lopdy_coef, lopdy_z, gr.C),
+      cur_stream());
 }
 
 // ---- stem: im2row (conv itself runs on the GEMM kernels) ---------------
@@ -194,7 +260,46 @@ void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
                    fp32_req(beta, "beta"), int(act), training,
                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                    dz.data_ptr(), ws.data_ptr<float>(),
-                   coef.data_ptr<float>(), zr.M, zr.C, cur_stream());
+                   coef.data_ptr<float>(), zr.M, zr.C, true, cur_stream());
+}
+
+// reduce-only backward: dgamma/dbeta + the coef[4C] used by LOP_BNBWD
+// consumers (no dz materialization)
+void bn_bwd_stats(const Tensor& dy, const Tensor& z, const Tensor& mean,
+                  const Tensor& invstd, const Tensor& gamma,
+                  const Tensor& beta, int64_t act, bool training,
+                  Tensor& dgamma, Tensor& dbeta, Tensor& coef) {
+  auto zr = rows_of(z, "z");
+  auto gr = rows_of(dy, "dy");
+  TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
+  TORCH_CHECK(coef.numel() == 4 * zr.C, "coef must be [4C]");
+  auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+  yams::bn_act_bwd(gr.ptr, zr.ptr, fp32_req(mean, "mean"),
+                   fp32_req(invstd, "invstd"), fp32_req(gamma, "gamma"),
+                   fp32_req(beta, "beta"), int(act), training,
+                   dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                   nullptr, ws.data_ptr<float>(), coef.data_ptr<float>(),
+                   zr.M, zr.C, false, cur_stream());
+}
+
+// combine conv-epilogue stats slabs [2, splits, C] into mean/invstd/ab
+void bn_finalize(const Tensor& ws, int64_t splits, Tensor& mean,
+                 Tensor& invstd, Tensor& ab, const Tensor& gamma,
+                 const Tensor& beta, c10::optional<Tensor> running_mean,
+                 c10::optional<Tensor> running_var, double momentum,
+                 double eps, int64_t M) {
+  const int C = int(mean.numel());
+  float* rm = nullptr;
+  float* rv = nullptr;
+  if (running_mean.has_value()) {
+    rm = running_mean->data_ptr<float>();
+    rv = running_var->data_ptr<float>();
+  }
+  yams::bn_finalize(fp32_req(ws, "ws"), int(splits),
+                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                    ab.data_ptr<float>(), fp32_req(gamma, "gamma"),
+                    fp32_req(beta, "beta"), rm, rv, M, C, float(momentum),
+                    float(eps), cur_stream());
 }
 
 // ---- misc --------------------------------------------------------------
@@ -224,25 +329,37 @@ void bias_act_bwd(const Tensor& dy, const Tensor& z,
                      zr.M, zr.C, cur_stream());
 }
 
-void rowscale_fwd(const Tensor& x, const Tensor& scale, Tensor& y) {
+void rowscale_fwd(const Tensor& x, const Tensor& scale, Tensor& y,
+                  int64_t lop_kind, int64_t lop_act,
+                  c10::optional<Tensor> lop_coef,
+                  c10::optional<Tensor> lop_z) {
   auto xr = rows_of(x, "x");
   yams::rowscale_fwd(xr.ptr, scale.data_ptr(), y.data_ptr(),
                      int(x.size(0)), int(x.size(2) * x.size(3)), xr.C,
+                     make_lop(lop_kind, lop_act, lop_coef, lop_z, xr.C),
                      cur_stream());
 }
 
 void rowscale_bwd(const Tensor& dy, const Tensor& x, const Tensor& scale,
-                  Tensor& dx, Tensor& dscale) {
+                  Tensor& dx, Tensor& dscale, int64_t lop_kind,
+                  int64_t lop_act, c10::optional<Tensor> lop_coef,
+                  c10::optional<Tensor> lop_z) {
   auto xr = rows_of(x, "x");
   yams::rowscale_bwd(dy.data_ptr(), xr.ptr, scale.data_ptr(), dx.data_ptr(),
                      dscale.data_ptr(), int(x.size(0)),
-                     int(x.size(2) * x.size(3)), xr.C, cur_stream());
+                     int(x.size(2) * x.size(3)), xr.C,
+                     make_lop(lop_kind, lop_act, lop_coef, lop_z, xr.C),
+                     cur_stream());
 }
 
-void avgpool_fwd(const Tensor& x, Tensor& out) {
+void avgpool_fwd(const Tensor& x, Tensor& out, int64_t lop_kind,
+                 int64_t lop_act, c10::optional<Tensor> lop_coef,
+                 c10::optional<Tensor> lop_z) {
   auto xr = rows_of(x, "x");
   yams::avgpool_fwd(xr.ptr, out.data_ptr(), int(x.size(0)),
-                    int(x.size(2) * x.size(3)), xr.C, cur_stream());
+                    int(x.size(2) * x.size(3)), xr.C,
+                    make_lop(lop_kind, lop_act, lop_coef, lop_z, xr.C),
+                    cur_stream());
 }
 
 void avgpool_bwd(const Tensor& dpooled, Tensor& dx) {
@@ -273,20 +390,51 @@ void label_smooth_ce(const Tensor& logits, const Tensor& target,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X (gfx950) CDNA4 kernel engine for the MobileNet zoo";
-  m.def("pw_conv", &pw_conv, "y[M,N] = x[M,K] @ w[N,K]^T (MFMA bf16)");
-  m.def("pw_conv_wgrad", &pw_conv_wgrad, "dw[N1,N2] = a^T @ b (split-M)");
-  m.def("dw_conv_fwd", &dw_conv_fwd);
-  m.def("dw_conv_dgrad", &dw_conv_dgrad);
-  m.def("dw_conv_wgrad", &dw_conv_wgrad);
+  using namespace pybind11::literals;
+  auto nolop = py::arg_v("lop_kind", 0);
+  m.def("pw_conv", &pw_conv, "y[M,N] = x[M,K] @ w[N,K]^T (MFMA bf16)",
+        "x"_a, "w"_a, "y"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
+        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt,
+        "stats_ws"_a = c10::nullopt);
+  m.def("pw_stats_mblocks", &pw_stats_mblocks);
+  m.def("pw_conv_wgrad", &pw_conv_wgrad, "dw[N1,N2] = a^T @ b (split-M)",
+        "a"_a, "b"_a, "dw"_a, "lopa_kind"_a = 0, "lopa_act"_a = 0,
+        "lopa_coef"_a = c10::nullopt, "lopa_z"_a = c10::nullopt,
+        "lopb_kind"_a = 0, "lopb_act"_a = 0, "lopb_coef"_a = c10::nullopt,
+        "lopb_z"_a = c10::nullopt);
+  m.def("dw_conv_fwd", &dw_conv_fwd, "x"_a, "w"_a, "y"_a, "k"_a, "stride"_a,
+        "pad"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
+        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt,
+        "stats_ws"_a = c10::nullopt);
+  m.def("dw_stats_chunks", &dw_stats_chunks);
+  m.def("dw_conv_dgrad", &dw_conv_dgrad, "dy"_a, "w"_a, "dx"_a, "k"_a,
+        "stride"_a, "pad"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
+        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt);
+  m.def("dw_conv_wgrad", &dw_conv_wgrad, "x"_a, "dy"_a, "dw"_a, "k"_a,
+        "stride"_a, "pad"_a, "lopx_kind"_a = 0, "lopx_act"_a = 0,
+        "lopx_coef"_a = c10::nullopt, "lopx_z"_a = c10::nullopt,
+        "lopdy_kind"_a = 0, "lopdy_act"_a = 0,
+        "lopdy_coef"_a = c10::nullopt, "lopdy_z"_a = c10::nullopt);
   m.def("im2row", &im2row);
   m.def("bn_stats", &bn_stats);
+  m.def("bn_finalize", &bn_finalize, "ws"_a, "splits"_a, "mean"_a,
+        "invstd"_a, "ab"_a, "gamma"_a, "beta"_a,
+        "running_mean"_a = c10::nullopt, "running_var"_a = c10::nullopt,
+        "momentum"_a = 0.1, "eps"_a = 1e-5, "M"_a = 1);
   m.def("bn_act_fwd", &bn_act_fwd);
   m.def("bn_act_bwd", &bn_act_bwd);
+  m.def("bn_bwd_stats", &bn_bwd_stats);
   m.def("bias_act_fwd", &bias_act_fwd);
   m.def("bias_act_bwd", &bias_act_bwd);
-  m.def("rowscale_fwd", &rowscale_fwd);
-  m.def("rowscale_bwd", &rowscale_bwd);
-  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("rowscale_fwd", &rowscale_fwd, "x"_a, "scale"_a, "y"_a,
+        "lop_kind"_a = 0, "lop_act"_a = 0, "lop_coef"_a = c10::nullopt,
+        "lop_z"_a = c10::nullopt);
+  m.def("rowscale_bwd", &rowscale_bwd, "dy"_a, "x"_a, "scale"_a, "dx"_a,
+        "dscale"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
+        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt);
+  m.def("avgpool_fwd", &avgpool_fwd, "x"_a, "out"_a, "lop_kind"_a = 0,
+        "lop_act"_a = 0, "lop_coef"_a = c10::nullopt,
+        "lop_z"_a = c10::nullopt);
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("avgpool_bwd_add", &avgpool_bwd_add);
   m.def("label_smooth_ce", &label_smooth_ce);

@@ -344,6 +344,15 @@ void bn_stats(const void* z, float* mean, float* invstd, float* ab,
       g.chunks, momentum, eps);
 }
 
+void bn_finalize(const float* ws, int splits, float* mean, float* invstd,
+                 float* ab, const float* gamma, const float* beta,
+                 float* running_mean, float* running_var, int64_t M, int C,
+                 float momentum, float eps, hipStream_t s) {
+  bn_stats_finalize_kernel<<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+      ws, mean, invstd, ab, gamma, beta, running_mean, running_var, M, C,
+      splits, momentum, eps);
+}
+
 void bn_act_fwd(const void* z, const float* ab, int act, void* y,
                 const void* residual, int64_t M, int C, hipStream_t s) {
   const int64_t total8 = M * (C >> 3);
@@ -364,7 +373,7 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
                 void* dz, float* ws, float* coef, int64_t M, int C,
-                hipStream_t s) {
+                bool apply, hipStream_t s) {
   const RedGeom g = red_geom(M, C);
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* zp = static_cast<const ushort_t*>(z);
@@ -381,8 +390,9 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
       bn_bwd_finalize_kernel<false><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
           ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
           g.chunks);
-    bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
-        gp, zp, coef, dzp, total8, C >> 3, C);
+    if (apply)
+      bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
+          gp, zp, coef, dzp, total8, C >> 3, C);
   });
 }
 

@@ -65,6 +65,66 @@ DEV_INLINE float act_bwd(float v) {
   return 1.f;
 }
 
+// runtime-act variants (wave-uniform switch: used by load-op paths where
+// templating every act would explode the kernel instantiation count)
+DEV_INLINE float act_fwd_rt(float v, int act) {
+  switch (act) {
+    case ACT_RELU: return fmaxf(v, 0.f);
+    case ACT_RELU6: return fminf(fmaxf(v, 0.f), 6.f);
+    case ACT_HSWISH:
+      return v * fminf(fmaxf(v + 3.f, 0.f), 6.f) * (1.f / 6.f);
+    case ACT_HSIGMOID: return fminf(fmaxf(v + 3.f, 0.f), 6.f) * (1.f / 6.f);
+    default: return v;
+  }
+}
+
+DEV_INLINE float act_bwd_rt(float v, int act) {
+  switch (act) {
+    case ACT_RELU: return v > 0.f ? 1.f : 0.f;
+    case ACT_RELU6: return (v > 0.f && v < 6.f) ? 1.f : 0.f;
+    case ACT_HSWISH:
+      return v <= -3.f ? 0.f : (v >= 3.f ? 1.f : (2.f * v + 3.f) * (1.f / 6.f));
+    case ACT_HSIGMOID: return (v > -3.f && v < 3.f) ? (1.f / 6.f) : 0.f;
+    default: return 1.f;
+  }
+}
+
+// Load-op: per-element transform applied while STAGING an operand stream
+// (the fusion mechanism: BN-normalize+activation on forward loads,
+// BN-backward dz computation on gradient loads — SURVEY.md §7 stage 4).
+#define LOP_NONE 0
+#define LOP_BNACT 1   // v' = act(a[c]*v + b[c]);           coef = ab[2C]
+#define LOP_BNBWD 2   // v' = dz(dy=v, z=zv);               coef = [4C] {a,b,K0,K2}
+
+struct Lop {
+  int kind = LOP_NONE;
+  int act = ACT_IDENTITY;
+  const float* coef = nullptr;   // [2C] or [4C], SoA
+  const ushort_t* z = nullptr;   // paired z stream for LOP_BNBWD
+  int C = 0;
+};
+
+DEV_INLINE float lop_apply(const Lop& L, float v, float zv, int c) {
+  if (L.kind == LOP_BNACT)
+    return act_fwd_rt(fmaf(v, L.coef[c], L.coef[L.C + c]), L.act);
+  if (L.kind == LOP_BNBWD) {
+    const float a = L.coef[c];
+    const float da = v * act_bwd_rt(fmaf(zv, a, L.coef[L.C + c]), L.act);
+    return fmaf(a, da, L.coef[2 * L.C + c]) - L.coef[3 * L.C + c] * zv;
+  }
+  return v;
+}
+
+// bf16 load + lop for one element of a stream at flat offset `off`,
+// channel `c` (zv fetched only when the lop needs it)
+DEV_INLINE float lop_load(const Lop& L, const ushort_t* p, int64_t off,
+                          int c) {
+  const float v = bf2f(p[off]);
+  if (L.kind == LOP_NONE) return v;
+  const float zv = L.kind == LOP_BNBWD ? bf2f(L.z[off]) : 0.f;
+  return lop_apply(L, v, zv, c);
+}
+
 // dispatch an `auto fn = [&](auto act_tag){...}` over a runtime act code
 #define ACT_SWITCH(ACT_CODE, ...)                                      \
   switch (ACT_CODE) {                                                  \

@@ -21,79 +21,152 @@ namespace {
 
 constexpr int MAX_BLOCKS = 4096;
 
+inline Lop to_lop(const LopDesc& d) {
+  Lop L;
+  L.kind = d.kind;
+  L.act = d.act;
+  L.coef = d.coef;
+  L.z = static_cast<const ushort_t*>(d.z);
+  L.C = d.C;
+  return L;
+}
+
 // forward: 4 consecutive wo outputs per thread share the x row window in
-// registers (k*k -> ~k*(k+3)/4 x loads per output).  FLIP reuses the same
-// kernel as the stride-1 dgrad (correlation with the flipped filter).
-template <int K, int S, bool FLIP>
+// registers.  Grid = (c8_tiles x chunks) so the block owns a channel range
+// and can fold output BN-stats partials (STATS) through LDS; lopx fuses
+// the producer's BN+act into the x loads.  FLIP reuses the kernel as the
+// stride-1 dgrad (correlation with the flipped filter; lopx then is the
+// BNBWD load-op on dy).
+template <int K, int S, bool FLIP, bool STATS>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
-    int pad, int64_t total8) {
+    int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
+  __shared__ float red[STATS ? 256 * 8 : 8];
   constexpr int XW = 3 * S + K;
   const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
+  const int chunk = blockIdx.x / c_tiles;
+  const int mlane = int(threadIdx.x) / CT8;
+  const int c = c8 * 8;
+
   const int Wo4 = (Wo + 3) / 4;
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c8 = int(idx % C8);
-    int64_t rest = idx / C8;
-    const int w4 = int(rest % Wo4);
-    rest /= Wo4;
-    const int ho = int(rest % Ho);
-    const int n = int(rest / Ho);
-    const int c = c8 * 8;
-    const int wo0 = w4 * 4;
-    float acc[4][8] = {};
+  const int64_t P = int64_t(N) * Ho * Wo4;
+  const int64_t per = (P + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi_ = std::min<int64_t>(lo + per, P);
+
+  float s1[8] = {}, s2[8] = {};
+  if (c8 < C8 && mlane < ML) {
+    for (int64_t p = lo + mlane; p < hi_; p += ML) {
+      const int w4 = int(p % Wo4);
+      int64_t rest = p / Wo4;
+      const int ho = int(rest % Ho);
+      const int n = int(rest / Ho);
+      const int wo0 = w4 * 4;
+      float acc[4][8] = {};
 #pragma unroll
-    for (int i = 0; i < K; ++i) {
-      const int hi = ho * S + i - pad;
-      if (hi < 0 || hi >= H) continue;
-      float xr[XW][8];
-      const int wbase = wo0 * S - pad;
+      for (int i = 0; i < K; ++i) {
+        const int hi = ho * S + i - pad;
+        if (hi < 0 || hi >= H) continue;
+        float xr[XW][8];
+        const int wbase = wo0 * S - pad;
 #pragma unroll
-      for (int t = 0; t < XW; ++t) {
-        const int wi = wbase + t;
-        if (wi >= 0 && wi < W) {
-          short8_t v = *reinterpret_cast<const short8_t*>(
-              &x[idx4(n, hi, wi, c, H, W, C)]);
+        for (int t = 0; t < XW; ++t) {
+          const int wi = wbase + t;
+          if (wi >= 0 && wi < W) {
+            const int64_t off = idx4(n, hi, wi, c, H, W, C);
+            short8_t v = *reinterpret_cast<const short8_t*>(&x[off]);
+            if (lopx.kind == LOP_NONE) {
 #pragma unroll
-          for (int e = 0; e < 8; ++e) xr[t][e] = bf2f(ushort_t(v[e]));
-        } else {
+              for (int e = 0; e < 8; ++e) xr[t][e] = bf2f(ushort_t(v[e]));
+            } else if (lopx.kind == LOP_BNACT) {
 #pragma unroll
-          for (int e = 0; e < 8; ++e) xr[t][e] = 0.f;
+              for (int e = 0; e < 8; ++e)
+                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(v[e])), 0.f, c + e);
+            } else {
+              short8_t zv = *reinterpret_cast<const short8_t*>(&lopx.z[off]);
+#pragma unroll
+              for (int e = 0; e < 8; ++e)
+                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(v[e])),
+                                     bf2f(ushort_t(zv[e])), c + e);
+            }
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) xr[t][e] = 0.f;
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < K; ++j) {
+          const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+          short8_t wv = *reinterpret_cast<const short8_t*>(&w[tap * C + c]);
+          float wf[8];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) wf[e] = bf2f(ushort_t(wv[e]));
+#pragma unroll
+          for (int q = 0; q < 4; ++q)
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              acc[q][e] = fmaf(xr[q * S + j][e], wf[e], acc[q][e]);
         }
       }
 #pragma unroll
-      for (int j = 0; j < K; ++j) {
-        const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-        short8_t wv = *reinterpret_cast<const short8_t*>(&w[tap * C + c]);
-        float wf[8];
+      for (int q = 0; q < 4; ++q) {
+        if (wo0 + q >= Wo) break;
+        short8_t out;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) wf[e] = bf2f(ushort_t(wv[e]));
-#pragma unroll
-        for (int q = 0; q < 4; ++q)
-#pragma unroll
-          for (int e = 0; e < 8; ++e)
-            acc[q][e] = fmaf(xr[q * S + j][e], wf[e], acc[q][e]);
+        for (int e = 0; e < 8; ++e) {
+          const ushort_t o = f2bf(acc[q][e]);
+          out[e] = short(o);
+          if (STATS) {
+            const float v = bf2f(o);
+            s1[e] += v;
+            s2[e] = fmaf(v, v, s2[e]);
+          }
+        }
+        *reinterpret_cast<short8_t*>(
+            &y[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) = out;
       }
     }
+  }
+  if (STATS) {
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      if (wo0 + q >= Wo) break;
-      short8_t out;
+    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s1[e];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT8) && c8 < C8) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[q][e]));
-      *reinterpret_cast<short8_t*>(
-          &y[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) = out;
+      for (int e = 0; e < 8; ++e) {
+        float t = 0.f;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 8 + e];
+        stats_ws[int64_t(chunk) * C + c + e] = t;
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s2[e];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT8) && c8 < C8) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float t = 0.f;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 8 + e];
+        stats_ws[(int64_t(chunks) + chunk) * C + c + e] = t;
+      }
     }
   }
 }
 
-// dgrad for stride 2 (stride 1 routes through dw_fwd_kernel<K,1,FLIP=true>)
+// dgrad for stride 2 (stride 1 routes through dw_fwd_kernel FLIP=true)
 template <int K>
 __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
-    int stride, int pad, int64_t total8) {
+    int stride, int pad, Lop lopdy, int64_t total8) {
   const int C8 = C >> 3;
   for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
        idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
@@ -117,13 +190,23 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
         if (wnum < 0 || wnum % stride) continue;
         const int wo = wnum / stride;
         if (wo >= Wo) continue;
-        short8_t gv = *reinterpret_cast<const short8_t*>(
-            &dy[idx4(n, ho, wo, c, Ho, Wo, C)]);
+        const int64_t off = idx4(n, ho, wo, c, Ho, Wo, C);
+        short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
         short8_t wv = *reinterpret_cast<const short8_t*>(
             &w[(i * K + j) * C + c]);
+        if (lopdy.kind == LOP_NONE) {
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          acc[e] = fmaf(bf2f(ushort_t(gv[e])), bf2f(ushort_t(wv[e])), acc[e]);
+          for (int e = 0; e < 8; ++e)
+            acc[e] = fmaf(bf2f(ushort_t(gv[e])), bf2f(ushort_t(wv[e])),
+                          acc[e]);
+        } else {
+          short8_t zv = *reinterpret_cast<const short8_t*>(&lopdy.z[off]);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc[e] = fmaf(lop_apply(lopdy, bf2f(ushort_t(gv[e])),
+                                    bf2f(ushort_t(zv[e])), c + e),
+                          bf2f(ushort_t(wv[e])), acc[e]);
+        }
       }
     }
     short8_t out;
@@ -140,7 +223,7 @@ template <int K, int S>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
-    int pad, int chunks) {
+    int pad, Lop lopx, Lop lopdy, int chunks) {
   __shared__ float red[256 * 4];
   constexpr int XW = 3 * S + K;  // x window for 4 outputs
   const int C4 = C >> 2;
@@ -170,10 +253,18 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (wo0 + q < Wo) {
-          short4_t g4 = *reinterpret_cast<const short4_t*>(
-              &dy[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]);
+          const int64_t off = idx4(n, ho, wo0 + q, c, Ho, Wo, C);
+          short4_t g4 = *reinterpret_cast<const short4_t*>(&dy[off]);
+          if (lopdy.kind == LOP_NONE) {
 #pragma unroll
-          for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
+            for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
+          } else {
+            short4_t z4 = *reinterpret_cast<const short4_t*>(&lopdy.z[off]);
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+              g[q][e] = lop_apply(lopdy, bf2f(ushort_t(g4[e])),
+                                  bf2f(ushort_t(z4[e])), c + e);
+          }
         } else {
 #pragma unroll
           for (int e = 0; e < 4; ++e) g[q][e] = 0.f;
@@ -189,10 +280,16 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
           if (wi >= 0 && wi < W) {
-            short4_t x4 = *reinterpret_cast<const short4_t*>(
-                &x[idx4(n, hi2, wi, c, H, W, C)]);
+            const int64_t off = idx4(n, hi2, wi, c, H, W, C);
+            short4_t x4 = *reinterpret_cast<const short4_t*>(&x[off]);
+            if (lopx.kind == LOP_NONE) {
 #pragma unroll
-            for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
+              for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
+            } else {
+#pragma unroll
+              for (int e = 0; e < 4; ++e)
+                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(x4[e])), 0.f, c + e);
+            }
           } else {
 #pragma unroll
             for (int e = 0; e < 4; ++e) xr[t][e] = 0.f;
@@ -273,53 +370,72 @@ int grid_for(int64_t total) {
 
 }  // namespace
 
+int dw_stats_chunks(int N, int Ho, int Wo, int C) {
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
+  int64_t want = 2048 / c_tiles;
+  int64_t maxc = P;
+  int c = int(std::min(std::max<int64_t>(want, 1), maxc));
+  return c < 1 ? 1 : c;
+}
+
 void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
                  int C, int Ho, int Wo, int k, int stride, int pad,
-                 hipStream_t s) {
-  const int64_t total8 = int64_t(N) * Ho * ((Wo + 3) / 4) * (C >> 3);
+                 LopDesc lopx, float* stats_ws, hipStream_t s) {
   auto* xp = static_cast<const ushort_t*>(x);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
-  dim3 g(grid_for(total8));
-  if (k == 3 && stride == 1)
-    dw_fwd_kernel<3, 1, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
-                                                 Wo, pad, total8);
-  else if (k == 3 && stride == 2)
-    dw_fwd_kernel<3, 2, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
-                                                 Wo, pad, total8);
-  else if (k == 5 && stride == 1)
-    dw_fwd_kernel<5, 1, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
-                                                 Wo, pad, total8);
-  else
-    dw_fwd_kernel<5, 2, false><<<g, 256, 0, s>>>(xp, wp, yp, N, H, W, C, Ho,
-                                                 Wo, pad, total8);
+  const Lop L = to_lop(lopx);
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int chunks = dw_stats_chunks(N, Ho, Wo, C);
+  dim3 g(c_tiles * chunks);
+#define DWF_CASE(KK, SS)                                                   \
+  if (k == KK && stride == SS) {                                           \
+    if (stats_ws != nullptr)                                               \
+      dw_fwd_kernel<KK, SS, false, true><<<g, 256, 0, s>>>(                \
+          xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);       \
+    else                                                                   \
+      dw_fwd_kernel<KK, SS, false, false><<<g, 256, 0, s>>>(               \
+          xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, nullptr, chunks);        \
+    return;                                                                \
+  }
+  DWF_CASE(3, 1) DWF_CASE(3, 2) DWF_CASE(5, 1) DWF_CASE(5, 2)
+#undef DWF_CASE
 }
 
 void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
                    int W, int C, int Ho, int Wo, int k, int stride, int pad,
-                   hipStream_t s) {
+                   LopDesc lopdy, hipStream_t s) {
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* xp = static_cast<ushort_t*>(dx);
+  const Lop L = to_lop(lopdy);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
-    const int64_t total8 = int64_t(N) * H * ((W + 3) / 4) * (C >> 3);
-    dim3 g(grid_for(total8));
+    const int C8 = C >> 3;
+    const int CT8 = C8 < 256 ? C8 : 256;
+    const int c_tiles = (C8 + CT8 - 1) / CT8;
+    const int chunks = dw_stats_chunks(N, H, W, C);
+    dim3 g(c_tiles * chunks);
     if (k == 3)
-      dw_fwd_kernel<3, 1, true><<<g, 256, 0, s>>>(gp, wp, xp, N, Ho, Wo, C,
-                                                  H, W, pad, total8);
+      dw_fwd_kernel<3, 1, true, false><<<g, 256, 0, s>>>(
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);
     else
-      dw_fwd_kernel<5, 1, true><<<g, 256, 0, s>>>(gp, wp, xp, N, Ho, Wo, C,
-                                                  H, W, pad, total8);
+      dw_fwd_kernel<5, 1, true, false><<<g, 256, 0, s>>>(
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);
     return;
   }
   const int64_t total8 = int64_t(N) * H * W * (C >> 3);
   if (k == 3)
     dw_dgrad_s2_kernel<3><<<grid_for(total8), 256, 0, s>>>(
-        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
+        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);
   else
     dw_dgrad_s2_kernel<5><<<grid_for(total8), 256, 0, s>>>(
-        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, total8);
+        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);
 }
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
@@ -336,26 +452,29 @@ int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k) {
 
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo, int k,
-                   int stride, int pad, hipStream_t s) {
+                   int stride, int pad, LopDesc lopx, LopDesc lopdy,
+                   hipStream_t s) {
   const int C4 = C >> 2;
   const int CT4 = C4 < 256 ? C4 : 256;
   const int c_tiles = (C4 + CT4 - 1) / CT4;
   const int chunks = (dw_wgrad_chunks(N, Ho, Wo) + c_tiles - 1) / c_tiles;
   auto* xp = static_cast<const ushort_t*>(x);
   auto* gp = static_cast<const ushort_t*>(dy);
+  const Lop Lx = to_lop(lopx);
+  const Lop Ldy = to_lop(lopdy);
   dim3 grid(c_tiles * chunks);
   if (k == 3 && stride == 1)
     dw_wgrad_kernel<3, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, chunks);
+                                               Wo, pad, Lx, Ldy, chunks);
   else if (k == 3 && stride == 2)
     dw_wgrad_kernel<3, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, chunks);
+                                               Wo, pad, Lx, Ldy, chunks);
   else if (k == 5 && stride == 1)
     dw_wgrad_kernel<5, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, chunks);
+                                               Wo, pad, Lx, Ldy, chunks);
   else
     dw_wgrad_kernel<5, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, chunks);
+                                               Wo, pad, Lx, Ldy, chunks);
   slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }
 

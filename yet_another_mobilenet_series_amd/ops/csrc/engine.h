@@ -10,17 +10,35 @@
 
 namespace yams {
 
+// Host-side descriptor of a load-op (see common.h Lop): a per-element
+// transform applied while an operand stream is staged.  kind 0 = none,
+// 1 = bnact (coef = ab[2C]), 2 = bnbwd (coef = [4C] {a,b,K0,K2}, z = the
+// paired conv-output stream with the main stream's layout).
+struct LopDesc {
+  int kind = 0;
+  int act = 0;
+  const float* coef = nullptr;
+  const void* z = nullptr;
+  int C = 0;
+};
+
 // --- GEMM family (pointwise conv K2, FC K10) ----------------------------
 // y[M,N] = x[M,K] @ w[N,K]^T   (both operands K-major; w is the natural
-// [Cout, Cin] torch layout for forward, its transpose for dgrad)
+// [Cout, Cin] torch layout for forward, its transpose for dgrad).
+// lopa transforms the A stream (channel = K index).  stats_ws, if
+// non-null, receives per-m-block partial column sums/sumsq of the OUTPUT
+// ([2, ceil(M/128), N] fp32) for BN-stats fusion; combine with
+// bn_finalize(splits = pw_stats_mblocks(M)).
 void pw_conv(const void* x, const void* w, void* y,
-             int64_t M, int K, int N, hipStream_t s);
+             int64_t M, int K, int N, LopDesc lopa, float* stats_ws,
+             hipStream_t s);
+int pw_stats_mblocks(int64_t M);
 
 // dw[N1,N2] = a[M,N1]^T @ b[M,N2], fp32 out, deterministic split-M slabs
-// workspace: caller-allocated fp32 of size splits(M)*N1*N2 (see
-// pw_wgrad_workspace_elems)
+// (lopa/lopb transform the streams; channel = the N1/N2 index)
 void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
-                   int64_t M, int N1, int N2, hipStream_t s);
+                   int64_t M, int N1, int N2, LopDesc lopa, LopDesc lopb,
+                   hipStream_t s);
 int pw_wgrad_splits(int64_t M, int N1, int N2);
 int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2);
 
@@ -29,17 +47,21 @@ void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s);
 
 // --- depthwise conv (K3/K4) --------------------------------------------
-// w layout [k*k, C] (C fastest)
+// w layout [k*k, C] (C fastest).  lopx transforms the input stream;
+// stats_ws ([2, dw_stats_chunks(...), C]) receives output stats partials.
 void dw_conv_fwd(const void* x, const void* w, void* y,
                  int N, int H, int W, int C, int Ho, int Wo,
-                 int k, int stride, int pad, hipStream_t s);
+                 int k, int stride, int pad, LopDesc lopx, float* stats_ws,
+                 hipStream_t s);
+int dw_stats_chunks(int N, int Ho, int Wo, int C);
 void dw_conv_dgrad(const void* dy, const void* w, void* dx,
                    int N, int H, int W, int C, int Ho, int Wo,
-                   int k, int stride, int pad, hipStream_t s);
+                   int k, int stride, int pad, LopDesc lopdy, hipStream_t s);
 // dwgt fp32 [k*k, C]; ws fp32 [chunks, k*k, C]
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo,
-                   int k, int stride, int pad, hipStream_t s);
+                   int k, int stride, int pad, LopDesc lopx, LopDesc lopdy,
+                   hipStream_t s);
 int dw_wgrad_chunks(int N, int Ho, int Wo);
 int64_t dw_wgrad_ws_elems(int N, int Ho, int Wo, int C, int k);
 
@@ -58,16 +80,25 @@ void bn_stats(const void* z, float* mean, float* invstd, float* ab,
 int bn_stats_chunks(int64_t M);
 int64_t bn_stats_ws_elems(int64_t M, int C);
 
+// finalize-only entry: combine sum/sumsq slabs ws[2, splits, C] produced
+// by a conv epilogue into mean/invstd/ab (+running update)
+void bn_finalize(const float* ws, int splits, float* mean, float* invstd,
+                 float* ab, const float* gamma, const float* beta,
+                 float* running_mean, float* running_var, int64_t M, int C,
+                 float momentum, float eps, hipStream_t s);
+
 // y = act(a*z + b) [+ residual], ab as produced by bn_stats
 void bn_act_fwd(const void* z, const float* ab, int act, void* y,
                 const void* residual, int64_t M, int C, hipStream_t s);
 
 // backward: dgamma/dbeta + dz; ws fp32 [2, chunks, C]; coef fp32 [4C]
+// apply=false computes only dgamma/dbeta and the coef[4C] (for LOP_BNBWD
+// consumers); dz may be null then
 void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
                 void* dz, float* ws, float* coef, int64_t M, int C,
-                hipStream_t s);
+                bool apply, hipStream_t s);
 
 // --- small fused elementwise/reduction kernels --------------------------
 // y = act(z + bias) on [M, C] (bias nullable)
@@ -80,17 +111,17 @@ void bias_act_bwd(const void* dy, const void* z, const float* bias, int act,
 int bias_bwd_chunks(int64_t M);
 int64_t bias_bwd_ws_elems(int64_t M, int C);
 
-// y[n,hw,c] = x[n,hw,c] * scale[n,c]
+// y[n,hw,c] = lop(x[n,hw,c]) * scale[n,c]
 void rowscale_fwd(const void* x, const void* scale, void* y,
-                  int N, int HW, int C, hipStream_t s);
-// dx = dy*scale ; dscale[n,c] = sum_hw dy*x
+                  int N, int HW, int C, LopDesc lopx, hipStream_t s);
+// dx = dy*scale ; dscale[n,c] = sum_hw dy*lop(x)
 void rowscale_bwd(const void* dy, const void* x, const void* scale,
-                  void* dx, void* dscale, int N, int HW, int C,
+                  void* dx, void* dscale, int N, int HW, int C, LopDesc lopx,
                   hipStream_t s);
 
-// out[n,c] = mean_hw x[n,hw,c]
+// out[n,c] = mean_hw lop(x[n,hw,c])
 void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
-                 hipStream_t s);
+                 LopDesc lopx, hipStream_t s);
 // dx[n,hw,c] = dpooled[n,c]/HW   (accumulate=true adds into dx)
 void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,
                  bool accumulate, hipStream_t s);

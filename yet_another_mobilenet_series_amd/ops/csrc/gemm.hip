@@ -29,9 +29,21 @@ constexpr int BM = 128, BN = 64, BK = 32;
 constexpr int LDA = BK + 8;  // +16B pad: conflict-free b128 column reads
 constexpr int FRAG_M = 4, FRAG_N = 2;  // 16x16 frags per wave
 
+inline Lop to_lop(const LopDesc& d) {
+  Lop L;
+  L.kind = d.kind;
+  L.act = d.act;
+  L.coef = d.coef;
+  L.z = static_cast<const ushort_t*>(d.z);
+  L.C = d.C;
+  return L;
+}
+
+template <bool STATS>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
-    ushort_t* __restrict__ C, int64_t M, int K, int N) {
+    ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA,
+    float* __restrict__ stats_ws) {
   __shared__ ushort_t Alds[BM * LDA];
   __shared__ ushort_t Blds[BN * LDA];
 
@@ -52,21 +64,49 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
   const int b_row = tid >> 2, b_seg = (tid & 3) * 8;    // 4 thr/row x 64
 
   for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A tile [BM][BK]
+    // stage A tile [BM][BK] (with load-op)
     {
       const int64_t gm = bm + a_row;
       const int gk = k0 + a_seg;
       ushort_t* dst = &Alds[a_row * LDA + a_seg];
       if (gm < M && gk + 16 <= K) {
-        const ushort_t* src = &A[gm * K + gk];
-        *reinterpret_cast<short8_t*>(dst) =
-            *reinterpret_cast<const short8_t*>(src);
-        *reinterpret_cast<short8_t*>(dst + 8) =
-            *reinterpret_cast<const short8_t*>(src + 8);
+        const int64_t off = gm * K + gk;
+        if (lopA.kind == LOP_NONE) {
+          *reinterpret_cast<short8_t*>(dst) =
+              *reinterpret_cast<const short8_t*>(&A[off]);
+          *reinterpret_cast<short8_t*>(dst + 8) =
+              *reinterpret_cast<const short8_t*>(&A[off + 8]);
+        } else {
+          short8_t v0 = *reinterpret_cast<const short8_t*>(&A[off]);
+          short8_t v1 = *reinterpret_cast<const short8_t*>(&A[off + 8]);
+          short8_t z0, z1;
+          if (lopA.kind == LOP_BNBWD) {
+            z0 = *reinterpret_cast<const short8_t*>(&lopA.z[off]);
+            z1 = *reinterpret_cast<const short8_t*>(&lopA.z[off + 8]);
+          }
+          short8_t o0, o1;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            o0[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v0[j])),
+                                         lopA.kind == LOP_BNBWD
+                                             ? bf2f(ushort_t(z0[j])) : 0.f,
+                                         gk + j)));
+            o1[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v1[j])),
+                                         lopA.kind == LOP_BNBWD
+                                             ? bf2f(ushort_t(z1[j])) : 0.f,
+                                         gk + 8 + j)));
+          }
+          *reinterpret_cast<short8_t*>(dst) = o0;
+          *reinterpret_cast<short8_t*>(dst + 8) = o1;
+        }
       } else {
 #pragma unroll
-        for (int j = 0; j < 16; ++j)
-          dst[j] = (gm < M && gk + j < K) ? A[gm * K + gk + j] : ushort_t(0);
+        for (int j = 0; j < 16; ++j) {
+          const int gkj = gk + j;
+          dst[j] = (gm < M && gkj < K)
+                       ? ushort_t(f2bf(lop_load(lopA, A, gm * K + gkj, gkj)))
+                       : ushort_t(0);
+        }
       }
     }
     // stage B tile [BN][BK] from Bt[N][K]
@@ -121,16 +161,69 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       }
     }
   }
+
+  if (STATS) {
+    // per-block partial column sums/sumsq of the (bf16-rounded) output.
+    // contributors per col: 2 M-waves x 4 row-groups = 8 slots.
+    float* red = reinterpret_cast<float*>(Alds);  // 512 floats
+    float s1[FRAG_N] = {}, s2[FRAG_N] = {};
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j)
+#pragma unroll
+      for (int i = 0; i < FRAG_M; ++i)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
+          if (row < M) {
+            const float v = bf2f(f2bf(acc[i][j][r]));
+            s1[j] += v;
+            s2[j] = fmaf(v, v, s2[j]);
+          }
+        }
+    const int slot = (wave >> 1) * 4 + (lane >> 4);  // 0..7
+    const int mblocks = gridDim.x;
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j)
+      red[slot * 64 + wcol0 + j * 16 + col_l] = s1[j];
+    __syncthreads();
+    if (tid < 64 && bn + tid < N) {
+      float t = 0.f;
+#pragma unroll
+      for (int l = 0; l < 8; ++l) t += red[l * 64 + tid];
+      stats_ws[int64_t(blockIdx.x) * N + bn + tid] = t;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int j = 0; j < FRAG_N; ++j)
+      red[slot * 64 + wcol0 + j * 16 + col_l] = s2[j];
+    __syncthreads();
+    if (tid < 64 && bn + tid < N) {
+      float t = 0.f;
+#pragma unroll
+      for (int l = 0; l < 8; ++l) t += red[l * 64 + tid];
+      stats_ws[(int64_t(mblocks) + blockIdx.x) * N + bn + tid] = t;
+    }
+  }
 }
 
 }  // namespace
 
+int pw_stats_mblocks(int64_t M) { return ceil_div(M, BM); }
+
 void pw_conv(const void* x, const void* w, void* y,
-             int64_t M, int K, int N, hipStream_t s) {
+             int64_t M, int K, int N, LopDesc lopa, float* stats_ws,
+             hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
-  pw_conv_kernel<<<grid, 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
-      static_cast<ushort_t*>(y), M, K, N);
+  const Lop L = to_lop(lopa);
+  if (stats_ws != nullptr)
+    pw_conv_kernel<true><<<grid, 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
+        static_cast<ushort_t*>(y), M, K, N, L, stats_ws);
+  else
+    pw_conv_kernel<false><<<grid, 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
+        static_cast<ushort_t*>(y), M, K, N, L, nullptr);
 }
 
 // ---------------------------------------------------------------------------
@@ -149,7 +242,8 @@ constexpr int MT = 32;  // m rows per LDS pass
 template <int W1, int W2>
 __global__ __launch_bounds__(256) void atb_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
-    float* __restrict__ ws, int64_t M, int N1, int N2, int splits) {
+    float* __restrict__ ws, int64_t M, int N1, int N2, int splits,
+    Lop lopA, Lop lopB) {
   constexpr int V1 = W1 / 16, V2 = W2 / 16;
   constexpr int LD1 = W1 + 4, LD2 = W2 + 4;
   __shared__ ushort_t Al[MT * LD1];
@@ -181,14 +275,16 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Al[r * LD1 + s_seg];
         const int gn = n1_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N1) {
+        if (gm < m_hi && gn + 8 <= N1 && lopA.kind == LOP_NONE) {
           *reinterpret_cast<short8_t*>(dst) =
               *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            dst[j] = (gm < m_hi && gn + j < N1) ? A[gm * N1 + gn + j]
-                                                : ushort_t(0);
+            dst[j] = (gm < m_hi && gn + j < N1)
+                         ? ushort_t(f2bf(lop_load(lopA, A,
+                                                  gm * N1 + gn + j, gn + j)))
+                         : ushort_t(0);
         }
       }
     }
@@ -198,14 +294,16 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Bl[r * LD2 + s_seg];
         const int gn = n2_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N2) {
+        if (gm < m_hi && gn + 8 <= N2 && lopB.kind == LOP_NONE) {
           *reinterpret_cast<short8_t*>(dst) =
               *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            dst[j] = (gm < m_hi && gn + j < N2) ? B[gm * N2 + gn + j]
-                                                : ushort_t(0);
+            dst[j] = (gm < m_hi && gn + j < N2)
+                         ? ushort_t(f2bf(lop_load(lopB, B,
+                                                  gm * N2 + gn + j, gn + j)))
+                         : ushort_t(0);
         }
       }
     }
@@ -264,12 +362,14 @@ __global__ __launch_bounds__(256) void atb_kernel(
 int tile_of(int n) { return n <= 32 ? 32 : (n <= 64 ? 64 : 128); }
 
 void atb_launch(const ushort_t* a, const ushort_t* b, float* ws, int64_t M,
-                int N1, int N2, int splits, int blocks, hipStream_t s) {
+                int N1, int N2, int splits, int blocks, Lop lopA, Lop lopB,
+                hipStream_t s) {
   const int w1 = tile_of(N1), w2 = tile_of(N2);
   dim3 g(blocks);
 #define ATB_CASE(X, Y)                                                   \
   if (w1 == X && w2 == Y) {                                              \
-    atb_kernel<X, Y><<<g, 256, 0, s>>>(a, b, ws, M, N1, N2, splits);     \
+    atb_kernel<X, Y><<<g, 256, 0, s>>>(a, b, ws, M, N1, N2, splits,      \
+                                       lopA, lopB);                      \
     return;                                                              \
   }
   ATB_CASE(32, 32) ATB_CASE(32, 64) ATB_CASE(32, 128)
@@ -332,12 +432,13 @@ int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2) {
 }
 
 void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
-                   int64_t M, int N1, int N2, hipStream_t s) {
+                   int64_t M, int N1, int N2, LopDesc lopa, LopDesc lopb,
+                   hipStream_t s) {
   const int splits = pw_wgrad_splits(M, N1, N2);
   const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   atb_launch(static_cast<const ushort_t*>(a),
              static_cast<const ushort_t*>(b), ws, M, N1, N2, splits,
-             tiles * splits, s);
+             tiles * splits, to_lop(lopa), to_lop(lopb), s);
   slab_reduce(ws, dw, int64_t(N1) * N2, splits, s);
 }
 

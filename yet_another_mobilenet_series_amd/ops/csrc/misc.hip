@@ -12,6 +12,35 @@ namespace yams {
 
 namespace {
 
+inline Lop to_lop(const LopDesc& d) {
+  Lop L;
+  L.kind = d.kind;
+  L.act = d.act;
+  L.coef = d.coef;
+  L.z = static_cast<const ushort_t*>(d.z);
+  L.C = d.C;
+  return L;
+}
+
+DEV_INLINE void load8_lop(const Lop& L, const ushort_t* p, int64_t off,
+                          int c, float* out) {
+  short8_t v = *reinterpret_cast<const short8_t*>(&p[off]);
+  if (L.kind == LOP_NONE) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = bf2f(ushort_t(v[e]));
+  } else if (L.kind == LOP_BNACT) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = lop_apply(L, bf2f(ushort_t(v[e])), 0.f, c + e);
+  } else {
+    short8_t z = *reinterpret_cast<const short8_t*>(&L.z[off]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = lop_apply(L, bf2f(ushort_t(v[e])), bf2f(ushort_t(z[e])),
+                         c + e);
+  }
+}
+
 int elem_grid(int64_t total) {
   return int(std::min<int64_t>((total + 255) / 256, 4096));
 }
@@ -66,18 +95,19 @@ __global__ __launch_bounds__(256) void bias_act_bwd_kernel(
 // --- SE channel scale: y[n,hw,c] = x * s[n,c] ---------------------------
 __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ scale,
-    ushort_t* __restrict__ y, int64_t total8, int HW, int C8) {
+    ushort_t* __restrict__ y, int64_t total8, int HW, int C8, Lop lopx) {
   for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
        idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
     const int c8 = int(idx % C8);
     const int n = int(idx / (int64_t(C8) * HW));
-    short8_t xv = *reinterpret_cast<const short8_t*>(&x[idx * 8]);
+    float xv[8];
+    load8_lop(lopx, x, idx * 8, c8 * 8, xv);
     short8_t sv = *reinterpret_cast<const short8_t*>(
         &scale[(int64_t(n) * C8 + c8) * 8]);
     short8_t out;
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      out[e] = short(f2bf(bf2f(ushort_t(xv[e])) * bf2f(ushort_t(sv[e]))));
+      out[e] = short(f2bf(xv[e] * bf2f(ushort_t(sv[e]))));
     *reinterpret_cast<short8_t*>(&y[idx * 8]) = out;
   }
 }
@@ -87,7 +117,7 @@ __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
 __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ scale, ushort_t* __restrict__ dx,
-    ushort_t* __restrict__ dscale, int HW, int C, int c8_tiles) {
+    ushort_t* __restrict__ dscale, int HW, int C, int c8_tiles, Lop lopx) {
   __shared__ float red[256 * 8];
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
@@ -107,12 +137,13 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
     for (int hw = hwlane; hw < HW; hw += ML) {
       const int64_t off = base + int64_t(hw) * C + c;
       short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
-      short8_t x8 = *reinterpret_cast<const short8_t*>(&x[off]);
+      float x8[8];
+      load8_lop(lopx, x, off, c, x8);
       short8_t o8;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const float g = bf2f(ushort_t(g8[e]));
-        acc[e] = fmaf(g, bf2f(ushort_t(x8[e])), acc[e]);
+        acc[e] = fmaf(g, x8[e], acc[e]);
         o8[e] = short(f2bf(g * sv[e]));
       }
       *reinterpret_cast<short8_t*>(&dx[off]) = o8;
@@ -136,7 +167,7 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
 // --- global average pool ------------------------------------------------
 __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
     const ushort_t* __restrict__ x, ushort_t* __restrict__ out, int HW,
-    int C, int c8_tiles) {
+    int C, int c8_tiles, Lop lopx) {
   __shared__ float red[256 * 8];
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
@@ -149,10 +180,10 @@ __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
   float acc[8] = {};
   if (c8 < C8 && hwlane < ML) {
     for (int hw = hwlane; hw < HW; hw += ML) {
-      short8_t x8 = *reinterpret_cast<const short8_t*>(
-          &x[base + int64_t(hw) * C + c]);
+      float x8[8];
+      load8_lop(lopx, x, base + int64_t(hw) * C + c, c, x8);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += bf2f(ushort_t(x8[e]));
+      for (int e = 0; e < 8; ++e) acc[e] += x8[e];
     }
   }
 #pragma unroll
@@ -309,32 +340,33 @@ void bias_act_bwd(const void* dy, const void* z, const float* bias, int act,
 }
 
 void rowscale_fwd(const void* x, const void* scale, void* y, int N, int HW,
-                  int C, hipStream_t s) {
+                  int C, LopDesc lopx, hipStream_t s) {
   const int64_t total8 = int64_t(N) * HW * (C >> 3);
   rowscale_fwd_kernel<<<elem_grid(total8), 256, 0, s>>>(
       static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(scale),
-      static_cast<ushort_t*>(y), total8, HW, C >> 3);
+      static_cast<ushort_t*>(y), total8, HW, C >> 3, to_lop(lopx));
 }
 
 void rowscale_bwd(const void* dy, const void* x, const void* scale, void* dx,
-                  void* dscale, int N, int HW, int C, hipStream_t s) {
+                  void* dscale, int N, int HW, int C, LopDesc lopx,
+                  hipStream_t s) {
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c8_tiles = (C8 + CT8 - 1) / CT8;
   rowscale_bwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
       static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
       static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
-      static_cast<ushort_t*>(dscale), HW, C, c8_tiles);
+      static_cast<ushort_t*>(dscale), HW, C, c8_tiles, to_lop(lopx));
 }
 
 void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
-                 hipStream_t s) {
+                 LopDesc lopx, hipStream_t s) {
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c8_tiles = (C8 + CT8 - 1) / CT8;
   avgpool_fwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
       static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C,
-      c8_tiles);
+      c8_tiles, to_lop(lopx));
 }
 
 void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,

@@ -1,0 +1,322 @@
+"""Fused inverted-residual block (SURVEY.md §7 stage 4: block-level fusion).
+
+One autograd Function runs the whole block {pw expand -> BN+act -> dw ->
+BN+act -> [SE] -> pw project -> BN [+residual]} with:
+
+- BN statistics computed in the conv epilogues (per-block partial slabs
+  combined by ``bn_finalize``) — no separate stats pass over z;
+- BN-normalize+activation applied as LOAD-OPS in each consumer kernel —
+  the activated tensors y1/y2 are never materialized (the only tensors
+  written per block are the raw conv outputs z1/z2/z3, the SE gate
+  path, and the block output);
+- backward computes dz on-the-fly as a BNBWD load-op inside the dgrad /
+  wgrad kernels — dz never materializes either.
+
+Traffic per conv drops from {z:1W+2R, y:1W+1R} to {z:1W+1R} forward and
+{dz:1W+2R} to zero backward: the fusion-depth lever SURVEY.md §6 says
+sets achievable img/s on this HBM-bound family.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .backend import ext
+from .hip_ops import (ACT_HSIGMOID, ACT_IDENTITY, ACT_RELU, BF16, CL,
+                      _empty_like_nhwc, _nhwc, _packed)
+
+F32 = torch.float32
+
+
+def _bn_bufs(c: int, dev) -> tuple:
+    mean = torch.empty(c, dtype=F32, device=dev)
+    invstd = torch.empty(c, dtype=F32, device=dev)
+    ab = torch.empty(2 * c, dtype=F32, device=dev)
+    return mean, invstd, ab
+
+
+def _eval_ab(mean, invstd, gamma, beta, ab) -> None:
+    a = gamma.float() * invstd
+    ab[:gamma.numel()] = a
+    ab[gamma.numel():] = beta.float() - mean * a
+
+
+class FusedInvertedResidualFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, ew, eg, ebb, erm, erv, dww, dg, db, drm, drv,
+                sw1, sb1, sw2, sb2, pw, pg, pb, prm, prv,
+                k, stride, act, training, momentum, eps, use_res):
+        e = ext()
+        x = _nhwc(x)
+        dev = x.device
+        n, cin, h, wdt = x.shape
+        pad = k // 2
+        has_expand = ew is not None
+        has_se = sw1 is not None
+
+        # ---- expand pw ------------------------------------------------
+        if has_expand:
+            ce = ew.shape[0]
+            ewp = _packed(ew, "pw", lambda w: w.view(ce, cin).to(BF16)
+                          .contiguous())
+            z1 = _empty_like_nhwc(x, c=ce)
+            m1 = n * h * wdt
+            mean1, invstd1, ab1 = _bn_bufs(ce, dev)
+            if training:
+                mb = e.pw_stats_mblocks(m1)
+                ws = torch.empty(2 * mb * ce, dtype=F32, device=dev)
+                e.pw_conv(x, ewp, z1, stats_ws=ws)
+                e.bn_finalize(ws, mb, mean1, invstd1, ab1, eg, ebb,
+                              erm, erv, momentum, eps, m1)
+            else:
+                e.pw_conv(x, ewp, z1)
+                mean1.copy_(erm)
+                invstd1.copy_((erv.float() + eps).rsqrt())
+                _eval_ab(mean1, invstd1, eg, ebb, ab1)
+            dw_in, dw_lop = z1, (1, act, ab1)
+        else:
+            z1, mean1, invstd1, ab1 = x, None, None, None
+            dw_in, dw_lop = x, (0, 0, None)
+            ce = cin
+
+        # ---- depthwise ------------------------------------------------
+        cd = dww.shape[0]
+        dwp = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
+                      .contiguous())
+        ho = (h + 2 * pad - k) // stride + 1
+        wo = (wdt + 2 * pad - k) // stride + 1
+        z2 = torch.empty((n, cd, ho, wo), dtype=BF16, device=dev,
+                         memory_format=CL)
+        m2 = n * ho * wo
+        mean2, invstd2, ab2 = _bn_bufs(cd, dev)
+        if training:
+            chunks = e.dw_stats_chunks(n, ho, wo, cd)
+            ws = torch.empty(2 * chunks * cd, dtype=F32, device=dev)
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
+                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
+                          lop_coef=dw_lop[2], stats_ws=ws)
+            e.bn_finalize(ws, chunks, mean2, invstd2, ab2, dg, db,
+                          drm, drv, momentum, eps, m2)
+        else:
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
+                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
+                          lop_coef=dw_lop[2])
+            mean2.copy_(drm)
+            invstd2.copy_((drv.float() + eps).rsqrt())
+            _eval_ab(mean2, invstd2, dg, db, ab2)
+
+        # ---- squeeze-excite -------------------------------------------
+        if has_se:
+            r = sw1.shape[0]
+            pooled = torch.empty(n, cd, dtype=BF16, device=dev)
+            e.avgpool_fwd(z2, pooled, lop_kind=1, lop_act=act, lop_coef=ab2)
+            w1m = _packed(sw1, "se", lambda w: w.view(r, cd).to(BF16)
+                          .contiguous())
+            hz = torch.empty(n, r, dtype=BF16, device=dev)
+            e.pw_conv(pooled.view(n, cd, 1, 1), w1m, hz.view(n, r, 1, 1))
+            ha = torch.empty_like(hz)
+            e.bias_act_fwd(hz, sb1.float(), ACT_RELU, ha)
+            w2m = _packed(sw2, "se", lambda w: w.view(cd, r).to(BF16)
+                          .contiguous())
+            sz = torch.empty(n, cd, dtype=BF16, device=dev)
+            e.pw_conv(ha.view(n, r, 1, 1), w2m, sz.view(n, cd, 1, 1))
+            sgate = torch.empty_like(sz)
+            e.bias_act_fwd(sz, sb2.float(), ACT_HSIGMOID, sgate)
+            y2s = torch.empty_like(z2)
+            e.rowscale_fwd(z2, sgate, y2s, lop_kind=1, lop_act=act,
+                           lop_coef=ab2)
+            pj_in, pj_lop = y2s, (0, 0, None)
+        else:
+            pooled = hz = ha = sz = sgate = y2s = None
+            pj_in, pj_lop = z2, (1, act, ab2)
+
+        # ---- project pw -----------------------------------------------
+        cp = pw.shape[0]
+        pwp = _packed(pw, "pw", lambda w: w.view(cp, cd).to(BF16)
+                      .contiguous())
+        z3 = torch.empty((n, cp, ho, wo), dtype=BF16, device=dev,
+                         memory_format=CL)
+        mean3, invstd3, ab3 = _bn_bufs(cp, dev)
+        if training:
+            mb = e.pw_stats_mblocks(m2)
+            ws = torch.empty(2 * mb * cp, dtype=F32, device=dev)
+            e.pw_conv(pj_in, pwp, z3, lop_kind=pj_lop[0], lop_act=pj_lop[1],
+                      lop_coef=pj_lop[2], stats_ws=ws)
+            e.bn_finalize(ws, mb, mean3, invstd3, ab3, pg, pb, prm, prv,
+                          momentum, eps, m2)
+        else:
+            e.pw_conv(pj_in, pwp, z3, lop_kind=pj_lop[0], lop_act=pj_lop[1],
+                      lop_coef=pj_lop[2])
+            mean3.copy_(prm)
+            invstd3.copy_((prv.float() + eps).rsqrt())
+            _eval_ab(mean3, invstd3, pg, pb, ab3)
+
+        # ---- block output: BN (linear) + residual ----------------------
+        yb = torch.empty_like(z3)
+        e.bn_act_fwd(z3, ab3, ACT_IDENTITY, yb, x if use_res else None)
+
+        ctx.save_for_backward(
+            x, z1 if has_expand else None, z2, z3,
+            mean1, invstd1, ab1, mean2, invstd2, ab2, mean3, invstd3,
+            pooled, hz, ha, sz, sgate, y2s,
+            ew, eg, ebb, dww, dg, db, sw1, sb1, sw2, sb2, pw, pg, pb)
+        ctx.cfg = (k, stride, act, training, has_expand, has_se, use_res,
+                   n, cin, h, wdt, ho, wo)
+        return yb
+
+    @staticmethod
+    def backward(ctx, dyb):
+        e = ext()
+        (x, z1, z2, z3, mean1, invstd1, ab1, mean2, invstd2, ab2, mean3,
+         invstd3, pooled, hz, ha, sz, sgate, y2s, ew, eg, ebb, dww, dg, db,
+         sw1, sb1, sw2, sb2, pw, pg, pb) = ctx.saved_tensors
+        (k, stride, act, training, has_expand, has_se, use_res, n, cin, h,
+         wdt, ho, wo) = ctx.cfg
+        dev = x.device
+        pad = k // 2
+        dyb = dyb.to(BF16).contiguous(memory_format=CL)
+        cd = dww.shape[0]
+        cp = pw.shape[0]
+        ce = ew.shape[0] if has_expand else cin
+
+        # ---- project bn backward stats + coef --------------------------
+        dg3 = torch.empty(cp, dtype=F32, device=dev)
+        db3 = torch.empty_like(dg3)
+        coef3 = torch.empty(4 * cp, dtype=F32, device=dev)
+        e.bn_bwd_stats(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
+                       bool(training), dg3, db3, coef3)
+
+        # dgrad: d_in3 = dz3 @ Wp (dz3 computed on load)
+        pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
+                      .contiguous())
+        d_in3 = torch.empty_like(z2)
+        e.pw_conv(dyb, pwt, d_in3, lop_kind=2, lop_act=ACT_IDENTITY,
+                  lop_coef=coef3, lop_z=z3)
+        # wgrad: dWp = dz3^T @ in3
+        dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
+        if has_se:
+            e.pw_conv_wgrad(dyb, y2s, dwp3, lopa_kind=2,
+                            lopa_act=ACT_IDENTITY, lopa_coef=coef3,
+                            lopa_z=z3)
+        else:
+            e.pw_conv_wgrad(dyb, z2, dwp3, lopa_kind=2,
+                            lopa_act=ACT_IDENTITY, lopa_coef=coef3,
+                            lopa_z=z3, lopb_kind=1, lopb_act=act,
+                            lopb_coef=ab2)
+        d_pw = dwp3.reshape(cp, cd, 1, 1)
+
+        # ---- SE backward ------------------------------------------------
+        if has_se:
+            r = sw1.shape[0]
+            d_y2 = torch.empty_like(z2)
+            ds = torch.empty(n, cd, dtype=BF16, device=dev)
+            e.rowscale_bwd(d_in3, z2, sgate, d_y2, ds, lop_kind=1,
+                           lop_act=act, lop_coef=ab2)
+            dsz = torch.empty_like(ds)
+            db2se = torch.empty(cd, dtype=F32, device=dev)
+            e.bias_act_bwd(ds, sz, sb2.float(), ACT_HSIGMOID, dsz, db2se)
+            w2t = _packed(sw2, "se_t", lambda w: w.view(cd, r).t().to(BF16)
+                          .contiguous())
+            dha = torch.empty_like(ha)
+            e.pw_conv(dsz.view(n, cd, 1, 1), w2t, dha.view(n, r, 1, 1))
+            dw2p = torch.empty(cd, r, dtype=F32, device=dev)
+            e.pw_conv_wgrad(dsz.view(n, cd, 1, 1), ha.view(n, r, 1, 1),
+                            dw2p)
+            dhz = torch.empty_like(dha)
+            db1se = torch.empty(r, dtype=F32, device=dev)
+            e.bias_act_bwd(dha, hz, sb1.float(), ACT_RELU, dhz, db1se)
+            w1t = _packed(sw1, "se_t", lambda w: w.view(r, cd).t().to(BF16)
+                          .contiguous())
+            dpooled = torch.empty_like(pooled)
+            e.pw_conv(dhz.view(n, r, 1, 1), w1t, dpooled.view(n, cd, 1, 1))
+            dw1p = torch.empty(r, cd, dtype=F32, device=dev)
+            e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, cd, 1, 1),
+                            dw1p)
+            e.avgpool_bwd_add(dpooled, d_y2)
+            d_sw1 = dw1p.reshape_as(sw1)
+            d_sw2 = dw2p.reshape_as(sw2)
+        else:
+            d_y2 = d_in3
+            d_sw1 = d_sw2 = db1se = db2se = None
+
+        # ---- depthwise bn backward + grads ------------------------------
+        dg2 = torch.empty(cd, dtype=F32, device=dev)
+        db2_ = torch.empty_like(dg2)
+        coef2 = torch.empty(4 * cd, dtype=F32, device=dev)
+        e.bn_bwd_stats(d_y2, z2, mean2, invstd2, dg, db, act,
+                       bool(training), dg2, db2_, coef2)
+        dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
+                       .contiguous())
+        dw_in = z1 if has_expand else x
+        d_z1out = torch.empty_like(dw_in)
+        e.dw_conv_dgrad(d_y2, dwpk, d_z1out, k, stride, pad, lop_kind=2,
+                        lop_act=act, lop_coef=coef2, lop_z=z2)
+        dwgt = torch.empty(k * k, cd, dtype=F32, device=dev)
+        if has_expand:
+            e.dw_conv_wgrad(dw_in, d_y2, dwgt, k, stride, pad,
+                            lopx_kind=1, lopx_act=act, lopx_coef=ab1,
+                            lopdy_kind=2, lopdy_act=act, lopdy_coef=coef2,
+                            lopdy_z=z2)
+        else:
+            e.dw_conv_wgrad(dw_in, d_y2, dwgt, k, stride, pad,
+                            lopdy_kind=2, lopdy_act=act, lopdy_coef=coef2,
+                            lopdy_z=z2)
+        d_dww = dwgt.t().reshape(cd, 1, k, k)
+
+        # ---- expand backward --------------------------------------------
+        if has_expand:
+            dg1 = torch.empty(ce, dtype=F32, device=dev)
+            db1_ = torch.empty_like(dg1)
+            coef1 = torch.empty(4 * ce, dtype=F32, device=dev)
+            e.bn_bwd_stats(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                           bool(training), dg1, db1_, coef1)
+            ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
+                          .contiguous())
+            d_x = _empty_like_nhwc(x)
+            e.pw_conv(d_z1out, ewt, d_x, lop_kind=2, lop_act=act,
+                      lop_coef=coef1, lop_z=z1)
+            dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
+            e.pw_conv_wgrad(d_z1out, x, dwp1, lopa_kind=2, lopa_act=act,
+                            lopa_coef=coef1, lopa_z=z1)
+            d_ew = dwp1.reshape(ce, cin, 1, 1)
+        else:
+            d_x = d_z1out
+            d_ew = dg1 = db1_ = None
+
+        if use_res:
+            d_x = d_x + dyb
+
+        return (d_x, d_ew, dg1, db1_, None, None,
+                d_dww, dg2, db2_, None, None,
+                d_sw1, db1se, d_sw2, db2se,
+                d_pw, dg3, db3, None, None,
+                None, None, None, None, None, None, None)
+
+
+def fused_inverted_residual(x, block):
+    """Run one InvertedResidual module through the fused HIP path."""
+    ew = eg = ebb = erm = erv = None
+    if block.expand is not None:
+        ew = block.expand.conv.weight
+        eg, ebb = block.expand.bn.weight, block.expand.bn.bias
+        erm, erv = block.expand.bn.running_mean, block.expand.bn.running_var
+    sw1 = sb1 = sw2 = sb2 = None
+    if block.se is not None:
+        sw1, sb1 = block.se.fc1.weight, block.se.fc1.bias
+        sw2, sb2 = block.se.fc2.weight, block.se.fc2.bias
+    dwm, pjm = block.dw, block.project
+    if block.training and dwm.bn.track_running_stats:
+        for m in (block.expand, dwm, pjm):
+            if m is not None:
+                m.bn.num_batches_tracked += 1
+    return FusedInvertedResidualFn.apply(
+        x, ew, eg, ebb, erm, erv,
+        dwm.conv.weight, dwm.bn.weight, dwm.bn.bias,
+        dwm.bn.running_mean, dwm.bn.running_var,
+        sw1, sb1, sw2, sb2,
+        pjm.conv.weight, pjm.bn.weight, pjm.bn.bias,
+        pjm.bn.running_mean, pjm.bn.running_var,
+        dwm.conv.weight.shape[2], dwm.stride, dwm.act, block.training,
+        dwm.bn.momentum, dwm.bn.eps, block.use_residual)
