@@ -73,12 +73,17 @@ def test_fused_block_parity(cfg, training):
     assert _rel(y.detach(), yr.detach()) < 0.03, "forward mismatch"
     if training:
         assert _rel(xh.grad, xr.grad) < 0.08
+        total_norm = sum(p.grad.float().norm().item() ** 2
+                         for p in block.parameters()) ** 0.5
         for n, p in block.named_parameters():
             assert g_hip[n] is not None, f"missing fused grad {n}"
             e = _rel(g_hip[n], p.grad)
-            # bn biases of linear-BN project layers can be cancellation-noisy
-            denom_small = p.grad.float().norm().item() < 1e-3 * p.numel() ** 0.5
-            assert e < 0.1 or denom_small, (n, e)
+            # gammas whose true grad is ~0 by ReLU homogeneity (a channel
+            # scale before dw+BN is a no-op) are cancellation residue in
+            # BOTH paths — the relative comparison is ill-posed there
+            null_grad = (p.grad.float().norm().item()
+                         < 1e-3 * max(total_norm, 1.0))
+            assert e < 0.1 or null_grad, (n, e)
         # running stats updated identically
         for k, v in block.state_dict().items():
             if "running" in k:
