@@ -344,10 +344,50 @@ void bn_stats(const void* z, float* mean, float* invstd, float* ab,
       g.chunks, momentum, eps);
 }
 
+// parallel pre-fold for very large slab counts (e.g. 112^2 layers produce
+// ~25k GEMM m-blocks): out[2][folded][C] = grouped sums of ws[2][splits][C]
+__global__ void bn_fold_kernel(const float* __restrict__ ws,
+                               float* __restrict__ out, int splits,
+                               int folded, int C) {
+  const int64_t total = 2LL * folded * C;
+  const int g = (splits + folded - 1) / folded;
+  for (int64_t i = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < total; i += int64_t(gridDim.x) * blockDim.x) {
+    const int c = int(i % C);
+    const int f = int((i / C) % folded);
+    const int plane = int(i / (int64_t(folded) * C));
+    const int lo = f * g;
+    const int hi = lo + g < splits ? lo + g : splits;
+    float v = 0.f;
+    for (int sp = lo; sp < hi; ++sp)
+      v += ws[(int64_t(plane) * splits + sp) * C + c];
+    out[(int64_t(plane) * folded + f) * C + c] = v;
+  }
+}
+
 void bn_finalize(const float* ws, int splits, float* mean, float* invstd,
                  float* ab, const float* gamma, const float* beta,
                  float* running_mean, float* running_var, int64_t M, int C,
                  float momentum, float eps, hipStream_t s) {
+  static float* fold_buf = nullptr;
+  static size_t fold_cap = 0;
+  if (splits > 1024) {
+    const int folded = 256;
+    const size_t need = 2 * folded * size_t(C);
+    if (need > fold_cap) {
+      if (fold_buf) (void)hipFree(fold_buf);
+      (void)hipMalloc(&fold_buf, need * sizeof(float));
+      fold_cap = need;
+    }
+    const int64_t total = 2LL * folded * C;
+    const int blocks = int(std::min<int64_t>((total + 255) / 256, 4096));
+    bn_fold_kernel<<<dim3(blocks), 256, 0, s>>>(ws, fold_buf, splits,
+                                                folded, C);
+    bn_stats_finalize_kernel<<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+        fold_buf, mean, invstd, ab, gamma, beta, running_mean, running_var,
+        M, C, folded, momentum, eps);
+    return;
+  }
   bn_stats_finalize_kernel<<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
       ws, mean, invstd, ab, gamma, beta, running_mean, running_var, M, C,
       splits, momentum, eps);

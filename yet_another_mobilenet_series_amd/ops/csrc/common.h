@@ -115,6 +115,29 @@ DEV_INLINE float lop_apply(const Lop& L, float v, float zv, int c) {
   return v;
 }
 
+// compile-time-kind variants (kernels template on the lop KIND so the
+// no-op path keeps its register budget; act stays a runtime switch)
+template <int LOPK>
+DEV_INLINE float lop_applyT(const Lop& L, float v, float zv, int c) {
+  if (LOPK == LOP_BNACT)
+    return act_fwd_rt(fmaf(v, L.coef[c], L.coef[L.C + c]), L.act);
+  if (LOPK == LOP_BNBWD) {
+    const float a = L.coef[c];
+    const float da = v * act_bwd_rt(fmaf(zv, a, L.coef[L.C + c]), L.act);
+    return fmaf(a, da, L.coef[2 * L.C + c]) - L.coef[3 * L.C + c] * zv;
+  }
+  return v;
+}
+
+template <int LOPK>
+DEV_INLINE float lop_loadT(const Lop& L, const ushort_t* p, int64_t off,
+                           int c) {
+  const float v = bf2f(p[off]);
+  if (LOPK == LOP_NONE) return v;
+  const float zv = LOPK == LOP_BNBWD ? bf2f(L.z[off]) : 0.f;
+  return lop_applyT<LOPK>(L, v, zv, c);
+}
+
 // bf16 load + lop for one element of a stream at flat offset `off`,
 // channel `c` (zv fetched only when the lop needs it)
 DEV_INLINE float lop_load(const Lop& L, const ushort_t* p, int64_t off,

@@ -37,7 +37,7 @@ inline Lop to_lop(const LopDesc& d) {
 // the producer's BN+act into the x loads.  FLIP reuses the kernel as the
 // stride-1 dgrad (correlation with the flipped filter; lopx then is the
 // BNBWD load-op on dy).
-template <int K, int S, bool FLIP, bool STATS>
+template <int K, int S, bool FLIP, bool STATS, int LOPK>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
@@ -80,19 +80,21 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
           if (wi >= 0 && wi < W) {
             const int64_t off = idx4(n, hi, wi, c, H, W, C);
             short8_t v = *reinterpret_cast<const short8_t*>(&x[off]);
-            if (lopx.kind == LOP_NONE) {
+            if (LOPK == LOP_NONE) {
 #pragma unroll
               for (int e = 0; e < 8; ++e) xr[t][e] = bf2f(ushort_t(v[e]));
-            } else if (lopx.kind == LOP_BNACT) {
+            } else if (LOPK == LOP_BNACT) {
 #pragma unroll
               for (int e = 0; e < 8; ++e)
-                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(v[e])), 0.f, c + e);
+                xr[t][e] = lop_applyT<LOP_BNACT>(lopx, bf2f(ushort_t(v[e])),
+                                                 0.f, c + e);
             } else {
               short8_t zv = *reinterpret_cast<const short8_t*>(&lopx.z[off]);
 #pragma unroll
               for (int e = 0; e < 8; ++e)
-                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(v[e])),
-                                     bf2f(ushort_t(zv[e])), c + e);
+                xr[t][e] = lop_applyT<LOP_BNBWD>(lopx, bf2f(ushort_t(v[e])),
+                                                 bf2f(ushort_t(zv[e])),
+                                                 c + e);
             }
           } else {
 #pragma unroll
@@ -162,7 +164,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
 }
 
 // dgrad for stride 2 (stride 1 routes through dw_fwd_kernel FLIP=true)
-template <int K>
+template <int K, int LOPK>
 __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
@@ -194,7 +196,7 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
         short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
         short8_t wv = *reinterpret_cast<const short8_t*>(
             &w[(i * K + j) * C + c]);
-        if (lopdy.kind == LOP_NONE) {
+        if (LOPK == LOP_NONE) {
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             acc[e] = fmaf(bf2f(ushort_t(gv[e])), bf2f(ushort_t(wv[e])),
@@ -203,8 +205,8 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
           short8_t zv = *reinterpret_cast<const short8_t*>(&lopdy.z[off]);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(lop_apply(lopdy, bf2f(ushort_t(gv[e])),
-                                    bf2f(ushort_t(zv[e])), c + e),
+            acc[e] = fmaf(lop_applyT<LOP_BNBWD>(lopdy, bf2f(ushort_t(gv[e])),
+                                                bf2f(ushort_t(zv[e])), c + e),
                           bf2f(ushort_t(wv[e])), acc[e]);
         }
       }
@@ -219,7 +221,7 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
 // wgrad: thread = 4-channel vector lane; 4 consecutive wo outputs share
 // one x row window in registers; per-chunk fp32 slab after an LDS fold
 // over the m lanes.
-template <int K, int S>
+template <int K, int S, int LX, int LD>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
@@ -255,15 +257,15 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
         if (wo0 + q < Wo) {
           const int64_t off = idx4(n, ho, wo0 + q, c, Ho, Wo, C);
           short4_t g4 = *reinterpret_cast<const short4_t*>(&dy[off]);
-          if (lopdy.kind == LOP_NONE) {
+          if (LD == LOP_NONE) {
 #pragma unroll
             for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
           } else {
             short4_t z4 = *reinterpret_cast<const short4_t*>(&lopdy.z[off]);
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              g[q][e] = lop_apply(lopdy, bf2f(ushort_t(g4[e])),
-                                  bf2f(ushort_t(z4[e])), c + e);
+              g[q][e] = lop_applyT<LOP_BNBWD>(lopdy, bf2f(ushort_t(g4[e])),
+                                              bf2f(ushort_t(z4[e])), c + e);
           }
         } else {
 #pragma unroll
@@ -282,13 +284,14 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
           if (wi >= 0 && wi < W) {
             const int64_t off = idx4(n, hi2, wi, c, H, W, C);
             short4_t x4 = *reinterpret_cast<const short4_t*>(&x[off]);
-            if (lopx.kind == LOP_NONE) {
+            if (LX == LOP_NONE) {
 #pragma unroll
               for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
             } else {
 #pragma unroll
               for (int e = 0; e < 4; ++e)
-                xr[t][e] = lop_apply(lopx, bf2f(ushort_t(x4[e])), 0.f, c + e);
+                xr[t][e] = lop_applyT<LOP_BNACT>(lopx, bf2f(ushort_t(x4[e])),
+                                                 0.f, c + e);
             }
           } else {
 #pragma unroll
@@ -393,17 +396,19 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int chunks = dw_stats_chunks(N, Ho, Wo, C);
   dim3 g(c_tiles * chunks);
-#define DWF_CASE(KK, SS)                                                   \
-  if (k == KK && stride == SS) {                                           \
-    if (stats_ws != nullptr)                                               \
-      dw_fwd_kernel<KK, SS, false, true><<<g, 256, 0, s>>>(                \
-          xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);       \
-    else                                                                   \
-      dw_fwd_kernel<KK, SS, false, false><<<g, 256, 0, s>>>(               \
-          xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, nullptr, chunks);        \
+  // fwd lop kinds in practice: none or bnact
+#define DWF_CASE(KK, SS, ST, LK)                                           \
+  if (k == KK && stride == SS && (stats_ws != nullptr) == ST &&            \
+      lopx.kind == LK) {                                                   \
+    dw_fwd_kernel<KK, SS, false, ST, LK><<<g, 256, 0, s>>>(                \
+        xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);         \
     return;                                                                \
   }
-  DWF_CASE(3, 1) DWF_CASE(3, 2) DWF_CASE(5, 1) DWF_CASE(5, 2)
+#define DWF_KS(KK, SS)                                                     \
+  DWF_CASE(KK, SS, false, 0) DWF_CASE(KK, SS, false, 1)                    \
+  DWF_CASE(KK, SS, true, 0) DWF_CASE(KK, SS, true, 1)
+  DWF_KS(3, 1) DWF_KS(3, 2) DWF_KS(5, 1) DWF_KS(5, 2)
+#undef DWF_KS
 #undef DWF_CASE
 }
 
@@ -421,21 +426,26 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     const int c_tiles = (C8 + CT8 - 1) / CT8;
     const int chunks = dw_stats_chunks(N, H, W, C);
     dim3 g(c_tiles * chunks);
-    if (k == 3)
-      dw_fwd_kernel<3, 1, true, false><<<g, 256, 0, s>>>(
-          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);
-    else
-      dw_fwd_kernel<5, 1, true, false><<<g, 256, 0, s>>>(
-          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);
+#define DWD_CASE(KK, LK)                                                   \
+    if (k == KK && lopdy.kind == LK) {                                     \
+      dw_fwd_kernel<KK, 1, true, false, LK><<<g, 256, 0, s>>>(             \
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);        \
+      return;                                                              \
+    }
+    DWD_CASE(3, 0) DWD_CASE(3, 2) DWD_CASE(5, 0) DWD_CASE(5, 2)
+#undef DWD_CASE
     return;
   }
   const int64_t total8 = int64_t(N) * H * W * (C >> 3);
-  if (k == 3)
-    dw_dgrad_s2_kernel<3><<<grid_for(total8), 256, 0, s>>>(
-        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);
-  else
-    dw_dgrad_s2_kernel<5><<<grid_for(total8), 256, 0, s>>>(
-        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);
+  dim3 g2(grid_for(total8));
+#define DWD2_CASE(KK, LK)                                                  \
+  if (k == KK && lopdy.kind == LK) {                                       \
+    dw_dgrad_s2_kernel<KK, LK><<<g2, 256, 0, s>>>(                         \
+        gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);           \
+    return;                                                                \
+  }
+  DWD2_CASE(3, 0) DWD2_CASE(3, 2) DWD2_CASE(5, 0) DWD2_CASE(5, 2)
+#undef DWD2_CASE
 }
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
@@ -463,18 +473,19 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   const Lop Lx = to_lop(lopx);
   const Lop Ldy = to_lop(lopdy);
   dim3 grid(c_tiles * chunks);
-  if (k == 3 && stride == 1)
-    dw_wgrad_kernel<3, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, Lx, Ldy, chunks);
-  else if (k == 3 && stride == 2)
-    dw_wgrad_kernel<3, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, Lx, Ldy, chunks);
-  else if (k == 5 && stride == 1)
-    dw_wgrad_kernel<5, 1><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, Lx, Ldy, chunks);
-  else
-    dw_wgrad_kernel<5, 2><<<grid, 256, 0, s>>>(xp, gp, ws, N, H, W, C, Ho,
-                                               Wo, pad, Lx, Ldy, chunks);
+#define DWW_CASE(KK, SS, LXX, LDD)                                         \
+  if (k == KK && stride == SS && lopx.kind == LXX &&                       \
+      lopdy.kind == LDD) {                                                 \
+    dw_wgrad_kernel<KK, SS, LXX, LDD><<<grid, 256, 0, s>>>(                \
+        xp, gp, ws, N, H, W, C, Ho, Wo, pad, Lx, Ldy, chunks);             \
+    return;                                                                \
+  }
+#define DWW_KS(KK, SS)                                                     \
+  DWW_CASE(KK, SS, 0, 0) DWW_CASE(KK, SS, 1, 0) DWW_CASE(KK, SS, 0, 2)     \
+  DWW_CASE(KK, SS, 1, 2)
+  DWW_KS(3, 1) DWW_KS(3, 2) DWW_KS(5, 1) DWW_KS(5, 2)
+#undef DWW_KS
+#undef DWW_CASE
   slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }
 

@@ -39,7 +39,7 @@ inline Lop to_lop(const LopDesc& d) {
   return L;
 }
 
-template <bool STATS>
+template <bool STATS, int LOPK>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
     ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA,
@@ -71,7 +71,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       ushort_t* dst = &Alds[a_row * LDA + a_seg];
       if (gm < M && gk + 16 <= K) {
         const int64_t off = gm * K + gk;
-        if (lopA.kind == LOP_NONE) {
+        if (LOPK == LOP_NONE) {
           *reinterpret_cast<short8_t*>(dst) =
               *reinterpret_cast<const short8_t*>(&A[off]);
           *reinterpret_cast<short8_t*>(dst + 8) =
@@ -80,7 +80,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
           short8_t v0 = *reinterpret_cast<const short8_t*>(&A[off]);
           short8_t v1 = *reinterpret_cast<const short8_t*>(&A[off + 8]);
           short8_t z0, z1;
-          if (lopA.kind == LOP_BNBWD) {
+          if (LOPK == LOP_BNBWD) {
             z0 = *reinterpret_cast<const short8_t*>(&lopA.z[off]);
             z1 = *reinterpret_cast<const short8_t*>(&lopA.z[off + 8]);
           }
@@ -88,11 +88,11 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             o0[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v0[j])),
-                                         lopA.kind == LOP_BNBWD
+                                         LOPK == LOP_BNBWD
                                              ? bf2f(ushort_t(z0[j])) : 0.f,
                                          gk + j)));
             o1[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v1[j])),
-                                         lopA.kind == LOP_BNBWD
+                                         LOPK == LOP_BNBWD
                                              ? bf2f(ushort_t(z1[j])) : 0.f,
                                          gk + 8 + j)));
           }
@@ -104,7 +104,8 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
         for (int j = 0; j < 16; ++j) {
           const int gkj = gk + j;
           dst[j] = (gm < M && gkj < K)
-                       ? ushort_t(f2bf(lop_load(lopA, A, gm * K + gkj, gkj)))
+                       ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
+                                                       gm * K + gkj, gkj)))
                        : ushort_t(0);
         }
       }
@@ -216,14 +217,18 @@ void pw_conv(const void* x, const void* w, void* y,
              hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   const Lop L = to_lop(lopa);
-  if (stats_ws != nullptr)
-    pw_conv_kernel<true><<<grid, 256, 0, s>>>(
-        static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
-        static_cast<ushort_t*>(y), M, K, N, L, stats_ws);
-  else
-    pw_conv_kernel<false><<<grid, 256, 0, s>>>(
-        static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(w),
-        static_cast<ushort_t*>(y), M, K, N, L, nullptr);
+  auto* ap = static_cast<const ushort_t*>(x);
+  auto* wp = static_cast<const ushort_t*>(w);
+  auto* yp = static_cast<ushort_t*>(y);
+#define PW_CASE(ST, LK)                                                   \
+  if ((stats_ws != nullptr) == ST && lopa.kind == LK) {                   \
+    pw_conv_kernel<ST, LK><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L,   \
+                                                stats_ws);                \
+    return;                                                               \
+  }
+  PW_CASE(false, 0) PW_CASE(false, 1) PW_CASE(false, 2)
+  PW_CASE(true, 0) PW_CASE(true, 1) PW_CASE(true, 2)
+#undef PW_CASE
 }
 
 // ---------------------------------------------------------------------------
@@ -239,7 +244,7 @@ constexpr int MT = 32;  // m rows per LDS pass
 // (e.g. dw 16->96 expands) waste up to 10x compute in a fixed 128x128
 // tile.  Thread layout 16x16, each thread V1 x V2 = (W1/16) x (W2/16)
 // fp32 accumulators; staging rows of max(W1,W2).
-template <int W1, int W2>
+template <int W1, int W2, int LA, int LB>
 __global__ __launch_bounds__(256) void atb_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     float* __restrict__ ws, int64_t M, int N1, int N2, int splits,
@@ -275,15 +280,16 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Al[r * LD1 + s_seg];
         const int gn = n1_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N1 && lopA.kind == LOP_NONE) {
+        if (gm < m_hi && gn + 8 <= N1 && LA == LOP_NONE) {
           *reinterpret_cast<short8_t*>(dst) =
               *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             dst[j] = (gm < m_hi && gn + j < N1)
-                         ? ushort_t(f2bf(lop_load(lopA, A,
-                                                  gm * N1 + gn + j, gn + j)))
+                         ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
+                                                       gm * N1 + gn + j,
+                                                       gn + j)))
                          : ushort_t(0);
         }
       }
@@ -294,15 +300,16 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Bl[r * LD2 + s_seg];
         const int gn = n2_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N2 && lopB.kind == LOP_NONE) {
+        if (gm < m_hi && gn + 8 <= N2 && LB == LOP_NONE) {
           *reinterpret_cast<short8_t*>(dst) =
               *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             dst[j] = (gm < m_hi && gn + j < N2)
-                         ? ushort_t(f2bf(lop_load(lopB, B,
-                                                  gm * N2 + gn + j, gn + j)))
+                         ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
+                                                       gm * N2 + gn + j,
+                                                       gn + j)))
                          : ushort_t(0);
         }
       }
@@ -363,18 +370,23 @@ int tile_of(int n) { return n <= 32 ? 32 : (n <= 64 ? 64 : 128); }
 
 void atb_launch(const ushort_t* a, const ushort_t* b, float* ws, int64_t M,
                 int N1, int N2, int splits, int blocks, Lop lopA, Lop lopB,
-                hipStream_t s) {
+                int ka, int kb, hipStream_t s) {
   const int w1 = tile_of(N1), w2 = tile_of(N2);
   dim3 g(blocks);
-#define ATB_CASE(X, Y)                                                   \
-  if (w1 == X && w2 == Y) {                                              \
-    atb_kernel<X, Y><<<g, 256, 0, s>>>(a, b, ws, M, N1, N2, splits,      \
-                                       lopA, lopB);                      \
+  // lop kinds used in practice: A in {none, bnbwd}, B in {none, bnact}
+#define ATB_CASE(X, Y, LA, LB)                                           \
+  if (w1 == X && w2 == Y && ka == LA && kb == LB) {                      \
+    atb_kernel<X, Y, LA, LB><<<g, 256, 0, s>>>(a, b, ws, M, N1, N2,      \
+                                               splits, lopA, lopB);      \
     return;                                                              \
   }
-  ATB_CASE(32, 32) ATB_CASE(32, 64) ATB_CASE(32, 128)
-  ATB_CASE(64, 32) ATB_CASE(64, 64) ATB_CASE(64, 128)
-  ATB_CASE(128, 32) ATB_CASE(128, 64) ATB_CASE(128, 128)
+#define ATB_TILE(X, Y)                                                   \
+  ATB_CASE(X, Y, 0, 0) ATB_CASE(X, Y, 2, 0) ATB_CASE(X, Y, 0, 1)         \
+  ATB_CASE(X, Y, 2, 1)
+  ATB_TILE(32, 32) ATB_TILE(32, 64) ATB_TILE(32, 128)
+  ATB_TILE(64, 32) ATB_TILE(64, 64) ATB_TILE(64, 128)
+  ATB_TILE(128, 32) ATB_TILE(128, 64) ATB_TILE(128, 128)
+#undef ATB_TILE
 #undef ATB_CASE
 }
 
@@ -438,7 +450,8 @@ void pw_conv_wgrad(const void* a, const void* b, float* dw, float* ws,
   const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   atb_launch(static_cast<const ushort_t*>(a),
              static_cast<const ushort_t*>(b), ws, M, N1, N2, splits,
-             tiles * splits, to_lop(lopa), to_lop(lopb), s);
+             tiles * splits, to_lop(lopa), to_lop(lopb), lopa.kind,
+             lopb.kind, s);
   slab_reduce(ws, dw, int64_t(N1) * N2, splits, s);
 }
 

@@ -22,22 +22,23 @@ inline Lop to_lop(const LopDesc& d) {
   return L;
 }
 
+template <int LOPK>
 DEV_INLINE void load8_lop(const Lop& L, const ushort_t* p, int64_t off,
                           int c, float* out) {
   short8_t v = *reinterpret_cast<const short8_t*>(&p[off]);
-  if (L.kind == LOP_NONE) {
+  if (LOPK == LOP_NONE) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) out[e] = bf2f(ushort_t(v[e]));
-  } else if (L.kind == LOP_BNACT) {
+  } else if (LOPK == LOP_BNACT) {
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      out[e] = lop_apply(L, bf2f(ushort_t(v[e])), 0.f, c + e);
+      out[e] = lop_applyT<LOP_BNACT>(L, bf2f(ushort_t(v[e])), 0.f, c + e);
   } else {
     short8_t z = *reinterpret_cast<const short8_t*>(&L.z[off]);
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      out[e] = lop_apply(L, bf2f(ushort_t(v[e])), bf2f(ushort_t(z[e])),
-                         c + e);
+      out[e] = lop_applyT<LOP_BNBWD>(L, bf2f(ushort_t(v[e])),
+                                     bf2f(ushort_t(z[e])), c + e);
   }
 }
 
@@ -93,6 +94,7 @@ __global__ __launch_bounds__(256) void bias_act_bwd_kernel(
 }
 
 // --- SE channel scale: y[n,hw,c] = x * s[n,c] ---------------------------
+template <int LOPK>
 __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ scale,
     ushort_t* __restrict__ y, int64_t total8, int HW, int C8, Lop lopx) {
@@ -101,7 +103,7 @@ __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
     const int c8 = int(idx % C8);
     const int n = int(idx / (int64_t(C8) * HW));
     float xv[8];
-    load8_lop(lopx, x, idx * 8, c8 * 8, xv);
+    load8_lop<LOPK>(lopx, x, idx * 8, c8 * 8, xv);
     short8_t sv = *reinterpret_cast<const short8_t*>(
         &scale[(int64_t(n) * C8 + c8) * 8]);
     short8_t out;
@@ -114,6 +116,7 @@ __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
 
 // dx = dy * s (broadcast over hw); grid (n x c8 tiles); vec8 channels,
 // hw lanes folded through LDS: ds[n,c] = sum_hw dy*x
+template <int LOPK>
 __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ scale, ushort_t* __restrict__ dx,
@@ -138,7 +141,7 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
       const int64_t off = base + int64_t(hw) * C + c;
       short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
       float x8[8];
-      load8_lop(lopx, x, off, c, x8);
+      load8_lop<LOPK>(lopx, x, off, c, x8);
       short8_t o8;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
@@ -165,6 +168,7 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
 }
 
 // --- global average pool ------------------------------------------------
+template <int LOPK>
 __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
     const ushort_t* __restrict__ x, ushort_t* __restrict__ out, int HW,
     int C, int c8_tiles, Lop lopx) {
@@ -181,7 +185,7 @@ __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
   if (c8 < C8 && hwlane < ML) {
     for (int hw = hwlane; hw < HW; hw += ML) {
       float x8[8];
-      load8_lop(lopx, x, base + int64_t(hw) * C + c, c, x8);
+      load8_lop<LOPK>(lopx, x, base + int64_t(hw) * C + c, c, x8);
 #pragma unroll
       for (int e = 0; e < 8; ++e) acc[e] += x8[e];
     }
@@ -342,9 +346,16 @@ void bias_act_bwd(const void* dy, const void* z, const float* bias, int act,
 void rowscale_fwd(const void* x, const void* scale, void* y, int N, int HW,
                   int C, LopDesc lopx, hipStream_t s) {
   const int64_t total8 = int64_t(N) * HW * (C >> 3);
-  rowscale_fwd_kernel<<<elem_grid(total8), 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<const ushort_t*>(scale),
-      static_cast<ushort_t*>(y), total8, HW, C >> 3, to_lop(lopx));
+  const Lop L = to_lop(lopx);
+  auto* xp = static_cast<const ushort_t*>(x);
+  auto* sp = static_cast<const ushort_t*>(scale);
+  auto* yp = static_cast<ushort_t*>(y);
+  if (lopx.kind == 0)
+    rowscale_fwd_kernel<0><<<elem_grid(total8), 256, 0, s>>>(
+        xp, sp, yp, total8, HW, C >> 3, L);
+  else
+    rowscale_fwd_kernel<1><<<elem_grid(total8), 256, 0, s>>>(
+        xp, sp, yp, total8, HW, C >> 3, L);
 }
 
 void rowscale_bwd(const void* dy, const void* x, const void* scale, void* dx,
@@ -353,10 +364,17 @@ void rowscale_bwd(const void* dy, const void* x, const void* scale, void* dx,
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c8_tiles = (C8 + CT8 - 1) / CT8;
-  rowscale_bwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
-      static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
-      static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
-      static_cast<ushort_t*>(dscale), HW, C, c8_tiles, to_lop(lopx));
+  const Lop L = to_lop(lopx);
+  if (lopx.kind == 0)
+    rowscale_bwd_kernel<0><<<dim3(N * c8_tiles), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
+        static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
+        static_cast<ushort_t*>(dscale), HW, C, c8_tiles, L);
+  else
+    rowscale_bwd_kernel<1><<<dim3(N * c8_tiles), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
+        static_cast<const ushort_t*>(scale), static_cast<ushort_t*>(dx),
+        static_cast<ushort_t*>(dscale), HW, C, c8_tiles, L);
 }
 
 void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
@@ -364,9 +382,15 @@ void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c8_tiles = (C8 + CT8 - 1) / CT8;
-  avgpool_fwd_kernel<<<dim3(N * c8_tiles), 256, 0, s>>>(
-      static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C,
-      c8_tiles, to_lop(lopx));
+  const Lop L = to_lop(lopx);
+  if (lopx.kind == 0)
+    avgpool_fwd_kernel<0><<<dim3(N * c8_tiles), 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C,
+        c8_tiles, L);
+  else
+    avgpool_fwd_kernel<1><<<dim3(N * c8_tiles), 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<ushort_t*>(out), HW, C,
+        c8_tiles, L);
 }
 
 void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,
