@@ -138,6 +138,88 @@ DEV_INLINE float lop_loadT(const Lop& L, const ushort_t* p, int64_t off,
   return lop_applyT<LOPK>(L, v, zv, c);
 }
 
+// Register-resident per-thread coefficient copy for loops where the
+// channel range is FIXED per thread (dw kernels, per-image reductions):
+// the coef pointer loads would otherwise be re-issued per element (the
+// compiler cannot hoist them past the output stores).
+template <int LOPK, int V>
+struct LopReg {
+  float a[V], b[V], k0[V], k2[V];
+  int act;
+  DEV_INLINE void init(const Lop& L, int c) {
+    act = L.act;
+    if (LOPK == LOP_NONE) return;
+#pragma unroll
+    for (int e = 0; e < V; ++e) {
+      a[e] = L.coef[c + e];
+      b[e] = L.coef[L.C + c + e];
+    }
+    if (LOPK == LOP_BNBWD) {
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        k0[e] = L.coef[2 * L.C + c + e];
+        k2[e] = L.coef[3 * L.C + c + e];
+      }
+    }
+  }
+  DEV_INLINE float apply(float v, float zv, int e) const {
+    if (LOPK == LOP_BNACT) return act_fwd_rt(fmaf(v, a[e], b[e]), act);
+    if (LOPK == LOP_BNBWD) {
+      const float da = v * act_bwd_rt(fmaf(zv, a[e], b[e]), act);
+      return fmaf(a[e], da, k0[e]) - k2[e] * zv;
+    }
+    return v;
+  }
+};
+
+// 8-wide vectorized lop for STREAMING channel ranges (GEMM staging):
+// coefs fetched as float4 pairs (requires c % 4 == 0, 8 in-bounds elems);
+// output written back as bf16.
+template <int LOPK>
+DEV_INLINE void lop_vec8(const Lop& L, const ushort_t* p, int64_t off,
+                         int c, ushort_t* dst) {
+  short8_t v = *reinterpret_cast<const short8_t*>(&p[off]);
+  if (LOPK == LOP_NONE) {
+    *reinterpret_cast<short8_t*>(dst) = v;
+    return;
+  }
+  float a[8], b[8];
+  *reinterpret_cast<float4_t*>(a) =
+      *reinterpret_cast<const float4_t*>(&L.coef[c]);
+  *reinterpret_cast<float4_t*>(a + 4) =
+      *reinterpret_cast<const float4_t*>(&L.coef[c + 4]);
+  *reinterpret_cast<float4_t*>(b) =
+      *reinterpret_cast<const float4_t*>(&L.coef[L.C + c]);
+  *reinterpret_cast<float4_t*>(b + 4) =
+      *reinterpret_cast<const float4_t*>(&L.coef[L.C + c + 4]);
+  short8_t out;
+  if (LOPK == LOP_BNACT) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      out[e] = short(f2bf(act_fwd_rt(
+          fmaf(bf2f(ushort_t(v[e])), a[e], b[e]), L.act)));
+  } else {
+    float k0[8], k2[8];
+    *reinterpret_cast<float4_t*>(k0) =
+        *reinterpret_cast<const float4_t*>(&L.coef[2 * L.C + c]);
+    *reinterpret_cast<float4_t*>(k0 + 4) =
+        *reinterpret_cast<const float4_t*>(&L.coef[2 * L.C + c + 4]);
+    *reinterpret_cast<float4_t*>(k2) =
+        *reinterpret_cast<const float4_t*>(&L.coef[3 * L.C + c]);
+    *reinterpret_cast<float4_t*>(k2 + 4) =
+        *reinterpret_cast<const float4_t*>(&L.coef[3 * L.C + c + 4]);
+    short8_t z = *reinterpret_cast<const short8_t*>(&L.z[off]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float zv = bf2f(ushort_t(z[e]));
+      const float da = bf2f(ushort_t(v[e])) *
+                       act_bwd_rt(fmaf(zv, a[e], b[e]), L.act);
+      out[e] = short(f2bf(fmaf(a[e], da, k0[e]) - k2[e] * zv));
+    }
+  }
+  *reinterpret_cast<short8_t*>(dst) = out;
+}
+
 // bf16 load + lop for one element of a stream at flat offset `off`,
 // channel `c` (zv fetched only when the lop needs it)
 DEV_INLINE float lop_load(const Lop& L, const ushort_t* p, int64_t off,

@@ -60,7 +60,9 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float s1[8] = {}, s2[8] = {};
+  LopReg<LOPK, 8> lr;
   if (c8 < C8 && mlane < ML) {
+    lr.init(lopx, c);
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
       const int w4 = int(p % Wo4);
       int64_t rest = p / Wo4;
@@ -80,21 +82,16 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
           if (wi >= 0 && wi < W) {
             const int64_t off = idx4(n, hi, wi, c, H, W, C);
             short8_t v = *reinterpret_cast<const short8_t*>(&x[off]);
-            if (LOPK == LOP_NONE) {
-#pragma unroll
-              for (int e = 0; e < 8; ++e) xr[t][e] = bf2f(ushort_t(v[e]));
-            } else if (LOPK == LOP_BNACT) {
+            if (LOPK != LOP_BNBWD) {
 #pragma unroll
               for (int e = 0; e < 8; ++e)
-                xr[t][e] = lop_applyT<LOP_BNACT>(lopx, bf2f(ushort_t(v[e])),
-                                                 0.f, c + e);
+                xr[t][e] = lr.apply(bf2f(ushort_t(v[e])), 0.f, e);
             } else {
               short8_t zv = *reinterpret_cast<const short8_t*>(&lopx.z[off]);
 #pragma unroll
               for (int e = 0; e < 8; ++e)
-                xr[t][e] = lop_applyT<LOP_BNBWD>(lopx, bf2f(ushort_t(v[e])),
-                                                 bf2f(ushort_t(zv[e])),
-                                                 c + e);
+                xr[t][e] = lr.apply(bf2f(ushort_t(v[e])),
+                                    bf2f(ushort_t(zv[e])), e);
             }
           } else {
 #pragma unroll
@@ -179,6 +176,8 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const int hi = int(rest % H);
     const int n = int(rest / H);
     const int c = c8 * 8;
+    LopReg<LOPK, 8> lr;
+    lr.init(lopdy, c);
     float acc[8] = {};
 #pragma unroll
     for (int i = 0; i < K; ++i) {
@@ -205,8 +204,8 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
           short8_t zv = *reinterpret_cast<const short8_t*>(&lopdy.z[off]);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(lop_applyT<LOP_BNBWD>(lopdy, bf2f(ushort_t(gv[e])),
-                                                bf2f(ushort_t(zv[e])), c + e),
+            acc[e] = fmaf(lr.apply(bf2f(ushort_t(gv[e])),
+                                   bf2f(ushort_t(zv[e])), e),
                           bf2f(ushort_t(wv[e])), acc[e]);
         }
       }
@@ -244,7 +243,11 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float acc[K * K][4] = {};
+  LopReg<LX, 4> lrx;
+  LopReg<LD, 4> lrd;
   if (c4 < C4 && mlane < ML) {
+    lrx.init(lopx, c);
+    lrd.init(lopdy, c);
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
       const int w4 = int(p % Wo4);
       int64_t rest = p / Wo4;
@@ -264,8 +267,8 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
             short4_t z4 = *reinterpret_cast<const short4_t*>(&lopdy.z[off]);
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              g[q][e] = lop_applyT<LOP_BNBWD>(lopdy, bf2f(ushort_t(g4[e])),
-                                              bf2f(ushort_t(z4[e])), c + e);
+              g[q][e] = lrd.apply(bf2f(ushort_t(g4[e])),
+                                  bf2f(ushort_t(z4[e])), e);
           }
         } else {
 #pragma unroll
@@ -290,8 +293,7 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
             } else {
 #pragma unroll
               for (int e = 0; e < 4; ++e)
-                xr[t][e] = lop_applyT<LOP_BNACT>(lopx, bf2f(ushort_t(x4[e])),
-                                                 0.f, c + e);
+                xr[t][e] = lrx.apply(bf2f(ushort_t(x4[e])), 0.f, e);
             }
           } else {
 #pragma unroll
@@ -478,7 +480,7 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
       lopdy.kind == LDD) {                                                 \
     dw_wgrad_kernel<KK, SS, LXX, LDD><<<grid, 256, 0, s>>>(                \
         xp, gp, ws, N, H, W, C, Ho, Wo, pad, Lx, Ldy, chunks);             \
-    return;                                                                \
+    goto combine;                                                          \
   }
 #define DWW_KS(KK, SS)                                                     \
   DWW_CASE(KK, SS, 0, 0) DWW_CASE(KK, SS, 1, 0) DWW_CASE(KK, SS, 0, 2)     \
@@ -486,6 +488,7 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   DWW_KS(3, 1) DWW_KS(3, 2) DWW_KS(5, 1) DWW_KS(5, 2)
 #undef DWW_KS
 #undef DWW_CASE
+combine:
   slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }
 

@@ -71,34 +71,8 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       ushort_t* dst = &Alds[a_row * LDA + a_seg];
       if (gm < M && gk + 16 <= K) {
         const int64_t off = gm * K + gk;
-        if (LOPK == LOP_NONE) {
-          *reinterpret_cast<short8_t*>(dst) =
-              *reinterpret_cast<const short8_t*>(&A[off]);
-          *reinterpret_cast<short8_t*>(dst + 8) =
-              *reinterpret_cast<const short8_t*>(&A[off + 8]);
-        } else {
-          short8_t v0 = *reinterpret_cast<const short8_t*>(&A[off]);
-          short8_t v1 = *reinterpret_cast<const short8_t*>(&A[off + 8]);
-          short8_t z0, z1;
-          if (LOPK == LOP_BNBWD) {
-            z0 = *reinterpret_cast<const short8_t*>(&lopA.z[off]);
-            z1 = *reinterpret_cast<const short8_t*>(&lopA.z[off + 8]);
-          }
-          short8_t o0, o1;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            o0[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v0[j])),
-                                         LOPK == LOP_BNBWD
-                                             ? bf2f(ushort_t(z0[j])) : 0.f,
-                                         gk + j)));
-            o1[j] = short(f2bf(lop_apply(lopA, bf2f(ushort_t(v1[j])),
-                                         LOPK == LOP_BNBWD
-                                             ? bf2f(ushort_t(z1[j])) : 0.f,
-                                         gk + 8 + j)));
-          }
-          *reinterpret_cast<short8_t*>(dst) = o0;
-          *reinterpret_cast<short8_t*>(dst + 8) = o1;
-        }
+        lop_vec8<LOPK>(lopA, A, off, gk, dst);
+        lop_vec8<LOPK>(lopA, A, off + 8, gk + 8, dst + 8);
       } else {
 #pragma unroll
         for (int j = 0; j < 16; ++j) {
@@ -280,9 +254,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Al[r * LD1 + s_seg];
         const int gn = n1_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N1 && LA == LOP_NONE) {
-          *reinterpret_cast<short8_t*>(dst) =
-              *reinterpret_cast<const short8_t*>(&A[gm * N1 + gn]);
+        if (gm < m_hi && gn + 8 <= N1) {
+          lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, dst);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
@@ -300,9 +273,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int64_t gm = m0 + r;
         ushort_t* dst = &Bl[r * LD2 + s_seg];
         const int gn = n2_0 + s_seg;
-        if (gm < m_hi && gn + 8 <= N2 && LB == LOP_NONE) {
-          *reinterpret_cast<short8_t*>(dst) =
-              *reinterpret_cast<const short8_t*>(&B[gm * N2 + gn]);
+        if (gm < m_hi && gn + 8 <= N2) {
+          lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, dst);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)

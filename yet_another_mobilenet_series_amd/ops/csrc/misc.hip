@@ -132,7 +132,9 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
   const int c = c8 * 8;
   float acc[8] = {};
   float sv[8];
+  LopReg<LOPK, 8> lr;
   if (c8 < C8 && hwlane < ML) {
+    lr.init(lopx, c);
     short8_t s8 = *reinterpret_cast<const short8_t*>(
         &scale[int64_t(n) * C + c]);
 #pragma unroll
@@ -141,7 +143,12 @@ __global__ __launch_bounds__(256) void rowscale_bwd_kernel(
       const int64_t off = base + int64_t(hw) * C + c;
       short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
       float x8[8];
-      load8_lop<LOPK>(lopx, x, off, c, x8);
+      {
+        short8_t xv = *reinterpret_cast<const short8_t*>(&x[off]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          x8[e] = lr.apply(bf2f(ushort_t(xv[e])), 0.f, e);
+      }
       short8_t o8;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
@@ -182,12 +189,15 @@ __global__ __launch_bounds__(256) void avgpool_fwd_kernel(
   const int64_t base = int64_t(n) * HW * C;
   const int c = c8 * 8;
   float acc[8] = {};
+  LopReg<LOPK, 8> lr;
   if (c8 < C8 && hwlane < ML) {
+    lr.init(lopx, c);
     for (int hw = hwlane; hw < HW; hw += ML) {
-      float x8[8];
-      load8_lop<LOPK>(lopx, x, base + int64_t(hw) * C + c, c, x8);
+      short8_t xv = *reinterpret_cast<const short8_t*>(
+          &x[base + int64_t(hw) * C + c]);
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] += x8[e];
+      for (int e = 0; e < 8; ++e) acc[e] += lr.apply(bf2f(ushort_t(xv[e])),
+                                                     0.f, e);
     }
   }
 #pragma unroll
