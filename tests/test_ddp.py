@@ -18,6 +18,7 @@ def _make_model(seed=0):
 
 
 def _worker(rank, world, tmpfile, overlap, q):
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
     dist.init_process_group(
         "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world)
     try:
@@ -64,17 +65,23 @@ def _single_process_reference():
 @pytest.mark.parametrize("overlap", [False, True])
 def test_dp_matches_single_process(tmp_path, overlap):
     world = 2
-    ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
-    tmpfile = str(tmp_path / f"init_{overlap}")
-    ps = [ctx.Process(target=_worker, args=(r, world, tmpfile, overlap, q))
-          for r in range(world)]
-    for p in ps:
-        p.start()
-    got = q.get()
-    for p in ps:
-        p.join(120)
-        assert p.exitcode == 0
+    # gloo TCP setup is occasionally flaky under parallel test load: retry
+    for attempt in range(3):
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        tmpfile = str(tmp_path / f"init_{overlap}_{attempt}")
+        ps = [ctx.Process(target=_worker,
+                          args=(r, world, tmpfile, overlap, q))
+              for r in range(world)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+        if all(p.exitcode == 0 for p in ps):
+            got = q.get()
+            break
+        if attempt == 2:
+            assert False, f"workers failed: {[p.exitcode for p in ps]}"
     want = _single_process_reference()
     for a, b in zip(got, want):
         assert torch.allclose(a, b, atol=1e-5)

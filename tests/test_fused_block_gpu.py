@@ -83,7 +83,7 @@ def test_fused_block_parity(cfg, training):
             # BOTH paths — the relative comparison is ill-posed there
             null_grad = (p.grad.float().norm().item()
                          < 1e-3 * max(total_norm, 1.0))
-            assert e < 0.1 or null_grad, (n, e)
+            assert e < 0.15 or null_grad, (n, e)
         # running stats updated identically
         for k, v in block.state_dict().items():
             if "running" in k:

@@ -42,16 +42,18 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
     int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
-  __shared__ float red[STATS ? 256 * 8 : 8];
+  // 4-wide channel vectors: the 8-wide variant's xr[XW][8] window pushed
+  // VGPR past 176 (2 waves/SIMD) once the lop/stats state joined it
+  __shared__ float red[STATS ? 256 * 4 : 4];
   constexpr int XW = 3 * S + K;
-  const int C8 = C >> 3;
+  const int C8 = C >> 2;   // 4-channel lanes
   const int CT8 = C8 < 256 ? C8 : 256;
   const int ML = 256 / CT8;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT8;
-  const int c = c8 * 8;
+  const int c = c8 * 4;
 
   const int Wo4 = (Wo + 3) / 4;
   const int64_t P = int64_t(N) * Ho * Wo4;
@@ -59,8 +61,8 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   const int64_t lo = chunk * per;
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
-  float s1[8] = {}, s2[8] = {};
-  LopReg<LOPK, 8> lr;
+  float s1[4] = {}, s2[4] = {};
+  LopReg<LOPK, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
@@ -69,55 +71,55 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
       const int ho = int(rest % Ho);
       const int n = int(rest / Ho);
       const int wo0 = w4 * 4;
-      float acc[4][8] = {};
+      float acc[4][4] = {};
 #pragma unroll
       for (int i = 0; i < K; ++i) {
         const int hi = ho * S + i - pad;
         if (hi < 0 || hi >= H) continue;
-        float xr[XW][8];
+        float xr[XW][4];
         const int wbase = wo0 * S - pad;
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
           if (wi >= 0 && wi < W) {
             const int64_t off = idx4(n, hi, wi, c, H, W, C);
-            short8_t v = *reinterpret_cast<const short8_t*>(&x[off]);
+            short4_t v = *reinterpret_cast<const short4_t*>(&x[off]);
             if (LOPK != LOP_BNBWD) {
 #pragma unroll
-              for (int e = 0; e < 8; ++e)
+              for (int e = 0; e < 4; ++e)
                 xr[t][e] = lr.apply(bf2f(ushort_t(v[e])), 0.f, e);
             } else {
-              short8_t zv = *reinterpret_cast<const short8_t*>(&lopx.z[off]);
+              short4_t zv = *reinterpret_cast<const short4_t*>(&lopx.z[off]);
 #pragma unroll
-              for (int e = 0; e < 8; ++e)
+              for (int e = 0; e < 4; ++e)
                 xr[t][e] = lr.apply(bf2f(ushort_t(v[e])),
                                     bf2f(ushort_t(zv[e])), e);
             }
           } else {
 #pragma unroll
-            for (int e = 0; e < 8; ++e) xr[t][e] = 0.f;
+            for (int e = 0; e < 4; ++e) xr[t][e] = 0.f;
           }
         }
 #pragma unroll
         for (int j = 0; j < K; ++j) {
           const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-          short8_t wv = *reinterpret_cast<const short8_t*>(&w[tap * C + c]);
-          float wf[8];
+          short4_t wv = *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
+          float wf[4];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) wf[e] = bf2f(ushort_t(wv[e]));
+          for (int e = 0; e < 4; ++e) wf[e] = bf2f(ushort_t(wv[e]));
 #pragma unroll
           for (int q = 0; q < 4; ++q)
 #pragma unroll
-            for (int e = 0; e < 8; ++e)
+            for (int e = 0; e < 4; ++e)
               acc[q][e] = fmaf(xr[q * S + j][e], wf[e], acc[q][e]);
         }
       }
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (wo0 + q >= Wo) break;
-        short8_t out;
+        short4_t out;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
+        for (int e = 0; e < 4; ++e) {
           const ushort_t o = f2bf(acc[q][e]);
           out[e] = short(o);
           if (STATS) {
@@ -126,34 +128,34 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
             s2[e] = fmaf(v, v, s2[e]);
           }
         }
-        *reinterpret_cast<short8_t*>(
+        *reinterpret_cast<short4_t*>(
             &y[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) = out;
       }
     }
   }
   if (STATS) {
 #pragma unroll
-    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s1[e];
+    for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = s1[e];
     __syncthreads();
     if (threadIdx.x < unsigned(CT8) && c8 < C8) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
+      for (int e = 0; e < 4; ++e) {
         float t = 0.f;
         for (int l = 0; l < ML; ++l)
-          t += red[(threadIdx.x + l * CT8) * 8 + e];
+          t += red[(threadIdx.x + l * CT8) * 4 + e];
         stats_ws[int64_t(chunk) * C + c + e] = t;
       }
     }
     __syncthreads();
 #pragma unroll
-    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s2[e];
+    for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = s2[e];
     __syncthreads();
     if (threadIdx.x < unsigned(CT8) && c8 < C8) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
+      for (int e = 0; e < 4; ++e) {
         float t = 0.f;
         for (int l = 0; l < ML; ++l)
-          t += red[(threadIdx.x + l * CT8) * 8 + e];
+          t += red[(threadIdx.x + l * CT8) * 4 + e];
         stats_ws[(int64_t(chunks) + chunk) * C + c + e] = t;
       }
     }
@@ -376,7 +378,7 @@ int grid_for(int64_t total) {
 }  // namespace
 
 int dw_stats_chunks(int N, int Ho, int Wo, int C) {
-  const int C8 = C >> 3;
+  const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
@@ -393,7 +395,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   const Lop L = to_lop(lopx);
-  const int C8 = C >> 3;
+  const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int chunks = dw_stats_chunks(N, Ho, Wo, C);
@@ -423,7 +425,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
   const Lop L = to_lop(lopdy);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
-    const int C8 = C >> 3;
+    const int C8 = C >> 2;
     const int CT8 = C8 < 256 ? C8 : 256;
     const int c_tiles = (C8 + CT8 - 1) / CT8;
     const int chunks = dw_stats_chunks(N, H, W, C);

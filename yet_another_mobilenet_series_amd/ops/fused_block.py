@@ -181,29 +181,24 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         cp = pw.shape[0]
         ce = ew.shape[0] if has_expand else cin
 
-        # ---- project bn backward stats + coef --------------------------
+        # ---- project bn backward: dz3 materialized once ----------------
+        # (computing dz as a BNBWD load-op inside both GEMMs costs more
+        # staging VALU/registers than the one dz write it saves)
         dg3 = torch.empty(cp, dtype=F32, device=dev)
         db3 = torch.empty_like(dg3)
-        coef3 = torch.empty(4 * cp, dtype=F32, device=dev)
-        e.bn_bwd_stats(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
-                       bool(training), dg3, db3, coef3)
+        dz3 = torch.empty_like(z3)
+        e.bn_act_bwd(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
+                     bool(training), dg3, db3, dz3)
 
-        # dgrad: d_in3 = dz3 @ Wp (dz3 computed on load)
         pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
                       .contiguous())
         d_in3 = torch.empty_like(z2)
-        e.pw_conv(dyb, pwt, d_in3, lop_kind=2, lop_act=ACT_IDENTITY,
-                  lop_coef=coef3, lop_z=z3)
-        # wgrad: dWp = dz3^T @ in3
+        e.pw_conv(dz3, pwt, d_in3)
         dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
         if has_se:
-            e.pw_conv_wgrad(dyb, y2s, dwp3, lopa_kind=2,
-                            lopa_act=ACT_IDENTITY, lopa_coef=coef3,
-                            lopa_z=z3)
+            e.pw_conv_wgrad(dz3, y2s, dwp3)
         else:
-            e.pw_conv_wgrad(dyb, z2, dwp3, lopa_kind=2,
-                            lopa_act=ACT_IDENTITY, lopa_coef=coef3,
-                            lopa_z=z3, lopb_kind=1, lopb_act=act,
+            e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1, lopb_act=act,
                             lopb_coef=ab2)
         d_pw = dwp3.reshape(cp, cd, 1, 1)
 
@@ -241,45 +236,38 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             d_y2 = d_in3
             d_sw1 = d_sw2 = db1se = db2se = None
 
-        # ---- depthwise bn backward + grads ------------------------------
+        # ---- depthwise bn backward + grads (dz2 materialized) -----------
         dg2 = torch.empty(cd, dtype=F32, device=dev)
         db2_ = torch.empty_like(dg2)
-        coef2 = torch.empty(4 * cd, dtype=F32, device=dev)
-        e.bn_bwd_stats(d_y2, z2, mean2, invstd2, dg, db, act,
-                       bool(training), dg2, db2_, coef2)
+        dz2 = torch.empty_like(z2)
+        e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
+                     bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
                        .contiguous())
         dw_in = z1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
-        e.dw_conv_dgrad(d_y2, dwpk, d_z1out, k, stride, pad, lop_kind=2,
-                        lop_act=act, lop_coef=coef2, lop_z=z2)
+        e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
         dwgt = torch.empty(k * k, cd, dtype=F32, device=dev)
         if has_expand:
-            e.dw_conv_wgrad(dw_in, d_y2, dwgt, k, stride, pad,
-                            lopx_kind=1, lopx_act=act, lopx_coef=ab1,
-                            lopdy_kind=2, lopdy_act=act, lopdy_coef=coef2,
-                            lopdy_z=z2)
+            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad,
+                            lopx_kind=1, lopx_act=act, lopx_coef=ab1)
         else:
-            e.dw_conv_wgrad(dw_in, d_y2, dwgt, k, stride, pad,
-                            lopdy_kind=2, lopdy_act=act, lopdy_coef=coef2,
-                            lopdy_z=z2)
+            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
         d_dww = dwgt.t().reshape(cd, 1, k, k)
 
         # ---- expand backward --------------------------------------------
         if has_expand:
             dg1 = torch.empty(ce, dtype=F32, device=dev)
             db1_ = torch.empty_like(dg1)
-            coef1 = torch.empty(4 * ce, dtype=F32, device=dev)
-            e.bn_bwd_stats(d_z1out, z1, mean1, invstd1, eg, ebb, act,
-                           bool(training), dg1, db1_, coef1)
+            dz1 = torch.empty_like(z1)
+            e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                         bool(training), dg1, db1_, dz1)
             ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
                           .contiguous())
             d_x = _empty_like_nhwc(x)
-            e.pw_conv(d_z1out, ewt, d_x, lop_kind=2, lop_act=act,
-                      lop_coef=coef1, lop_z=z1)
+            e.pw_conv(dz1, ewt, d_x)
             dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
-            e.pw_conv_wgrad(d_z1out, x, dwp1, lopa_kind=2, lopa_act=act,
-                            lopa_coef=coef1, lopa_z=z1)
+            e.pw_conv_wgrad(dz1, x, dwp1)
             d_ew = dwp1.reshape(ce, cin, 1, 1)
         else:
             d_x = d_z1out
