@@ -91,8 +91,8 @@ class InvertedResidual(nn.Module):
         self.project = ConvBNAct(expand_ch, cout, 1, act="identity", **bn_kw)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        from ..ops.backend import use_hip
-        if use_hip(x):
+        from ..ops.backend import use_fused_blocks, use_hip
+        if use_hip(x) and use_fused_blocks():
             from ..ops.fused_block import fused_inverted_residual
             return fused_inverted_residual(x, self)
         y = self.expand(x) if self.expand is not None else x

@@ -70,6 +70,20 @@ def use_hip(x) -> bool:
     return x.is_cuda and _ENGINE_MODE == "hip"
 
 
+_FUSED_BLOCKS = os.environ.get("YAMS_FUSED", "1") != "0"
+
+
+def use_fused_blocks() -> bool:
+    """Block-level fusion toggle (YAMS_FUSED=0 falls back to per-layer
+    HIP ops — used for A/B benchmarking, not a different backend)."""
+    return _FUSED_BLOCKS
+
+
+def set_fused_blocks(on: bool) -> None:
+    global _FUSED_BLOCKS
+    _FUSED_BLOCKS = on
+
+
 def ext():
     """The extension module; raises if missing (GPU path must not fall back)."""
     return load_extension(required=True)
