@@ -65,31 +65,48 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   LopReg<LOPK, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
+    // weights for this thread's 4 channels, preloaded once
+    float wf[K * K][4];
+#pragma unroll
+    for (int i = 0; i < K; ++i)
+#pragma unroll
+      for (int j = 0; j < K; ++j) {
+        const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+        short4_t wv = *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          wf[i * K + j][e] = bf2f(ushort_t(wv[e]));
+      }
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
       const int w4 = int(p % Wo4);
       int64_t rest = p / Wo4;
       const int ho = int(rest % Ho);
       const int n = int(rest / Ho);
       const int wo0 = w4 * 4;
+      const int wbase = wo0 * S - pad;
+      const int64_t imgbase = int64_t(n) * H * W * C + c;
       float acc[4][4] = {};
 #pragma unroll
       for (int i = 0; i < K; ++i) {
         const int hi = ho * S + i - pad;
         if (hi < 0 || hi >= H) continue;
+        const ushort_t* xrow = x + imgbase + int64_t(hi) * W * C;
+        const ushort_t* zrow =
+            LOPK == LOP_BNBWD ? lopx.z + imgbase + int64_t(hi) * W * C
+                              : nullptr;
         float xr[XW][4];
-        const int wbase = wo0 * S - pad;
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
           if (wi >= 0 && wi < W) {
-            const int64_t off = idx4(n, hi, wi, c, H, W, C);
-            short4_t v = *reinterpret_cast<const short4_t*>(&x[off]);
+            const int off = wi * C;
+            short4_t v = *reinterpret_cast<const short4_t*>(&xrow[off]);
             if (LOPK != LOP_BNBWD) {
 #pragma unroll
               for (int e = 0; e < 4; ++e)
                 xr[t][e] = lr.apply(bf2f(ushort_t(v[e])), 0.f, e);
             } else {
-              short4_t zv = *reinterpret_cast<const short4_t*>(&lopx.z[off]);
+              short4_t zv = *reinterpret_cast<const short4_t*>(&zrow[off]);
 #pragma unroll
               for (int e = 0; e < 4; ++e)
                 xr[t][e] = lr.apply(bf2f(ushort_t(v[e])),
@@ -101,19 +118,16 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
           }
         }
 #pragma unroll
-        for (int j = 0; j < K; ++j) {
-          const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-          short4_t wv = *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
-          float wf[4];
-#pragma unroll
-          for (int e = 0; e < 4; ++e) wf[e] = bf2f(ushort_t(wv[e]));
+        for (int j = 0; j < K; ++j)
 #pragma unroll
           for (int q = 0; q < 4; ++q)
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              acc[q][e] = fmaf(xr[q * S + j][e], wf[e], acc[q][e]);
-        }
+              acc[q][e] = fmaf(xr[q * S + j][e], wf[i * K + j][e],
+                               acc[q][e]);
       }
+      ushort_t* yout =
+          y + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + int64_t(wo0) * C + c;
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (wo0 + q >= Wo) break;
@@ -128,8 +142,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
             s2[e] = fmaf(v, v, s2[e]);
           }
         }
-        *reinterpret_cast<short4_t*>(
-            &y[idx4(n, ho, wo0 + q, c, Ho, Wo, C)]) = out;
+        *reinterpret_cast<short4_t*>(&yout[q * C]) = out;
       }
     }
   }
@@ -256,17 +269,23 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
       const int ho = int(rest % Ho);
       const int n = int(rest / Ho);
       const int wo0 = w4 * 4;
+      const ushort_t* dyrow =
+          dy + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + c;
+      const ushort_t* dzrow =
+          LD == LOP_BNBWD
+              ? lopdy.z + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + c
+              : nullptr;
       float g[4][4];
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (wo0 + q < Wo) {
-          const int64_t off = idx4(n, ho, wo0 + q, c, Ho, Wo, C);
-          short4_t g4 = *reinterpret_cast<const short4_t*>(&dy[off]);
+          const int off = (wo0 + q) * C;
+          short4_t g4 = *reinterpret_cast<const short4_t*>(&dyrow[off]);
           if (LD == LOP_NONE) {
 #pragma unroll
             for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
           } else {
-            short4_t z4 = *reinterpret_cast<const short4_t*>(&lopdy.z[off]);
+            short4_t z4 = *reinterpret_cast<const short4_t*>(&dzrow[off]);
 #pragma unroll
             for (int e = 0; e < 4; ++e)
               g[q][e] = lrd.apply(bf2f(ushort_t(g4[e])),
@@ -277,18 +296,19 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
           for (int e = 0; e < 4; ++e) g[q][e] = 0.f;
         }
       }
+      const int wbase = wo0 * S - pad;
+      const int64_t imgbase = int64_t(n) * H * W * C + c;
 #pragma unroll
       for (int i = 0; i < K; ++i) {
         const int hi2 = ho * S + i - pad;
         if (hi2 < 0 || hi2 >= H) continue;
+        const ushort_t* xrow = x + imgbase + int64_t(hi2) * W * C;
         float xr[XW][4];
-        const int wbase = wo0 * S - pad;
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
           if (wi >= 0 && wi < W) {
-            const int64_t off = idx4(n, hi2, wi, c, H, W, C);
-            short4_t x4 = *reinterpret_cast<const short4_t*>(&x[off]);
+            short4_t x4 = *reinterpret_cast<const short4_t*>(&xrow[wi * C]);
             if (LX == LOP_NONE) {
 #pragma unroll
               for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
