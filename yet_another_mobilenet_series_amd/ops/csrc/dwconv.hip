@@ -31,22 +31,22 @@ inline Lop to_lop(const LopDesc& d) {
   return L;
 }
 
-// forward: 4 consecutive wo outputs per thread share the x row window in
-// registers.  Grid = (c8_tiles x chunks) so the block owns a channel range
-// and can fold output BN-stats partials (STATS) through LDS; lopx fuses
-// the producer's BN+act into the x loads.  FLIP reuses the kernel as the
-// stride-1 dgrad (correlation with the flipped filter; lopx then is the
-// BNBWD load-op on dy).
+// forward: each thread computes a TH x 4 (ho x wo) output tile for 4
+// channels, so every input row is loaded ONCE per thread and reused by
+// the K output rows it feeds (the row-per-thread version re-read x up to
+// K times through L2 — the dominant cost on these latency-bound shapes).
+// FLIP reuses the kernel as the stride-1 dgrad; lopx fuses the producer
+// BN+act (or the BNBWD dz for dgrad) into the loads.
 template <int K, int S, bool FLIP, bool STATS, int LOPK>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
     int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
-  // 4-wide channel vectors: the 8-wide variant's xr[XW][8] window pushed
-  // VGPR past 176 (2 waves/SIMD) once the lop/stats state joined it
   __shared__ float red[STATS ? 256 * 4 : 4];
-  constexpr int XW = 3 * S + K;
-  const int C8 = C >> 2;   // 4-channel lanes
+  constexpr int XW = 3 * S + K;        // x cols for 4 wo outputs
+  constexpr int TH = (S == 1) ? 4 : 2; // output rows per thread
+  constexpr int RR = (TH - 1) * S + K; // input rows touched
+  const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int ML = 256 / CT8;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
@@ -56,7 +56,8 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   const int c = c8 * 4;
 
   const int Wo4 = (Wo + 3) / 4;
-  const int64_t P = int64_t(N) * Ho * Wo4;
+  const int HoG = (Ho + TH - 1) / TH;
+  const int64_t P = int64_t(N) * HoG * Wo4;
   const int64_t per = (P + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
@@ -65,30 +66,19 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   LopReg<LOPK, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
-    // weights for this thread's 4 channels, preloaded once
-    float wf[K * K][4];
-#pragma unroll
-    for (int i = 0; i < K; ++i)
-#pragma unroll
-      for (int j = 0; j < K; ++j) {
-        const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-        short4_t wv = *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
-#pragma unroll
-        for (int e = 0; e < 4; ++e)
-          wf[i * K + j][e] = bf2f(ushort_t(wv[e]));
-      }
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
       const int w4 = int(p % Wo4);
       int64_t rest = p / Wo4;
-      const int ho = int(rest % Ho);
-      const int n = int(rest / Ho);
+      const int hg = int(rest % HoG);
+      const int n = int(rest / HoG);
+      const int ho0 = hg * TH;
       const int wo0 = w4 * 4;
       const int wbase = wo0 * S - pad;
       const int64_t imgbase = int64_t(n) * H * W * C + c;
-      float acc[4][4] = {};
+      float acc[TH][4][4] = {};
 #pragma unroll
-      for (int i = 0; i < K; ++i) {
-        const int hi = ho * S + i - pad;
+      for (int r = 0; r < RR; ++r) {
+        const int hi = ho0 * S + r - pad;
         if (hi < 0 || hi >= H) continue;
         const ushort_t* xrow = x + imgbase + int64_t(hi) * W * C;
         const ushort_t* zrow =
@@ -118,31 +108,45 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
           }
         }
 #pragma unroll
-        for (int j = 0; j < K; ++j)
+        for (int t = 0; t < TH; ++t) {
+          const int i = r - t * S;
+          if (i < 0 || i >= K) continue;
 #pragma unroll
-          for (int q = 0; q < 4; ++q)
+          for (int j = 0; j < K; ++j) {
+            const int tap =
+                FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+            short4_t wv =
+                *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
 #pragma unroll
-            for (int e = 0; e < 4; ++e)
-              acc[q][e] = fmaf(xr[q * S + j][e], wf[i * K + j][e],
-                               acc[q][e]);
-      }
-      ushort_t* yout =
-          y + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + int64_t(wo0) * C + c;
+            for (int q = 0; q < 4; ++q)
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
-        if (wo0 + q >= Wo) break;
-        short4_t out;
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const ushort_t o = f2bf(acc[q][e]);
-          out[e] = short(o);
-          if (STATS) {
-            const float v = bf2f(o);
-            s1[e] += v;
-            s2[e] = fmaf(v, v, s2[e]);
+              for (int e = 0; e < 4; ++e)
+                acc[t][q][e] = fmaf(xr[q * S + j][e],
+                                    bf2f(ushort_t(wv[e])), acc[t][q][e]);
           }
         }
-        *reinterpret_cast<short4_t*>(&yout[q * C]) = out;
+      }
+#pragma unroll
+      for (int t = 0; t < TH; ++t) {
+        if (ho0 + t >= Ho) break;
+        ushort_t* yout = y + (int64_t(n) * Ho + ho0 + t) * int64_t(Wo) * C +
+                         int64_t(wo0) * C + c;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          if (wo0 + q >= Wo) break;
+          short4_t out;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const ushort_t o = f2bf(acc[t][q][e]);
+            out[e] = short(o);
+            if (STATS) {
+              const float v = bf2f(o);
+              s1[e] += v;
+              s2[e] = fmaf(v, v, s2[e]);
+            }
+          }
+          *reinterpret_cast<short4_t*>(&yout[q * C]) = out;
+        }
       }
     }
   }
@@ -401,7 +405,7 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C) {
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
-  const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
+  const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
   int64_t want = 2048 / c_tiles;
   int64_t maxc = P;
   int c = int(std::min(std::max<int64_t>(want, 1), maxc));
