@@ -79,32 +79,35 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < RR; ++r) {
         const int hi = ho0 * S + r - pad;
-        if (hi < 0 || hi >= H) continue;
-        const ushort_t* xrow = x + imgbase + int64_t(hi) * W * C;
+        const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
+        const bool row_ok = hi >= 0 && hi < H;
+        const ushort_t* xrow = x + imgbase + int64_t(hic) * W * C;
         const ushort_t* zrow =
-            LOPK == LOP_BNBWD ? lopx.z + imgbase + int64_t(hi) * W * C
+            LOPK == LOP_BNBWD ? lopx.z + imgbase + int64_t(hic) * W * C
                               : nullptr;
         float xr[XW][4];
+        // branchless: always load from a CLAMPED in-bounds address and
+        // zero by value select — a per-element branch around a load makes
+        // hipcc wait vmcnt(0) per load (serial HBM round trips)
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
-          if (wi >= 0 && wi < W) {
-            const int off = wi * C;
-            short4_t v = *reinterpret_cast<const short4_t*>(&xrow[off]);
-            if (LOPK != LOP_BNBWD) {
+          const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+          const bool ok = wi >= 0 && wi < W;
+          const int off = wic * C;
+          short4_t v = *reinterpret_cast<const short4_t*>(&xrow[off]);
+          const bool keep = ok && row_ok;
+          if (LOPK != LOP_BNBWD) {
 #pragma unroll
-              for (int e = 0; e < 4; ++e)
-                xr[t][e] = lr.apply(bf2f(ushort_t(v[e])), 0.f, e);
-            } else {
-              short4_t zv = *reinterpret_cast<const short4_t*>(&zrow[off]);
-#pragma unroll
-              for (int e = 0; e < 4; ++e)
-                xr[t][e] = lr.apply(bf2f(ushort_t(v[e])),
-                                    bf2f(ushort_t(zv[e])), e);
-            }
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])), 0.f, e) : 0.f;
           } else {
+            short4_t zv = *reinterpret_cast<const short4_t*>(&zrow[off]);
 #pragma unroll
-            for (int e = 0; e < 4; ++e) xr[t][e] = 0.f;
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])),
+                                         bf2f(ushort_t(zv[e])), e)
+                              : 0.f;
           }
         }
 #pragma unroll
@@ -282,22 +285,22 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
       float g[4][4];
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
-        if (wo0 + q < Wo) {
-          const int off = (wo0 + q) * C;
-          short4_t g4 = *reinterpret_cast<const short4_t*>(&dyrow[off]);
-          if (LD == LOP_NONE) {
+        const int wo = wo0 + q;
+        const int woc = wo < Wo ? wo : Wo - 1;
+        const bool ok = wo < Wo;
+        const int off = woc * C;
+        short4_t g4 = *reinterpret_cast<const short4_t*>(&dyrow[off]);
+        if (LD == LOP_NONE) {
 #pragma unroll
-            for (int e = 0; e < 4; ++e) g[q][e] = bf2f(ushort_t(g4[e]));
-          } else {
-            short4_t z4 = *reinterpret_cast<const short4_t*>(&dzrow[off]);
-#pragma unroll
-            for (int e = 0; e < 4; ++e)
-              g[q][e] = lrd.apply(bf2f(ushort_t(g4[e])),
-                                  bf2f(ushort_t(z4[e])), e);
-          }
+          for (int e = 0; e < 4; ++e)
+            g[q][e] = ok ? bf2f(ushort_t(g4[e])) : 0.f;
         } else {
+          short4_t z4 = *reinterpret_cast<const short4_t*>(&dzrow[off]);
 #pragma unroll
-          for (int e = 0; e < 4; ++e) g[q][e] = 0.f;
+          for (int e = 0; e < 4; ++e)
+            g[q][e] = ok ? lrd.apply(bf2f(ushort_t(g4[e])),
+                                     bf2f(ushort_t(z4[e])), e)
+                         : 0.f;
         }
       }
       const int wbase = wo0 * S - pad;
@@ -305,25 +308,25 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
 #pragma unroll
       for (int i = 0; i < K; ++i) {
         const int hi2 = ho * S + i - pad;
-        if (hi2 < 0 || hi2 >= H) continue;
-        const ushort_t* xrow = x + imgbase + int64_t(hi2) * W * C;
+        const int hic = hi2 < 0 ? 0 : (hi2 >= H ? H - 1 : hi2);
+        const bool row_ok = hi2 >= 0 && hi2 < H;
+        const ushort_t* xrow = x + imgbase + int64_t(hic) * W * C;
         float xr[XW][4];
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
           const int wi = wbase + t;
-          if (wi >= 0 && wi < W) {
-            short4_t x4 = *reinterpret_cast<const short4_t*>(&xrow[wi * C]);
-            if (LX == LOP_NONE) {
+          const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+          const bool keep = row_ok && wi >= 0 && wi < W;
+          short4_t x4 = *reinterpret_cast<const short4_t*>(&xrow[wic * C]);
+          if (LX == LOP_NONE) {
 #pragma unroll
-              for (int e = 0; e < 4; ++e) xr[t][e] = bf2f(ushort_t(x4[e]));
-            } else {
-#pragma unroll
-              for (int e = 0; e < 4; ++e)
-                xr[t][e] = lrx.apply(bf2f(ushort_t(x4[e])), 0.f, e);
-            }
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? bf2f(ushort_t(x4[e])) : 0.f;
           } else {
 #pragma unroll
-            for (int e = 0; e < 4; ++e) xr[t][e] = 0.f;
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? lrx.apply(bf2f(ushort_t(x4[e])), 0.f, e)
+                              : 0.f;
           }
         }
 #pragma unroll
@@ -376,12 +379,16 @@ __global__ __launch_bounds__(256) void im2row_kernel(
     int t = 0;
     for (int i = 0; i < k; ++i) {
       const int hi = ho * stride + i - pad;
+      const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
       for (int j = 0; j < k; ++j) {
         const int wi = wo * stride + j - pad;
+        const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
         const bool ok = hi >= 0 && hi < H && wi >= 0 && wi < W;
 #pragma unroll
-        for (int ci = 0; ci < 3; ++ci)
-          buf[t++] = ok ? x[idx4(n, hi, wi, ci, H, W, Cin)] : ushort_t(0);
+        for (int ci = 0; ci < 3; ++ci) {
+          const ushort_t v = x[idx4(n, hic, wic, ci, H, W, Cin)];
+          buf[t++] = ok ? v : ushort_t(0);
+        }
       }
     }
     for (; t < KP; ++t) buf[t] = ushort_t(0);
