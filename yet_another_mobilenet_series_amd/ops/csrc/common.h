@@ -172,6 +172,35 @@ struct LopReg {
   }
 };
 
+// compile-time (kind, act) register-resident variant
+template <int LOPK, int ACT, int V>
+struct LopRegT {
+  float a[V], b[V], k0[V], k2[V];
+  DEV_INLINE void init(const Lop& L, int c) {
+    if (LOPK == LOP_NONE) return;
+#pragma unroll
+    for (int e = 0; e < V; ++e) {
+      a[e] = L.coef[c + e];
+      b[e] = L.coef[L.C + c + e];
+    }
+    if (LOPK == LOP_BNBWD) {
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        k0[e] = L.coef[2 * L.C + c + e];
+        k2[e] = L.coef[3 * L.C + c + e];
+      }
+    }
+  }
+  DEV_INLINE float apply(float v, float zv, int e) const {
+    if (LOPK == LOP_BNACT) return act_fwd<ACT>(fmaf(v, a[e], b[e]));
+    if (LOPK == LOP_BNBWD) {
+      const float da = v * act_bwd<ACT>(fmaf(zv, a[e], b[e]));
+      return fmaf(a[e], da, k0[e]) - k2[e] * zv;
+    }
+    return v;
+  }
+};
+
 // 8-wide vectorized lop for STREAMING channel ranges (GEMM staging):
 // coefs fetched as float4 pairs (requires c % 4 == 0, 8 in-bounds elems);
 // output written back as bf16.

@@ -37,14 +37,14 @@ inline Lop to_lop(const LopDesc& d) {
 // K times through L2 — the dominant cost on these latency-bound shapes).
 // FLIP reuses the kernel as the stride-1 dgrad; lopx fuses the producer
 // BN+act (or the BNBWD dz for dgrad) into the loads.
-template <int K, int S, bool FLIP, bool STATS, int LOPK>
+template <int K, int S, bool FLIP, bool STATS, int LOPK, int ACT>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
     int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
   __shared__ float red[STATS ? 256 * 4 : 4];
   constexpr int XW = 3 * S + K;        // x cols for 4 wo outputs
-  constexpr int TH = (S == 1) ? 4 : 2; // output rows per thread
+  constexpr int TH = (S == 1 && K == 3) ? 4 : 2;  // rows per thread
   constexpr int RR = (TH - 1) * S + K; // input rows touched
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
@@ -63,7 +63,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float s1[4] = {}, s2[4] = {};
-  LopReg<LOPK, 4> lr;
+  LopRegT<LOPK, ACT, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
     for (int64_t p = lo + mlane; p < hi_; p += ML) {
@@ -183,7 +183,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
 }
 
 // dgrad for stride 2 (stride 1 routes through dw_fwd_kernel FLIP=true)
-template <int K, int LOPK>
+template <int K, int LOPK, int ACT>
 __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
@@ -198,7 +198,7 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const int hi = int(rest % H);
     const int n = int(rest / H);
     const int c = c8 * 8;
-    LopReg<LOPK, 8> lr;
+    LopRegT<LOPK, ACT, 8> lr;
     lr.init(lopdy, c);
     float acc[8] = {};
 #pragma unroll
@@ -242,7 +242,7 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
 // wgrad: thread = 4-channel vector lane; 4 consecutive wo outputs share
 // one x row window in registers; per-chunk fp32 slab after an LDS fold
 // over the m lanes.
-template <int K, int S, int LX, int LD>
+template <int K, int S, int LX, int LXACT, int LD>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
@@ -265,8 +265,8 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
   float acc[K * K][4] = {};
-  LopReg<LX, 4> lrx;
-  LopReg<LD, 4> lrd;
+  LopRegT<LX, LXACT, 4> lrx;
+  LopRegT<LD, 0, 4> lrd;
   if (c4 < C4 && mlane < ML) {
     lrx.init(lopx, c);
     lrd.init(lopdy, c);
@@ -431,19 +431,21 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int chunks = dw_stats_chunks(N, Ho, Wo, C);
   dim3 g(c_tiles * chunks);
-  // fwd lop kinds in practice: none or bnact
-#define DWF_CASE(KK, SS, ST, LK)                                           \
+  // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
+#define DWF_CASE(KK, SS, ST, LK, AC)                                       \
   if (k == KK && stride == SS && (stats_ws != nullptr) == ST &&            \
-      lopx.kind == LK) {                                                   \
-    dw_fwd_kernel<KK, SS, false, ST, LK><<<g, 256, 0, s>>>(                \
+      lopx.kind == LK && (LK == 0 || lopx.act == AC)) {                    \
+    dw_fwd_kernel<KK, SS, false, ST, LK, AC><<<g, 256, 0, s>>>(            \
         xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);         \
     return;                                                                \
   }
-#define DWF_KS(KK, SS)                                                     \
-  DWF_CASE(KK, SS, false, 0) DWF_CASE(KK, SS, false, 1)                    \
-  DWF_CASE(KK, SS, true, 0) DWF_CASE(KK, SS, true, 1)
+#define DWF_ST(KK, SS, ST)                                                 \
+  DWF_CASE(KK, SS, ST, 0, 0) DWF_CASE(KK, SS, ST, 1, 1)                    \
+  DWF_CASE(KK, SS, ST, 1, 2) DWF_CASE(KK, SS, ST, 1, 3)
+#define DWF_KS(KK, SS) DWF_ST(KK, SS, false) DWF_ST(KK, SS, true)
   DWF_KS(3, 1) DWF_KS(3, 2) DWF_KS(5, 1) DWF_KS(5, 2)
 #undef DWF_KS
+#undef DWF_ST
 #undef DWF_CASE
 }
 
@@ -461,25 +463,26 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     const int c_tiles = (C8 + CT8 - 1) / CT8;
     const int chunks = dw_stats_chunks(N, H, W, C);
     dim3 g(c_tiles * chunks);
-#define DWD_CASE(KK, LK)                                                   \
-    if (k == KK && lopdy.kind == LK) {                                     \
-      dw_fwd_kernel<KK, 1, true, false, LK><<<g, 256, 0, s>>>(             \
+    // dgrad supports only lop none (dz is materialized by bn_act_bwd)
+#define DWD_CASE(KK)                                                       \
+    if (k == KK && lopdy.kind == 0) {                                      \
+      dw_fwd_kernel<KK, 1, true, false, 0, 0><<<g, 256, 0, s>>>(           \
           gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);        \
       return;                                                              \
     }
-    DWD_CASE(3, 0) DWD_CASE(3, 2) DWD_CASE(5, 0) DWD_CASE(5, 2)
+    DWD_CASE(3) DWD_CASE(5)
 #undef DWD_CASE
     return;
   }
   const int64_t total8 = int64_t(N) * H * W * (C >> 3);
   dim3 g2(grid_for(total8));
-#define DWD2_CASE(KK, LK)                                                  \
-  if (k == KK && lopdy.kind == LK) {                                       \
-    dw_dgrad_s2_kernel<KK, LK><<<g2, 256, 0, s>>>(                         \
+#define DWD2_CASE(KK)                                                      \
+  if (k == KK && lopdy.kind == 0) {                                        \
+    dw_dgrad_s2_kernel<KK, 0, 0><<<g2, 256, 0, s>>>(                       \
         gp, wp, xp, N, H, W, C, Ho, Wo, stride, pad, L, total8);           \
     return;                                                                \
   }
-  DWD2_CASE(3, 0) DWD2_CASE(3, 2) DWD2_CASE(5, 0) DWD2_CASE(5, 2)
+  DWD2_CASE(3) DWD2_CASE(5)
 #undef DWD2_CASE
 }
 
@@ -508,16 +511,17 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   const Lop Lx = to_lop(lopx);
   const Lop Ldy = to_lop(lopdy);
   dim3 grid(c_tiles * chunks);
-#define DWW_CASE(KK, SS, LXX, LDD)                                         \
+  // wgrad x-stream lop: none or bnact (act 1..3); dy-stream: none only
+#define DWW_CASE(KK, SS, LXX, AC)                                          \
   if (k == KK && stride == SS && lopx.kind == LXX &&                       \
-      lopdy.kind == LDD) {                                                 \
-    dw_wgrad_kernel<KK, SS, LXX, LDD><<<grid, 256, 0, s>>>(                \
+      (LXX == 0 || lopx.act == AC) && lopdy.kind == 0) {                   \
+    dw_wgrad_kernel<KK, SS, LXX, AC, 0><<<grid, 256, 0, s>>>(              \
         xp, gp, ws, N, H, W, C, Ho, Wo, pad, Lx, Ldy, chunks);             \
     goto combine;                                                          \
   }
 #define DWW_KS(KK, SS)                                                     \
-  DWW_CASE(KK, SS, 0, 0) DWW_CASE(KK, SS, 1, 0) DWW_CASE(KK, SS, 0, 2)     \
-  DWW_CASE(KK, SS, 1, 2)
+  DWW_CASE(KK, SS, 0, 0) DWW_CASE(KK, SS, 1, 1) DWW_CASE(KK, SS, 1, 2)     \
+  DWW_CASE(KK, SS, 1, 3)
   DWW_KS(3, 1) DWW_KS(3, 2) DWW_KS(5, 1) DWW_KS(5, 2)
 #undef DWW_KS
 #undef DWW_CASE
