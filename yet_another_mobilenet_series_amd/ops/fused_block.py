@@ -92,7 +92,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         m2 = n * ho * wo
         mean2, invstd2, ab2 = _bn_bufs(cd, dev)
         if training:
-            chunks = e.dw_stats_chunks(n, ho, wo, cd)
+            chunks = e.dw_stats_chunks(n, ho, wo, cd, stride)
             ws = torch.empty(2 * chunks * cd, dtype=F32, device=dev)
             e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
                           lop_kind=dw_lop[0], lop_act=dw_lop[1],
