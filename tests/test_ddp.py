@@ -17,7 +17,7 @@ def _make_model(seed=0):
     return nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 4))
 
 
-def _worker(rank, world, tmpfile, overlap, q):
+def _worker(rank, world, tmpfile, overlap, out_path):
     os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
     dist.init_process_group(
         "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world)
@@ -39,7 +39,8 @@ def _worker(rank, world, tmpfile, overlap, q):
                 for p in model.parameters():
                     p -= 0.1 * p.grad
         if rank == 0:
-            q.put([p.detach().clone() for p in model.parameters()])
+            torch.save([p.detach().clone() for p in model.parameters()],
+                       out_path)
     finally:
         dist.destroy_process_group()
 
@@ -68,17 +69,17 @@ def test_dp_matches_single_process(tmp_path, overlap):
     # gloo TCP setup is occasionally flaky under parallel test load: retry
     for attempt in range(3):
         ctx = mp.get_context("spawn")
-        q = ctx.SimpleQueue()
         tmpfile = str(tmp_path / f"init_{overlap}_{attempt}")
+        out_path = str(tmp_path / f"out_{overlap}_{attempt}.pt")
         ps = [ctx.Process(target=_worker,
-                          args=(r, world, tmpfile, overlap, q))
+                          args=(r, world, tmpfile, overlap, out_path))
               for r in range(world)]
         for p in ps:
             p.start()
         for p in ps:
             p.join(120)
-        if all(p.exitcode == 0 for p in ps):
-            got = q.get()
+        if all(p.exitcode == 0 for p in ps) and os.path.exists(out_path):
+            got = torch.load(out_path)
             break
         if attempt == 2:
             assert False, f"workers failed: {[p.exitcode for p in ps]}"

@@ -50,7 +50,7 @@ for (c, h, k, s) in [(72, 56, 5, 2), (120, 28, 5, 1), (240, 28, 3, 2),
     w = (torch.randn(k * k, c, device=dev) * 0.3).to(BF16)
     y = torch.empty(N, c, ho, ho, device=dev, dtype=BF16, memory_format=CL)
     ab = torch.rand(2 * c, device=dev)
-    chunks = e.dw_stats_chunks(N, ho, ho, c)
+    chunks = e.dw_stats_chunks(N, ho, ho, c, s)
     ws = torch.empty(2 * chunks * c, device=dev)
     base = timeit(lambda: e.dw_conv_fwd(x, w, y, k, s, p))
     lop = timeit(lambda: e.dw_conv_fwd(x, w, y, k, s, p, lop_kind=1,

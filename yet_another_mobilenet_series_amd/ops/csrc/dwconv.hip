@@ -31,172 +31,152 @@ inline Lop to_lop(const LopDesc& d) {
   return L;
 }
 
-// LDS-tiled depthwise forward (the CDNA4-native shape for this op):
-// a block stages one (input-rows x input-cols x 64ch) tile plus the taps
-// into LDS, then 16 channel-lanes x 16 workers compute the output tile
-// from LDS.  Staging costs few registers and the compute loop has no HBM
-// latency to hide, so occupancy stays high — the register-tiled variants
-// needed 200-256 VGPRs to keep their window loads in flight (1-2
-// waves/SIMD) and sat 10-20x off the bandwidth roofline.
-// FLIP reuses the kernel as the stride-1 dgrad; the lop (producer BN+act)
-// applies once per staged element.
+// forward: each thread computes a TH x 4 (ho x wo) output tile for 4
+// channels, so every input row is loaded ONCE per thread and reused by
+// the K output rows it feeds (the row-per-thread version re-read x up to
+// K times through L2 — the dominant cost on these latency-bound shapes).
+// FLIP reuses the kernel as the stride-1 dgrad; lopx fuses the producer
+// BN+act (or the BNBWD dz for dgrad) into the loads.
 template <int K, int S, bool FLIP, bool STATS, int LOPK, int ACT>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
-    int pad, Lop lopx, float* __restrict__ stats_ws, int sp_tiles) {
-  constexpr int TO = (S == 1) ? 8 : 6;         // output tile side
-  constexpr int TI = (TO - 1) * S + K;         // input tile side
-  constexpr int NO = (TO * TO + 15) / 16;      // outputs per worker
-  __shared__ ushort_t xt[TI * TI * 64];
-  __shared__ ushort_t wt[K * K * 64];
-  __shared__ float red[STATS ? 16 * 64 * 2 : 1];
+    int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
+  __shared__ float red[STATS ? 256 * 4 : 4];
+  constexpr int XW = 3 * S + K;        // x cols for 4 wo outputs
+  constexpr int TH = (S == 1 && K == 3) ? 4 : 2;  // rows per thread
+  constexpr int RR = (TH - 1) * S + K; // input rows touched
+  const int C8 = C >> 2;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
+  const int chunk = blockIdx.x / c_tiles;
+  const int mlane = int(threadIdx.x) / CT8;
+  const int c = c8 * 4;
 
-  const int tid = threadIdx.x;
-  const int c_tiles = (C + 63) / 64;
-  const int ct = blockIdx.x % c_tiles;
-  const int sp = blockIdx.x / c_tiles;
-  const int WoT = (Wo + TO - 1) / TO;
-  const int HoT = (Ho + TO - 1) / TO;
-  const int wo_t = sp % WoT;
-  int rest = sp / WoT;
-  const int ho_t = rest % HoT;
-  const int n = rest / HoT;
-  const int c0 = ct * 64;
+  const int Wo4 = (Wo + 3) / 4;
+  const int HoG = (Ho + TH - 1) / TH;
+  const int64_t P = int64_t(N) * HoG * Wo4;
+  const int64_t per = (P + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
-  // ---- stage taps (zero-padded past C) ----
-  for (int u = tid; u < K * K * 16; u += 256) {
-    const int tap = u / 16, c4 = u % 16;
-    const int c = c0 + c4 * 4;
-    short4_t v{0, 0, 0, 0};
-    if (c + 4 <= C)
-      v = *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
-    *reinterpret_cast<short4_t*>(&wt[tap * 64 + c4 * 4]) = v;
-  }
-
-  // ---- stage input tile (lop applied once per element) ----
-  LopRegT<LOPK, ACT, 4> lr;
-  {
-    const int c4s = tid % 16;
-    const int cs = c0 + c4s * 4;
-    lr.init(lopx, cs < C ? cs : 0);
-    const int64_t imgbase = int64_t(n) * H * W * C;
-    for (int u = tid; u < TI * TI * 16; u += 256) {
-      const int c4 = u % 16;  // == c4s every iteration (stride 256 = 16*16)
-      const int col = (u / 16) % TI;
-      const int row = u / (16 * TI);
-      const int hi = ho_t * TO * S + row - pad;
-      const int wi = wo_t * TO * S + col - pad;
-      const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
-      const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
-      const int c = c0 + c4 * 4;
-      const bool ok = hi >= 0 && hi < H && wi >= 0 && wi < W && c + 4 <= C;
-      short4_t v = *reinterpret_cast<const short4_t*>(
-          &x[imgbase + (int64_t(hic) * W + wic) * C + (c + 4 <= C ? c : 0)]);
-      short4_t sv{0, 0, 0, 0};
-      if (LOPK == LOP_BNBWD) {
-        short4_t zv = *reinterpret_cast<const short4_t*>(
-            &lopx.z[imgbase + (int64_t(hic) * W + wic) * C +
-                    (c + 4 <= C ? c : 0)]);
-#pragma unroll
-        for (int e = 0; e < 4; ++e)
-          sv[e] = ok ? short(f2bf(lr.apply(bf2f(ushort_t(v[e])),
-                                           bf2f(ushort_t(zv[e])), e)))
-                     : short(0);
-      } else if (LOPK == LOP_BNACT) {
-#pragma unroll
-        for (int e = 0; e < 4; ++e)
-          sv[e] = ok ? short(f2bf(lr.apply(bf2f(ushort_t(v[e])), 0.f, e)))
-                     : short(0);
-      } else {
-#pragma unroll
-        for (int e = 0; e < 4; ++e) sv[e] = ok ? v[e] : short(0);
-      }
-      *reinterpret_cast<short4_t*>(&xt[u * 4]) = sv;
-    }
-  }
-  __syncthreads();
-
-  // ---- compute: 16 channel-lanes x 16 workers, tap-outer so each
-  //      worker's NO output accumulators stay in registers ----
-  const int c4 = tid % 16;
-  const int worker = tid / 16;
-  const int c = c0 + c4 * 4;
-  float acc[NO][4];
-#pragma unroll
-  for (int o = 0; o < NO; ++o)
-#pragma unroll
-    for (int e = 0; e < 4; ++e) acc[o][e] = 0.f;
-
-#pragma unroll
-  for (int i = 0; i < K; ++i)
-#pragma unroll
-    for (int j = 0; j < K; ++j) {
-      const int tap = FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-      short4_t wv =
-          *reinterpret_cast<const short4_t*>(&wt[tap * 64 + c4 * 4]);
-      float wf[4];
-#pragma unroll
-      for (int e = 0; e < 4; ++e) wf[e] = bf2f(ushort_t(wv[e]));
-#pragma unroll
-      for (int o = 0; o < NO; ++o) {
-        const int op = worker + o * 16;
-        const int oh = op / TO, ow = op % TO;
-        short4_t xv = *reinterpret_cast<const short4_t*>(
-            &xt[((oh * S + i) * TI + ow * S + j) * 64 + c4 * 4]);
-#pragma unroll
-        for (int e = 0; e < 4; ++e)
-          acc[o][e] = fmaf(bf2f(ushort_t(xv[e])), wf[e], acc[o][e]);
-      }
-    }
-
-  // ---- store + per-block stats ----
   float s1[4] = {}, s2[4] = {};
+  LopRegT<LOPK, ACT, 4> lr;
+  if (c8 < C8 && mlane < ML) {
+    lr.init(lopx, c);
+    for (int64_t p = lo + mlane; p < hi_; p += ML) {
+      const int w4 = int(p % Wo4);
+      int64_t rest = p / Wo4;
+      const int hg = int(rest % HoG);
+      const int n = int(rest / HoG);
+      const int ho0 = hg * TH;
+      const int wo0 = w4 * 4;
+      const int wbase = wo0 * S - pad;
+      const int64_t imgbase = int64_t(n) * H * W * C + c;
+      float acc[TH][4][4] = {};
 #pragma unroll
-  for (int o = 0; o < NO; ++o) {
-    const int op = worker + o * 16;
-    const int oh = ho_t * TO + op / TO;
-    const int ow = wo_t * TO + op % TO;
-    if (op < TO * TO && oh < Ho && ow < Wo && c + 4 <= C) {
-      short4_t out;
+      for (int r = 0; r < RR; ++r) {
+        const int hi = ho0 * S + r - pad;
+        const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
+        const bool row_ok = hi >= 0 && hi < H;
+        const ushort_t* xrow = x + imgbase + int64_t(hic) * W * C;
+        const ushort_t* zrow =
+            LOPK == LOP_BNBWD ? lopx.z + imgbase + int64_t(hic) * W * C
+                              : nullptr;
+        float xr[XW][4];
+        // branchless: always load from a CLAMPED in-bounds address and
+        // zero by value select — a per-element branch around a load makes
+        // hipcc wait vmcnt(0) per load (serial HBM round trips)
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        const ushort_t ov = f2bf(acc[o][e]);
-        out[e] = short(ov);
-        if (STATS) {
-          const float v = bf2f(ov);
-          s1[e] += v;
-          s2[e] = fmaf(v, v, s2[e]);
+        for (int t = 0; t < XW; ++t) {
+          const int wi = wbase + t;
+          const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+          const bool ok = wi >= 0 && wi < W;
+          const int off = wic * C;
+          short4_t v = *reinterpret_cast<const short4_t*>(&xrow[off]);
+          const bool keep = ok && row_ok;
+          if (LOPK != LOP_BNBWD) {
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])), 0.f, e) : 0.f;
+          } else {
+            short4_t zv = *reinterpret_cast<const short4_t*>(&zrow[off]);
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])),
+                                         bf2f(ushort_t(zv[e])), e)
+                              : 0.f;
+          }
+        }
+#pragma unroll
+        for (int t = 0; t < TH; ++t) {
+          const int i = r - t * S;
+          if (i < 0 || i >= K) continue;
+#pragma unroll
+          for (int j = 0; j < K; ++j) {
+            const int tap =
+                FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+            short4_t wv =
+                *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+#pragma unroll
+              for (int e = 0; e < 4; ++e)
+                acc[t][q][e] = fmaf(xr[q * S + j][e],
+                                    bf2f(ushort_t(wv[e])), acc[t][q][e]);
+          }
         }
       }
-      *reinterpret_cast<short4_t*>(
-          &y[(int64_t(n) * Ho + oh) * int64_t(Wo) * C + int64_t(ow) * C +
-             c]) = out;
+#pragma unroll
+      for (int t = 0; t < TH; ++t) {
+        if (ho0 + t >= Ho) break;
+        ushort_t* yout = y + (int64_t(n) * Ho + ho0 + t) * int64_t(Wo) * C +
+                         int64_t(wo0) * C + c;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+          if (wo0 + q >= Wo) break;
+          short4_t out;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const ushort_t o = f2bf(acc[t][q][e]);
+            out[e] = short(o);
+            if (STATS) {
+              const float v = bf2f(o);
+              s1[e] += v;
+              s2[e] = fmaf(v, v, s2[e]);
+            }
+          }
+          *reinterpret_cast<short4_t*>(&yout[q * C]) = out;
+        }
+      }
     }
   }
   if (STATS) {
-    // fold 16 workers per channel-lane; slab row = spatial tile index
 #pragma unroll
-    for (int e = 0; e < 4; ++e) red[(worker * 16 + c4) * 4 + e] = s1[e];
+    for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = s1[e];
     __syncthreads();
-    if (tid < 16 && c + 4 <= C) {
+    if (threadIdx.x < unsigned(CT8) && c8 < C8) {
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         float t = 0.f;
-        for (int l = 0; l < 16; ++l) t += red[(l * 16 + tid) * 4 + e];
-        stats_ws[int64_t(sp) * C + c0 + tid * 4 + e] = t;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 4 + e];
+        stats_ws[int64_t(chunk) * C + c + e] = t;
       }
     }
     __syncthreads();
 #pragma unroll
-    for (int e = 0; e < 4; ++e) red[(worker * 16 + c4) * 4 + e] = s2[e];
+    for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = s2[e];
     __syncthreads();
-    if (tid < 16 && c + 4 <= C) {
+    if (threadIdx.x < unsigned(CT8) && c8 < C8) {
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
         float t = 0.f;
-        for (int l = 0; l < 16; ++l) t += red[(l * 16 + tid) * 4 + e];
-        stats_ws[(int64_t(sp_tiles) + sp) * C + c0 + tid * 4 + e] = t;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 4 + e];
+        stats_ws[(int64_t(chunks) + chunk) * C + c + e] = t;
       }
     }
   }
@@ -428,17 +408,16 @@ int grid_for(int64_t total) {
 
 }  // namespace
 
-namespace {
-int dw_sp_tiles(int N, int Ho, int Wo, int S) {
-  const int to = (S == 1) ? 8 : 6;
-  return N * ((Ho + to - 1) / to) * ((Wo + to - 1) / to);
-}
-}  // namespace
-
 int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride) {
-  // slab rows = spatial tiles of the LDS-tiled forward
-  (void)C;
-  return dw_sp_tiles(N, Ho, Wo, stride);
+  (void)stride;
+  const int C8 = C >> 2;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
+  int64_t want = 2048 / c_tiles;
+  int64_t maxc = P;
+  int c = int(std::min(std::max<int64_t>(want, 1), maxc));
+  return c < 1 ? 1 : c;
 }
 
 void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
@@ -448,15 +427,17 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   const Lop L = to_lop(lopx);
-  const int c_tiles = (C + 63) / 64;
-  const int sp_tiles = dw_sp_tiles(N, Ho, Wo, stride);
-  dim3 g(c_tiles * sp_tiles);
+  const int C8 = C >> 2;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride);
+  dim3 g(c_tiles * chunks);
   // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
 #define DWF_CASE(KK, SS, ST, LK, AC)                                       \
   if (k == KK && stride == SS && (stats_ws != nullptr) == ST &&            \
       lopx.kind == LK && (LK == 0 || lopx.act == AC)) {                    \
     dw_fwd_kernel<KK, SS, false, ST, LK, AC><<<g, 256, 0, s>>>(            \
-        xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, sp_tiles);       \
+        xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);         \
     return;                                                                \
   }
 #define DWF_ST(KK, SS, ST)                                                 \
@@ -478,14 +459,16 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
   const Lop L = to_lop(lopdy);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
-    const int c_tiles = (C + 63) / 64;
-    const int sp_tiles = dw_sp_tiles(N, H, W, 1);
-    dim3 g(c_tiles * sp_tiles);
+    const int C8 = C >> 2;
+    const int CT8 = C8 < 256 ? C8 : 256;
+    const int c_tiles = (C8 + CT8 - 1) / CT8;
+    const int chunks = dw_stats_chunks(N, H, W, C, 1);
+    dim3 g(c_tiles * chunks);
     // dgrad supports only lop none (dz is materialized by bn_act_bwd)
 #define DWD_CASE(KK)                                                       \
     if (k == KK && lopdy.kind == 0) {                                      \
       dw_fwd_kernel<KK, 1, true, false, 0, 0><<<g, 256, 0, s>>>(           \
-          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, sp_tiles);      \
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);        \
       return;                                                              \
     }
     DWD_CASE(3) DWD_CASE(5)

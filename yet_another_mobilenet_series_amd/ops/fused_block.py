@@ -75,10 +75,15 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                 mean1.copy_(erm)
                 invstd1.copy_((erv.float() + eps).rsqrt())
                 _eval_ab(mean1, invstd1, eg, ebb, ab1)
-            dw_in, dw_lop = z1, (1, act, ab1)
+            # materialize y1: the dw window-load lop costs more registers
+            # than this memory-bound pass costs time (measured)
+            y1 = torch.empty_like(z1)
+            e.bn_act_fwd(z1, ab1, act, y1, None)
+            dw_in = y1
         else:
             z1, mean1, invstd1, ab1 = x, None, None, None
-            dw_in, dw_lop = x, (0, 0, None)
+            y1 = None
+            dw_in = x
             ce = cin
 
         # ---- depthwise ------------------------------------------------
@@ -94,24 +99,23 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         if training:
             chunks = e.dw_stats_chunks(n, ho, wo, cd, stride)
             ws = torch.empty(2 * chunks * cd, dtype=F32, device=dev)
-            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
-                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
-                          lop_coef=dw_lop[2], stats_ws=ws)
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad, stats_ws=ws)
             e.bn_finalize(ws, chunks, mean2, invstd2, ab2, dg, db,
                           drm, drv, momentum, eps, m2)
         else:
-            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
-                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
-                          lop_coef=dw_lop[2])
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad)
             mean2.copy_(drm)
             invstd2.copy_((drv.float() + eps).rsqrt())
             _eval_ab(mean2, invstd2, dg, db, ab2)
 
         # ---- squeeze-excite -------------------------------------------
         if has_se:
+            # y2 materialized: three consumers (pool, scale, wgrad)
+            y2 = torch.empty_like(z2)
+            e.bn_act_fwd(z2, ab2, act, y2, None)
             r = sw1.shape[0]
             pooled = torch.empty(n, cd, dtype=BF16, device=dev)
-            e.avgpool_fwd(z2, pooled, lop_kind=1, lop_act=act, lop_coef=ab2)
+            e.avgpool_fwd(y2, pooled)
             w1m = _packed(sw1, "se", lambda w: w.view(r, cd).to(BF16)
                           .contiguous())
             hz = torch.empty(n, r, dtype=BF16, device=dev)
@@ -125,11 +129,12 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             sgate = torch.empty_like(sz)
             e.bias_act_fwd(sz, sb2.float(), ACT_HSIGMOID, sgate)
             y2s = torch.empty_like(z2)
-            e.rowscale_fwd(z2, sgate, y2s, lop_kind=1, lop_act=act,
-                           lop_coef=ab2)
+            e.rowscale_fwd(y2, sgate, y2s)
             pj_in, pj_lop = y2s, (0, 0, None)
         else:
-            pooled = hz = ha = sz = sgate = y2s = None
+            # no-SE: project reads z2 with the BN+act applied on load —
+            # y2 is never materialized (single consumer)
+            y2 = pooled = hz = ha = sz = sgate = y2s = None
             pj_in, pj_lop = z2, (1, act, ab2)
 
         # ---- project pw -----------------------------------------------
@@ -161,6 +166,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             x, z1 if has_expand else None, z2, z3,
             mean1, invstd1, ab1, mean2, invstd2, ab2, mean3, invstd3,
             pooled, hz, ha, sz, sgate, y2s,
+            y1 if has_expand else None, y2,
             ew, eg, ebb, dww, dg, db, sw1, sb1, sw2, sb2, pw, pg, pb)
         ctx.cfg = (k, stride, act, training, has_expand, has_se, use_res,
                    n, cin, h, wdt, ho, wo)
@@ -170,8 +176,8 @@ class FusedInvertedResidualFn(torch.autograd.Function):
     def backward(ctx, dyb):
         e = ext()
         (x, z1, z2, z3, mean1, invstd1, ab1, mean2, invstd2, ab2, mean3,
-         invstd3, pooled, hz, ha, sz, sgate, y2s, ew, eg, ebb, dww, dg, db,
-         sw1, sb1, sw2, sb2, pw, pg, pb) = ctx.saved_tensors
+         invstd3, pooled, hz, ha, sz, sgate, y2s, y1, y2, ew, eg, ebb, dww,
+         dg, db, sw1, sb1, sw2, sb2, pw, pg, pb) = ctx.saved_tensors
         (k, stride, act, training, has_expand, has_se, use_res, n, cin, h,
          wdt, ho, wo) = ctx.cfg
         dev = x.device
@@ -207,8 +213,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             r = sw1.shape[0]
             d_y2 = torch.empty_like(z2)
             ds = torch.empty(n, cd, dtype=BF16, device=dev)
-            e.rowscale_bwd(d_in3, z2, sgate, d_y2, ds, lop_kind=1,
-                           lop_act=act, lop_coef=ab2)
+            e.rowscale_bwd(d_in3, y2, sgate, d_y2, ds)
             dsz = torch.empty_like(ds)
             db2se = torch.empty(cd, dtype=F32, device=dev)
             e.bias_act_bwd(ds, sz, sb2.float(), ACT_HSIGMOID, dsz, db2se)
@@ -244,15 +249,11 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                      bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
                        .contiguous())
-        dw_in = z1 if has_expand else x
+        dw_in = y1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
         e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
         dwgt = torch.empty(k * k, cd, dtype=F32, device=dev)
-        if has_expand:
-            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad,
-                            lopx_kind=1, lopx_act=act, lopx_coef=ab1)
-        else:
-            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
+        e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
         d_dww = dwgt.t().reshape(cd, 1, k, k)
 
         # ---- expand backward --------------------------------------------
