@@ -26,8 +26,9 @@ from yet_another_mobilenet_series_amd.ops.backend import (  # noqa: E402
     load_extension, set_engine_mode)
 from yet_another_mobilenet_series_amd.parallel import (  # noqa: E402
     DataParallelEngine, distributed as udist)
+from yet_another_mobilenet_series_amd.utils.config import default_config  # noqa: E402
 from yet_another_mobilenet_series_amd.utils.optim import (  # noqa: E402
-    RMSpropTF, split_param_groups)
+    get_optimizer)
 
 
 def parse_args():
@@ -64,12 +65,13 @@ def main():
     if on_gpu:
         model = model.to(memory_format=torch.channels_last)
     engine = DataParallelEngine(model, bucket_mb=args.bucket_mb, overlap=True)
+    ocfg = default_config().optim
+    ocfg.name = args.optimizer
     if args.optimizer == "sgd":
-        opt = torch.optim.SGD(split_param_groups(model, 4e-5), lr=0.4,
-                              momentum=0.9, nesterov=True)
+        ocfg.lr, ocfg.weight_decay = 0.4, 4e-5
     else:
-        opt = RMSpropTF(split_param_groups(model, 1e-5), lr=0.064,
-                        alpha=0.9, eps=1e-3, momentum=0.9)
+        ocfg.lr, ocfg.weight_decay = 0.064, 1e-5
+    opt = get_optimizer(model, ocfg)  # fused multi-tensor kernels on GPU
     model.train()
 
     x = torch.randn(args.batch, 3, 224, 224, device=dev)
@@ -139,7 +141,8 @@ def main():
 
     if rank == 0:
         print(json.dumps({
-            "metric": "images/sec (whole node) MobileNetV3-Large 224 bf16 train",
+            "metric": f"images/sec (whole node) {args.model} 224 "
+                      f"{'bf16' if on_gpu else 'fp32'} train",
             "value": round(imgs_per_sec, 1),
             "unit": "img/s",
             "n_gpus": world,

@@ -18,7 +18,8 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "yet_another_mobilenet_series_amd", "ops", "csrc")
 
 sources = [os.path.join(CSRC, f) for f in
-           ("bindings.cpp", "gemm.hip", "dwconv.hip", "bn.hip", "misc.hip")]
+           ("bindings.cpp", "gemm.hip", "dwconv.hip", "bn.hip", "misc.hip",
+            "update.hip")]
 
 setup(
     name="yams_hip",

@@ -341,3 +341,61 @@ def test_label_smooth_ce(ext):
     ((0.9 * (-logp.gather(1, target.view(-1, 1)).squeeze(1))
       + 0.1 * (-logp.mean(dim=1))).mean()).backward()
     assert_close(dlogits, lr.grad, rtol=1e-3, atol=1e-5)
+
+
+# ------------------------------------------------------- fused updates
+
+def test_fused_sgd_matches_torch(ext):
+    torch.manual_seed(0)
+    shapes = [(64, 32), (128,), (16, 8, 3, 3), (1000, 1280)]
+    ps1 = [torch.randn(*s_, device=dev()) for s_ in shapes]
+    gs = [torch.randn(*s_, device=dev()) for s_ in shapes]
+    ps2 = [p.clone() for p in ps1]
+    m1 = [torch.zeros_like(p) for p in ps1]
+    for _ in range(3):
+        ext.fused_sgd(ps1, gs, m1, [1e-4, 0.0, 1e-4, 1e-4], 0.1, 0.9, True)
+    opt = torch.optim.SGD([
+        {"params": [ps2[0], ps2[2], ps2[3]], "weight_decay": 1e-4},
+        {"params": [ps2[1]], "weight_decay": 0.0}], lr=0.1, momentum=0.9,
+        nesterov=True)
+    for _ in range(3):
+        for p, g in zip(ps2, gs):
+            p.grad = g.clone()
+        opt.step()
+    for a, b in zip(ps1, ps2):
+        torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-6)
+
+
+def test_fused_rmsprop_tf_matches_cpu_impl(ext):
+    from yet_another_mobilenet_series_amd.utils.optim import RMSpropTF
+    torch.manual_seed(1)
+    shapes = [(48, 24), (96,)]
+    ps1 = [torch.randn(*s_, device=dev()) for s_ in shapes]
+    gs = [torch.randn(*s_, device=dev()) for s_ in shapes]
+    ps2 = [torch.nn.Parameter(p.clone()) for p in ps1]
+    ms = [torch.ones_like(p) for p in ps1]
+    mom = [torch.zeros_like(p) for p in ps1]
+    for _ in range(3):
+        ext.fused_rmsprop_tf(ps1, gs, ms, mom, [1e-5, 0.0], 0.064, 0.9,
+                             1e-3, 0.9)
+    opt = RMSpropTF([{"params": [ps2[0]], "weight_decay": 1e-5},
+                     {"params": [ps2[1]], "weight_decay": 0.0}],
+                    lr=0.064, alpha=0.9, eps=1e-3, momentum=0.9)
+    for _ in range(3):
+        for p, g in zip(ps2, gs):
+            p.grad = g.clone()
+        opt.step()
+    for a, b in zip(ps1, ps2):
+        torch.testing.assert_close(a, b.detach(), rtol=1e-5, atol=1e-6)
+
+
+def test_fused_ema_matches_reference(ext):
+    torch.manual_seed(2)
+    vals = [torch.randn(100, device=dev()), torch.randn(7, 9, device=dev())]
+    sh1 = [torch.randn_like(v) for v in vals]
+    sh2 = [s_.clone() for s_ in sh1]
+    ext.fused_ema(vals, sh1, 0.99)
+    for v, s_ in zip(vals, sh2):
+        s_.mul_(0.99).add_(v, alpha=0.01)
+    for a, b in zip(sh1, sh2):
+        torch.testing.assert_close(a, b, rtol=1e-6, atol=1e-7)

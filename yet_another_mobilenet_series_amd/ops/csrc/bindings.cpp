@@ -388,6 +388,90 @@ void label_smooth_ce(const Tensor& logits, const Tensor& target,
                         int(lr.M), lr.C, cur_stream());
 }
 
+// ---- fused multi-tensor updates ----------------------------------------
+
+void check_f32(const Tensor& t) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                  t.is_contiguous(),
+              "update tensors must be fp32 cuda contiguous");
+}
+
+void fused_sgd(const std::vector<Tensor>& params,
+               const std::vector<Tensor>& grads,
+               const std::vector<Tensor>& moms,
+               const std::vector<double>& wds, double lr, double momentum,
+               bool nesterov) {
+  const int maxc = yams::update_max_chunks();
+  const int n = int(params.size());
+  for (int base = 0; base < n; base += maxc) {
+    const int cnt = std::min(maxc, n - base);
+    const float* g[32];
+    float* p[32];
+    float* m[32];
+    float wd[32];
+    int numel[32];
+    for (int c = 0; c < cnt; ++c) {
+      check_f32(params[base + c]);
+      g[c] = grads[base + c].data_ptr<float>();
+      p[c] = params[base + c].data_ptr<float>();
+      m[c] = moms[base + c].data_ptr<float>();
+      wd[c] = float(wds[base + c]);
+      numel[c] = int(params[base + c].numel());
+    }
+    yams::fused_sgd(g, p, m, wd, numel, cnt, float(lr), float(momentum),
+                    nesterov, cur_stream());
+  }
+}
+
+void fused_rmsprop_tf(const std::vector<Tensor>& params,
+                      const std::vector<Tensor>& grads,
+                      const std::vector<Tensor>& mss,
+                      const std::vector<Tensor>& moms,
+                      const std::vector<double>& wds, double lr,
+                      double alpha, double eps, double momentum) {
+  const int maxc = yams::update_max_chunks();
+  const int n = int(params.size());
+  for (int base = 0; base < n; base += maxc) {
+    const int cnt = std::min(maxc, n - base);
+    const float* g[32];
+    float* p[32];
+    float* ms[32];
+    float* mm[32];
+    float wd[32];
+    int numel[32];
+    for (int c = 0; c < cnt; ++c) {
+      check_f32(params[base + c]);
+      g[c] = grads[base + c].data_ptr<float>();
+      p[c] = params[base + c].data_ptr<float>();
+      ms[c] = mss[base + c].data_ptr<float>();
+      mm[c] = moms[base + c].data_ptr<float>();
+      wd[c] = float(wds[base + c]);
+      numel[c] = int(params[base + c].numel());
+    }
+    yams::fused_rmsprop_tf(g, p, ms, mm, wd, numel, cnt, float(lr),
+                           float(alpha), float(eps), float(momentum),
+                           cur_stream());
+  }
+}
+
+void fused_ema(const std::vector<Tensor>& values,
+               const std::vector<Tensor>& shadows, double decay) {
+  const int maxc = yams::update_max_chunks();
+  const int n = int(values.size());
+  for (int base = 0; base < n; base += maxc) {
+    const int cnt = std::min(maxc, n - base);
+    const float* v[32];
+    float* s[32];
+    int numel[32];
+    for (int c = 0; c < cnt; ++c) {
+      v[c] = values[base + c].data_ptr<float>();
+      s[c] = shadows[base + c].data_ptr<float>();
+      numel[c] = int(values[base + c].numel());
+    }
+    yams::fused_ema(v, s, numel, cnt, float(decay), cur_stream());
+  }
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -440,4 +524,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("avgpool_bwd_add", &avgpool_bwd_add);
   m.def("label_smooth_ce", &label_smooth_ce);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_rmsprop_tf", &fused_rmsprop_tf);
+  m.def("fused_ema", &fused_ema);
 }

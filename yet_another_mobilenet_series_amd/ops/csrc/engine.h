@@ -132,4 +132,16 @@ void label_smooth_ce(const void* logits, const int64_t* target,
                      float smoothing, float* loss, float* dlogits,
                      float* rowloss, int N, int C, hipStream_t s);
 
+// --- fused multi-tensor updates (K12/K13) -------------------------------
+int update_max_chunks();
+void fused_sgd(const float** g, float** p, float** m, const float* wd,
+               const int* numel, int count, float lr, float momentum,
+               bool nesterov, hipStream_t s);
+void fused_rmsprop_tf(const float** g, float** p, float** ms, float** mom,
+                      const float* wd, const int* numel, int count,
+                      float lr, float alpha, float eps, float momentum,
+                      hipStream_t s);
+void fused_ema(const float** v, float** s_, const int* numel, int count,
+               float decay, hipStream_t st);
+
 }  // namespace yams

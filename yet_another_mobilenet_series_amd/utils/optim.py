@@ -106,12 +106,101 @@ def split_param_groups(model: nn.Module, weight_decay: float,
             {"params": no_decay, "weight_decay": 0.0}]
 
 
+class FusedSGD(torch.optim.Optimizer):
+    """SGD(momentum, nesterov, L2 wd) on the fused multi-tensor HIP kernel
+    (K12): one launch per ~32 tensors instead of per-tensor torch ops.
+    Semantics match torch.optim.SGD for fp32 CUDA params."""
+
+    def __init__(self, params, lr: float, momentum: float = 0.0,
+                 weight_decay: float = 0.0, nesterov: bool = False) -> None:
+        super().__init__(params, dict(lr=lr, momentum=momentum,
+                                      weight_decay=weight_decay,
+                                      nesterov=nesterov))
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        from ..ops.backend import ext
+        e = ext()
+        for group in self.param_groups:
+            ps, gs, ms, wds = [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if "momentum_buffer" not in state:
+                    state["momentum_buffer"] = torch.zeros_like(p)
+                ps.append(p)
+                gs.append(p.grad.contiguous())
+                ms.append(state["momentum_buffer"])
+                wds.append(group["weight_decay"])
+            if ps:
+                e.fused_sgd(ps, gs, ms, wds, group["lr"],
+                            group["momentum"], group["nesterov"])
+        return loss
+
+
+class FusedRMSpropTF(torch.optim.Optimizer):
+    """TF-semantics RMSprop on the fused HIP kernel (see RMSpropTF)."""
+
+    def __init__(self, params, lr: float, alpha: float = 0.9,
+                 eps: float = 1e-3, momentum: float = 0.9,
+                 weight_decay: float = 0.0) -> None:
+        super().__init__(params, dict(lr=lr, alpha=alpha, eps=eps,
+                                      momentum=momentum,
+                                      weight_decay=weight_decay))
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        from ..ops.backend import ext
+        e = ext()
+        for group in self.param_groups:
+            ps, gs, mss, moms, wds = [], [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if "square_avg" not in state:
+                    state["square_avg"] = torch.ones_like(p)
+                    state["momentum_buffer"] = torch.zeros_like(p)
+                ps.append(p)
+                gs.append(p.grad.contiguous())
+                mss.append(state["square_avg"])
+                moms.append(state["momentum_buffer"])
+                wds.append(group["weight_decay"])
+            if ps:
+                e.fused_rmsprop_tf(ps, gs, mss, moms, wds, group["lr"],
+                                   group["alpha"], group["eps"],
+                                   group["momentum"])
+        return loss
+
+
 def get_optimizer(model: nn.Module, cfg) -> torch.optim.Optimizer:
     groups = split_param_groups(model, cfg.weight_decay, cfg.no_wd_on_bn_bias)
+    on_gpu = next(model.parameters()).is_cuda
     if cfg.name == "sgd":
+        if on_gpu:
+            from ..ops.backend import extension_available
+            if extension_available():
+                return FusedSGD(groups, lr=cfg.lr, momentum=cfg.momentum,
+                                nesterov=cfg.nesterov)
         return torch.optim.SGD(groups, lr=cfg.lr, momentum=cfg.momentum,
                                nesterov=cfg.nesterov)
     if cfg.name in ("rmsprop", "rmsprop_tf"):
+        if on_gpu:
+            from ..ops.backend import extension_available
+            if extension_available():
+                return FusedRMSpropTF(groups, lr=cfg.lr,
+                                      alpha=cfg.rmsprop_alpha,
+                                      eps=cfg.rmsprop_eps,
+                                      momentum=cfg.momentum)
         return RMSpropTF(groups, lr=cfg.lr, alpha=cfg.rmsprop_alpha,
                          eps=cfg.rmsprop_eps, momentum=cfg.momentum)
     raise ValueError(f"unknown optimizer {cfg.name!r}")
@@ -196,6 +285,15 @@ class ExponentialMovingAverage:
     def update(self, model: Optional[nn.Module] = None) -> None:
         model = model or self.model
         d = self.decay
+        items = list(self._named_tensors(model))
+        if items and items[0][1].is_cuda and items[0][1].dtype == torch.float32:
+            from ..ops.backend import extension_available, load_extension
+            if extension_available():
+                e = load_extension()
+                vals = [v for _, v in items]
+                shadows = [self.shadow[k] for k, _ in items]
+                e.fused_ema(vals, shadows, d)  # K13: one fused launch
+                return
         for k, v in self._named_tensors(model):
             s = self.shadow[k]
             s.mul_(d).add_(v.float(), alpha=1 - d)
