@@ -391,9 +391,12 @@ void label_smooth_ce(const Tensor& logits, const Tensor& target,
 // ---- fused multi-tensor updates ----------------------------------------
 
 void check_f32(const Tensor& t) {
+  // dense (possibly channels_last) fp32: the flat update only needs every
+  // element covered once; the Python side guarantees grads/buffers share
+  // the param's memory layout
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
-                  t.is_contiguous(),
-              "update tensors must be fp32 cuda contiguous");
+                  t.is_non_overlapping_and_dense(),
+              "update tensors must be fp32 cuda dense");
 }
 
 void fused_sgd(const std::vector<Tensor>& params,

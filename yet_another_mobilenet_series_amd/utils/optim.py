@@ -21,6 +21,17 @@ import torch.nn as nn
 from ..ops import functional as OF
 
 
+def _grad_like(p: torch.Tensor) -> torch.Tensor:
+    """Gradient re-laid-out to match the param's memory format (flat fused
+    updates walk raw storage, so element order must agree)."""
+    g = p.grad
+    if p.is_contiguous():
+        return g if g.is_contiguous() else g.contiguous()
+    if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last):
+        return g.contiguous(memory_format=torch.channels_last)
+    return g.contiguous()  # caller must ensure param is contiguous then
+
+
 class CrossEntropyLabelSmooth(nn.Module):
     def __init__(self, smoothing: float = 0.1) -> None:
         super().__init__()
@@ -134,7 +145,7 @@ class FusedSGD(torch.optim.Optimizer):
                 if "momentum_buffer" not in state:
                     state["momentum_buffer"] = torch.zeros_like(p)
                 ps.append(p)
-                gs.append(p.grad.contiguous())
+                gs.append(_grad_like(p))
                 ms.append(state["momentum_buffer"])
                 wds.append(group["weight_decay"])
             if ps:
@@ -171,7 +182,7 @@ class FusedRMSpropTF(torch.optim.Optimizer):
                     state["square_avg"] = torch.ones_like(p)
                     state["momentum_buffer"] = torch.zeros_like(p)
                 ps.append(p)
-                gs.append(p.grad.contiguous())
+                gs.append(_grad_like(p))
                 mss.append(state["square_avg"])
                 moms.append(state["momentum_buffer"])
                 wds.append(group["weight_decay"])
@@ -288,7 +299,9 @@ class ExponentialMovingAverage:
         items = list(self._named_tensors(model))
         if items and items[0][1].is_cuda and items[0][1].dtype == torch.float32:
             from ..ops.backend import extension_available, load_extension
-            if extension_available():
+            if extension_available() and all(
+                    v.stride() == self.shadow[k].stride()
+                    for k, v in items):
                 e = load_extension()
                 vals = [v for _, v in items]
                 shadows = [self.shadow[k] for k, _ in items]
