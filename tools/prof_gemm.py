@@ -1,10 +1,11 @@
 #!/usr/bin/env python3
 """Tight loop over the hot pw-GEMM shapes for PMC counter collection."""
+import os
 import sys
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from yet_another_mobilenet_series_amd.ops.backend import ext
 
 e = ext()

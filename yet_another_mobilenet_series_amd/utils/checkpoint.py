@@ -41,10 +41,11 @@ def _rng_state() -> Dict:
 def _load_rng_state(state: Dict) -> None:
     random.setstate(state["python"])
     np.random.set_state(state["numpy"])
-    torch.set_rng_state(state["torch"])
+    # map_location may have moved the ByteTensors to GPU: bring them back
+    torch.set_rng_state(state["torch"].cpu())
     if torch.cuda.is_available() and "cuda" in state:
         try:
-            torch.cuda.set_rng_state_all(state["cuda"])
+            torch.cuda.set_rng_state_all([t.cpu() for t in state["cuda"]])
         except RuntimeError:
             pass  # different device count than at save time
 
