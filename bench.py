@@ -37,6 +37,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--model", default="mobilenet_v3_large")
+    p.add_argument("--width", type=float, default=1.0)
     p.add_argument("--batch", type=int, default=256,
                    help="per-GPU batch (weak scaling)")
     p.add_argument("--engine", default="hip", choices=["hip", "reference"],
@@ -61,7 +62,8 @@ def main():
     set_engine_mode(args.engine)
 
     torch.manual_seed(1234 + rank)
-    model = models.build_model(args.model, dropout=0.0).to(dev)
+    model = models.build_model(args.model, dropout=0.0,
+                               width_mult=args.width).to(dev)
     if on_gpu:
         model = model.to(memory_format=torch.channels_last)
     engine = DataParallelEngine(model, bucket_mb=args.bucket_mb, overlap=True)
@@ -155,7 +157,8 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "loss": round(float(loss), 4) if loss is not None else None,
-            "config": {"model": args.model, "global_batch": args.batch * world,
+            "config": {"model": args.model, "width_mult": args.width,
+                       "global_batch": args.batch * world,
                        "image_size": 224, "parallelism": f"dp{world}",
                        "engine": args.engine, "graph": graphed is not None,
                        "optimizer": args.optimizer},

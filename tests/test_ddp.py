@@ -111,3 +111,46 @@ def test_grad_order_reverse_registration():
     first_bucket_params = set(id(p) for p in eng.buckets[0].params)
     last_param = list(model.parameters())[-1]
     assert id(last_param) in first_bucket_params
+
+
+def _dying_worker(rank, world, tmpfile, out_path):
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    import datetime
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world,
+        timeout=datetime.timedelta(seconds=5))
+    try:
+        if rank == 1:
+            os._exit(7)  # simulate a dead rank mid-step
+        t = torch.ones(4)
+        try:
+            dist.all_reduce(t)  # must NOT hang forever: timeout or error
+            result = "completed"
+        except Exception:
+            result = "raised"
+        torch.save(result, out_path)
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_dead_rank_does_not_hang(tmp_path):
+    """Failure detection: a dead rank must surface as a
+    timeout/error on the survivors within the configured window, not a
+    silent hang."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    tmpfile = str(tmp_path / "init_dead")
+    out_path = str(tmp_path / "survivor.pt")
+    ps = [ctx.Process(target=_dying_worker,
+                      args=(r, world, tmpfile, out_path))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(60)
+        if p.is_alive():
+            p.terminate()
+            assert False, "survivor hung past the collective timeout"
+    assert os.path.exists(out_path)
+    assert torch.load(out_path) == "raised"

@@ -94,17 +94,41 @@ class Trainer:
             sampler.set_epoch(epoch)
         meters = {k: AverageMeter(k) for k in ("loss", "top1", "top5", "img_s")}
         t_last = time.time()
+        # per-phase hipEvent timing, printed every log
+        # interval when train.perf_breakdown is on
+        breakdown = (getattr(cfg.train, "perf_breakdown", False)
+                     and self.device.type == "cuda")
         for it, (x, y) in enumerate(self.train_loader):
             x, y = self._to_device(x, y)
             lr = self.lr_sched.step(self.global_step)
+            if breakdown:
+                evs = [torch.cuda.Event(enable_timing=True)
+                       for _ in range(5)]
+                evs[0].record()
             self.engine.zero_grad()
             logits = self.engine(x)
             loss = self.criterion(logits, y)
+            if breakdown:
+                evs[1].record()
             loss.backward()
+            if breakdown:
+                evs[2].record()
             self.engine.finish()
+            if breakdown:
+                evs[3].record()
             self.optimizer.step()
             if self.ema is not None:
                 self.ema.update()
+            if breakdown:
+                evs[4].record()
+                if it % cfg.train.log_interval == 0:
+                    torch.cuda.synchronize()
+                    names = ("fwd", "bwd", "comm", "opt")
+                    ms = {n: evs[i].elapsed_time(evs[i + 1])
+                          for i, n in enumerate(names)}
+                    self.jsonl.write(kind="phase", step=self.global_step,
+                                     **{f"{n}_ms": round(v, 3)
+                                        for n, v in ms.items()})
             self.global_step += 1
 
             with torch.no_grad():

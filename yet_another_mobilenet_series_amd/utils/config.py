@@ -138,7 +138,7 @@ DEFAULTS: Dict[str, Any] = {
               "grad_accum_steps": 1, "dtype": "bf16",
               "resume": "", "pretrained": "",
               "save_interval_epochs": 1, "channels_last": True,
-              "deterministic": False,
+              "deterministic": False, "perf_breakdown": False,
               "engine": "auto", "use_cuda_graph": True},
     "dist": {"backend": "auto", "bucket_mb": 4, "overlap": True,
              "grad_dtype": "fp32", "timeout_sec": 300},
