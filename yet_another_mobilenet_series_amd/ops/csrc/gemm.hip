@@ -44,8 +44,12 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
     ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA,
     float* __restrict__ stats_ws) {
-  __shared__ ushort_t Alds[BM * LDA];
-  __shared__ ushort_t Blds[BN * LDA];
+  // double-buffered register staging (guide T14: issue the next tile's
+  // global loads before the current tile's MFMAs, write LDS after the
+  // barrier) — the single-buffered loop parked 47% of wave cycles on
+  // staging waits (profiles/gemm_counter_collection.csv)
+  __shared__ ushort_t Alds[2][BM * LDA];
+  __shared__ ushort_t Blds[2][BN * LDA];
 
   const int tid = threadIdx.x;
   const int64_t bm = int64_t(blockIdx.x) * BM;
@@ -62,63 +66,80 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 
   const int a_row = tid >> 1, a_seg = (tid & 1) * 16;   // 2 thr/row x 128
   const int b_row = tid >> 2, b_seg = (tid & 3) * 8;    // 4 thr/row x 64
+  const int64_t gm = bm + a_row;
+  const int gn = bn + b_row;
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A tile [BM][BK] (with load-op)
-    {
-      const int64_t gm = bm + a_row;
-      const int gk = k0 + a_seg;
-      ushort_t* dst = &Alds[a_row * LDA + a_seg];
-      if (gm < M && gk + 16 <= K) {
-        const int64_t off = gm * K + gk;
-        lop_vec8<LOPK>(lopA, A, off, gk, dst);
-        lop_vec8<LOPK>(lopA, A, off + 8, gk + 8, dst + 8);
-      } else {
+  ushort_t areg[16];
+  ushort_t breg[8];
+
+  auto stage_regs = [&](int k0) {
+    const int gka = k0 + a_seg;
+    if (gm < M && gka + 16 <= K) {
+      const int64_t off = gm * K + gka;
+      lop_vec8<LOPK>(lopA, A, off, gka, areg);
+      lop_vec8<LOPK>(lopA, A, off + 8, gka + 8, areg + 8);
+    } else {
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int gkj = gk + j;
-          dst[j] = (gm < M && gkj < K)
-                       ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
-                                                       gm * K + gkj, gkj)))
-                       : ushort_t(0);
-        }
+      for (int j = 0; j < 16; ++j) {
+        const int gkj = gka + j;
+        areg[j] = (gm < M && gkj < K)
+                      ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
+                                                      gm * K + gkj, gkj)))
+                      : ushort_t(0);
       }
     }
-    // stage B tile [BN][BK] from Bt[N][K]
-    {
-      const int gn = bn + b_row;
-      const int gk = k0 + b_seg;
-      ushort_t* dst = &Blds[b_row * LDA + b_seg];
-      if (gn < N && gk + 8 <= K) {
-        *reinterpret_cast<short4_t*>(dst) =
-            *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gk]);
-        *reinterpret_cast<short4_t*>(dst + 4) =
-            *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gk + 4]);
-      } else {
+    const int gkb = k0 + b_seg;
+    if (gn < N && gkb + 8 <= K) {
+      *reinterpret_cast<short4_t*>(breg) =
+          *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gkb]);
+      *reinterpret_cast<short4_t*>(breg + 4) =
+          *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gkb + 4]);
+    } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          dst[j] = (gn < N && gk + j < K)
-                       ? Bt[int64_t(gn) * K + gk + j] : ushort_t(0);
-      }
+      for (int j = 0; j < 8; ++j)
+        breg[j] = (gn < N && gkb + j < K)
+                      ? Bt[int64_t(gn) * K + gkb + j] : ushort_t(0);
     }
-    __syncthreads();
+  };
 
+  auto write_lds = [&](int buf) {
+    ushort_t* da = &Alds[buf][a_row * LDA + a_seg];
+    *reinterpret_cast<short8_t*>(da) =
+        *reinterpret_cast<const short8_t*>(areg);
+    *reinterpret_cast<short8_t*>(da + 8) =
+        *reinterpret_cast<const short8_t*>(areg + 8);
+    ushort_t* db = &Blds[buf][b_row * LDA + b_seg];
+    *reinterpret_cast<short8_t*>(db) =
+        *reinterpret_cast<const short8_t*>(breg);
+  };
+
+  const int nt = (K + BK - 1) / BK;
+  stage_regs(0);
+  write_lds(0);
+  __syncthreads();
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) stage_regs((t + 1) * BK);  // loads fly under the MFMAs
+    const int cur = t & 1;
     bf16x8 a_frag[FRAG_M], b_frag[FRAG_N];
 #pragma unroll
     for (int i = 0; i < FRAG_M; ++i)
       a_frag[i] = *reinterpret_cast<const bf16x8*>(
-          &Alds[(wrow0 + i * 16 + fr) * LDA + fk]);
+          &Alds[cur][(wrow0 + i * 16 + fr) * LDA + fk]);
 #pragma unroll
     for (int j = 0; j < FRAG_N; ++j)
       b_frag[j] = *reinterpret_cast<const bf16x8*>(
-          &Blds[(wcol0 + j * 16 + fr) * LDA + fk]);
+          &Blds[cur][(wcol0 + j * 16 + fr) * LDA + fk]);
 #pragma unroll
     for (int i = 0; i < FRAG_M; ++i)
 #pragma unroll
       for (int j = 0; j < FRAG_N; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
-    __syncthreads();
+    if (t + 1 < nt) {
+      __syncthreads();       // everyone past buf[cur^1] reads of tile t-1
+      write_lds(cur ^ 1);
+      __syncthreads();       // tile t+1 visible
+    }
   }
 
   // epilogue: C/D frag layout col = lane&15, row = (lane>>4)*4 + r
@@ -139,7 +160,6 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 
   if (STATS) {
     // per-block partial column sums/sumsq of the (bf16-rounded) output.
-    // contributors per col: 2 M-waves x 4 row-groups = 8 slots.
     float* red = reinterpret_cast<float*>(Alds);  // 512 floats
     float s1[FRAG_N] = {}, s2[FRAG_N] = {};
 #pragma unroll
@@ -212,25 +232,31 @@ void pw_conv(const void* x, const void* w, void* y,
 // ---------------------------------------------------------------------------
 namespace {
 
-constexpr int MT = 32;  // m rows per LDS pass
+// MFMA version: C[N1,N2] = A^T B with A:[M,N1], B:[M,N2].  Both operands
+// are M-major (M is the reduction axis), so staging TRANSPOSES tiles into
+// K-major LDS rows ([n][k], k = m-chunk) and then the frag-read + MFMA
+// structure is identical to pw_conv_kernel.  The transposed scatter writes
+// use an XOR chunk swizzle (phys k-chunk = (k>>3) ^ (n&7)) so the 8
+// column-writes of one thread land in 8 distinct bank groups while frag
+// reads stay single aligned b128s.  Replaces the VALU fp32-FMA version,
+// which was compute-bound at ~1/16 of MFMA throughput.
+constexpr int KT = 64;        // m rows per LDS pass
+constexpr int LDK = KT + 8;   // shorts per LDS row (pad keeps b128 align)
 
-// Adaptive output tile (W1 x W2 in {32,64,128}): small-channel conv wgrads
-// (e.g. dw 16->96 expands) waste up to 10x compute in a fixed 128x128
-// tile.  Thread layout 16x16, each thread V1 x V2 = (W1/16) x (W2/16)
-// fp32 accumulators; staging rows of max(W1,W2).
 template <int W1, int W2, int LA, int LB>
 __global__ __launch_bounds__(256) void atb_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ B,
     float* __restrict__ ws, int64_t M, int N1, int N2, int splits,
     Lop lopA, Lop lopB) {
-  constexpr int V1 = W1 / 16, V2 = W2 / 16;
-  constexpr int LD1 = W1 + 4, LD2 = W2 + 4;
-  __shared__ ushort_t Al[MT * LD1];
-  __shared__ ushort_t Bl[MT * LD2];
+  constexpr int HW1 = W1 / 2, HW2 = W2 / 2;   // per-wave sub-tile (2x2 waves)
+  constexpr int F1 = HW1 / 16, F2 = HW2 / 16; // 16x16 frags per wave
+  __shared__ ushort_t Al[W1 * LDK];
+  __shared__ ushort_t Bl[W2 * LDK];
 
   const int tid = threadIdx.x;
-  const int t1 = (tid & 15) * V1;
-  const int t2 = (tid >> 4) * V2;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wrow0 = (wave >> 1) * HW1, wcol0 = (wave & 1) * HW2;
+  const int fr = lane & 15, fk = (lane >> 4) * 8;
   const int n1_tiles = ceil_div(N1, W1);
   const int n_tiles = n1_tiles * ceil_div(N2, W2);
   const int tile = blockIdx.x % n_tiles;
@@ -242,99 +268,105 @@ __global__ __launch_bounds__(256) void atb_kernel(
   const int64_t m_lo = split * m_per;
   const int64_t m_hi = m_lo + m_per < M ? m_lo + m_per : M;
 
-  float acc[V1][V2] = {};
-  // staging: each thread loads  W/32 bf16x? ...  rows = 256/(W/8) per pass
-  constexpr int TPR1 = W1 / 8;   // threads per row (8 elems each)
+  f32x4 acc[F1][F2];
+#pragma unroll
+  for (int i = 0; i < F1; ++i)
+#pragma unroll
+    for (int j = 0; j < F2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  constexpr int TPR1 = W1 / 8;  // threads per m-row (vec8 along n)
   constexpr int TPR2 = W2 / 8;
 
-  for (int64_t m0 = m_lo; m0 < m_hi; m0 += MT) {
+  for (int64_t m0 = m_lo; m0 < m_hi; m0 += KT) {
+    __syncthreads();
     {
-      const int s_row = tid / TPR1, s_seg = (tid % TPR1) * 8;
-      for (int r = s_row; r < MT; r += 256 / TPR1) {
+      const int s_m0 = tid / TPR1, s_n = (tid % TPR1) * 8;
+      ushort_t tmp[8];
+      for (int r = s_m0; r < KT; r += 256 / TPR1) {
         const int64_t gm = m0 + r;
-        ushort_t* dst = &Al[r * LD1 + s_seg];
-        const int gn = n1_0 + s_seg;
+        const int gn = n1_0 + s_n;
         if (gm < m_hi && gn + 8 <= N1) {
-          lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, dst);
+          lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, tmp);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            dst[j] = (gm < m_hi && gn + j < N1)
+            tmp[j] = (gm < m_hi && gn + j < N1)
                          ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
                                                        gm * N1 + gn + j,
                                                        gn + j)))
                          : ushort_t(0);
         }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int n = s_n + j;
+          Al[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+        }
       }
     }
     {
-      const int s_row = tid / TPR2, s_seg = (tid % TPR2) * 8;
-      for (int r = s_row; r < MT; r += 256 / TPR2) {
+      const int s_m0 = tid / TPR2, s_n = (tid % TPR2) * 8;
+      ushort_t tmp[8];
+      for (int r = s_m0; r < KT; r += 256 / TPR2) {
         const int64_t gm = m0 + r;
-        ushort_t* dst = &Bl[r * LD2 + s_seg];
-        const int gn = n2_0 + s_seg;
+        const int gn = n2_0 + s_n;
         if (gm < m_hi && gn + 8 <= N2) {
-          lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, dst);
+          lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, tmp);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            dst[j] = (gm < m_hi && gn + j < N2)
+            tmp[j] = (gm < m_hi && gn + j < N2)
                          ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
                                                        gm * N2 + gn + j,
                                                        gn + j)))
                          : ushort_t(0);
         }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int n = s_n + j;
+          Bl[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+        }
       }
     }
     __syncthreads();
 
-    const int mm_end = (m_hi - m0) < MT ? int(m_hi - m0) : MT;
-    for (int mm = 0; mm < mm_end; ++mm) {
-      float a[V1], b[V2];
-      {  // V-wide vector loads (V*2 bytes, 8B-aligned for V>=4)
-        if (V1 >= 4) {
 #pragma unroll
-          for (int e = 0; e < V1; e += 4) {
-            short4_t v = *reinterpret_cast<const short4_t*>(
-                &Al[mm * LD1 + t1 + e]);
+    for (int ks = 0; ks < KT / 32; ++ks) {
+      bf16x8 a_frag[F1], b_frag[F2];
 #pragma unroll
-            for (int q = 0; q < 4; ++q) a[e + q] = bf2f(ushort_t(v[q]));
-          }
-        } else {
-#pragma unroll
-          for (int e = 0; e < V1; ++e) a[e] = bf2f(Al[mm * LD1 + t1 + e]);
-        }
-        if (V2 >= 4) {
-#pragma unroll
-          for (int e = 0; e < V2; e += 4) {
-            short4_t v = *reinterpret_cast<const short4_t*>(
-                &Bl[mm * LD2 + t2 + e]);
-#pragma unroll
-            for (int q = 0; q < 4; ++q) b[e + q] = bf2f(ushort_t(v[q]));
-          }
-        } else {
-#pragma unroll
-          for (int e = 0; e < V2; ++e) b[e] = bf2f(Bl[mm * LD2 + t2 + e]);
-        }
+      for (int i = 0; i < F1; ++i) {
+        const int n = wrow0 + i * 16 + fr;
+        const int kc = ((ks * 32 + fk) >> 3) ^ (n & 7);
+        a_frag[i] = *reinterpret_cast<const bf16x8*>(&Al[n * LDK + kc * 8]);
       }
 #pragma unroll
-      for (int i = 0; i < V1; ++i)
+      for (int j = 0; j < F2; ++j) {
+        const int n = wcol0 + j * 16 + fr;
+        const int kc = ((ks * 32 + fk) >> 3) ^ (n & 7);
+        b_frag[j] = *reinterpret_cast<const bf16x8*>(&Bl[n * LDK + kc * 8]);
+      }
 #pragma unroll
-        for (int j = 0; j < V2; ++j) acc[i][j] = fmaf(a[i], b[j], acc[i][j]);
+      for (int i = 0; i < F1; ++i)
+#pragma unroll
+        for (int j = 0; j < F2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
   }
 
+  // D frag layout: row (=n1) = (lane>>4)*4 + r, col (=n2) = lane&15
   float* out = ws + int64_t(split) * N1 * N2;
+  const int row_l4 = (lane >> 4) * 4;
 #pragma unroll
-  for (int i = 0; i < V1; ++i) {
-    const int n1 = n1_0 + t1 + i;
-    if (n1 >= N1) continue;
+  for (int j = 0; j < F2; ++j) {
+    const int n2 = n2_0 + wcol0 + j * 16 + fr;
+    if (n2 >= N2) continue;
 #pragma unroll
-    for (int j = 0; j < V2; ++j) {
-      const int n2 = n2_0 + t2 + j;
-      if (n2 < N2) out[int64_t(n1) * N2 + n2] = acc[i][j];
-    }
+    for (int i = 0; i < F1; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int n1 = n1_0 + wrow0 + i * 16 + row_l4 + r;
+        if (n1 < N1) out[int64_t(n1) * N2 + n2] = acc[i][j][r];
+      }
   }
 }
 
