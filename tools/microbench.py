@@ -101,7 +101,8 @@ for (m_hw, cin, cout) in [(112, 16, 64), (56, 24, 72), (28, 40, 120),
           f"stats-pass {statp:7.1f}")
 
 # ---- dw wgrad lopx cost ------------------------------------------------
-for (c, h, k, s) in [(120, 28, 5, 1), (480, 14, 3, 1), (960, 7, 5, 1)]:
+for (c, h, k, s) in [(120, 28, 5, 1), (480, 14, 3, 1), (960, 7, 5, 1),
+                     (72, 56, 5, 2), (672, 14, 5, 2), (64, 112, 3, 2)]:
     p = k // 2
     ho = (h + 2 * p - k) // s + 1
     x = nhwc(N, c, h, h)

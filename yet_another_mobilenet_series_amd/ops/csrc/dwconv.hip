@@ -242,7 +242,11 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
 // wgrad: thread = 4-channel vector lane; 4 consecutive wo outputs share
 // one x row window in registers; per-chunk fp32 slab after an LDS fold
 // over the m lanes.
-template <int K, int S, int LX, int LXACT, int LD>
+// IP: kernel-row parts — K=5 accumulates 25 taps x 4ch = 100 VGPRs in one
+// block, capping occupancy at 2 waves/SIMD; splitting the K rows across IP
+// blocks (rows re-read dy, x-row traffic unchanged) trades a little dy
+// bandwidth for 2x occupancy on the latency-bound load stream.
+template <int K, int S, int LX, int LXACT, int LD, int IP>
 __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
     float* __restrict__ ws, int N, int H, int W, int C, int Ho, int Wo,
@@ -253,10 +257,13 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int CT4 = C4 < 256 ? C4 : 256;
   const int ML = 256 / CT4;
   const int c_tiles = (C4 + CT4 - 1) / CT4;
+  constexpr int KR = (K + IP - 1) / IP;  // kernel rows per part
   const int c4 = (blockIdx.x % c_tiles) * CT4 + int(threadIdx.x) % CT4;
-  const int chunk = blockIdx.x / c_tiles;
+  const int part = int(blockIdx.x / c_tiles) % IP;
+  const int chunk = blockIdx.x / (c_tiles * IP);
   const int mlane = int(threadIdx.x) / CT4;
   const int c = c4 * 4;
+  const int i0 = part * KR;
 
   const int Wo4 = (Wo + 3) / 4;
   const int64_t P = int64_t(N) * Ho * Wo4;
@@ -264,7 +271,7 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int64_t lo = chunk * per;
   const int64_t hi_ = std::min<int64_t>(lo + per, P);
 
-  float acc[K * K][4] = {};
+  float acc[KR * K][4] = {};
   LopRegT<LX, LXACT, 4> lrx;
   LopRegT<LD, 0, 4> lrd;
   if (c4 < C4 && mlane < ML) {
@@ -306,7 +313,9 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
       const int wbase = wo0 * S - pad;
       const int64_t imgbase = int64_t(n) * H * W * C + c;
 #pragma unroll
-      for (int i = 0; i < K; ++i) {
+      for (int ii = 0; ii < KR; ++ii) {
+        const int i = i0 + ii;
+        if (IP > 1 && i >= K) break;
         const int hi2 = ho * S + i - pad;
         const int hic = hi2 < 0 ? 0 : (hi2 >= H ? H - 1 : hi2);
         const bool row_ok = hi2 >= 0 && hi2 < H;
@@ -337,14 +346,16 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
 #pragma unroll
             for (int q = 0; q < 4; ++q)
               sum = fmaf(g[q][e], xr[q * S + j][e], sum);
-            acc[i * K + j][e] += sum;
+            acc[ii * K + j][e] += sum;
           }
       }
     }
   }
   float* slab = ws + int64_t(chunk) * K * K * C;
 #pragma unroll
-  for (int t = 0; t < K * K; ++t) {
+  for (int t = 0; t < KR * K; ++t) {
+    const int tap = (i0 + t / K) * K + t % K;
+    if (IP > 1 && tap >= K * K) break;
 #pragma unroll
     for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = acc[t][e];
     __syncthreads();
@@ -353,7 +364,7 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
       for (int e = 0; e < 4; ++e) {
         float v = 0.f;
         for (int l = 0; l < ML; ++l) v += red[(threadIdx.x + l * CT4) * 4 + e];
-        slab[t * C + c + e] = v;
+        slab[tap * C + c + e] = v;
       }
     }
     __syncthreads();
@@ -511,12 +522,13 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   auto* gp = static_cast<const ushort_t*>(dy);
   const Lop Lx = to_lop(lopx);
   const Lop Ldy = to_lop(lopdy);
-  dim3 grid(c_tiles * chunks);
   // wgrad x-stream lop: none or bnact (act 1..3); dy-stream: none only
 #define DWW_CASE(KK, SS, LXX, AC)                                          \
   if (k == KK && stride == SS && lopx.kind == LXX &&                       \
       (LXX == 0 || lopx.act == AC) && lopdy.kind == 0) {                   \
-    dw_wgrad_kernel<KK, SS, LXX, AC, 0><<<grid, 256, 0, s>>>(              \
+    constexpr int IP = (KK == 5) ? 2 : 1;                                  \
+    dim3 grid(c_tiles * chunks * IP);                                      \
+    dw_wgrad_kernel<KK, SS, LXX, AC, 0, IP><<<grid, 256, 0, s>>>(          \
         xp, gp, ws, N, H, W, C, Ho, Wo, pad, Lx, Ldy, chunks);             \
     goto combine;                                                          \
   }

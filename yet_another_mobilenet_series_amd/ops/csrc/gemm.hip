@@ -394,26 +394,44 @@ void atb_launch(const ushort_t* a, const ushort_t* b, float* ws, int64_t M,
 #undef ATB_CASE
 }
 
-// split-parallel: 256 threads = EL element lanes x SL split lanes
-template <int EL>
-__global__ __launch_bounds__(256) void slab_reduce_kernel(
+// vec4 variant: thread = 4 consecutive fp32 elems x one of SL split lanes;
+// reads are 16B/lane fully coalesced (the scalar kernel read 4B chunks and
+// serialized the split loop -> 4-5x off the slab-stream floor).
+template <int SL>
+__global__ __launch_bounds__(256) void slab_reduce4_kernel(
     const float* __restrict__ ws, float* __restrict__ out, int64_t elems,
     int splits) {
-  constexpr int SL = 256 / EL;
-  __shared__ float red[256];
+  constexpr int EL = 256 / SL;          // float4 lanes per block
+  __shared__ float red[256][4];
   const int el = int(threadIdx.x) % EL;
   const int sl = int(threadIdx.x) / EL;
-  for (int64_t base = int64_t(blockIdx.x) * EL; base < elems;
-       base += int64_t(gridDim.x) * EL) {
-    const int64_t i = base + el;
-    float v = 0.f;
-    if (i < elems)
-      for (int sp = sl; sp < splits; sp += SL) v += ws[int64_t(sp) * elems + i];
-    red[threadIdx.x] = v;
+  for (int64_t base = int64_t(blockIdx.x) * EL * 4; base < elems;
+       base += int64_t(gridDim.x) * EL * 4) {
+    const int64_t i = base + el * 4;
+    float v[4] = {};
+    if (i + 4 <= elems) {
+      for (int sp = sl; sp < splits; sp += SL) {
+        const float4 w4 = *reinterpret_cast<const float4*>(
+            &ws[int64_t(sp) * elems + i]);
+        v[0] += w4.x; v[1] += w4.y; v[2] += w4.z; v[3] += w4.w;
+      }
+    } else if (i < elems) {
+      for (int sp = sl; sp < splits; sp += SL)
+        for (int e = 0; e < 4; ++e)
+          if (i + e < elems) v[e] += ws[int64_t(sp) * elems + i + e];
+    }
+#pragma unroll
+    for (int e = 0; e < 4; ++e) red[threadIdx.x][e] = v[e];
     __syncthreads();
     if (sl == 0 && i < elems) {
-      for (int l = 1; l < SL; ++l) v += red[el + l * EL];
-      out[i] = v;
+      for (int l = 1; l < SL; ++l)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) v[e] += red[el + l * EL][e];
+      if (i + 4 <= elems) {
+        *reinterpret_cast<float4*>(&out[i]) = float4{v[0], v[1], v[2], v[3]};
+      } else {
+        for (int e = 0; e < 4 && i + e < elems; ++e) out[i + e] = v[e];
+      }
     }
     __syncthreads();
   }
@@ -423,15 +441,18 @@ __global__ __launch_bounds__(256) void slab_reduce_kernel(
 
 void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s) {
-  if (splits >= 256) {
-    const int blocks = int(std::min<int64_t>((elems + 7) / 8, 8192));
-    slab_reduce_kernel<8><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
-                                                       splits);
-  } else {
-    const int blocks = int(std::min<int64_t>((elems + 15) / 16, 8192));
-    slab_reduce_kernel<16><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
+  const int sl = splits >= 64 ? 64 : (splits >= 16 ? 16 : 4);
+  const int el4 = (256 / sl) * 4;
+  const int blocks = int(std::min<int64_t>((elems + el4 - 1) / el4, 8192));
+  if (sl == 64)
+    slab_reduce4_kernel<64><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
+                                                         splits);
+  else if (sl == 16)
+    slab_reduce4_kernel<16><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
+                                                         splits);
+  else
+    slab_reduce4_kernel<4><<<dim3(blocks), 256, 0, s>>>(ws, out, elems,
                                                         splits);
-  }
 }
 
 int pw_wgrad_splits(int64_t M, int N1, int N2) {
