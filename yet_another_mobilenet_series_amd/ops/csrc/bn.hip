@@ -247,26 +247,26 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     const float* __restrict__ beta, float* __restrict__ dgamma,
     float* __restrict__ dbeta, float* __restrict__ coef, int64_t M, int C,
     int splits) {
-  __shared__ float red[256];
+  __shared__ float red[256][2];
   const int cl = int(threadIdx.x) % 16;
   const int sl = int(threadIdx.x) / 16;
   const int c = blockIdx.x * 16 + cl;
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
+#pragma unroll 4
     for (int sp = sl; sp < splits; sp += 16) {
       s1 += ws[int64_t(sp) * C + c];
       s2 += ws[int64_t(splits + sp) * C + c];
     }
   }
-  red[threadIdx.x] = s1;
-  __syncthreads();
-  if (sl == 0 && c < C)
-    for (int l = 1; l < 16; ++l) s1 += red[cl + l * 16];
-  __syncthreads();
-  red[threadIdx.x] = s2;
+  red[threadIdx.x][0] = s1;
+  red[threadIdx.x][1] = s2;
   __syncthreads();
   if (sl == 0 && c < C) {
-    for (int l = 1; l < 16; ++l) s2 += red[cl + l * 16];
+    for (int l = 1; l < 16; ++l) {
+      s1 += red[cl + l * 16][0];
+      s2 += red[cl + l * 16][1];
+    }
     const float mu = mean[c], inv = invstd[c];
     const float dg = (s2 - mu * s1) * inv;  // sum(da*zhat)
     dgamma[c] = dg;

@@ -79,13 +79,15 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       lop_vec8<LOPK>(lopA, A, off, gka, areg);
       lop_vec8<LOPK>(lopA, A, off + 8, gka + 8, areg + 8);
     } else {
+      // branchless: clamped in-bounds addresses + value select (guide trap c:
+      // per-element branches around loads serialize on vmcnt(0))
+      const int64_t gmc = gm < M ? gm : M - 1;
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int gkj = gka + j;
-        areg[j] = (gm < M && gkj < K)
-                      ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
-                                                      gm * K + gkj, gkj)))
-                      : ushort_t(0);
+        const int gkc = gkj < K ? gkj : K - 1;
+        const float v = lop_loadT<LOPK>(lopA, A, gmc * K + gkc, gkc);
+        areg[j] = (gm < M && gkj < K) ? ushort_t(f2bf(v)) : ushort_t(0);
       }
     }
     const int gkb = k0 + b_seg;
@@ -95,10 +97,14 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       *reinterpret_cast<short4_t*>(breg + 4) =
           *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gkb + 4]);
     } else {
+      const int gnc = gn < N ? gn : N - 1;
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        breg[j] = (gn < N && gkb + j < K)
-                      ? Bt[int64_t(gn) * K + gkb + j] : ushort_t(0);
+      for (int j = 0; j < 8; ++j) {
+        const int gkj = gkb + j;
+        const int gkc = gkj < K ? gkj : K - 1;
+        const ushort_t v = Bt[int64_t(gnc) * K + gkc];
+        breg[j] = (gn < N && gkj < K) ? v : ushort_t(0);
+      }
     }
   };
 
@@ -288,13 +294,15 @@ __global__ __launch_bounds__(256) void atb_kernel(
         if (gm < m_hi && gn + 8 <= N1) {
           lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, tmp);
         } else {
+          const int64_t gmc = gm < m_hi ? gm : m_hi - 1;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            tmp[j] = (gm < m_hi && gn + j < N1)
-                         ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
-                                                       gm * N1 + gn + j,
-                                                       gn + j)))
-                         : ushort_t(0);
+          for (int j = 0; j < 8; ++j) {
+            const int gnj = gn + j;
+            const int gnc = gnj < N1 ? gnj : N1 - 1;
+            const float v = lop_loadT<LA>(lopA, A, gmc * N1 + gnc, gnc);
+            tmp[j] = (gm < m_hi && gnj < N1) ? ushort_t(f2bf(v))
+                                             : ushort_t(0);
+          }
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -312,13 +320,15 @@ __global__ __launch_bounds__(256) void atb_kernel(
         if (gm < m_hi && gn + 8 <= N2) {
           lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, tmp);
         } else {
+          const int64_t gmc = gm < m_hi ? gm : m_hi - 1;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            tmp[j] = (gm < m_hi && gn + j < N2)
-                         ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
-                                                       gm * N2 + gn + j,
-                                                       gn + j)))
-                         : ushort_t(0);
+          for (int j = 0; j < 8; ++j) {
+            const int gnj = gn + j;
+            const int gnc = gnj < N2 ? gnj : N2 - 1;
+            const float v = lop_loadT<LB>(lopB, B, gmc * N2 + gnc, gnc);
+            tmp[j] = (gm < m_hi && gnj < N2) ? ushort_t(f2bf(v))
+                                             : ushort_t(0);
+          }
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
