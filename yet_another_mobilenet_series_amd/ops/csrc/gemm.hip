@@ -78,16 +78,20 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       const int64_t off = gm * K + gka;
       lop_vec8<LOPK>(lopA, A, off, gka, areg);
       lop_vec8<LOPK>(lopA, A, off + 8, gka + 8, areg + 8);
+    } else if (gm >= M || gka >= K) {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) areg[j] = ushort_t(0);
     } else {
-      // branchless: clamped in-bounds addresses + value select (guide trap c:
-      // per-element branches around loads serialize on vmcnt(0))
-      const int64_t gmc = gm < M ? gm : M - 1;
+      // partial segment: per-element conditional loads (the segment-level
+      // all-OOB cases above are wave-uniform and skip loads entirely;
+      // always-executed clamped loads measured slower for lop variants)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const int gkj = gka + j;
-        const int gkc = gkj < K ? gkj : K - 1;
-        const float v = lop_loadT<LOPK>(lopA, A, gmc * K + gkc, gkc);
-        areg[j] = (gm < M && gkj < K) ? ushort_t(f2bf(v)) : ushort_t(0);
+        areg[j] = (gm < M && gkj < K)
+                      ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
+                                                      gm * K + gkj, gkj)))
+                      : ushort_t(0);
       }
     }
     const int gkb = k0 + b_seg;
@@ -96,15 +100,14 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
           *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gkb]);
       *reinterpret_cast<short4_t*>(breg + 4) =
           *reinterpret_cast<const short4_t*>(&Bt[int64_t(gn) * K + gkb + 4]);
-    } else {
-      const int gnc = gn < N ? gn : N - 1;
+    } else if (gn >= N || gkb >= K) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int gkj = gkb + j;
-        const int gkc = gkj < K ? gkj : K - 1;
-        const ushort_t v = Bt[int64_t(gnc) * K + gkc];
-        breg[j] = (gn < N && gkj < K) ? v : ushort_t(0);
-      }
+      for (int j = 0; j < 8; ++j) breg[j] = ushort_t(0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        breg[j] = (gn < N && gkb + j < K)
+                      ? Bt[int64_t(gn) * K + gkb + j] : ushort_t(0);
     }
   };
 
@@ -293,16 +296,17 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int gn = n1_0 + s_n;
         if (gm < m_hi && gn + 8 <= N1) {
           lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, tmp);
-        } else {
-          const int64_t gmc = gm < m_hi ? gm : m_hi - 1;
+        } else if (gm >= m_hi || gn >= N1) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int gnj = gn + j;
-            const int gnc = gnj < N1 ? gnj : N1 - 1;
-            const float v = lop_loadT<LA>(lopA, A, gmc * N1 + gnc, gnc);
-            tmp[j] = (gm < m_hi && gnj < N1) ? ushort_t(f2bf(v))
-                                             : ushort_t(0);
-          }
+          for (int j = 0; j < 8; ++j) tmp[j] = ushort_t(0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tmp[j] = (gm < m_hi && gn + j < N1)
+                         ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
+                                                       gm * N1 + gn + j,
+                                                       gn + j)))
+                         : ushort_t(0);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -319,16 +323,17 @@ __global__ __launch_bounds__(256) void atb_kernel(
         const int gn = n2_0 + s_n;
         if (gm < m_hi && gn + 8 <= N2) {
           lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, tmp);
-        } else {
-          const int64_t gmc = gm < m_hi ? gm : m_hi - 1;
+        } else if (gm >= m_hi || gn >= N2) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int gnj = gn + j;
-            const int gnc = gnj < N2 ? gnj : N2 - 1;
-            const float v = lop_loadT<LB>(lopB, B, gmc * N2 + gnc, gnc);
-            tmp[j] = (gm < m_hi && gnj < N2) ? ushort_t(f2bf(v))
-                                             : ushort_t(0);
-          }
+          for (int j = 0; j < 8; ++j) tmp[j] = ushort_t(0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tmp[j] = (gm < m_hi && gn + j < N2)
+                         ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
+                                                       gm * N2 + gn + j,
+                                                       gn + j)))
+                         : ushort_t(0);
         }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
