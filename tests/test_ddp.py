@@ -154,3 +154,57 @@ def test_dead_rank_does_not_hang(tmp_path):
             assert False, "survivor hung past the collective timeout"
     assert os.path.exists(out_path)
     assert torch.load(out_path) == "raised"
+
+def _accum_worker(rank, world, tmpfile, out_path):
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world)
+    try:
+        model = _make_model(seed=7)
+        eng = DataParallelEngine(model, bucket_mb=0.0001, overlap=True)
+        torch.manual_seed(1)
+        xs = torch.randn(world * 8, 8)
+        ys = torch.randn(world * 8, 4)
+        # 2 micro-batches of 4 per rank, one optimizer boundary
+        for micro in range(2):
+            lo = rank * 8 + micro * 4
+            eng.set_no_sync(micro < 1)
+            loss = ((eng(xs[lo:lo + 4]) - ys[lo:lo + 4]) ** 2).mean() / 2
+            loss.backward()
+        eng.finish()
+        if rank == 0:
+            torch.save([p.grad.detach().clone()
+                        for p in model.parameters()], out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_grad_accumulation_matches_big_batch(tmp_path):
+    """k=2 accumulation over gloo world-2 == one global 16-sample step."""
+    world = 2
+    for attempt in range(3):
+        ctx = mp.get_context("spawn")
+        tmpfile = str(tmp_path / f"init_acc_{attempt}")
+        out_path = str(tmp_path / f"out_acc_{attempt}.pt")
+        ps = [ctx.Process(target=_accum_worker,
+                          args=(r, world, tmpfile, out_path))
+              for r in range(world)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+        if all(p.exitcode == 0 for p in ps) and os.path.exists(out_path):
+            got = torch.load(out_path)
+            break
+        if attempt == 2:
+            assert False, f"workers failed: {[p.exitcode for p in ps]}"
+    model = _make_model(seed=7)
+    torch.manual_seed(1)
+    xs = torch.randn(16, 8)
+    ys = torch.randn(16, 4)
+    # mean over the 4 micro-losses (2 ranks x 2 micros), matching /2 + /world
+    loss = sum(((model(xs[i * 4:(i + 1) * 4]) - ys[i * 4:(i + 1) * 4]) ** 2
+                ).mean() for i in range(4)) / 4
+    loss.backward()
+    for a, p in zip(got, model.parameters()):
+        assert torch.allclose(a, p.grad, atol=1e-5), (a - p.grad).abs().max()

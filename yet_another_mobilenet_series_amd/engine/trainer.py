@@ -98,21 +98,31 @@ class Trainer:
         # interval when train.perf_breakdown is on
         breakdown = (getattr(cfg.train, "perf_breakdown", False)
                      and self.device.type == "cuda")
+        accum = max(int(getattr(cfg.train, "grad_accum_steps", 1)), 1)
         for it, (x, y) in enumerate(self.train_loader):
             x, y = self._to_device(x, y)
             lr = self.lr_sched.step(self.global_step)
+            micro = it % accum
+            boundary = micro == accum - 1
             if breakdown:
                 evs = [torch.cuda.Event(enable_timing=True)
                        for _ in range(5)]
                 evs[0].record()
-            self.engine.zero_grad()
+            if micro == 0:
+                self.engine.zero_grad()
+            if accum > 1:
+                self.engine.set_no_sync(not boundary)
             logits = self.engine(x)
             loss = self.criterion(logits, y)
+            if accum > 1:
+                loss = loss / accum
             if breakdown:
                 evs[1].record()
             loss.backward()
             if breakdown:
                 evs[2].record()
+            if not boundary:
+                continue
             self.engine.finish()
             if breakdown:
                 evs[3].record()
@@ -135,7 +145,7 @@ class Trainer:
                 top1, top5 = accuracy(logits.detach(), y, (1, 5))
             bs = x.shape[0]
             now = time.time()
-            meters["loss"].update(loss.item(), bs)
+            meters["loss"].update(loss.item() * accum, bs)
             meters["top1"].update(top1.item(), bs)
             meters["top5"].update(top5.item(), bs)
             meters["img_s"].update(bs * self.world / max(now - t_last, 1e-9))

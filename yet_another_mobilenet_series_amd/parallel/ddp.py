@@ -78,6 +78,7 @@ class DataParallelEngine(nn.Module):
             self._hooks = []
             self._launched = []
             self._params = params
+            self._no_sync = False
             return
         device = params[0].device if params else torch.device("cpu")
         # reverse registration order ≈ grad production order during backward
@@ -104,8 +105,21 @@ class DataParallelEngine(nn.Module):
                 h = p.register_post_accumulate_grad_hook(self._grad_hook)
                 self._hooks.append(h)
         self._launched: List[_Bucket] = []
+        self._no_sync = False
 
     # -- training-step API -------------------------------------------------
+
+    def set_no_sync(self, flag: bool) -> None:
+        """Gradient accumulation: while True, backward accumulates into the
+        flat buckets but no all-reduce is launched (DDP.no_sync semantics).
+        Call with False before the boundary micro-step so its backward
+        triggers the (overlapped) reduce of the accumulated sums."""
+        if self._no_sync and not flag and self._bucketed:
+            for b in self.buckets:   # re-arm fill counters for the boundary
+                b.pending = len(b.params)
+                b.handle = None
+            self._launched.clear()
+        self._no_sync = flag
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
@@ -124,6 +138,8 @@ class DataParallelEngine(nn.Module):
         self._launched.clear()
 
     def _grad_hook(self, p: nn.Parameter) -> None:
+        if self._no_sync:
+            return
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0 and self.world_size > 1:
@@ -132,7 +148,7 @@ class DataParallelEngine(nn.Module):
 
     def finish(self) -> None:
         """Wait for in-flight reduces (or run them now) and average."""
-        if self.world_size <= 1:
+        if self.world_size <= 1 or self._no_sync:
             return
         if self.overlap:
             for b in self.buckets:
