@@ -57,20 +57,22 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
 
   const int Wo4 = (Wo + 3) / 4;
   const int HoG = (Ho + TH - 1) / TH;
-  const int64_t P = int64_t(N) * HoG * Wo4;
-  const int64_t per = (P + chunks - 1) / chunks;
-  const int64_t lo = chunk * per;
-  const int64_t hi_ = std::min<int64_t>(lo + per, P);
+  // 32-bit position decode: P = N*HoG*Wo4 < 2^31 always for these nets,
+  // and a 64-bit div is ~4x the instructions of a 32-bit one on gfx9
+  const int P = N * HoG * Wo4;
+  const int per = (P + chunks - 1) / chunks;
+  const int lo = chunk * per;
+  const int hi_ = min(lo + per, P);
 
   float s1[4] = {}, s2[4] = {};
   LopRegT<LOPK, ACT, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
-    for (int64_t p = lo + mlane; p < hi_; p += ML) {
-      const int w4 = int(p % Wo4);
-      int64_t rest = p / Wo4;
-      const int hg = int(rest % HoG);
-      const int n = int(rest / HoG);
+    for (int p = lo + mlane; p < hi_; p += ML) {
+      const int w4 = p % Wo4;
+      const int rest = p / Wo4;
+      const int hg = rest % HoG;
+      const int n = rest / HoG;
       const int ho0 = hg * TH;
       const int wo0 = w4 * 4;
       const int wbase = wo0 * S - pad;
@@ -266,10 +268,10 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   const int i0 = part * KR;
 
   const int Wo4 = (Wo + 3) / 4;
-  const int64_t P = int64_t(N) * Ho * Wo4;
-  const int64_t per = (P + chunks - 1) / chunks;
-  const int64_t lo = chunk * per;
-  const int64_t hi_ = std::min<int64_t>(lo + per, P);
+  const int P = N * Ho * Wo4;   // < 2^31; 32-bit div is 4x cheaper
+  const int per = (P + chunks - 1) / chunks;
+  const int lo = chunk * per;
+  const int hi_ = min(lo + per, P);
 
   float acc[KR * K][4] = {};
   LopRegT<LX, LXACT, 4> lrx;
@@ -277,11 +279,11 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
   if (c4 < C4 && mlane < ML) {
     lrx.init(lopx, c);
     lrd.init(lopdy, c);
-    for (int64_t p = lo + mlane; p < hi_; p += ML) {
-      const int w4 = int(p % Wo4);
-      int64_t rest = p / Wo4;
-      const int ho = int(rest % Ho);
-      const int n = int(rest / Ho);
+    for (int p = lo + mlane; p < hi_; p += ML) {
+      const int w4 = p % Wo4;
+      const int rest = p / Wo4;
+      const int ho = rest % Ho;
+      const int n = rest / Ho;
       const int wo0 = w4 * 4;
       const ushort_t* dyrow =
           dy + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + c;
