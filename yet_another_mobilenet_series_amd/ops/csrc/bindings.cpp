@@ -94,7 +94,9 @@ yams::LopDesc make_lop(int64_t kind, int64_t act,
 void pw_conv(const Tensor& x, const Tensor& w, Tensor& y,
              int64_t lop_kind, int64_t lop_act,
              c10::optional<Tensor> lop_coef, c10::optional<Tensor> lop_z,
-             c10::optional<Tensor> stats_ws) {
+             c10::optional<Tensor> stats_ws,
+             int64_t bstats_act, c10::optional<Tensor> bstats_coef,
+             c10::optional<Tensor> bstats_z) {
   auto xr = rows_of(x, "x");
   auto yr = rows_of(y, "y");
   check_w(w);
@@ -110,8 +112,19 @@ void pw_conv(const Tensor& x, const Tensor& w, Tensor& y,
                 "stats_ws too small");
     sw = stats_ws->data_ptr<float>();
   }
+  // bstats: the output is a dgrad dy; emit the consumer BN's backward
+  // partial sums into stats_ws (EPI==2) instead of forward sum/sumsq
+  yams::LopDesc epi{};
+  if (bstats_coef.has_value()) {
+    TORCH_CHECK(sw != nullptr, "bstats requires stats_ws");
+    TORCH_CHECK(bstats_z.has_value() &&
+                    bstats_z->scalar_type() == torch::kBFloat16,
+                "bstats requires bf16 z");
+    epi = make_lop(1, bstats_act, bstats_coef, bstats_z, yr.C);
+    epi.z = bstats_z->data_ptr();
+  }
   yams::pw_conv(xr.ptr, w.data_ptr(), y.data_ptr(), xr.M, xr.C, yr.C, lop,
-                sw, cur_stream());
+                epi, sw, cur_stream());
 }
 
 int64_t pw_stats_mblocks(int64_t M) { return yams::pw_stats_mblocks(M); }
@@ -164,17 +177,32 @@ int64_t dw_stats_chunks(int64_t N, int64_t Ho, int64_t Wo, int64_t C,
 void dw_conv_dgrad(const Tensor& dy, const Tensor& w, Tensor& dx, int64_t k,
                    int64_t stride, int64_t pad, int64_t lop_kind,
                    int64_t lop_act, c10::optional<Tensor> lop_coef,
-                   c10::optional<Tensor> lop_z) {
+                   c10::optional<Tensor> lop_z,
+                   c10::optional<Tensor> stats_ws, int64_t bstats_act,
+                   c10::optional<Tensor> bstats_coef,
+                   c10::optional<Tensor> bstats_z) {
   auto gr = rows_of(dy, "dy");
   auto xr = rows_of(dx, "dx");
   check_w(w);
   TORCH_CHECK(gr.C == xr.C, "dw channel mismatch");
+  float* sw = nullptr;
+  yams::LopDesc epi{};
+  if (stats_ws.has_value()) {
+    TORCH_CHECK(stride == 1 && bstats_coef.has_value(),
+                "dgrad bstats: stride-1 only, needs coef");
+    TORCH_CHECK(bstats_z.has_value() &&
+                    bstats_z->scalar_type() == torch::kBFloat16,
+                "bstats requires bf16 z");
+    sw = stats_ws->data_ptr<float>();
+    epi = make_lop(1, bstats_act, bstats_coef, bstats_z, xr.C);
+    epi.z = bstats_z->data_ptr();
+  }
   yams::dw_conv_dgrad(gr.ptr, w.data_ptr(), dx.data_ptr(), int(dx.size(0)),
                       int(dx.size(2)), int(dx.size(3)), xr.C,
                       int(dy.size(2)), int(dy.size(3)), int(k), int(stride),
                       int(pad),
                       make_lop(lop_kind, lop_act, lop_coef, lop_z, gr.C),
-                      cur_stream());
+                      epi, sw, cur_stream());
 }
 
 void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
@@ -251,18 +279,29 @@ void bn_act_fwd(const Tensor& z, const Tensor& ab, int64_t act, Tensor& y,
 void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
                 const Tensor& invstd, const Tensor& gamma,
                 const Tensor& beta, int64_t act, bool training,
-                Tensor& dgamma, Tensor& dbeta, Tensor& dz) {
+                Tensor& dgamma, Tensor& dbeta, Tensor& dz,
+                c10::optional<Tensor> partial_ws, int64_t partial_splits) {
   auto zr = rows_of(z, "z");
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
-  auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
   auto coef = f32_ws(z, 4 * zr.C);
+  float* wp;
+  Tensor ws;
+  if (partial_ws.has_value()) {
+    TORCH_CHECK(partial_splits > 0 &&
+                    partial_ws->numel() >= 2 * partial_splits * zr.C,
+                "partial_ws too small");
+    wp = partial_ws->data_ptr<float>();
+  } else {
+    ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+    wp = ws.data_ptr<float>();
+  }
   yams::bn_act_bwd(gr.ptr, zr.ptr, fp32_req(mean, "mean"),
                    fp32_req(invstd, "invstd"), fp32_req(gamma, "gamma"),
                    fp32_req(beta, "beta"), int(act), training,
                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                   dz.data_ptr(), ws.data_ptr<float>(),
-                   coef.data_ptr<float>(), zr.M, zr.C, true, cur_stream());
+                   dz.data_ptr(), wp, coef.data_ptr<float>(), zr.M, zr.C,
+                   true, int(partial_splits), cur_stream());
 }
 
 // reduce-only backward: dgamma/dbeta + the coef[4C] used by LOP_BNBWD
@@ -281,7 +320,7 @@ void bn_bwd_stats(const Tensor& dy, const Tensor& z, const Tensor& mean,
                    fp32_req(beta, "beta"), int(act), training,
                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                    nullptr, ws.data_ptr<float>(), coef.data_ptr<float>(),
-                   zr.M, zr.C, false, cur_stream());
+                   zr.M, zr.C, false, 0, cur_stream());
 }
 
 // combine conv-epilogue stats slabs [2, splits, C] into mean/invstd/ab
@@ -484,7 +523,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pw_conv", &pw_conv, "y[M,N] = x[M,K] @ w[N,K]^T (MFMA bf16)",
         "x"_a, "w"_a, "y"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
         "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt,
-        "stats_ws"_a = c10::nullopt);
+        "stats_ws"_a = c10::nullopt, "bstats_act"_a = 0,
+        "bstats_coef"_a = c10::nullopt, "bstats_z"_a = c10::nullopt);
   m.def("pw_stats_mblocks", &pw_stats_mblocks);
   m.def("pw_conv_wgrad", &pw_conv_wgrad, "dw[N1,N2] = a^T @ b (split-M)",
         "a"_a, "b"_a, "dw"_a, "lopa_kind"_a = 0, "lopa_act"_a = 0,
@@ -498,7 +538,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dw_stats_chunks", &dw_stats_chunks);
   m.def("dw_conv_dgrad", &dw_conv_dgrad, "dy"_a, "w"_a, "dx"_a, "k"_a,
         "stride"_a, "pad"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
-        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt);
+        "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt,
+        "stats_ws"_a = c10::nullopt, "bstats_act"_a = 0,
+        "bstats_coef"_a = c10::nullopt, "bstats_z"_a = c10::nullopt);
   m.def("dw_conv_wgrad", &dw_conv_wgrad, "x"_a, "dy"_a, "dw"_a, "k"_a,
         "stride"_a, "pad"_a, "lopx_kind"_a = 0, "lopx_act"_a = 0,
         "lopx_coef"_a = c10::nullopt, "lopx_z"_a = c10::nullopt,
@@ -511,7 +553,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "running_mean"_a = c10::nullopt, "running_var"_a = c10::nullopt,
         "momentum"_a = 0.1, "eps"_a = 1e-5, "M"_a = 1);
   m.def("bn_act_fwd", &bn_act_fwd);
-  m.def("bn_act_bwd", &bn_act_bwd);
+  m.def("bn_act_bwd", &bn_act_bwd, "dy"_a, "z"_a, "mean"_a, "invstd"_a,
+        "gamma"_a, "beta"_a, "act"_a, "training"_a, "dgamma"_a, "dbeta"_a,
+        "dz"_a, "partial_ws"_a = c10::nullopt, "partial_splits"_a = 0);
   m.def("bn_bwd_stats", &bn_bwd_stats);
   m.def("bias_act_fwd", &bias_act_fwd);
   m.def("bias_act_bwd", &bias_act_bwd);

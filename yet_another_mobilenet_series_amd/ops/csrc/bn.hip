@@ -413,23 +413,28 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
                 void* dz, float* ws, float* coef, int64_t M, int C,
-                bool apply, hipStream_t s) {
+                bool apply, int ext_splits, hipStream_t s) {
+  // ext_splits > 0: the partial sums were already produced by the kernel
+  // that WROTE dy (conv-dgrad epilogue, EPI==2) into ws as [2, ext_splits,
+  // C] — skip the bn_bwd_partial pass over dy+z entirely.
   const RedGeom g = red_geom(M, C);
+  const int splits = ext_splits > 0 ? ext_splits : g.chunks;
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* zp = static_cast<const ushort_t*>(z);
   auto* dzp = static_cast<ushort_t*>(dz);
   const int64_t total8 = M * (C >> 3);
   ACT_SWITCH(act, {
-    bn_bwd_partial_kernel<kAct><<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
-        gp, zp, mean, invstd, gamma, beta, ws, M, C, g.chunks);
+    if (ext_splits <= 0)
+      bn_bwd_partial_kernel<kAct><<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
+          gp, zp, mean, invstd, gamma, beta, ws, M, C, g.chunks);
     if (training)
       bn_bwd_finalize_kernel<true><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
           ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
-          g.chunks);
+          splits);
     else
       bn_bwd_finalize_kernel<false><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
           ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
-          g.chunks);
+          splits);
     if (apply)
       bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
           gp, zp, coef, dzp, total8, C >> 3, C);

@@ -37,12 +37,16 @@ inline Lop to_lop(const LopDesc& d) {
 // K times through L2 — the dominant cost on these latency-bound shapes).
 // FLIP reuses the kernel as the stride-1 dgrad; lopx fuses the producer
 // BN+act (or the BNBWD dz for dgrad) into the loads.
-template <int K, int S, bool FLIP, bool STATS, int LOPK, int ACT>
+// EPI: 0 none, 1 forward BN stats of the output, 2 BN-BACKWARD stats:
+// FLIP dgrad output dy feeds a BN backward, and the epilogue accumulates
+// (sum da, sum da*z) with da = dy * act'(a*z+b) so no separate
+// bn_bwd_partial pass reads dy+z again.  lopE carries z/ab/act for EPI==2.
+template <int K, int S, bool FLIP, int EPI, int LOPK, int ACT>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
-    int pad, Lop lopx, float* __restrict__ stats_ws, int chunks) {
-  __shared__ float red[STATS ? 256 * 4 : 4];
+    int pad, Lop lopx, Lop lopE, float* __restrict__ stats_ws, int chunks) {
+  __shared__ float red[EPI ? 256 * 4 : 4];
   constexpr int XW = 3 * S + K;        // x cols for 4 wo outputs
   constexpr int TH = (S == 1 && K == 3) ? 4 : 2;  // rows per thread
   constexpr int RR = (TH - 1) * S + K; // input rows touched
@@ -65,9 +69,17 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   const int hi_ = min(lo + per, P);
 
   float s1[4] = {}, s2[4] = {};
+  float ea[4], eb[4];
   LopRegT<LOPK, ACT, 4> lr;
   if (c8 < C8 && mlane < ML) {
     lr.init(lopx, c);
+    if (EPI == 2) {
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        ea[e] = lopE.coef[c + e];
+        eb[e] = lopE.coef[C + c + e];
+      }
+    }
     for (int p = lo + mlane; p < hi_; p += ML) {
       const int w4 = p % Wo4;
       const int rest = p / Wo4;
@@ -134,20 +146,32 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
 #pragma unroll
       for (int t = 0; t < TH; ++t) {
         if (ho0 + t >= Ho) break;
-        ushort_t* yout = y + (int64_t(n) * Ho + ho0 + t) * int64_t(Wo) * C +
-                         int64_t(wo0) * C + c;
+        const int64_t obase =
+            (int64_t(n) * Ho + ho0 + t) * int64_t(Wo) * C +
+            int64_t(wo0) * C + c;
+        ushort_t* yout = y + obase;
 #pragma unroll
         for (int q = 0; q < 4; ++q) {
           if (wo0 + q >= Wo) break;
           short4_t out;
+          short4_t zq;
+          if (EPI == 2)
+            zq = *reinterpret_cast<const short4_t*>(&lopE.z[obase + q * C]);
 #pragma unroll
           for (int e = 0; e < 4; ++e) {
             const ushort_t o = f2bf(acc[t][q][e]);
             out[e] = short(o);
-            if (STATS) {
+            if (EPI == 1) {
               const float v = bf2f(o);
               s1[e] += v;
               s2[e] = fmaf(v, v, s2[e]);
+            } else if (EPI == 2) {
+              const float v = bf2f(o);
+              const float zv = bf2f(ushort_t(zq[e]));
+              const float da =
+                  v * act_bwd_rt(fmaf(zv, ea[e], eb[e]), lopE.act);
+              s1[e] += da;
+              s2[e] = fmaf(da, zv, s2[e]);
             }
           }
           *reinterpret_cast<short4_t*>(&yout[q * C]) = out;
@@ -155,7 +179,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
       }
     }
   }
-  if (STATS) {
+  if (EPI) {
 #pragma unroll
     for (int e = 0; e < 4; ++e) red[threadIdx.x * 4 + e] = s1[e];
     __syncthreads();
@@ -446,11 +470,12 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride);
   dim3 g(c_tiles * chunks);
   // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
+  Lop LE{};
 #define DWF_CASE(KK, SS, ST, LK, AC)                                       \
-  if (k == KK && stride == SS && (stats_ws != nullptr) == ST &&            \
+  if (k == KK && stride == SS && (stats_ws != nullptr) == (ST != 0) &&     \
       lopx.kind == LK && (LK == 0 || lopx.act == AC)) {                    \
     dw_fwd_kernel<KK, SS, false, ST, LK, AC><<<g, 256, 0, s>>>(            \
-        xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, stats_ws, chunks);         \
+        xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, LE, stats_ws, chunks);     \
     return;                                                                \
   }
 #define DWF_ST(KK, SS, ST)                                                 \
@@ -465,11 +490,13 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
 
 void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
                    int W, int C, int Ho, int Wo, int k, int stride, int pad,
-                   LopDesc lopdy, hipStream_t s) {
+                   LopDesc lopdy, LopDesc epi, float* stats_ws,
+                   hipStream_t s) {
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* xp = static_cast<ushort_t*>(dx);
   const Lop L = to_lop(lopdy);
+  const Lop LE = to_lop(epi);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
     const int C8 = C >> 2;
@@ -477,11 +504,17 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     const int c_tiles = (C8 + CT8 - 1) / CT8;
     const int chunks = dw_stats_chunks(N, H, W, C, 1);
     dim3 g(c_tiles * chunks);
-    // dgrad supports only lop none (dz is materialized by bn_act_bwd)
+    // dgrad supports only lop none (dz is materialized by bn_act_bwd);
+    // with stats_ws + epi it also emits the consumer BN's backward
+    // partial sums (EPI==2) so bn_bwd_partial is skipped
 #define DWD_CASE(KK)                                                       \
     if (k == KK && lopdy.kind == 0) {                                      \
-      dw_fwd_kernel<KK, 1, true, false, 0, 0><<<g, 256, 0, s>>>(           \
-          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, nullptr, chunks);        \
+      if (stats_ws != nullptr)                                             \
+        dw_fwd_kernel<KK, 1, true, 2, 0, 0><<<g, 256, 0, s>>>(             \
+            gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, stats_ws, chunks); \
+      else                                                                 \
+        dw_fwd_kernel<KK, 1, true, 0, 0, 0><<<g, 256, 0, s>>>(             \
+            gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, nullptr, chunks);  \
       return;                                                              \
     }
     DWD_CASE(3) DWD_CASE(5)

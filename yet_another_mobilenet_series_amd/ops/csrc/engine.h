@@ -29,9 +29,12 @@ struct LopDesc {
 // non-null, receives per-m-block partial column sums/sumsq of the OUTPUT
 // ([2, ceil(M/128), N] fp32) for BN-stats fusion; combine with
 // bn_finalize(splits = pw_stats_mblocks(M)).
+// epi: kind!=0 => stats_ws receives BN-BACKWARD partials (sum da, sum
+// da*z) of the consumer BN instead of forward sum/sumsq; epi carries the
+// z tensor, ab[2C] coefs and act (see gemm.hip EPI==2).
 void pw_conv(const void* x, const void* w, void* y,
-             int64_t M, int K, int N, LopDesc lopa, float* stats_ws,
-             hipStream_t s);
+             int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
+             float* stats_ws, hipStream_t s);
 int pw_stats_mblocks(int64_t M);
 
 // dw[N1,N2] = a[M,N1]^T @ b[M,N2], fp32 out, deterministic split-M slabs
@@ -56,7 +59,8 @@ void dw_conv_fwd(const void* x, const void* w, void* y,
 int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride);
 void dw_conv_dgrad(const void* dy, const void* w, void* dx,
                    int N, int H, int W, int C, int Ho, int Wo,
-                   int k, int stride, int pad, LopDesc lopdy, hipStream_t s);
+                   int k, int stride, int pad, LopDesc lopdy, LopDesc epi,
+                   float* stats_ws, hipStream_t s);
 // dwgt fp32 [k*k, C]; ws fp32 [chunks, k*k, C]
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo,
@@ -98,7 +102,7 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 const float* invstd, const float* gamma, const float* beta,
                 int act, bool training, float* dgamma, float* dbeta,
                 void* dz, float* ws, float* coef, int64_t M, int C,
-                bool apply, hipStream_t s);
+                bool apply, int ext_splits, hipStream_t s);
 
 // --- small fused elementwise/reduction kernels --------------------------
 // y = act(z + bias) on [M, C] (bias nullable)

@@ -39,10 +39,16 @@ inline Lop to_lop(const LopDesc& d) {
   return L;
 }
 
-template <bool STATS, int LOPK>
+// EPI: output epilogue — 0 none, 1 forward BN stats (sum, sumsq of the
+// bf16-rounded output per column), 2 BN-BACKWARD stats: the output is a
+// dgrad dy feeding a BN backward, and the epilogue computes that BN's
+// partial sums (sum da, sum da*z with da = dy * act'(a*z+b)) so the
+// separate bn_bwd_partial pass over dy+z never runs.  lopE carries the
+// z tensor, the ab[2C] coefs and the act for EPI==2.
+template <int EPI, int LOPK>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
-    ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA,
+    ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA, Lop lopE,
     float* __restrict__ stats_ws) {
   // double-buffered register staging (guide T14: issue the next tile's
   // global loads before the current tile's MFMAs, write LDS after the
@@ -167,12 +173,19 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
     }
   }
 
-  if (STATS) {
-    // per-block partial column sums/sumsq of the (bf16-rounded) output.
+  if (EPI) {
+    // per-block partial column sums of the (bf16-rounded) output.
     float* red = reinterpret_cast<float*>(Alds);  // 512 floats
     float s1[FRAG_N] = {}, s2[FRAG_N] = {};
 #pragma unroll
-    for (int j = 0; j < FRAG_N; ++j)
+    for (int j = 0; j < FRAG_N; ++j) {
+      const int col = bn + wcol0 + j * 16 + col_l;
+      float ea = 0.f, eb = 0.f;
+      if (EPI == 2) {
+        if (col >= N) continue;  // s1/s2 stay 0; red slot still written
+        ea = lopE.coef[col];
+        eb = lopE.coef[N + col];
+      }
 #pragma unroll
       for (int i = 0; i < FRAG_M; ++i)
 #pragma unroll
@@ -180,10 +193,18 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
           const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
           if (row < M) {
             const float v = bf2f(f2bf(acc[i][j][r]));
-            s1[j] += v;
-            s2[j] = fmaf(v, v, s2[j]);
+            if (EPI == 1) {
+              s1[j] += v;
+              s2[j] = fmaf(v, v, s2[j]);
+            } else {
+              const float zv = bf2f(lopE.z[row * N + col]);
+              const float da = v * act_bwd_rt(fmaf(zv, ea, eb), lopE.act);
+              s1[j] += da;
+              s2[j] = fmaf(da, zv, s2[j]);
+            }
           }
         }
+    }
     const int slot = (wave >> 1) * 4 + (lane >> 4);  // 0..7
     const int mblocks = gridDim.x;
     __syncthreads();
@@ -216,21 +237,24 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 int pw_stats_mblocks(int64_t M) { return ceil_div(M, BM); }
 
 void pw_conv(const void* x, const void* w, void* y,
-             int64_t M, int K, int N, LopDesc lopa, float* stats_ws,
-             hipStream_t s) {
+             int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
+             float* stats_ws, hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   const Lop L = to_lop(lopa);
+  const Lop LE = to_lop(epi);
+  const int ep = stats_ws == nullptr ? 0 : (epi.kind == LOP_BNACT ? 2 : 1);
   auto* ap = static_cast<const ushort_t*>(x);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
-#define PW_CASE(ST, LK)                                                   \
-  if ((stats_ws != nullptr) == ST && lopa.kind == LK) {                   \
-    pw_conv_kernel<ST, LK><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L,   \
-                                                stats_ws);                \
+#define PW_CASE(EP, LK)                                                   \
+  if (ep == EP && lopa.kind == LK) {                                      \
+    pw_conv_kernel<EP, LK><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L,   \
+                                                LE, stats_ws);            \
     return;                                                               \
   }
-  PW_CASE(false, 0) PW_CASE(false, 1) PW_CASE(false, 2)
-  PW_CASE(true, 0) PW_CASE(true, 1) PW_CASE(true, 2)
+  PW_CASE(0, 0) PW_CASE(0, 1) PW_CASE(0, 2)
+  PW_CASE(1, 0) PW_CASE(1, 1) PW_CASE(1, 2)
+  PW_CASE(2, 0)
 #undef PW_CASE
 }
 

@@ -199,7 +199,17 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
                       .contiguous())
         d_in3 = torch.empty_like(z2)
-        e.pw_conv(dz3, pwt, d_in3)
+        if not has_se:
+            # d_in3 IS the dw-BN's backward dy: emit that BN's partial sums
+            # from the GEMM epilogue (EPI==2) so bn_bwd_partial never runs
+            m2 = n * ho * wo
+            sp2 = int(e.pw_stats_mblocks(m2))
+            ws2 = torch.empty(2 * sp2 * cd, dtype=F32, device=dev)
+            e.pw_conv(dz3, pwt, d_in3, stats_ws=ws2, bstats_act=act,
+                      bstats_coef=ab2, bstats_z=z2)
+        else:
+            ws2 = sp2 = None
+            e.pw_conv(dz3, pwt, d_in3)
         dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
         if has_se:
             e.pw_conv_wgrad(dz3, y2s, dwp3)
@@ -245,13 +255,27 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         dg2 = torch.empty(cd, dtype=F32, device=dev)
         db2_ = torch.empty_like(dg2)
         dz2 = torch.empty_like(z2)
-        e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
-                     bool(training), dg2, db2_, dz2)
+        if ws2 is not None:
+            e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
+                         bool(training), dg2, db2_, dz2,
+                         partial_ws=ws2, partial_splits=sp2)
+        else:
+            e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
+                         bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
                        .contiguous())
         dw_in = y1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
-        e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
+        if has_expand and stride == 1:
+            # d_z1out feeds the expand BN's backward: fuse its partials
+            sp1 = int(e.dw_stats_chunks(n, h, wdt, ce, 1))
+            ws1 = torch.empty(2 * sp1 * ce, dtype=F32, device=dev)
+            e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad,
+                            stats_ws=ws1, bstats_act=act, bstats_coef=ab1,
+                            bstats_z=z1)
+        else:
+            ws1 = sp1 = None
+            e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
         dwgt = torch.empty(k * k, cd, dtype=F32, device=dev)
         e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
         d_dww = dwgt.t().reshape(cd, 1, k, k)
@@ -261,8 +285,13 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             dg1 = torch.empty(ce, dtype=F32, device=dev)
             db1_ = torch.empty_like(dg1)
             dz1 = torch.empty_like(z1)
-            e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
-                         bool(training), dg1, db1_, dz1)
+            if ws1 is not None:
+                e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                             bool(training), dg1, db1_, dz1,
+                             partial_ws=ws1, partial_splits=sp1)
+            else:
+                e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                             bool(training), dg1, db1_, dz1)
             ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
                           .contiguous())
             d_x = _empty_like_nhwc(x)
