@@ -304,6 +304,31 @@ void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
                    true, int(partial_splits), cur_stream());
 }
 
+// SE path: bn backward with the avg-pool grad added to dy in the
+// partial pass (replaces avgpool_bwd_add + bn_act_bwd)
+void bn_act_bwd_addpool(Tensor& dy, const Tensor& z, const Tensor& dpooled,
+                        const Tensor& mean, const Tensor& invstd,
+                        const Tensor& gamma, const Tensor& beta,
+                        int64_t act, bool training, Tensor& dgamma,
+                        Tensor& dbeta, Tensor& dz) {
+  auto zr = rows_of(z, "z");
+  auto gr = rows_of(dy, "dy");
+  TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
+  TORCH_CHECK(dpooled.scalar_type() == torch::kBFloat16 &&
+                  dpooled.numel() == int64_t(dy.size(0)) * zr.C,
+              "dpooled must be bf16 [N, C]");
+  const int HW = int(zr.M / dy.size(0));
+  auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+  auto coef = f32_ws(z, 4 * zr.C);
+  yams::bn_act_bwd_addpool(
+      const_cast<void*>(gr.ptr), zr.ptr, dpooled.data_ptr(), HW,
+      fp32_req(mean, "mean"), fp32_req(invstd, "invstd"),
+      fp32_req(gamma, "gamma"), fp32_req(beta, "beta"), int(act), training,
+      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), dz.data_ptr(),
+      ws.data_ptr<float>(), coef.data_ptr<float>(), zr.M, zr.C,
+      cur_stream());
+}
+
 // reduce-only backward: dgamma/dbeta + the coef[4C] used by LOP_BNBWD
 // consumers (no dz materialization)
 void bn_bwd_stats(const Tensor& dy, const Tensor& z, const Tensor& mean,
@@ -557,6 +582,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gamma"_a, "beta"_a, "act"_a, "training"_a, "dgamma"_a, "dbeta"_a,
         "dz"_a, "partial_ws"_a = c10::nullopt, "partial_splits"_a = 0);
   m.def("bn_bwd_stats", &bn_bwd_stats);
+  m.def("bn_act_bwd_addpool", &bn_act_bwd_addpool);
   m.def("bias_act_fwd", &bias_act_fwd);
   m.def("bias_act_bwd", &bias_act_bwd);
   m.def("rowscale_fwd", &rowscale_fwd, "x"_a, "scale"_a, "y"_a,

@@ -175,12 +175,18 @@ __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
 }
 
 // --- backward reduce: vec8 channels, per-chunk sum(da), sum(da*z) -------
-template <int ACT>
+// ADDPOOL: dy is the SE rowscale grad and the global-avg-pool backward
+// contribution (dpooled/HW per image) still has to be added; doing the
+// add HERE (writing the sum back to dy) removes the separate
+// avgpool_bwd_add pass over the tensor.
+template <int ACT, bool ADDPOOL = false>
 __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
-    const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
+    ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
-    float* __restrict__ ws, int64_t M, int C, int chunks) {
+    float* __restrict__ ws, int64_t M, int C, int chunks,
+    const ushort_t* __restrict__ dpooled = nullptr, int HW = 1,
+    float inv_hw = 0.f) {
   __shared__ float red[256 * 8];
   const int C8 = C >> 3;
   const int CT8 = C8 < 256 ? C8 : 256;
@@ -204,6 +210,16 @@ __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
     for (int64_t m = lo + mlane; m < hi; m += ML) {
       short8_t z8 = *reinterpret_cast<const short8_t*>(&z[m * C + c]);
       short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[m * C + c]);
+      if (ADDPOOL) {
+        const int n = int(m / HW);
+        short8_t p8 = *reinterpret_cast<const short8_t*>(
+            &dpooled[int64_t(n) * C + c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          g8[e] = short(f2bf(bf2f(ushort_t(g8[e])) +
+                             bf2f(ushort_t(p8[e])) * inv_hw));
+        *reinterpret_cast<short8_t*>(&dy[m * C + c]) = g8;
+      }
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const float zv = bf2f(ushort_t(z8[e]));
@@ -416,9 +432,28 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 bool apply, int ext_splits, hipStream_t s) {
   // ext_splits > 0: the partial sums were already produced by the kernel
   // that WROTE dy (conv-dgrad epilogue, EPI==2) into ws as [2, ext_splits,
-  // C] — skip the bn_bwd_partial pass over dy+z entirely.
+  // C] — skip the bn_bwd_partial pass over dy+z entirely.  Very large
+  // split counts (pw mblocks at 56²+ can be >6k) would serialize the
+  // underfilled finalize, so pre-fold them with the parallel fold kernel.
   const RedGeom g = red_geom(M, C);
-  const int splits = ext_splits > 0 ? ext_splits : g.chunks;
+  int splits = ext_splits > 0 ? ext_splits : g.chunks;
+  if (ext_splits > 1024) {
+    constexpr int FOLDED = 256;
+    static float* fold_buf = nullptr;
+    static size_t fold_cap = 0;
+    const size_t need = 2LL * FOLDED * C;
+    if (need > fold_cap) {
+      if (fold_buf) (void)hipFree(fold_buf);
+      (void)hipMalloc(&fold_buf, need * sizeof(float));
+      fold_cap = need;
+    }
+    const int blocks = int(std::min<int64_t>((2LL * FOLDED * C + 255) / 256,
+                                             2048));
+    bn_fold_kernel<<<dim3(blocks), 256, 0, s>>>(ws, fold_buf, splits,
+                                                FOLDED, C);
+    ws = fold_buf;
+    splits = FOLDED;
+  }
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* zp = static_cast<const ushort_t*>(z);
   auto* dzp = static_cast<ushort_t*>(dz);
@@ -426,7 +461,8 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
   ACT_SWITCH(act, {
     if (ext_splits <= 0)
       bn_bwd_partial_kernel<kAct><<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
-          gp, zp, mean, invstd, gamma, beta, ws, M, C, g.chunks);
+          const_cast<ushort_t*>(gp), zp, mean, invstd, gamma, beta, ws, M,
+          C, g.chunks);
     if (training)
       bn_bwd_finalize_kernel<true><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
           ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
@@ -438,6 +474,38 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
     if (apply)
       bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
           gp, zp, coef, dzp, total8, C >> 3, C);
+  });
+}
+
+// bn backward where dy still needs the avg-pool grad added (SE blocks):
+// partial does add+reduce in one pass, then finalize + apply as usual.
+void bn_act_bwd_addpool(void* dy, const void* z, const void* dpooled,
+                        int HW, const float* mean, const float* invstd,
+                        const float* gamma, const float* beta, int act,
+                        bool training, float* dgamma, float* dbeta,
+                        void* dz, float* ws, float* coef, int64_t M, int C,
+                        hipStream_t s) {
+  const RedGeom g = red_geom(M, C);
+  auto* gp = static_cast<ushort_t*>(dy);
+  auto* zp = static_cast<const ushort_t*>(z);
+  auto* pp = static_cast<const ushort_t*>(dpooled);
+  auto* dzp = static_cast<ushort_t*>(dz);
+  const int64_t total8 = M * (C >> 3);
+  ACT_SWITCH(act, {
+    bn_bwd_partial_kernel<kAct, true>
+        <<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
+            gp, zp, mean, invstd, gamma, beta, ws, M, C, g.chunks, pp, HW,
+            1.f / float(HW));
+    if (training)
+      bn_bwd_finalize_kernel<true><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+          ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
+          g.chunks);
+    else
+      bn_bwd_finalize_kernel<false><<<dim3(ceil_div(C, 16)), 256, 0, s>>>(
+          ws, mean, invstd, gamma, beta, dgamma, dbeta, coef, M, C,
+          g.chunks);
+    bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
+        gp, zp, coef, dzp, total8, C >> 3, C);
   });
 }
 

@@ -168,8 +168,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
             } else if (EPI == 2) {
               const float v = bf2f(o);
               const float zv = bf2f(ushort_t(zq[e]));
-              const float da =
-                  v * act_bwd_rt(fmaf(zv, ea[e], eb[e]), lopE.act);
+              const float da = v * act_bwd<ACT>(fmaf(zv, ea[e], eb[e]));
               s1[e] += da;
               s2[e] = fmaf(da, zv, s2[e]);
             }
@@ -507,18 +506,23 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     // dgrad supports only lop none (dz is materialized by bn_act_bwd);
     // with stats_ws + epi it also emits the consumer BN's backward
     // partial sums (EPI==2) so bn_bwd_partial is skipped
+#define DWD_EPI(KK, AC)                                                    \
+    if (stats_ws != nullptr && k == KK && lopdy.kind == 0 &&               \
+        epi.act == AC) {                                                   \
+      dw_fwd_kernel<KK, 1, true, 2, 0, AC><<<g, 256, 0, s>>>(              \
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, stats_ws, chunks);   \
+      return;                                                              \
+    }
 #define DWD_CASE(KK)                                                       \
+    DWD_EPI(KK, 0) DWD_EPI(KK, 1) DWD_EPI(KK, 2) DWD_EPI(KK, 3)            \
     if (k == KK && lopdy.kind == 0) {                                      \
-      if (stats_ws != nullptr)                                             \
-        dw_fwd_kernel<KK, 1, true, 2, 0, 0><<<g, 256, 0, s>>>(             \
-            gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, stats_ws, chunks); \
-      else                                                                 \
-        dw_fwd_kernel<KK, 1, true, 0, 0, 0><<<g, 256, 0, s>>>(             \
-            gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, nullptr, chunks);  \
+      dw_fwd_kernel<KK, 1, true, 0, 0, 0><<<g, 256, 0, s>>>(               \
+          gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, nullptr, chunks);    \
       return;                                                              \
     }
     DWD_CASE(3) DWD_CASE(5)
 #undef DWD_CASE
+#undef DWD_EPI
     return;
   }
   const int64_t total8 = int64_t(N) * H * W * (C >> 3);

@@ -45,7 +45,7 @@ inline Lop to_lop(const LopDesc& d) {
 // partial sums (sum da, sum da*z with da = dy * act'(a*z+b)) so the
 // separate bn_bwd_partial pass over dy+z never runs.  lopE carries the
 // z tensor, the ab[2C] coefs and the act for EPI==2.
-template <int EPI, int LOPK>
+template <int EPI, int LOPK, int EACT = 0>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
     ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA, Lop lopE,
@@ -198,7 +198,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
               s2[j] = fmaf(v, v, s2[j]);
             } else {
               const float zv = bf2f(lopE.z[row * N + col]);
-              const float da = v * act_bwd_rt(fmaf(zv, ea, eb), lopE.act);
+              const float da = v * act_bwd<EACT>(fmaf(zv, ea, eb));
               s1[j] += da;
               s2[j] = fmaf(da, zv, s2[j]);
             }
@@ -252,9 +252,16 @@ void pw_conv(const void* x, const void* w, void* y,
                                                 LE, stats_ws);            \
     return;                                                               \
   }
+#define PW_EPI2(AC)                                                       \
+  if (ep == 2 && lopa.kind == 0 && epi.act == AC) {                       \
+    pw_conv_kernel<2, 0, AC><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L, \
+                                                  LE, stats_ws);          \
+    return;                                                               \
+  }
   PW_CASE(0, 0) PW_CASE(0, 1) PW_CASE(0, 2)
   PW_CASE(1, 0) PW_CASE(1, 1) PW_CASE(1, 2)
-  PW_CASE(2, 0)
+  PW_EPI2(0) PW_EPI2(1) PW_EPI2(2) PW_EPI2(3)
+#undef PW_EPI2
 #undef PW_CASE
 }
 

@@ -244,7 +244,6 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             dw1p = torch.empty(r, cd, dtype=F32, device=dev)
             e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, cd, 1, 1),
                             dw1p)
-            e.avgpool_bwd_add(dpooled, d_y2)
             d_sw1 = dw1p.reshape_as(sw1)
             d_sw2 = dw2p.reshape_as(sw2)
         else:
@@ -259,6 +258,10 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2,
                          partial_ws=ws2, partial_splits=sp2)
+        elif has_se:
+            # pool-grad add + partial reduce in one pass over d_y2
+            e.bn_act_bwd_addpool(d_y2, z2, dpooled, mean2, invstd2, dg, db,
+                                 act, bool(training), dg2, db2_, dz2)
         else:
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2)
