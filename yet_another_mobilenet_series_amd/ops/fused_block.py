@@ -21,6 +21,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 
 from .backend import ext
@@ -41,6 +43,13 @@ def _eval_ab(mean, invstd, gamma, beta, ab) -> None:
     a = gamma.float() * invstd
     ab[:gamma.numel()] = a
     ab[gamma.numel():] = beta.float() - mean * a
+
+
+# backward-fusion knobs (A/B measurement; all default ON):
+#   YAMS_BWD_EPI=0   disable conv-dgrad BN-partial epilogues (pw + dw)
+#   YAMS_BWD_POOL=0  disable the pool-add+partial fusion on SE blocks
+_EPI_ON = os.environ.get("YAMS_BWD_EPI", "1") != "0"
+_POOL_ON = os.environ.get("YAMS_BWD_POOL", "1") != "0"
 
 
 class FusedInvertedResidualFn(torch.autograd.Function):
@@ -199,7 +208,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
                       .contiguous())
         d_in3 = torch.empty_like(z2)
-        if not has_se:
+        if not has_se and _EPI_ON:
             # d_in3 IS the dw-BN's backward dy: emit that BN's partial sums
             # from the GEMM epilogue (EPI==2) so bn_bwd_partial never runs
             m2 = n * ho * wo
@@ -244,6 +253,8 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             dw1p = torch.empty(r, cd, dtype=F32, device=dev)
             e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, cd, 1, 1),
                             dw1p)
+            if not _POOL_ON:
+                e.avgpool_bwd_add(dpooled, d_y2)
             d_sw1 = dw1p.reshape_as(sw1)
             d_sw2 = dw2p.reshape_as(sw2)
         else:
@@ -258,7 +269,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2,
                          partial_ws=ws2, partial_splits=sp2)
-        elif has_se:
+        elif has_se and _POOL_ON:
             # pool-grad add + partial reduce in one pass over d_y2
             e.bn_act_bwd_addpool(d_y2, z2, dpooled, mean2, invstd2, dg, db,
                                  act, bool(training), dg2, db2_, dz2)
@@ -269,7 +280,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                        .contiguous())
         dw_in = y1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
-        if has_expand and stride == 1:
+        if has_expand and stride == 1 and _EPI_ON:
             # d_z1out feeds the expand BN's backward: fuse its partials
             sp1 = int(e.dw_stats_chunks(n, h, wdt, ce, 1))
             ws1 = torch.empty(2 * sp1 * ce, dtype=F32, device=dev)
