@@ -45,10 +45,16 @@ def _eval_ab(mean, invstd, gamma, beta, ab) -> None:
     ab[gamma.numel():] = beta.float() - mean * a
 
 
-# backward-fusion knobs (A/B measurement; all default ON):
-#   YAMS_BWD_EPI=0   disable conv-dgrad BN-partial epilogues (pw + dw)
-#   YAMS_BWD_POOL=0  disable the pool-add+partial fusion on SE blocks
-_EPI_ON = os.environ.get("YAMS_BWD_EPI", "1") != "0"
+# backward-fusion knobs (A/B measured on MI355X, V3-Large B256 bf16):
+#   YAMS_BWD_EPI=1   emit BN-backward partials from the conv-dgrad
+#                    epilogues (pw EPI==2 / dw FLIP EPI==2) instead of the
+#                    separate bn_bwd_partial pass.  Measured NET-NEGATIVE
+#                    (+0.9 ms/step): the scalar z loads + per-frag
+#                    reductions slow the stream-optimal producer kernels
+#                    more than the removed partial pass costs.  Default off.
+#   YAMS_BWD_POOL=0  disable the SE pool-add+partial fusion (default on;
+#                    saves a full pass over d_y2 and one launch per block).
+_EPI_ON = os.environ.get("YAMS_BWD_EPI", "0") != "0"
 _POOL_ON = os.environ.get("YAMS_BWD_POOL", "1") != "0"
 
 
