@@ -565,7 +565,7 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
 #define DWW_CASE(KK, SS, LXX, AC)                                          \
   if (k == KK && stride == SS && lopx.kind == LXX &&                       \
       (LXX == 0 || lopx.act == AC) && lopdy.kind == 0) {                   \
-    constexpr int IP = (KK == 5) ? 2 : 1;                                  \
+    constexpr int IP = (KK == 5) ? (SS == 2 ? 3 : 2) : 1;                  \
     dim3 grid(c_tiles * chunks * IP);                                      \
     dw_wgrad_kernel<KK, SS, LXX, AC, 0, IP><<<grid, 256, 0, s>>>(          \
         xp, gp, ws, N, H, W, C, Ho, Wo, pad, Lx, Ldy, chunks);             \
