@@ -317,87 +317,64 @@ __global__ __launch_bounds__(256) void atb_kernel(
   constexpr int TPR1 = W1 / 8;  // threads per m-row (vec8 along n)
   constexpr int TPR2 = W2 / 8;
 
-  // register-staged double buffer: issue pass t+1's global loads before
-  // pass t's MFMAs (same T14 pattern as pw_conv_kernel)
-  const int s_m0a = tid / TPR1, s_na = (tid % TPR1) * 8;
-  const int s_m0b = tid / TPR2, s_nb = (tid % TPR2) * 8;
-  constexpr int RA = KT / (256 / TPR1);  // staging rows per thread (A)
-  constexpr int RB = KT / (256 / TPR2);
-  ushort_t rega[RA][8];
-  ushort_t regb[RB][8];
-
-  auto stage_regs = [&](int64_t m0) {
-#pragma unroll
-    for (int ri = 0; ri < RA; ++ri) {
-      const int r = s_m0a + ri * (256 / TPR1);
-      const int64_t gm = m0 + r;
-      const int gn = n1_0 + s_na;
-      if (gm < m_hi && gn + 8 <= N1) {
-        lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, rega[ri]);
-      } else if (gm >= m_hi || gn >= N1) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) rega[ri][j] = ushort_t(0);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          rega[ri][j] = (gn + j < N1)
-                            ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
-                                                          gm * N1 + gn + j,
-                                                          gn + j)))
-                            : ushort_t(0);
-      }
-    }
-#pragma unroll
-    for (int ri = 0; ri < RB; ++ri) {
-      const int r = s_m0b + ri * (256 / TPR2);
-      const int64_t gm = m0 + r;
-      const int gn = n2_0 + s_nb;
-      if (gm < m_hi && gn + 8 <= N2) {
-        lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, regb[ri]);
-      } else if (gm >= m_hi || gn >= N2) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) regb[ri][j] = ushort_t(0);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          regb[ri][j] = (gn + j < N2)
-                            ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
-                                                          gm * N2 + gn + j,
-                                                          gn + j)))
-                            : ushort_t(0);
-      }
-    }
-  };
-
-  auto write_lds = [&]() {
-#pragma unroll
-    for (int ri = 0; ri < RA; ++ri) {
-      const int r = s_m0a + ri * (256 / TPR1);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int n = s_na + j;
-        Al[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = rega[ri][j];
-      }
-    }
-#pragma unroll
-    for (int ri = 0; ri < RB; ++ri) {
-      const int r = s_m0b + ri * (256 / TPR2);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int n = s_nb + j;
-        Bl[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = regb[ri][j];
-      }
-    }
-  };
-
-  const int64_t npass = (m_hi - m_lo + KT - 1) / KT;
-  if (npass > 0) {
-    stage_regs(m_lo);
-    write_lds();
+  for (int64_t m0 = m_lo; m0 < m_hi; m0 += KT) {
     __syncthreads();
-  }
-  for (int64_t t = 0; t < npass; ++t) {
-    if (t + 1 < npass) stage_regs(m_lo + (t + 1) * KT);
+    {
+      const int s_m0 = tid / TPR1, s_n = (tid % TPR1) * 8;
+      ushort_t tmp[8];
+      for (int r = s_m0; r < KT; r += 256 / TPR1) {
+        const int64_t gm = m0 + r;
+        const int gn = n1_0 + s_n;
+        if (gm < m_hi && gn + 8 <= N1) {
+          lop_vec8<LA>(lopA, A, gm * N1 + gn, gn, tmp);
+        } else if (gm >= m_hi || gn >= N1) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) tmp[j] = ushort_t(0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tmp[j] = (gm < m_hi && gn + j < N1)
+                         ? ushort_t(f2bf(lop_loadT<LA>(lopA, A,
+                                                       gm * N1 + gn + j,
+                                                       gn + j)))
+                         : ushort_t(0);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int n = s_n + j;
+          Al[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+        }
+      }
+    }
+    {
+      const int s_m0 = tid / TPR2, s_n = (tid % TPR2) * 8;
+      ushort_t tmp[8];
+      for (int r = s_m0; r < KT; r += 256 / TPR2) {
+        const int64_t gm = m0 + r;
+        const int gn = n2_0 + s_n;
+        if (gm < m_hi && gn + 8 <= N2) {
+          lop_vec8<LB>(lopB, B, gm * N2 + gn, gn, tmp);
+        } else if (gm >= m_hi || gn >= N2) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) tmp[j] = ushort_t(0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            tmp[j] = (gm < m_hi && gn + j < N2)
+                         ? ushort_t(f2bf(lop_loadT<LB>(lopB, B,
+                                                       gm * N2 + gn + j,
+                                                       gn + j)))
+                         : ushort_t(0);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int n = s_n + j;
+          Bl[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+        }
+      }
+    }
+    __syncthreads();
+
 #pragma unroll
     for (int ks = 0; ks < KT / 32; ++ks) {
       bf16x8 a_frag[F1], b_frag[F2];
@@ -419,11 +396,6 @@ __global__ __launch_bounds__(256) void atb_kernel(
         for (int j = 0; j < F2; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
-    }
-    if (t + 1 < npass) {
-      __syncthreads();
-      write_lds();
-      __syncthreads();
     }
   }
 
