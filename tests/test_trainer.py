@@ -77,3 +77,13 @@ def test_trainer_determinism(tmp_path):
         return [m["loss"] for m in ms if m["kind"] == "train"]
 
     assert losses(tmp_path / "r1") == losses(tmp_path / "r2")
+
+
+def test_trainer_trace_flag(tmp_path):
+    """train.trace writes a chrome trace for the first steps (§5.1c)."""
+    cfg = _smoke_cfg(tmp_path, **{"train.trace": True, "train.trace_steps": 2,
+                                  "data.synthetic_size": 32})
+    Trainer(cfg).fit()
+    import glob
+    assert glob.glob(str(tmp_path / "run" / "trace_rank*.json")), \
+        "no chrome trace"

@@ -98,6 +98,21 @@ class Trainer:
         # interval when train.perf_breakdown is on
         breakdown = (getattr(cfg.train, "perf_breakdown", False)
                      and self.device.type == "cuda")
+        # optional per-kernel timeline — profile the
+        # first trace_steps steps of the first epoch into a chrome trace
+        prof = None
+        if getattr(cfg.train, "trace", False) and epoch == self.start_epoch:
+            acts = [torch.profiler.ProfilerActivity.CPU]
+            if self.device.type == "cuda":
+                acts.append(torch.profiler.ProfilerActivity.CUDA)
+            n_trace = int(getattr(cfg.train, "trace_steps", 3))
+            path = os.path.join(cfg.log.dir, f"trace_rank{self.rank}.json")
+            prof = torch.profiler.profile(
+                activities=acts,
+                schedule=torch.profiler.schedule(wait=1, warmup=1,
+                                                 active=n_trace),
+                on_trace_ready=lambda p: p.export_chrome_trace(path))
+            prof.__enter__()
         accum = max(int(getattr(cfg.train, "grad_accum_steps", 1)), 1)
         for it, (x, y) in enumerate(self.train_loader):
             x, y = self._to_device(x, y)
@@ -140,6 +155,13 @@ class Trainer:
                                      **{f"{n}_ms": round(v, 3)
                                         for n, v in ms.items()})
             self.global_step += 1
+            if prof is not None:
+                prof.step()
+                if it >= int(getattr(cfg.train, "trace_steps", 3)) + 2:
+                    prof.__exit__(None, None, None)
+                    prof = None
+                    self.logger.info("chrome trace written to %s",
+                                     cfg.log.dir)
 
             with torch.no_grad():
                 top1, top5 = accuracy(logits.detach(), y, (1, 5))
@@ -158,6 +180,9 @@ class Trainer:
                 self.jsonl.write(kind="train", epoch=epoch, step=self.global_step,
                                  loss=meters["loss"].avg, top1=meters["top1"].avg,
                                  lr=lr, img_s=meters["img_s"].val)
+
+        if prof is not None:
+            prof.__exit__(None, None, None)
 
     @torch.no_grad()
     def validate(self, epoch: int, use_ema: Optional[bool] = None) -> float:

@@ -139,6 +139,7 @@ DEFAULTS: Dict[str, Any] = {
               "resume": "", "pretrained": "",
               "save_interval_epochs": 1, "channels_last": True,
               "deterministic": False, "perf_breakdown": False,
+              "trace": False, "trace_steps": 3,
               "engine": "auto", "use_cuda_graph": True},
     "dist": {"backend": "auto", "bucket_mb": 4, "overlap": True,
              "grad_dtype": "fp32", "timeout_sec": 300},
