@@ -111,3 +111,34 @@ def test_fused_block_determinism():
     a, b = run(), run()
     for t1, t2 in zip(a, b):
         assert torch.equal(t1, t2)
+
+
+@pytest.mark.parametrize("cfg", [CFGS[0], CFGS[1], CFGS[2]])
+def test_fused_block_parity_epi_fusion(cfg, monkeypatch):
+    """The (default-off) dgrad BN-partial epilogue path must stay exact:
+    same results as the default path for no-SE (pw EPI), SE (addpool) and
+    expand (dw-dgrad EPI) layouts."""
+    from yet_another_mobilenet_series_amd.ops import fused_block as fb
+    torch.manual_seed(3)
+    dev = torch.device("cuda:0")
+    block = InvertedResidual(**cfg).to(dev).to(
+        memory_format=torch.channels_last).bfloat16().train()
+    x = (torch.randn(8, cfg["cin"], 28, 28, device=dev).bfloat16()
+         .contiguous(memory_format=torch.channels_last))
+
+    def run():
+        for p in block.parameters():
+            p.grad = None
+        xi = x.clone().requires_grad_(True)
+        y = block(xi)
+        y.float().square().mean().backward()
+        return ([y.detach().float()] + [xi.grad.float()] +
+                [p.grad.float().clone() for p in block.parameters()
+                 if p.grad is not None])
+
+    monkeypatch.setattr(fb, "_EPI_ON", False)
+    base = run()
+    monkeypatch.setattr(fb, "_EPI_ON", True)
+    fused = run()
+    for a, b in zip(base, fused):
+        assert _rel(a, b) < 2e-2, _rel(a, b)
