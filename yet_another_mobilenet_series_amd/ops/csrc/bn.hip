@@ -150,9 +150,10 @@ __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
     const ushort_t* __restrict__ z, const float* __restrict__ ab,
     ushort_t* __restrict__ y, const ushort_t* __restrict__ res,
     int64_t total8, int C8, int C) {
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c = int(idx % C8) * 8;
+  const int tot = int(total8);  // < 2^31 for every supported shape
+  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
+       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
+    const int c = (idx % C8) * 8;
     const int64_t off = idx * 8;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
     float4_t a0 = *reinterpret_cast<const float4_t*>(&ab[c]);
@@ -307,9 +308,10 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
     const float* __restrict__ coef, ushort_t* __restrict__ dz,
     int64_t total8, int C8, int C) {
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c = int(idx % C8) * 8;
+  const int tot = int(total8);  // < 2^31 for every supported shape
+  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
+       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
+    const int c = (idx % C8) * 8;
     const int64_t off = idx * 8;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
     short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);

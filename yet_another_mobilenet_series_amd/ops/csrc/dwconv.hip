@@ -213,46 +213,56 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ dx, int N, int H, int W, int C, int Ho, int Wo,
     int stride, int pad, Lop lopdy, int64_t total8) {
+  // stride==2 only.  Tap parity is fixed by (hi+pad)&1 / (wi+pad)&1, so
+  // iterate ONLY the parity-valid taps (no per-tap %), and mask range
+  // edges with clamped in-bounds loads + value select (a per-tap branch
+  // around the load serializes on vmcnt).  Index decode is 32-bit.
   const int C8 = C >> 3;
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c8 = int(idx % C8);
-    int64_t rest = idx / C8;
-    const int wi = int(rest % W);
+  const int tot = int(total8);
+  for (int idx = int(blockIdx.x) * 256 + int(threadIdx.x); idx < tot;
+       idx += int(gridDim.x) * 256) {
+    const int c8 = idx % C8;
+    int rest = idx / C8;
+    const int wi = rest % W;
     rest /= W;
-    const int hi = int(rest % H);
-    const int n = int(rest / H);
+    const int hi = rest % H;
+    const int n = rest / H;
     const int c = c8 * 8;
     LopRegT<LOPK, ACT, 8> lr;
     lr.init(lopdy, c);
     float acc[8] = {};
+    const int ip0 = (hi + pad) & 1;
+    const int jp0 = (wi + pad) & 1;
 #pragma unroll
-    for (int i = 0; i < K; ++i) {
-      const int hnum = hi + pad - i;
-      if (hnum < 0 || hnum % stride) continue;
-      const int ho = hnum / stride;
-      if (ho >= Ho) continue;
+    for (int ii = 0; ii < (K + 1) / 2; ++ii) {
+      const int i = ip0 + ii * 2;
+      if (i >= K) break;
+      const int ho = (hi + pad - i) >> 1;      // exact by parity
+      const bool hok = ho >= 0 && ho < Ho;
+      const int hoc = ho < 0 ? 0 : (ho >= Ho ? Ho - 1 : ho);
 #pragma unroll
-      for (int j = 0; j < K; ++j) {
-        const int wnum = wi + pad - j;
-        if (wnum < 0 || wnum % stride) continue;
-        const int wo = wnum / stride;
-        if (wo >= Wo) continue;
-        const int64_t off = idx4(n, ho, wo, c, Ho, Wo, C);
+      for (int jj = 0; jj < (K + 1) / 2; ++jj) {
+        const int j = jp0 + jj * 2;
+        if (j >= K) break;
+        const int wo = (wi + pad - j) >> 1;
+        const bool ok = hok && wo >= 0 && wo < Wo;
+        const int woc = wo < 0 ? 0 : (wo >= Wo ? Wo - 1 : wo);
+        const int64_t off = idx4(n, hoc, woc, c, Ho, Wo, C);
         short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
         short8_t wv = *reinterpret_cast<const short8_t*>(
             &w[(i * K + j) * C + c]);
         if (LOPK == LOP_NONE) {
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(bf2f(ushort_t(gv[e])), bf2f(ushort_t(wv[e])),
-                          acc[e]);
+            acc[e] = fmaf(ok ? bf2f(ushort_t(gv[e])) : 0.f,
+                          bf2f(ushort_t(wv[e])), acc[e]);
         } else {
           short8_t zv = *reinterpret_cast<const short8_t*>(&lopdy.z[off]);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(lr.apply(bf2f(ushort_t(gv[e])),
-                                   bf2f(ushort_t(zv[e])), e),
+            acc[e] = fmaf(ok ? lr.apply(bf2f(ushort_t(gv[e])),
+                                        bf2f(ushort_t(zv[e])), e)
+                             : 0.f,
                           bf2f(ushort_t(wv[e])), acc[e]);
         }
       }

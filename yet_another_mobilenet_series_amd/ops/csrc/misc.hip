@@ -51,9 +51,10 @@ template <int ACT>
 __global__ __launch_bounds__(256) void bias_act_fwd_kernel(
     const ushort_t* __restrict__ z, const float* __restrict__ bias,
     ushort_t* __restrict__ y, int64_t total8, int C8) {
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c = int(idx % C8) * 8;
+  const int tot = int(total8);  // < 2^31 for every supported shape
+  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
+       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
+    const int c = (idx % C8) * 8;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[idx * 8]);
     short8_t out;
 #pragma unroll
@@ -98,10 +99,11 @@ template <int LOPK>
 __global__ __launch_bounds__(256) void rowscale_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ scale,
     ushort_t* __restrict__ y, int64_t total8, int HW, int C8, Lop lopx) {
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c8 = int(idx % C8);
-    const int n = int(idx / (int64_t(C8) * HW));
+  const int tot = int(total8);  // < 2^31 for every supported shape
+  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
+       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
+    const int c8 = idx % C8;
+    const int n = idx / (C8 * HW);
     float xv[8];
     load8_lop<LOPK>(lopx, x, idx * 8, c8 * 8, xv);
     short8_t sv = *reinterpret_cast<const short8_t*>(
@@ -219,10 +221,11 @@ template <bool ACCUM>
 __global__ __launch_bounds__(256) void avgpool_bwd_kernel(
     const ushort_t* __restrict__ dpooled, ushort_t* __restrict__ dx,
     int64_t total8, int HW, int C8, float inv_hw) {
-  for (int64_t idx = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
-       idx < total8; idx += int64_t(gridDim.x) * blockDim.x) {
-    const int c8 = int(idx % C8);
-    const int n = int(idx / (int64_t(C8) * HW));
+  const int tot = int(total8);  // < 2^31 for every supported shape
+  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
+       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
+    const int c8 = idx % C8;
+    const int n = idx / (C8 * HW);
     short8_t gv = *reinterpret_cast<const short8_t*>(
         &dpooled[(int64_t(n) * C8 + c8) * 8]);
     short8_t out;
