@@ -140,5 +140,11 @@ def test_fused_block_parity_epi_fusion(cfg, monkeypatch):
     base = run()
     monkeypatch.setattr(fb, "_EPI_ON", True)
     fused = run()
+    # relu-family homogeneity makes some BN-gamma grads ~null; a different
+    # (but still deterministic) summation order then has huge RELATIVE
+    # error on a ~zero tensor — compare against the overall grad scale
+    maxn = max(b.norm().item() for b in base)
     for a, b in zip(base, fused):
-        assert _rel(a, b) < 2e-2, _rel(a, b)
+        err = (a - b).norm().item()
+        assert err < 2e-2 * max(b.norm().item(), 1e-3 * maxn), \
+            (err, b.norm().item(), maxn)
