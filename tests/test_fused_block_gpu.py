@@ -122,7 +122,7 @@ def test_fused_block_parity_epi_fusion(cfg, monkeypatch):
     torch.manual_seed(3)
     dev = torch.device("cuda:0")
     block = InvertedResidual(**cfg).to(dev).to(
-        memory_format=torch.channels_last).bfloat16().train()
+        memory_format=torch.channels_last).train()
     x = (torch.randn(8, cfg["cin"], 28, 28, device=dev).bfloat16()
          .contiguous(memory_format=torch.channels_last))
 
