@@ -270,6 +270,7 @@ void bn_stats(const Tensor& z, Tensor& mean, Tensor& invstd, Tensor& ab,
 void bn_act_fwd(const Tensor& z, const Tensor& ab, int64_t act, Tensor& y,
                 c10::optional<Tensor> residual) {
   auto zr = rows_of(z, "z");
+  TORCH_CHECK(zr.C % 8 == 0, "C must be a multiple of 8");
   const void* res = nullptr;
   if (residual.has_value()) res = rows_of(*residual, "residual").ptr;
   yams::bn_act_fwd(zr.ptr, fp32_req(ab, "ab"), int(act), y.data_ptr(), res,
@@ -282,6 +283,7 @@ void bn_act_bwd(const Tensor& dy, const Tensor& z, const Tensor& mean,
                 Tensor& dgamma, Tensor& dbeta, Tensor& dz,
                 c10::optional<Tensor> partial_ws, int64_t partial_splits) {
   auto zr = rows_of(z, "z");
+  TORCH_CHECK(zr.C % 8 == 0, "C must be a multiple of 8");
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
   auto coef = f32_ws(z, 4 * zr.C);
@@ -312,6 +314,7 @@ void bn_act_bwd_addpool(Tensor& dy, const Tensor& z, const Tensor& dpooled,
                         int64_t act, bool training, Tensor& dgamma,
                         Tensor& dbeta, Tensor& dz) {
   auto zr = rows_of(z, "z");
+  TORCH_CHECK(zr.C % 8 == 0, "C must be a multiple of 8");
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
   TORCH_CHECK(dpooled.scalar_type() == torch::kBFloat16 &&
@@ -336,6 +339,7 @@ void bn_bwd_stats(const Tensor& dy, const Tensor& z, const Tensor& mean,
                   const Tensor& beta, int64_t act, bool training,
                   Tensor& dgamma, Tensor& dbeta, Tensor& coef) {
   auto zr = rows_of(z, "z");
+  TORCH_CHECK(zr.C % 8 == 0, "C must be a multiple of 8");
   auto gr = rows_of(dy, "dy");
   TORCH_CHECK(zr.M == gr.M && zr.C == gr.C, "bn bwd shape mismatch");
   TORCH_CHECK(coef.numel() == 4 * zr.C, "coef must be [4C]");
