@@ -8,6 +8,7 @@
 // deterministically via per-(chunk,mlane) fp32 slabs + slab_reduce.
 
 #include <algorithm>
+#include <cstdlib>
 
 #include "common.h"
 #include "engine.h"
@@ -456,11 +457,18 @@ int grid_for(int64_t total) {
 
 int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride) {
   (void)stride;
+  // measurement knob (roadmap §3): YAMS_DW_CHUNKS_DIV divides the block
+  // target so the work-per-wave vs launch-count tradeoff can be swept
+  // without a rebuild.  Unset => default behavior.
+  static int div_ = [] {
+    const char* e = getenv("YAMS_DW_CHUNKS_DIV");
+    return e ? atoi(e) : 1;
+  }();
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
-  int64_t want = 2048 / c_tiles;
+  int64_t want = 2048 / (int64_t(c_tiles) * (div_ > 0 ? div_ : 1));
   int64_t maxc = P;
   int c = int(std::min(std::max<int64_t>(want, 1), maxc));
   return c < 1 ? 1 : c;
