@@ -468,7 +468,10 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride) {
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
-  int64_t want = 2048 / (int64_t(c_tiles) * (div_ > 0 ? div_ : 1));
+  // 512-block default: the chunks sweep (tools/probe_dw_sweep.py, profiles)
+  // measured div=4 off the old 2048 target 10-30% faster on most shapes —
+  // per-wave prologue + launch ramp dominate when waves do ~1 iteration
+  int64_t want = 512 / (int64_t(c_tiles) * (div_ > 0 ? div_ : 1));
   int64_t maxc = P;
   int c = int(std::min(std::max<int64_t>(want, 1), maxc));
   return c < 1 ? 1 : c;
@@ -557,7 +560,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
   const int64_t P = int64_t(N) * Ho * ((Wo + 3) / 4);
-  int c = int(std::min<int64_t>((P + 3) / 4, 4096));
+  int c = int(std::min<int64_t>((P + 3) / 4, 1024));  // see dw_stats_chunks
   return c < 1 ? 1 : c;
 }
 
