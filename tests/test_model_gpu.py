@@ -69,10 +69,16 @@ def test_model_fwd_bwd_parity(name):
     # eager) is — layer-local exactness is covered by
     # test_layer_local_backward_parity
     bad = []
+    maxnorm = max(g.norm().item() for g in g_ref.values() if g is not None)
     for n in g_ref:
         if g_ref[n] is None:
             continue
         assert g_hip[n] is not None, f"missing grad {n}"
+        if g_ref[n].norm().item() < 1e-3 * maxnorm:
+            # relu-family homogeneity makes some true grads ~null (e.g. a
+            # channel scale directly before a dw+BN); relative error on a
+            # ~zero tensor is uninformative at batch 4
+            continue
         e_hip = _rel_err(g_hip[n], g_ref[n])
         e_ac = _rel_err(g_ac[n], g_ref[n])
         if e_ac > 1.0:
