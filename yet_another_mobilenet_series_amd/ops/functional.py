@@ -142,7 +142,9 @@ def label_smooth_ce(logits: torch.Tensor, target: torch.Tensor,
     """
     if use_hip(logits):
         return hip_ops.label_smooth_ce(logits, target, smoothing)
-    logp = F.log_softmax(logits.float(), dim=1)
+    if logits.dtype not in (torch.float32, torch.float64):
+        logits = logits.float()  # upcast bf16; keep fp64 for gradcheck
+    logp = F.log_softmax(logits, dim=1)
     nll = -logp.gather(1, target.view(-1, 1)).squeeze(1)
     smooth = -logp.mean(dim=1)
     return ((1.0 - smoothing) * nll + smoothing * smooth).mean()
