@@ -1,20 +1,22 @@
 """Fused inverted-residual block (SURVEY.md §7 stage 4: block-level fusion).
 
 One autograd Function runs the whole block {pw expand -> BN+act -> dw ->
-BN+act -> [SE] -> pw project -> BN [+residual]} with:
+BN+act -> [SE] -> pw project -> BN [+residual]}.  The fusion set below is
+the one that MEASURED fastest on MI355X (each candidate was A/B'd per
+kernel and per step; losers are documented in profiles/README.md):
 
-- BN statistics computed in the conv epilogues (per-block partial slabs
-  combined by ``bn_finalize``) — no separate stats pass over z;
-- BN-normalize+activation applied as LOAD-OPS in each consumer kernel —
-  the activated tensors y1/y2 are never materialized (the only tensors
-  written per block are the raw conv outputs z1/z2/z3, the SE gate
-  path, and the block output);
-- backward computes dz on-the-fly as a BNBWD load-op inside the dgrad /
-  wgrad kernels — dz never materializes either.
-
-Traffic per conv drops from {z:1W+2R, y:1W+1R} to {z:1W+1R} forward and
-{dz:1W+2R} to zero backward: the fusion-depth lever SURVEY.md §6 says
-sets achievable img/s on this HBM-bound family.
+- BN statistics always ride the conv epilogues (per-block partial slabs
+  combined by ``bn_finalize``) — no separate stats pass over any z;
+- the no-SE projection conv applies the dw-BN+act as a LOAD-OP, so its
+  activated input y2 is never written; y1 (expand activation) IS
+  materialized — the depthwise window-load lop costs more registers
+  (occupancy) than this memory-bound elementwise pass costs time;
+- backward materializes dz per BN via the fused
+  partial→finalize→apply chain (``bn_act_bwd``); computing dz as a
+  BNBWD load-op inside the dgrad/wgrad GEMMs, or emitting BN-backward
+  partials from dgrad epilogues (``YAMS_BWD_EPI``), both measured
+  net-negative.  The SE pool-grad add IS fused into the BN partial pass
+  (``bn_act_bwd_addpool``) — one less full pass over d_y2.
 """
 
 from __future__ import annotations
