@@ -5,9 +5,12 @@ logical / ``torch.channels_last`` memory (== NHWC rows, C fastest), compute
 dtype bf16, accumulation and statistics fp32.  Weight tensors are re-packed
 per call into the kernel-friendly layouts:
 
-- pointwise 1x1 [Cout, Cin, 1, 1]   -> GEMM B operand [Cin, Cout] bf16
+- pointwise 1x1 [Cout, Cin, 1, 1]   -> GEMM B operand [Cout, Cin] bf16
+  (K-major: the MFMA GEMM reads both operands along the reduction axis,
+  so the natural [Cout, Cin] order needs no transpose staging)
 - depthwise  [C, 1, k, k]           -> [k*k, C] bf16 (C fastest)
-- stem dense [Cout, 3, 3, 3]        -> [3, 3, 3, Cout] bf16 (Cout fastest)
+- stem dense [Cout, 3, k, k]        -> [Cout, k*k*3 padded to /8] bf16,
+  consumed by im2row patches through the same MFMA GEMM
 
 Gradients w.r.t. parameters are produced in fp32 (master-weight dtype);
 gradients w.r.t. activations in bf16.  All reductions (BN stats, wgrad)
@@ -91,12 +94,9 @@ def _conv_forward(x: torch.Tensor, w_packed: torch.Tensor, kind: str,
         ho, wo = _conv_out_hw(h, wdt, k, s, p)
         z = _empty_like_nhwc(x, h=ho, w=wo)
         e.dw_conv_fwd(x, w_packed, z, k, s, p)
-    elif kind == "stem":
-        cout = w_packed.shape[3]
-        ho, wo = _conv_out_hw(h, wdt, k, s, p)
-        z = _empty_like_nhwc(x, c=cout, h=ho, w=wo)
-        e.stem_conv_fwd(x, w_packed, z, k, s, p)
     else:
+        # the stem never reaches here: it runs im2row + pw_conv in
+        # _ConvBNActFn.forward directly (needs the patches for wgrad)
         raise ValueError(kind)
     return z
 
