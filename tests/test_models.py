@@ -68,3 +68,27 @@ def test_se_blocks_present_only_where_expected():
     se_blocks = [b for b in m.features
                  if isinstance(b, models.InvertedResidual) and b.se is not None]
     assert len(se_blocks) == 8  # 3+2+3 SE stages (App. A.6)
+
+
+def test_custom_num_classes_and_width():
+    for name in ("mobilenet_v2", "mobilenet_v3_small"):
+        m = models.build_model(name, num_classes=10, width_mult=0.5)
+        x = torch.randn(2, 3, 64, 64)
+        y = m(x)
+        assert y.shape == (2, 10)
+
+
+def test_dropout_only_in_train_mode():
+    torch.manual_seed(0)
+    m = models.build_model("mobilenet_v3_large", dropout=0.9)
+    m.eval()
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        a, b = m(x), m(x)
+    assert torch.equal(a, b)  # dropout inactive in eval
+    m.train()
+    torch.manual_seed(1)
+    c = m(x)
+    torch.manual_seed(2)
+    d = m(x)
+    assert not torch.equal(c, d)  # active (and BN batch stats) in train
