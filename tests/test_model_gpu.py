@@ -79,6 +79,8 @@ def test_model_fwd_bwd_parity(name):
             # channel scale directly before a dw+BN); relative error on a
             # ~zero tensor is uninformative at batch 4
             continue
+        if (g_hip[n] - g_ref[n]).norm().item() < 1e-3 * maxnorm:
+            continue  # absolute deviation negligible at model-grad scale
         e_hip = _rel_err(g_hip[n], g_ref[n])
         e_ac = _rel_err(g_ac[n], g_ref[n])
         if e_ac > 1.0:
