@@ -55,8 +55,14 @@ class Trainer:
         self.steps_per_epoch = len(self.train_loader)
         self.criterion = CrossEntropyLabelSmooth(cfg.optim.label_smoothing)
         self.optimizer = get_optimizer(model, cfg.optim)
+        # the LR schedule advances once per OPTIMIZER step, which with
+        # gradient accumulation is len(loader)/accum times per epoch —
+        # scheduling on microbatches would stretch warmup/decay by the
+        # accumulation factor (ADVICE.md round-1 medium finding)
+        accum = max(int(getattr(cfg.train, "grad_accum_steps", 1)), 1)
+        opt_steps_per_epoch = max((self.steps_per_epoch + accum - 1) // accum, 1)
         self.lr_sched = LRScheduler(self.optimizer, cfg.optim,
-                                    self.steps_per_epoch, cfg.train.epochs)
+                                    opt_steps_per_epoch, cfg.train.epochs)
         self.ema = (ExponentialMovingAverage(model, cfg.ema.decay)
                     if cfg.ema.enabled else None)
         self.start_epoch = 0
@@ -114,11 +120,15 @@ class Trainer:
                 on_trace_ready=lambda p: p.export_chrome_trace(path))
             prof.__enter__()
         accum = max(int(getattr(cfg.train, "grad_accum_steps", 1)), 1)
+        n_iters = len(self.train_loader)
         for it, (x, y) in enumerate(self.train_loader):
             x, y = self._to_device(x, y)
             lr = self.lr_sched.step(self.global_step)
             micro = it % accum
-            boundary = micro == accum - 1
+            # the last microbatch of the epoch always steps, so a trailing
+            # partial accumulation window is applied rather than silently
+            # dropped (ADVICE.md round-1 low finding)
+            boundary = micro == accum - 1 or it == n_iters - 1
             if breakdown:
                 evs = [torch.cuda.Event(enable_timing=True)
                        for _ in range(5)]

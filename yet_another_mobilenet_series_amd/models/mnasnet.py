@@ -12,14 +12,14 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, make_divisible)
+                             LinearAct, ZooModel, make_divisible)
 
 # (expand_ratio t, out_ch c, repeats n, stride s, kernel k, se_ratio_of_cin)
 _CFG = [(6, 24, 2, 2, 3, 0.0), (3, 40, 3, 2, 5, 0.25), (6, 80, 4, 2, 3, 0.0),
         (6, 112, 2, 1, 3, 0.25), (6, 160, 3, 2, 5, 0.25), (6, 320, 1, 1, 3, 0.0)]
 
 
-class MNasNetA1(nn.Module):
+class MNasNetA1(ZooModel):
     def __init__(self, num_classes: int = 1000, width_mult: float = 1.0,
                  dropout: float = 0.0, bn_momentum: float = 0.1,
                  bn_epsilon: float = 1e-5, **_: object) -> None:
@@ -48,6 +48,7 @@ class MNasNetA1(nn.Module):
         self.classifier = LinearAct(1280, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self.bump_bn_counters()
         x = self.features(x)
         x = self.pool(x)
         x = self.dropout(x)

@@ -18,6 +18,35 @@ from ..ops import functional as OF
 from ..utils.common import make_divisible
 
 
+class ZooModel(nn.Module):
+    """Base class for zoo models: one batched BN-counter bump per forward.
+
+    BatchNorm's ``num_batches_tracked`` counters are bookkeeping, but
+    incrementing them per-module launched ~46 eager ``add<long>`` kernels
+    every training step (~0.2 ms of pure launch overhead on MI355X).  The
+    zoo models instead call :meth:`bump_bn_counters` once per forward —
+    a single ``torch._foreach_add_`` over every counter in the model.
+    """
+
+    def bump_bn_counters(self) -> None:
+        if not self.training:
+            return
+        counters = getattr(self, "_bn_counters", None)
+        if counters is None:
+            counters = [m.num_batches_tracked for m in self.modules()
+                        if isinstance(m, nn.BatchNorm2d)
+                        and m.track_running_stats
+                        and m.num_batches_tracked is not None]
+            self._bn_counters = counters
+        if counters:
+            torch._foreach_add_(counters, 1)
+
+    def _apply(self, fn, recurse=True):
+        # .to()/.cuda() replace buffer tensors; drop the cached list
+        self._bn_counters = None
+        return super()._apply(fn, recurse)
+
+
 class ConvBNAct(nn.Module):
     """conv2d → BatchNorm → activation, fused on the HIP path.
 
@@ -40,8 +69,8 @@ class ConvBNAct(nn.Module):
 
     def forward(self, x: torch.Tensor,
                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:
-        if self.training and self.bn.track_running_stats:
-            self.bn.num_batches_tracked += 1
+        # num_batches_tracked is bumped once per model forward by
+        # ZooModel.bump_bn_counters (batched), not per module
         return OF.conv_bn_act(
             x, self.conv.weight, self.bn.weight, self.bn.bias,
             self.bn.running_mean, self.bn.running_var,
@@ -121,5 +150,5 @@ class LinearAct(nn.Module):
         return OF.linear(x, self.fc.weight, self.fc.bias, self.act)
 
 
-__all__ = ["ConvBNAct", "SqueezeExcite", "InvertedResidual",
+__all__ = ["ZooModel", "ConvBNAct", "SqueezeExcite", "InvertedResidual",
            "GlobalAvgPool", "LinearAct", "make_divisible"]

@@ -9,7 +9,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from .mobilenet_base import ConvBNAct, GlobalAvgPool, LinearAct, make_divisible
+from .mobilenet_base import (ConvBNAct, GlobalAvgPool, LinearAct,
+                             ZooModel, make_divisible)
 
 # (out_channels, stride) for each depthwise-separable stage
 _CFG = [(64, 1), (128, 2), (128, 1), (256, 2), (256, 1), (512, 2),
@@ -17,7 +18,7 @@ _CFG = [(64, 1), (128, 2), (128, 1), (256, 2), (256, 1), (512, 2),
         (1024, 1)]
 
 
-class MobileNetV1(nn.Module):
+class MobileNetV1(ZooModel):
     def __init__(self, num_classes: int = 1000, width_mult: float = 1.0,
                  dropout: float = 0.0, bn_momentum: float = 0.1,
                  bn_epsilon: float = 1e-5, **_: object) -> None:
@@ -38,6 +39,7 @@ class MobileNetV1(nn.Module):
         self.classifier = LinearAct(cin, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self.bump_bn_counters()
         x = self.features(x)
         x = self.pool(x)
         x = self.dropout(x)

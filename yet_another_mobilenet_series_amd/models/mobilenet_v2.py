@@ -9,14 +9,14 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, make_divisible)
+                             LinearAct, ZooModel, make_divisible)
 
 # (expand_ratio t, out_channels c, repeats n, first stride s)
 _CFG = [(1, 16, 1, 1), (6, 24, 2, 2), (6, 32, 3, 2), (6, 64, 4, 2),
         (6, 96, 3, 1), (6, 160, 3, 2), (6, 320, 1, 1)]
 
 
-class MobileNetV2(nn.Module):
+class MobileNetV2(ZooModel):
     def __init__(self, num_classes: int = 1000, width_mult: float = 1.0,
                  dropout: float = 0.0, bn_momentum: float = 0.1,
                  bn_epsilon: float = 1e-5, **_: object) -> None:
@@ -41,6 +41,7 @@ class MobileNetV2(nn.Module):
         self.classifier = LinearAct(last, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self.bump_bn_counters()
         x = self.features(x)
         x = self.pool(x)
         x = self.dropout(x)

@@ -14,7 +14,7 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, make_divisible)
+                             LinearAct, ZooModel, make_divisible)
 
 # (kernel, expand_ch, out_ch, use_se, act, stride)
 _LARGE: List[Tuple[int, int, int, bool, str, int]] = [
@@ -49,7 +49,7 @@ _SMALL: List[Tuple[int, int, int, bool, str, int]] = [
 ]
 
 
-class MobileNetV3(nn.Module):
+class MobileNetV3(ZooModel):
     def __init__(self, mode: str = "large", num_classes: int = 1000,
                  width_mult: float = 1.0, dropout: float = 0.2,
                  bn_momentum: float = 0.1, bn_epsilon: float = 1e-5,
@@ -80,6 +80,7 @@ class MobileNetV3(nn.Module):
         self.classifier = LinearAct(head_ch, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        self.bump_bn_counters()
         x = self.features(x)
         x = self.pool(x)
         x = self.head(x)

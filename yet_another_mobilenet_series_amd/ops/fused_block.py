@@ -346,11 +346,8 @@ def fused_inverted_residual(x, block):
     if block.se is not None:
         sw1, sb1 = block.se.fc1.weight, block.se.fc1.bias
         sw2, sb2 = block.se.fc2.weight, block.se.fc2.bias
+    # num_batches_tracked is bumped once per model forward (ZooModel)
     dwm, pjm = block.dw, block.project
-    if block.training and dwm.bn.track_running_stats:
-        for m in (block.expand, dwm, pjm):
-            if m is not None:
-                m.bn.num_batches_tracked += 1
     return FusedInvertedResidualFn.apply(
         x, ew, eg, ebb, erm, erv,
         dwm.conv.weight, dwm.bn.weight, dwm.bn.bias,

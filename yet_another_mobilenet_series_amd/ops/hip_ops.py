@@ -43,7 +43,7 @@ def _packed(param: torch.Tensor, key: str, fn):
     The pack ops re-run whenever the optimizer bumps the param version,
     and ALWAYS during CUDA-graph capture (so replay re-reads the updated
     fp32 master through the baked cast kernels)."""
-    if torch.cuda.is_current_stream_capturing():
+    if param.is_cuda and torch.cuda.is_current_stream_capturing():
         return fn(param)
     cache = getattr(param, "_yams_pack", None)
     if cache is None:

@@ -28,9 +28,15 @@ KEY_REMAP: Dict[str, str] = {}
 
 
 def _rng_state() -> Dict:
+    # stored as primitives + tensors only, so the whole checkpoint loads
+    # under torch.load(weights_only=True)
+    np_name, np_key, np_pos, np_gauss, np_cached = np.random.get_state()
     state = {
-        "python": random.getstate(),
-        "numpy": np.random.get_state(),
+        "python": list(random.getstate()[1]),
+        "numpy": {"name": str(np_name),
+                  "key": torch.from_numpy(np_key.astype(np.int64)),
+                  "pos": int(np_pos), "has_gauss": int(np_gauss),
+                  "cached": float(np_cached)},
         "torch": torch.get_rng_state(),
     }
     if torch.cuda.is_available():
@@ -39,8 +45,18 @@ def _rng_state() -> Dict:
 
 
 def _load_rng_state(state: Dict) -> None:
-    random.setstate(state["python"])
-    np.random.set_state(state["numpy"])
+    py = state["python"]
+    if isinstance(py, list):  # current format
+        random.setstate((3, tuple(py), None))
+    else:  # legacy tuple format
+        random.setstate(py)
+    nps = state["numpy"]
+    if isinstance(nps, dict):
+        np.random.set_state((nps["name"],
+                             nps["key"].numpy().astype(np.uint32),
+                             nps["pos"], nps["has_gauss"], nps["cached"]))
+    else:
+        np.random.set_state(nps)
     # map_location may have moved the ByteTensors to GPU: bring them back
     torch.set_rng_state(state["torch"].cpu())
     if torch.cuda.is_available() and "cuda" in state:
@@ -98,7 +114,13 @@ def remap_state_dict(sd: Dict[str, torch.Tensor], model) -> Dict[str, torch.Tens
 def load_checkpoint(path: str, model, optimizer=None, ema=None,
                     lr_scheduler=None, strict: bool = True,
                     restore_rng: bool = True, map_location="cpu") -> Dict:
-    state = torch.load(path, map_location=map_location, weights_only=False)
+    # our own resume checkpoints carry RNG state (numpy tuples etc.) that
+    # weights_only rejects; try the safe path first and fall back for the
+    # trusted-resume case only (ADVICE.md round-1 low finding)
+    try:
+        state = torch.load(path, map_location=map_location, weights_only=True)
+    except Exception:
+        state = torch.load(path, map_location=map_location, weights_only=False)
     sd = state["model"] if "model" in state else state
     model.load_state_dict(remap_state_dict(sd, model), strict=strict)
     if optimizer is not None and "optimizer" in state:
@@ -113,7 +135,12 @@ def load_checkpoint(path: str, model, optimizer=None, ema=None,
 
 
 def load_pretrained(path: str, model, strict: bool = False) -> None:
-    """Load weights only (the reference's ``pretrained`` config key)."""
-    state = torch.load(path, map_location="cpu", weights_only=False)
-    sd = state.get("model", state)
+    """Load weights only (the reference's ``pretrained`` config key).
+
+    Pretrained files are potentially FOREIGN (e.g. the reference repo's
+    released checkpoints), so this path never executes pickled code:
+    ``weights_only=True``, no fallback.
+    """
+    state = torch.load(path, map_location="cpu", weights_only=True)
+    sd = state.get("model", state) if isinstance(state, dict) else state
     model.load_state_dict(remap_state_dict(sd, model), strict=strict)

@@ -128,6 +128,31 @@ class ImageFolderDataset(Dataset):
         return _pil_to_tensor(img), label
 
 
+class EvalDistributedSampler(torch.utils.data.Sampler):
+    """Uneven-shard eval sampler: rank r gets indices r, r+W, r+2W, ...
+
+    ``DistributedSampler``'s default padding duplicates samples so every
+    rank has an equal shard, which biases summed top-1/top-5 counters when
+    the val-set size is not divisible by world size (ADVICE.md round-1 low
+    finding).  Eval metrics are reduced as COUNTERS, so unequal shard
+    lengths are fine — no padding needed.
+    """
+
+    def __init__(self, dataset, num_replicas: int, rank: int) -> None:
+        self.n = len(dataset)
+        self.num_replicas = num_replicas
+        self.rank = rank
+
+    def __iter__(self):
+        return iter(range(self.rank, self.n, self.num_replicas))
+
+    def __len__(self) -> int:
+        return (self.n - self.rank + self.num_replicas - 1) // self.num_replicas
+
+    def set_epoch(self, epoch: int) -> None:  # noqa: ARG002 — fixed order
+        pass
+
+
 def build_dataset(cfg, train: bool, seed: int = 0) -> Dataset:
     if cfg.dataset == "synthetic":
         size = cfg.synthetic_size if train else max(cfg.synthetic_size // 10, 64)
@@ -146,8 +171,13 @@ def get_data_loader(cfg, train: bool, batch_size: Optional[int] = None,
     ds = build_dataset(cfg, train, seed)
     if distributed is None:
         distributed = dist.is_initialized()
-    sampler = (DistributedSampler(ds, shuffle=train, seed=seed)
-               if distributed else None)
+    if not distributed:
+        sampler = None
+    elif train:
+        sampler = DistributedSampler(ds, shuffle=True, seed=seed)
+    else:
+        sampler = EvalDistributedSampler(ds, dist.get_world_size(),
+                                         dist.get_rank())
     bs = batch_size or cfg.batch_size_per_gpu
     return DataLoader(
         ds, batch_size=bs, shuffle=(train and sampler is None),

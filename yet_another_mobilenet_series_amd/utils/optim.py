@@ -21,6 +21,20 @@ import torch.nn as nn
 from ..ops import functional as OF
 
 
+def _mark_updated(params: Iterable[torch.Tensor]) -> None:
+    """Record that the fused kernels mutated ``params`` in place.
+
+    The fused update kernels (K12/K13) write through raw data pointers, so
+    the autograd version counter — which ``hip_ops._packed`` keys its
+    bf16 weight-pack cache on — does not advance on its own.  Without this
+    bump the next forward would reuse the pre-step packed weights and
+    conv/linear weights would silently stop training (ADVICE.md round-1
+    high finding).
+    """
+    for p in params:
+        torch.autograd.graph.increment_version(p)
+
+
 def _grad_like(p: torch.Tensor) -> torch.Tensor:
     """Gradient re-laid-out to match the param's memory format (flat fused
     updates walk raw storage, so element order must agree)."""
@@ -151,6 +165,7 @@ class FusedSGD(torch.optim.Optimizer):
             if ps:
                 e.fused_sgd(ps, gs, ms, wds, group["lr"],
                             group["momentum"], group["nesterov"])
+                _mark_updated(ps)
         return loss
 
 
@@ -190,6 +205,7 @@ class FusedRMSpropTF(torch.optim.Optimizer):
                 e.fused_rmsprop_tf(ps, gs, mss, moms, wds, group["lr"],
                                    group["alpha"], group["eps"],
                                    group["momentum"])
+                _mark_updated(ps)
         return loss
 
 
