@@ -8,6 +8,20 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <stdexcept>
+#include <string>
+
+// Hard-error for launcher dispatch fallthrough: an unsupported
+// (kernel, stride, lop, act, ...) combination must raise — propagated by
+// pybind as RuntimeError — never silently return with the output tensor
+// uninitialized (no silent corruption).  Host-only.
+[[noreturn]] inline void yams_launch_fail(const char* fn, long a, long b,
+                                          long c, long d) {
+  throw std::runtime_error(
+      std::string("yams: no kernel instantiation for ") + fn + "(" +
+      std::to_string(a) + ", " + std::to_string(b) + ", " +
+      std::to_string(c) + ", " + std::to_string(d) + ")");
+}
 
 #define DEV_INLINE __device__ __forceinline__
 
@@ -267,6 +281,7 @@ DEV_INLINE float lop_load(const Lop& L, const ushort_t* p, int64_t off,
     case ACT_RELU6:    { constexpr int kAct = ACT_RELU6;    __VA_ARGS__; break; } \
     case ACT_HSWISH:   { constexpr int kAct = ACT_HSWISH;   __VA_ARGS__; break; } \
     case ACT_HSIGMOID: { constexpr int kAct = ACT_HSIGMOID; __VA_ARGS__; break; } \
+    default: yams_launch_fail("act_switch", (ACT_CODE), 0, 0, 0);        \
   }
 
 DEV_INLINE int64_t idx4(int n, int h, int w, int c, int H, int W, int C) {

@@ -506,6 +506,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
 #undef DWF_KS
 #undef DWF_ST
 #undef DWF_CASE
+  yams_launch_fail("dw_conv_fwd", k, stride, lopx.kind, lopx.act);
 }
 
 void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
@@ -544,7 +545,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     DWD_CASE(3) DWD_CASE(5)
 #undef DWD_CASE
 #undef DWD_EPI
-    return;
+    yams_launch_fail("dw_conv_dgrad_s1", k, lopdy.kind, lopdy.act, epi.act);
   }
   const int64_t total8 = int64_t(N) * H * W * (C >> 3);
   dim3 g2(grid_for(total8));
@@ -556,6 +557,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
   }
   DWD2_CASE(3) DWD2_CASE(5)
 #undef DWD2_CASE
+  yams_launch_fail("dw_conv_dgrad_s2", k, stride, lopdy.kind, lopdy.act);
 }
 
 int dw_wgrad_chunks(int N, int Ho, int Wo) {
@@ -598,6 +600,7 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
   DWW_KS(3, 1) DWW_KS(3, 2) DWW_KS(5, 1) DWW_KS(5, 2)
 #undef DWW_KS
 #undef DWW_CASE
+  yams_launch_fail("dw_conv_wgrad", k, stride, lopx.kind, lopdy.kind);
 combine:
   slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
 }

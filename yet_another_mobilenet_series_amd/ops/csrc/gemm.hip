@@ -263,6 +263,7 @@ void pw_conv(const void* x, const void* w, void* y,
   PW_EPI2(0) PW_EPI2(1) PW_EPI2(2) PW_EPI2(3)
 #undef PW_EPI2
 #undef PW_CASE
+  yams_launch_fail("pw_conv", ep, lopa.kind, lopa.act, epi.act);
 }
 
 // ---------------------------------------------------------------------------
@@ -438,6 +439,7 @@ void atb_launch(const ushort_t* a, const ushort_t* b, float* ws, int64_t M,
   ATB_TILE(128, 32) ATB_TILE(128, 64) ATB_TILE(128, 128)
 #undef ATB_TILE
 #undef ATB_CASE
+  yams_launch_fail("atb", w1, w2, ka, kb);
 }
 
 // vec4 variant: thread = 4 consecutive fp32 elems x one of SL split lanes;
