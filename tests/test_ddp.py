@@ -208,3 +208,105 @@ def test_grad_accumulation_matches_big_batch(tmp_path):
     loss.backward()
     for a, p in zip(got, model.parameters()):
         assert torch.allclose(a, p.grad, atol=1e-5), (a - p.grad).abs().max()
+
+
+def test_auto_bucket_sizing():
+    """bucket_mb=None targets ~7 in-flight reduces (one per xGMI link) for
+    big models and a 1 MB floor for small ones."""
+    from yet_another_mobilenet_series_amd.parallel.ddp import (
+        MIN_BUCKET_MB, XGMI_LINKS, auto_bucket_mb)
+    v3l_bytes = 5_483_000 * 4  # V3-Large fp32 grads ≈ 21.9 MB
+    mb = auto_bucket_mb(v3l_bytes)
+    assert abs(mb - 21.9 / XGMI_LINKS) / mb < 0.05
+    assert auto_bucket_mb(100_000 * 4) == MIN_BUCKET_MB
+    # engine path: world-1 stays unbucketed; forced bucketing honors auto
+    model = _make_model()
+    eng = DataParallelEngine(model, bucket_mb=None, force_bucketed=True)
+    assert eng.bucket_mb == MIN_BUCKET_MB
+    assert len(eng.buckets) == 1  # tiny model -> one bucket
+
+
+def _bf16_reduce_worker(rank, world, tmpfile, out_path):
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world)
+    try:
+        model = _make_model(seed=100)
+        eng = DataParallelEngine(model, bucket_mb=0.0001, overlap=True,
+                                 reduce_dtype=torch.bfloat16)
+        torch.manual_seed(0)
+        xs = torch.randn(world * 4, 8)
+        ys = torch.randn(world * 4, 4)
+        eng.zero_grad()
+        loss = ((eng(xs[rank * 4:(rank + 1) * 4]) -
+                 ys[rank * 4:(rank + 1) * 4]) ** 2).mean()
+        loss.backward()
+        eng.finish()
+        if rank == 0:
+            torch.save([p.grad.detach().clone()
+                        for p in model.parameters()], out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_bf16_wire_reduction(tmp_path):
+    """reduce_dtype=bf16: gradients reduce over a bf16 wire buffer and land
+    back in the fp32 flat grads within bf16 rounding of the fp32 result."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    tmpfile = str(tmp_path / "init_bf16")
+    out_path = str(tmp_path / "grads.pt")
+    ps = [ctx.Process(target=_bf16_reduce_worker,
+                      args=(r, world, tmpfile, out_path))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(120)
+        assert p.exitcode == 0
+    got = torch.load(out_path)
+    # fp32 reference grads of the same global batch
+    model = _make_model(seed=100)
+    torch.manual_seed(0)
+    xs = torch.randn(8, 8)
+    ys = torch.randn(8, 4)
+    loss = 0.5 * (((model(xs[:4]) - ys[:4]) ** 2).mean() +
+                  ((model(xs[4:]) - ys[4:]) ** 2).mean())
+    loss.backward()
+    for g, p in zip(got, model.parameters()):
+        assert torch.allclose(g.float(), p.grad, rtol=2e-2, atol=1e-3)
+
+
+def test_dp_world8_gloo(tmp_path):
+    """World-8 fake-world parity: the bucket/overlap machinery at the real
+    node's rank count (VERDICT round-1 item 3)."""
+    world = 8
+    ctx = mp.get_context("spawn")
+    tmpfile = str(tmp_path / "init_w8")
+    out_path = str(tmp_path / "params_w8.pt")
+    ps = [ctx.Process(target=_worker,
+                      args=(r, world, tmpfile, True, out_path))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(180)
+        assert p.exitcode == 0
+    got = torch.load(out_path)
+    # single-process reference over the same global batch
+    model = _make_model(seed=100)
+    torch.manual_seed(0)
+    xs = torch.randn(world * 4, 8)
+    ys = torch.randn(world * 4, 4)
+    for _ in range(3):
+        for p in model.parameters():
+            p.grad = None
+        loss = sum(((model(xs[r * 4:(r + 1) * 4]) -
+                     ys[r * 4:(r + 1) * 4]) ** 2).mean()
+                   for r in range(world)) / world
+        loss.backward()
+        with torch.no_grad():
+            for p in model.parameters():
+                p -= 0.1 * p.grad
+    for a, b in zip(got, model.parameters()):
+        assert torch.allclose(a, b.detach(), rtol=1e-4, atol=1e-5)

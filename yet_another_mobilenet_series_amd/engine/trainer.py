@@ -46,9 +46,18 @@ class Trainer:
         model = model.to(self.device)
         if cfg.train.channels_last and self.device.type == "cuda":
             model = model.to(memory_format=torch.channels_last)
+        reduce_dtype = (torch.bfloat16
+                        if getattr(cfg.dist, "reduce_dtype", "fp32") == "bf16"
+                        else None)
         self.engine = DataParallelEngine(
-            model, bucket_mb=cfg.dist.bucket_mb, overlap=cfg.dist.overlap)
+            model, bucket_mb=cfg.dist.bucket_mb, overlap=cfg.dist.overlap,
+            reduce_dtype=reduce_dtype)
         self.model = model
+        # abort non-zero if a step wedges (watchdog_sec=0
+        # disables; the pg timeout still covers erroring collectives)
+        self.watchdog = udist.Watchdog(
+            float(getattr(cfg.dist, "watchdog_sec", 0) or 0),
+            logger=None)
 
         self.train_loader, self.val_loader = dataflow.get_data_loaders(
             cfg.data, seed=cfg.train.seed)
@@ -165,6 +174,7 @@ class Trainer:
                                      **{f"{n}_ms": round(v, 3)
                                         for n, v in ms.items()})
             self.global_step += 1
+            self.watchdog.kick()
             if prof is not None:
                 prof.step()
                 if it >= int(getattr(cfg.train, "trace_steps", 3)) + 2:
@@ -203,6 +213,8 @@ class Trainer:
         self.engine.eval()
         counters = torch.zeros(3, device=self.device)  # n, top1-correct, top5-correct
         for x, y in self.val_loader:
+            if hasattr(self, "watchdog"):
+                self.watchdog.kick()
             x, y = self._to_device(x, y)
             logits = self.model(x)
             top1, top5 = accuracy(logits, y, (1, 5))
@@ -238,11 +250,13 @@ class Trainer:
 
     def fit(self) -> None:
         cfg = self.cfg
-        for epoch in range(self.start_epoch, cfg.train.epochs):
-            self.train_one_epoch(epoch)
-            top1 = 0.0
-            if (epoch + 1) % cfg.val.interval_epochs == 0:
-                top1 = self.validate(epoch)
-            if (epoch + 1) % cfg.train.save_interval_epochs == 0:
-                self.save(epoch, top1)
+        self.watchdog.logger = self.logger
+        with self.watchdog:
+            for epoch in range(self.start_epoch, cfg.train.epochs):
+                self.train_one_epoch(epoch)
+                top1 = 0.0
+                if (epoch + 1) % cfg.val.interval_epochs == 0:
+                    top1 = self.validate(epoch)
+                if (epoch + 1) % cfg.train.save_interval_epochs == 0:
+                    self.save(epoch, top1)
         self.jsonl.close()

@@ -28,19 +28,64 @@ import torch.nn as nn
 from .distributed import broadcast_module, get_world_size
 
 
-class _Bucket:
-    __slots__ = ("params", "flat", "pending", "handle", "ready_event")
+#: number of point-to-point xGMI links per GPU on an 8-OAM MI355X node.
+#: The node is fully connected (no switch): a single ring all-reduce is
+#: bound by ONE link (~153 GB/s), so the engine keeps ~XGMI_LINKS buckets
+#: in flight — RCCL pipelines them across links/channels — and each bucket
+#: launches as soon as backward produces its last gradient.
+XGMI_LINKS = 7
 
-    def __init__(self, params: List[torch.nn.Parameter], device, dtype) -> None:
+#: floor on the bucket size: below ~1 MB the per-collective launch/sync
+#: overhead (~20 us) exceeds the wire time of the message itself
+#: (1 MB over one 153 GB/s link ≈ 7 us x 2(8-1)/8 steps ≈ 12 us).
+MIN_BUCKET_MB = 1.0
+
+
+def auto_bucket_mb(total_bytes: int) -> float:
+    """Bucket size targeting ~XGMI_LINKS in-flight reduces per step.
+
+    V3-Large (21.9 MB fp32 grads) -> ~3.1 MB buckets; small zoo models
+    degrade gracefully to fewer, MIN_BUCKET_MB-sized buckets.  The wire
+    math (xGMI 7x153 GB/s, ring per-link bound) is laid out in
+    docs/COMM.md with the measured tuning table.
+    """
+    mb = total_bytes / (1024.0 * 1024.0)
+    return max(mb / XGMI_LINKS, MIN_BUCKET_MB)
+
+
+class _Bucket:
+    __slots__ = ("params", "flat", "stage", "pending", "handle",
+                 "ready_event")
+
+    def __init__(self, params: List[torch.nn.Parameter], device, dtype,
+                 reduce_dtype) -> None:
         self.params = params
         total = sum(p.numel() for p in params)
         self.flat = torch.zeros(total, device=device, dtype=dtype)
+        # optional reduced-precision wire buffer (bf16 halves the per-link
+        # bytes; local accumulation stays fp32)
+        self.stage = (torch.zeros(total, device=device, dtype=reduce_dtype)
+                      if reduce_dtype is not None and reduce_dtype != dtype
+                      else None)
         off = 0
         for p in params:
             p.grad = self.flat[off:off + p.numel()].view_as(p)
             off += p.numel()
         self.pending = 0
         self.handle = None
+
+    def launch(self) -> None:
+        if self.stage is not None:
+            self.stage.copy_(self.flat)
+            self.handle = dist.all_reduce(self.stage, async_op=True)
+        else:
+            self.handle = dist.all_reduce(self.flat, async_op=True)
+
+    def wait(self) -> None:
+        self.handle.wait()
+        self.handle = None
+        if self.stage is not None:
+            self.flat.copy_(self.stage)
 
 
 class DataParallelEngine(nn.Module):
@@ -56,15 +101,22 @@ class DataParallelEngine(nn.Module):
         optimizer.step()
     """
 
-    def __init__(self, module: nn.Module, bucket_mb: float = 4.0,
+    def __init__(self, module: nn.Module, bucket_mb: Optional[float] = None,
                  overlap: bool = True, grad_dtype: torch.dtype = torch.float32,
                  broadcast_params: bool = True,
-                 force_bucketed: bool = False) -> None:
+                 force_bucketed: bool = False,
+                 reduce_dtype: Optional[torch.dtype] = None) -> None:
+        """``bucket_mb=None`` sizes buckets for the 7 xGMI links
+        (:func:`auto_bucket_mb`).  ``reduce_dtype=torch.bfloat16`` reduces
+        on a bf16 wire buffer (half the xGMI bytes; fp32 local accumulation
+        is unchanged) — off by default for bitwise-exact parity with the
+        single-process run (SURVEY.md §5.8 "fp32 first")."""
         super().__init__()
         self.module = module
         self.world_size = get_world_size()
         self.overlap = overlap and self.world_size > 1
         self.grad_dtype = grad_dtype
+        self.reduce_dtype = reduce_dtype
         if broadcast_params and self.world_size > 1:
             broadcast_module(module)
         # single-process: skip flat buckets entirely — preset grad views
@@ -83,6 +135,10 @@ class DataParallelEngine(nn.Module):
         device = params[0].device if params else torch.device("cpu")
         # reverse registration order ≈ grad production order during backward
         params = list(reversed(params))
+        if bucket_mb is None:
+            total_bytes = sum(p.numel() for p in params) * 4
+            bucket_mb = auto_bucket_mb(total_bytes)
+        self.bucket_mb = bucket_mb
         bucket_elems = max(int(bucket_mb * 1024 * 1024 / 4), 1)
         self.buckets: List[_Bucket] = []
         cur: List[nn.Parameter] = []
@@ -91,10 +147,12 @@ class DataParallelEngine(nn.Module):
             cur.append(p)
             cur_n += p.numel()
             if cur_n >= bucket_elems:
-                self.buckets.append(_Bucket(cur, device, grad_dtype))
+                self.buckets.append(_Bucket(cur, device, grad_dtype,
+                                            reduce_dtype))
                 cur, cur_n = [], 0
         if cur:
-            self.buckets.append(_Bucket(cur, device, grad_dtype))
+            self.buckets.append(_Bucket(cur, device, grad_dtype,
+                                        reduce_dtype))
         self._param_bucket: Dict[nn.Parameter, _Bucket] = {}
         for b in self.buckets:
             for p in b.params:
@@ -143,7 +201,7 @@ class DataParallelEngine(nn.Module):
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0 and self.world_size > 1:
-            b.handle = dist.all_reduce(b.flat, async_op=True)
+            b.launch()
             self._launched.append(b)
 
     def finish(self) -> None:
@@ -155,13 +213,14 @@ class DataParallelEngine(nn.Module):
                 if b.handle is None:
                     # a bucket whose params got no grads this step: reduce
                     # anyway so ranks stay collectively consistent
-                    b.handle = dist.all_reduce(b.flat, async_op=True)
+                    b.launch()
             for b in self.buckets:
-                b.handle.wait()
-                b.handle = None
+                b.wait()
         else:
             for b in self.buckets:
-                dist.all_reduce(b.flat)
+                b.launch()
+            for b in self.buckets:
+                b.wait()
         inv = 1.0 / self.world_size
         torch._foreach_mul_([b.flat for b in self.buckets], inv)
 

@@ -92,3 +92,79 @@ def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
 def destroy() -> None:
     if dist.is_initialized():
         dist.destroy_process_group()
+
+
+class Watchdog:
+    """Abort the process (exit code 124) if the training step stalls.
+
+    The process-group timeout covers collectives that RETURN an error, but
+    a wedged GPU kernel or a hung RCCL channel can stall a rank without any
+    API ever returning; every other rank then blocks in its next collective
+    forever.  The watchdog is a daemon thread: the step
+    loop calls :meth:`kick` once per step, and if no kick arrives within
+    ``timeout_sec`` the watchdog logs and hard-exits non-zero so the
+    launcher (torchrun) tears the job down and the checkpoint/resume path
+    takes over.
+
+    Use ``with Watchdog(120):`` around the step loop, or ``start()`` /
+    ``stop()`` explicitly.  ``timeout_sec <= 0`` disables it (no thread).
+    """
+
+    EXIT_CODE = 124  # same convention as coreutils `timeout`
+
+    def __init__(self, timeout_sec: float, name: str = "step",
+                 logger=None) -> None:
+        import threading
+        self.timeout_sec = float(timeout_sec)
+        self.name = name
+        self.logger = logger
+        self._lock = threading.Lock()
+        self._last = None
+        self._stop = threading.Event()
+        self._thread = None
+
+    def start(self) -> "Watchdog":
+        if self.timeout_sec <= 0 or self._thread is not None:
+            return self
+        import threading
+        import time
+        self._last = time.monotonic()
+        self._stop.clear()
+
+        def _watch():
+            while not self._stop.wait(min(self.timeout_sec / 4, 5.0)):
+                with self._lock:
+                    last = self._last
+                idle = time.monotonic() - last
+                if idle > self.timeout_sec:
+                    msg = (f"watchdog: no {self.name} progress for "
+                           f"{idle:.0f}s (> {self.timeout_sec:.0f}s) on "
+                           f"rank {get_rank()}; aborting")
+                    if self.logger is not None:
+                        self.logger.error(msg)
+                    print(msg, flush=True)
+                    os._exit(self.EXIT_CODE)
+
+        self._thread = __import__("threading").Thread(
+            target=_watch, name="yams-watchdog", daemon=True)
+        self._thread.start()
+        return self
+
+    def kick(self) -> None:
+        if self._thread is None:
+            return
+        import time
+        with self._lock:
+            self._last = time.monotonic()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=1.0)
+            self._thread = None
+
+    def __enter__(self) -> "Watchdog":
+        return self.start()
+
+    def __exit__(self, *exc) -> None:
+        self.stop()

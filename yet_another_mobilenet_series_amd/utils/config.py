@@ -141,8 +141,9 @@ DEFAULTS: Dict[str, Any] = {
               "deterministic": False, "perf_breakdown": False,
               "trace": False, "trace_steps": 3,
               "engine": "auto", "use_cuda_graph": True},
-    "dist": {"backend": "auto", "bucket_mb": 4, "overlap": True,
-             "grad_dtype": "fp32", "timeout_sec": 300},
+    "dist": {"backend": "auto", "bucket_mb": None, "overlap": True,
+             "grad_dtype": "fp32", "reduce_dtype": "fp32",
+             "timeout_sec": 300, "watchdog_sec": 0},
     "log": {"dir": "runs/default", "jsonl": True},
     "val": {"interval_epochs": 1, "batch_size_per_gpu": 256, "use_ema": True},
 }
