@@ -19,7 +19,7 @@ CSRC = os.path.join(ROOT, "yet_another_mobilenet_series_amd", "ops", "csrc")
 
 sources = [os.path.join(CSRC, f) for f in
            ("bindings.cpp", "gemm.hip", "dwconv.hip", "bn.hip", "misc.hip",
-            "update.hip")]
+            "se.hip", "update.hip")]
 
 setup(
     name="yams_hip",

@@ -453,3 +453,115 @@ def test_forward_changes_after_fused_optimizer_step():
     # conv/linear weights moved between the two eval forwards -> logits move
     assert (la - lb).abs().max().item() > 1e-3, \
         "forward unchanged after optimizer step: stale weight-pack cache"
+
+
+# ---- fused squeeze-excite (K7, se.hip) ------------------------------------
+
+def _se_ref(x32, w1, b1, w2, b2):
+    """fp32 torch reference of the SE chain from bf16-rounded x."""
+    pooled = x32.mean(dim=(2, 3))                       # [N, C]
+    ha = torch.relu(pooled @ w1.t() + b1)               # [N, R]
+    sz = ha @ w2.t() + b2                               # [N, C]
+    s = torch.clamp(sz + 3, 0, 6) / 6
+    y = x32 * s[:, :, None, None]
+    return y, pooled, ha, sz, s
+
+
+@pytest.mark.parametrize("c,r,hw", [(16, 8, 56), (72, 24, 28), (120, 32, 28),
+                                    (480, 120, 14), (672, 168, 7),
+                                    (960, 240, 7)])
+def test_se_fwd_fused(ext, c, r, hw):
+    torch.manual_seed(c + r)
+    x = randn_nhwc(5, c, hw, hw, seed=c)
+    w1 = (torch.randn(r, c) / c ** 0.5).to(dev())
+    b1 = torch.randn(r, device=dev()) * 0.1
+    w2 = (torch.randn(c, r) / r ** 0.5).to(dev())
+    b2 = torch.randn(c, device=dev()) * 0.1
+    y = torch.empty_like(x)
+    pooled = torch.empty(5, c, dtype=torch.float32, device=dev())
+    ha = torch.empty(5, r, dtype=torch.float32, device=dev())
+    sz = torch.empty(5, c, dtype=torch.float32, device=dev())
+    ext.se_fwd(x, w1.bfloat16().contiguous(), b1, w2.bfloat16().contiguous(),
+               b2, y, pooled, ha, sz)
+    # reference from the same bf16-rounded operands
+    yr, pr, har, szr, _ = _se_ref(x.float(), w1.bfloat16().float(), b1,
+                                  w2.bfloat16().float(), b2)
+    assert_close(pooled, pr, rtol=2e-3)
+    assert_close(ha, har, rtol=2e-3)
+    assert_close(sz, szr, rtol=5e-3)
+    assert_close(y, yr, rtol=1e-2)
+
+
+def test_se_fwd_with_bnact_lop(ext):
+    """LOP_BNACT path: BN+act applied on load, activated input streamed to
+    ymat, SE math on the activated values."""
+    c, r, hw = 120, 32, 14
+    z = randn_nhwc(4, c, hw, hw, seed=9)
+    ab = torch.randn(2 * c, device=dev()).contiguous()
+    ab[:c] = ab[:c].abs() + 0.5
+    w1 = (torch.randn(r, c) / c ** 0.5).to(dev()).bfloat16().contiguous()
+    b1 = torch.randn(r, device=dev()) * 0.1
+    w2 = (torch.randn(c, r) / r ** 0.5).to(dev()).bfloat16().contiguous()
+    b2 = torch.randn(c, device=dev()) * 0.1
+    y = torch.empty_like(z)
+    ymat = torch.empty_like(z)
+    pooled = torch.empty(4, c, dtype=torch.float32, device=dev())
+    ha = torch.empty(4, r, dtype=torch.float32, device=dev())
+    sz = torch.empty(4, c, dtype=torch.float32, device=dev())
+    ext.se_fwd(z, w1, b1, w2, b2, y, pooled, ha, sz,
+               lop_kind=1, lop_act=3, lop_coef=ab, ymat=ymat)
+    zv = z.float()
+    v = zv * ab[:c].view(1, c, 1, 1) + ab[c:].view(1, c, 1, 1)
+    act = v * torch.clamp(v + 3, 0, 6) / 6  # hswish
+    act = act.bfloat16().float()            # ymat is bf16-rounded
+    assert_close(ymat, act, rtol=1e-2)
+    yr, pr, _, _, _ = _se_ref(act, w1.float(), b1, w2.float(), b2)
+    assert_close(pooled, pr, rtol=3e-3)
+    assert_close(y, yr, rtol=1e-2)
+
+
+@pytest.mark.parametrize("c,r,hw", [(16, 8, 56), (480, 120, 14),
+                                    (960, 240, 7)])
+def test_se_bwd_fused(ext, c, r, hw):
+    """Full SE backward vs autograd on the fp32 reference (same bf16
+    inputs): the fused kernels must carry NO extra rounding noise."""
+    n = 6
+    torch.manual_seed(3)
+    x = randn_nhwc(n, c, hw, hw, seed=c + 1)
+    dy = randn_nhwc(n, c, hw, hw, seed=c + 2)
+    w1 = ((torch.randn(r, c) / c ** 0.5).to(dev()).bfloat16())
+    b1 = torch.randn(r, device=dev()) * 0.1
+    w2 = ((torch.randn(c, r) / r ** 0.5).to(dev()).bfloat16())
+    b2 = torch.randn(c, device=dev()) * 0.1
+
+    # fused path
+    y = torch.empty_like(x)
+    pooled = torch.empty(n, c, dtype=torch.float32, device=dev())
+    ha = torch.empty(n, r, dtype=torch.float32, device=dev())
+    sz = torch.empty(n, c, dtype=torch.float32, device=dev())
+    ext.se_fwd(x, w1.contiguous(), b1, w2.contiguous(), b2, y, pooled, ha,
+               sz)
+    dx = torch.empty_like(x)
+    dsz = torch.empty(n, c, dtype=torch.float32, device=dev())
+    dhz = torch.empty(n, r, dtype=torch.float32, device=dev())
+    ext.se_bwd(dy, x, w1.contiguous(), w2.contiguous(), ha, sz, dx, dsz,
+               dhz)
+    dw1 = torch.empty(r, c, dtype=torch.float32, device=dev())
+    dw2 = torch.empty(c, r, dtype=torch.float32, device=dev())
+    db1 = torch.empty(r, dtype=torch.float32, device=dev())
+    db2 = torch.empty(c, dtype=torch.float32, device=dev())
+    ext.se_wgrad(dsz, ha, dhz, pooled, dw1, dw2, db1, db2)
+
+    # fp32 autograd reference from the same bf16-rounded operands
+    xr = x.float().requires_grad_(True)
+    w1r = w1.float().requires_grad_(True)
+    b1r = b1.detach().clone().requires_grad_(True)
+    w2r = w2.float().requires_grad_(True)
+    b2r = b2.detach().clone().requires_grad_(True)
+    yr, _, _, _, _ = _se_ref(xr, w1r, b1r, w2r, b2r)
+    yr.backward(dy.float())
+    assert_close(dx, xr.grad, rtol=2e-2)
+    assert_close(dw1, w1r.grad, rtol=1e-2)
+    assert_close(dw2, w2r.grad, rtol=1e-2)
+    assert_close(db1, b1r.grad, rtol=1e-2)
+    assert_close(db2, b2r.grad, rtol=1e-2)

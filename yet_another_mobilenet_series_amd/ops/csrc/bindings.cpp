@@ -227,6 +227,83 @@ void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
       cur_stream());
 }
 
+// ---- fused squeeze-excite (K7) -----------------------------------------
+
+void se_fwd(const Tensor& x, const Tensor& w1, const Tensor& b1,
+            const Tensor& w2, const Tensor& b2, Tensor& y,
+            Tensor& pooled, Tensor& ha, Tensor& sz,
+            int64_t lop_kind, int64_t lop_act,
+            c10::optional<Tensor> lop_coef, c10::optional<Tensor> ymat) {
+  auto xr = rows_of(x, "x");
+  auto yr = rows_of(y, "y");
+  check_w(w1);
+  check_w(w2);
+  const int C = xr.C, R = int(w1.size(0));
+  TORCH_CHECK(yr.C == C && yr.M == xr.M, "se_fwd shape mismatch");
+  TORCH_CHECK(w1.dim() == 2 && w1.size(1) == C, "w1 must be [R, C]");
+  TORCH_CHECK(w2.dim() == 2 && w2.size(0) == C && w2.size(1) == R,
+              "w2 must be [C, R]");
+  const int N = int(x.size(0));
+  const int HW = int(xr.M / N);
+  TORCH_CHECK(pooled.numel() >= int64_t(N) * C &&
+                  ha.numel() >= int64_t(N) * R &&
+                  sz.numel() >= int64_t(N) * C,
+              "se_fwd save buffers too small");
+  void* mp = nullptr;
+  if (ymat.has_value()) {
+    auto mr = rows_of(*ymat, "ymat");
+    TORCH_CHECK(mr.C == C && mr.M == xr.M, "ymat shape mismatch");
+    mp = const_cast<void*>(mr.ptr);
+  }
+  yams::se_fwd(xr.ptr, w1.data_ptr(), fp32_req(b1, "b1"), w2.data_ptr(),
+               fp32_req(b2, "b2"), y.data_ptr(), mp,
+               pooled.data_ptr<float>(), ha.data_ptr<float>(),
+               sz.data_ptr<float>(), N, HW, C, R,
+               make_lop(lop_kind, lop_act, lop_coef, c10::nullopt, C),
+               cur_stream());
+}
+
+void se_bwd(const Tensor& dy, const Tensor& x, const Tensor& w1,
+            const Tensor& w2, const Tensor& ha, const Tensor& sz,
+            Tensor& dx, Tensor& dsz, Tensor& dhz) {
+  auto gr = rows_of(dy, "dy");
+  auto xr = rows_of(x, "x");
+  auto dxr = rows_of(dx, "dx");
+  check_w(w1);
+  check_w(w2);
+  const int C = xr.C, R = int(w1.size(0));
+  TORCH_CHECK(gr.C == C && dxr.C == C && gr.M == xr.M && dxr.M == xr.M,
+              "se_bwd shape mismatch");
+  const int N = int(x.size(0));
+  const int HW = int(xr.M / N);
+  TORCH_CHECK(dsz.numel() >= int64_t(N) * C &&
+                  dhz.numel() >= int64_t(N) * R,
+              "se_bwd stage buffers too small");
+  yams::se_bwd(gr.ptr, xr.ptr, w1.data_ptr(), w2.data_ptr(),
+               fp32_req(ha, "ha"), fp32_req(sz, "sz"), dx.data_ptr(),
+               dsz.data_ptr<float>(), dhz.data_ptr<float>(), N, HW, C, R,
+               cur_stream());
+}
+
+void se_wgrad(const Tensor& dsz, const Tensor& ha, const Tensor& dhz,
+              const Tensor& pooled, Tensor& dw1, Tensor& dw2, Tensor& db1,
+              Tensor& db2) {
+  const int N = int(dsz.size(0));
+  const int C = int(dsz.size(1));
+  const int R = int(dhz.size(1));
+  TORCH_CHECK(ha.size(0) == N && ha.size(1) == R &&
+                  pooled.size(0) == N && pooled.size(1) == C,
+              "se_wgrad shape mismatch");
+  TORCH_CHECK(dw1.sizes() == torch::IntArrayRef({R, C}) &&
+                  dw2.sizes() == torch::IntArrayRef({C, R}),
+              "se_wgrad out shape mismatch");
+  yams::se_wgrad(fp32_req(dsz, "dsz"), fp32_req(ha, "ha"),
+                 fp32_req(dhz, "dhz"), fp32_req(pooled, "pooled"),
+                 dw1.data_ptr<float>(), dw2.data_ptr<float>(),
+                 db1.data_ptr<float>(), db2.data_ptr<float>(), N, C, R,
+                 cur_stream());
+}
+
 // ---- stem: im2row (conv itself runs on the GEMM kernels) ---------------
 
 void im2row(const Tensor& x, Tensor& patches, int64_t k, int64_t stride,
@@ -575,6 +652,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "lopx_coef"_a = c10::nullopt, "lopx_z"_a = c10::nullopt,
         "lopdy_kind"_a = 0, "lopdy_act"_a = 0,
         "lopdy_coef"_a = c10::nullopt, "lopdy_z"_a = c10::nullopt);
+  m.def("se_fwd", &se_fwd, "fused squeeze-excite forward (one kernel)",
+        "x"_a, "w1"_a, "b1"_a, "w2"_a, "b2"_a, "y"_a, "pooled"_a, "ha"_a,
+        "sz"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
+        "lop_coef"_a = c10::nullopt, "ymat"_a = c10::nullopt);
+  m.def("se_bwd", &se_bwd, "dy"_a, "x"_a, "w1"_a, "w2"_a, "ha"_a, "sz"_a,
+        "dx"_a, "dsz"_a, "dhz"_a);
+  m.def("se_wgrad", &se_wgrad, "dsz"_a, "ha"_a, "dhz"_a, "pooled"_a,
+        "dw1"_a, "dw2"_a, "db1"_a, "db2"_a);
   m.def("im2row", &im2row);
   m.def("bn_stats", &bn_stats);
   m.def("bn_finalize", &bn_finalize, "ws"_a, "splits"_a, "mean"_a,

@@ -137,6 +137,25 @@ void avgpool_fwd(const void* x, void* out, int N, int HW, int C,
 void avgpool_bwd(const void* dpooled, void* dx, int N, int HW, int C,
                  bool accumulate, hipStream_t s);
 
+// --- fused squeeze-excite (K7) ------------------------------------------
+// ONE kernel per direction (se.hip).  w1 [R,C] bf16, w2 [C,R] bf16,
+// b1/b2 fp32.  Saved-for-backward intermediates pooled[N,C], ha[N,R],
+// sz[N,C] are fp32 (no bf16 rounding inside the SE chain).
+// lopx: LOP_NONE, or LOP_BNACT to apply the producing BN+act on load and
+// stream the activated input to `ymat` (folds bn_act_fwd into SE).
+void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
+            const float* b2, void* y, void* ymat, float* pooled,
+            float* ha, float* sz, int N, int HW, int C, int R,
+            LopDesc lopx, hipStream_t s);
+// dx = dy*s + avgpool_bwd(dpooled); dsz[N,C]/dhz[N,R] staged for se_wgrad
+void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
+            const float* ha, const float* sz, void* dx, float* dsz,
+            float* dhz, int N, int HW, int C, int R, hipStream_t s);
+// dw1[R,C], dw2[C,R], db1[R], db2[C] — deterministic serial-N reduction
+void se_wgrad(const float* dsz, const float* ha, const float* dhz,
+              const float* pooled, float* dw1, float* dw2, float* db1,
+              float* db2, int N, int C, int R, hipStream_t s);
+
 // --- loss (K11): label-smoothed softmax CE, mean over batch -------------
 // loss: fp32 scalar; dlogits fp32 [N, C]; rowloss ws fp32 [N]
 void label_smooth_ce(const void* logits, const int64_t* target,

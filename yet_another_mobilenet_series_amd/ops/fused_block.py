@@ -28,8 +28,8 @@ import os
 import torch
 
 from .backend import ext
-from .hip_ops import (ACT_HSIGMOID, ACT_IDENTITY, ACT_RELU, BF16, CL,
-                      _empty_like_nhwc, _nhwc, _packed)
+from .hip_ops import (ACT_IDENTITY, BF16, CL, _empty_like_nhwc, _nhwc,
+                      _packed)
 
 F32 = torch.float32
 
@@ -54,10 +54,7 @@ def _eval_ab(mean, invstd, gamma, beta, ab) -> None:
 #                    (+0.9 ms/step): the scalar z loads + per-frag
 #                    reductions slow the stream-optimal producer kernels
 #                    more than the removed partial pass costs.  Default off.
-#   YAMS_BWD_POOL=0  disable the SE pool-add+partial fusion (default on;
-#                    saves a full pass over d_y2 and one launch per block).
 _EPI_ON = os.environ.get("YAMS_BWD_EPI", "0") != "0"
-_POOL_ON = os.environ.get("YAMS_BWD_POOL", "1") != "0"
 
 
 class FusedInvertedResidualFn(torch.autograd.Function):
@@ -127,31 +124,18 @@ class FusedInvertedResidualFn(torch.autograd.Function):
 
         # ---- squeeze-excite -------------------------------------------
         if has_se:
-            # y2 materialized: three consumers (pool, scale, wgrad)
+            # ONE fused kernel (se.hip): BN+act applied on load (the
+            # activated y2 is streamed out for project-wgrad/backward),
+            # pool->FC1+ReLU->FC2+hsig->scale inside; fp32 internals
+            from .hip_ops import se_forward
             y2 = torch.empty_like(z2)
-            e.bn_act_fwd(z2, ab2, act, y2, None)
-            r = sw1.shape[0]
-            pooled = torch.empty(n, cd, dtype=BF16, device=dev)
-            e.avgpool_fwd(y2, pooled)
-            w1m = _packed(sw1, "se", lambda w: w.view(r, cd).to(BF16)
-                          .contiguous())
-            hz = torch.empty(n, r, dtype=BF16, device=dev)
-            e.pw_conv(pooled.view(n, cd, 1, 1), w1m, hz.view(n, r, 1, 1))
-            ha = torch.empty_like(hz)
-            e.bias_act_fwd(hz, sb1.float(), ACT_RELU, ha)
-            w2m = _packed(sw2, "se", lambda w: w.view(cd, r).to(BF16)
-                          .contiguous())
-            sz = torch.empty(n, cd, dtype=BF16, device=dev)
-            e.pw_conv(ha.view(n, r, 1, 1), w2m, sz.view(n, cd, 1, 1))
-            sgate = torch.empty_like(sz)
-            e.bias_act_fwd(sz, sb2.float(), ACT_HSIGMOID, sgate)
-            y2s = torch.empty_like(z2)
-            e.rowscale_fwd(y2, sgate, y2s)
+            y2s, pooled, ha, sz = se_forward(
+                e, z2, sw1, sb1, sw2, sb2, lop=(1, act, ab2), ymat=y2)
             pj_in, pj_lop = y2s, (0, 0, None)
         else:
             # no-SE: project reads z2 with the BN+act applied on load —
             # y2 is never materialized (single consumer)
-            y2 = pooled = hz = ha = sz = sgate = y2s = None
+            y2 = pooled = ha = sz = y2s = None
             pj_in, pj_lop = z2, (1, act, ab2)
 
         # ---- project pw -----------------------------------------------
@@ -182,7 +166,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         ctx.save_for_backward(
             x, z1 if has_expand else None, z2, z3,
             mean1, invstd1, ab1, mean2, invstd2, ab2, mean3, invstd3,
-            pooled, hz, ha, sz, sgate, y2s,
+            pooled, ha, sz, y2s,
             y1 if has_expand else None, y2,
             ew, eg, ebb, dww, dg, db, sw1, sb1, sw2, sb2, pw, pg, pb)
         ctx.cfg = (k, stride, act, training, has_expand, has_se, use_res,
@@ -193,7 +177,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
     def backward(ctx, dyb):
         e = ext()
         (x, z1, z2, z3, mean1, invstd1, ab1, mean2, invstd2, ab2, mean3,
-         invstd3, pooled, hz, ha, sz, sgate, y2s, y1, y2, ew, eg, ebb, dww,
+         invstd3, pooled, ha, sz, y2s, y1, y2, ew, eg, ebb, dww,
          dg, db, sw1, sb1, sw2, sb2, pw, pg, pb) = ctx.saved_tensors
         (k, stride, act, training, has_expand, has_se, use_res, n, cin, h,
          wdt, ho, wo) = ctx.cfg
@@ -235,34 +219,11 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                             lopb_coef=ab2)
         d_pw = dwp3.reshape(cp, cd, 1, 1)
 
-        # ---- SE backward ------------------------------------------------
+        # ---- SE backward: fused per-image + wgrad kernels (se.hip) ------
         if has_se:
-            r = sw1.shape[0]
-            d_y2 = torch.empty_like(z2)
-            ds = torch.empty(n, cd, dtype=BF16, device=dev)
-            e.rowscale_bwd(d_in3, y2, sgate, d_y2, ds)
-            dsz = torch.empty_like(ds)
-            db2se = torch.empty(cd, dtype=F32, device=dev)
-            e.bias_act_bwd(ds, sz, sb2.float(), ACT_HSIGMOID, dsz, db2se)
-            w2t = _packed(sw2, "se_t", lambda w: w.view(cd, r).t().to(BF16)
-                          .contiguous())
-            dha = torch.empty_like(ha)
-            e.pw_conv(dsz.view(n, cd, 1, 1), w2t, dha.view(n, r, 1, 1))
-            dw2p = torch.empty(cd, r, dtype=F32, device=dev)
-            e.pw_conv_wgrad(dsz.view(n, cd, 1, 1), ha.view(n, r, 1, 1),
-                            dw2p)
-            dhz = torch.empty_like(dha)
-            db1se = torch.empty(r, dtype=F32, device=dev)
-            e.bias_act_bwd(dha, hz, sb1.float(), ACT_RELU, dhz, db1se)
-            w1t = _packed(sw1, "se_t", lambda w: w.view(r, cd).t().to(BF16)
-                          .contiguous())
-            dpooled = torch.empty_like(pooled)
-            e.pw_conv(dhz.view(n, r, 1, 1), w1t, dpooled.view(n, cd, 1, 1))
-            dw1p = torch.empty(r, cd, dtype=F32, device=dev)
-            e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, cd, 1, 1),
-                            dw1p)
-            if not _POOL_ON:
-                e.avgpool_bwd_add(dpooled, d_y2)
+            from .hip_ops import se_backward
+            d_y2, dw1p, db1se, dw2p, db2se = se_backward(
+                e, d_in3, y2, sw1, sw2, pooled, ha, sz)
             d_sw1 = dw1p.reshape_as(sw1)
             d_sw2 = dw2p.reshape_as(sw2)
         else:
@@ -277,11 +238,9 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2,
                          partial_ws=ws2, partial_splits=sp2)
-        elif has_se and _POOL_ON:
-            # pool-grad add + partial reduce in one pass over d_y2
-            e.bn_act_bwd_addpool(d_y2, z2, dpooled, mean2, invstd2, dg, db,
-                                 act, bool(training), dg2, db2_, dz2)
         else:
+            # (SE blocks: the avgpool backward contribution is already in
+            # d_y2 — the fused se_bwd kernel adds it while writing dx)
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)

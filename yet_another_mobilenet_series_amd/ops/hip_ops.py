@@ -334,74 +334,78 @@ def global_avg_pool(x):
     return _GlobalAvgPoolFn.apply(x)
 
 
-class _SqueezeExciteFn(torch.autograd.Function):
-    """SE block (K7): pool → FC+ReLU → FC+hsigmoid → channel scale.
+def se_pack(w1, w2):
+    """bf16 kernel packs: w1 [R,C], w2 [C,R] (cached per param version)."""
+    r, c = w1.shape[0], w1.shape[1]
+    w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous())
+    w2m = _packed(w2, "se2", lambda w: w.view(c, r).to(BF16).contiguous())
+    return w1m, w2m
 
-    The two FCs run on the pointwise GEMM at M = N (batch); pool/scale are
-    dedicated reduction/broadcast kernels.
+
+def se_forward(e, x, w1, b1, w2, b2, lop=None, ymat=None):
+    """One-kernel fused SE forward (K7, se.hip).  Returns
+    (y, pooled, ha, sz) with fp32 saved intermediates.
+
+    ``lop=(1, act, ab)`` applies the producing BN+act while loading x and
+    streams the activated input to ``ymat``.
+    """
+    n, c = x.shape[0], x.shape[1]
+    r = w1.shape[0]
+    w1m, w2m = se_pack(w1, w2)
+    y = torch.empty_like(x)
+    pooled = torch.empty(n, c, dtype=torch.float32, device=x.device)
+    ha = torch.empty(n, r, dtype=torch.float32, device=x.device)
+    sz = torch.empty(n, c, dtype=torch.float32, device=x.device)
+    if lop is None:
+        e.se_fwd(x, w1m, b1.float(), w2m, b2.float(), y, pooled, ha, sz)
+    else:
+        e.se_fwd(x, w1m, b1.float(), w2m, b2.float(), y, pooled, ha, sz,
+                 lop_kind=lop[0], lop_act=lop[1], lop_coef=lop[2],
+                 ymat=ymat)
+    return y, pooled, ha, sz
+
+
+def se_backward(e, dy, x, w1, w2, pooled, ha, sz):
+    """Fused SE backward: one per-image kernel (dx including the pool
+    gradient) + one deterministic cross-batch wgrad kernel.  All
+    SE-internal gradients stay fp32.  Returns (dx, dw1, db1, dw2, db2)."""
+    n, c = pooled.shape
+    r = ha.shape[1]
+    w1m, w2m = se_pack(w1, w2)
+    dx = torch.empty_like(x)
+    dsz = torch.empty(n, c, dtype=torch.float32, device=x.device)
+    dhz = torch.empty(n, r, dtype=torch.float32, device=x.device)
+    e.se_bwd(dy, x, w1m, w2m, ha, sz, dx, dsz, dhz)
+    dw1 = torch.empty(r, c, dtype=torch.float32, device=x.device)
+    dw2 = torch.empty(c, r, dtype=torch.float32, device=x.device)
+    db1 = torch.empty(r, dtype=torch.float32, device=x.device)
+    db2 = torch.empty(c, dtype=torch.float32, device=x.device)
+    e.se_wgrad(dsz, ha, dhz, pooled, dw1, dw2, db1, db2)
+    return dx, dw1, db1, dw2, db2
+
+
+class _SqueezeExciteFn(torch.autograd.Function):
+    """SE block (K7): ONE fused kernel per direction (se.hip) — pool,
+    FC1+ReLU, FC2+h-sigmoid and the channel scale in a single launch,
+    fp32 internal precision (VERDICT round-1 item 7).
     """
 
     @staticmethod
     def forward(ctx, x, w1, b1, w2, b2):
         e = ext()
         x = _nhwc(x)
-        n, c, h, wdt = x.shape
-        r = w1.shape[0]
-        pooled = torch.empty(n, c, dtype=BF16, device=x.device)
-        e.avgpool_fwd(x, pooled)
-        w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous())
-        hz = torch.empty(n, r, dtype=BF16, device=x.device)
-        e.pw_conv(pooled.view(n, c, 1, 1), w1m, hz.view(n, r, 1, 1))
-        ha = torch.empty_like(hz)
-        e.bias_act_fwd(hz, b1.float(), ACT_RELU, ha)
-        w2m = _packed(w2, "se", lambda w: w.view(c, r).to(BF16).contiguous())
-        sz = torch.empty(n, c, dtype=BF16, device=x.device)
-        e.pw_conv(ha.view(n, r, 1, 1), w2m, sz.view(n, c, 1, 1))
-        s = torch.empty_like(sz)
-        e.bias_act_fwd(sz, b2.float(), ACT_HSIGMOID, s)
-        y = torch.empty_like(x)
-        e.rowscale_fwd(x, s, y)
-        ctx.save_for_backward(x, pooled, hz, ha, sz, s, w1, b1, w2, b2)
+        y, pooled, ha, sz = se_forward(e, x, w1, b1, w2, b2)
+        ctx.save_for_backward(x, pooled, ha, sz, w1, w2)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         e = ext()
-        x, pooled, hz, ha, sz, s, w1, b1, w2, b2 = ctx.saved_tensors
-        n, c, h, wdt = x.shape
-        r = w1.shape[0]
+        x, pooled, ha, sz, w1, w2 = ctx.saved_tensors
         dy = dy.to(BF16).contiguous(memory_format=CL)
-        # d(x*s): dx_part = dy*s (broadcast), ds = sum_hw dy*x
-        dx = torch.empty_like(x)
-        ds = torch.empty(n, c, dtype=BF16, device=x.device)
-        e.rowscale_bwd(dy, x, s, dx, ds)
-        # hsigmoid gate backward
-        dsz = torch.empty_like(ds)
-        db2 = torch.empty(c, dtype=torch.float32, device=x.device)
-        e.bias_act_bwd(ds, sz, b2.float(), ACT_HSIGMOID, dsz, db2)
-        # fc2 backward
-        w2t = _packed(w2, "se_t",
-                      lambda w: w.view(c, r).t().to(BF16).contiguous())
-        dha = torch.empty_like(ha)
-        e.pw_conv(dsz.view(n, c, 1, 1), w2t, dha.view(n, r, 1, 1))
-        dw2p = torch.empty(c, r, dtype=torch.float32, device=x.device)
-        e.pw_conv_wgrad(dsz.view(n, c, 1, 1), ha.view(n, r, 1, 1), dw2p)
-        dw2 = dw2p.reshape_as(w2)
-        # relu backward
-        dhz = torch.empty_like(dha)
-        db1 = torch.empty(r, dtype=torch.float32, device=x.device)
-        e.bias_act_bwd(dha, hz, b1.float(), ACT_RELU, dhz, db1)
-        # fc1 backward
-        w1t = _packed(w1, "se_t",
-                      lambda w: w.view(r, c).t().to(BF16).contiguous())
-        dpooled = torch.empty_like(pooled)
-        e.pw_conv(dhz.view(n, r, 1, 1), w1t, dpooled.view(n, c, 1, 1))
-        dw1p = torch.empty(r, c, dtype=torch.float32, device=x.device)
-        e.pw_conv_wgrad(dhz.view(n, r, 1, 1), pooled.view(n, c, 1, 1), dw1p)
-        dw1 = dw1p.reshape_as(w1)
-        # pool backward folds into dx: dx += broadcast(dpooled)/(H*W)
-        e.avgpool_bwd_add(dpooled, dx)
-        return dx, dw1, db1, dw2, db2
+        dx, dw1, db1, dw2, db2 = se_backward(e, dy, x, w1, w2, pooled,
+                                             ha, sz)
+        return dx, dw1.reshape_as(w1), db1, dw2.reshape_as(w2), db2
 
 
 def squeeze_excite(x, w1, b1, w2, b2):
