@@ -565,3 +565,45 @@ def test_se_bwd_fused(ext, c, r, hw):
     assert_close(dw2, w2r.grad, rtol=1e-2)
     assert_close(db1, b1r.grad, rtol=1e-2)
     assert_close(db2, b2r.grad, rtol=1e-2)
+
+
+# ---- dropout (K14) --------------------------------------------------------
+
+def test_dropout_kernel(ext):
+    torch.manual_seed(11)
+    x = torch.randn(64, 1280, device=dev()).bfloat16().contiguous()
+    y = torch.empty_like(x)
+    p = 0.2
+    ext.dropout(x, y, 12345, p, forward=True)
+    scale = 1.0 / (1.0 - p)
+    zero = (y == 0)
+    frac = zero.float().mean().item()
+    assert abs(frac - p) < 0.02, frac  # ~Bernoulli(p) at 82k samples
+    kept = ~zero
+    assert torch.allclose(y[kept].float(), (x[kept].float() * scale)
+                          .bfloat16().float())
+    # backward regenerates the SAME mask from the seed
+    dy = torch.randn_like(x.float()).bfloat16()
+    dx = torch.empty_like(dy)
+    ext.dropout(dy, dx, 12345, p, forward=False)
+    assert (dx[zero] == 0).all()
+    assert torch.allclose(dx[kept].float(), (dy[kept].float() * scale)
+                          .bfloat16().float())
+    # a different seed gives a different mask
+    y2 = torch.empty_like(x)
+    ext.dropout(x, y2, 54321, p, forward=True)
+    assert not torch.equal(y, y2)
+
+
+def test_dropout_autograd_path():
+    from yet_another_mobilenet_series_amd.ops import functional as OF
+    torch.manual_seed(0)
+    x = torch.randn(32, 256, device=dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = OF.dropout(x, 0.5, training=True)
+    dy = torch.ones_like(y)
+    y.backward(dy)
+    # grad is nonzero exactly where the output was kept
+    assert torch.equal((x.grad != 0), (y != 0).detach())
+    # eval mode: identity
+    assert OF.dropout(x, 0.5, training=False) is x

@@ -12,7 +12,7 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, ZooModel, make_divisible)
+                             LinearAct, Dropout, ZooModel, make_divisible)
 
 # (expand_ratio t, out_ch c, repeats n, stride s, kernel k, se_ratio_of_cin)
 _CFG = [(6, 24, 2, 2, 3, 0.0), (3, 40, 3, 2, 5, 0.25), (6, 80, 4, 2, 3, 0.0),
@@ -44,7 +44,7 @@ class MNasNetA1(ZooModel):
         layers.append(ConvBNAct(cin, 1280, 1, act="relu", **bn_kw))
         self.features = nn.Sequential(*layers)
         self.pool = GlobalAvgPool()
-        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self.dropout = Dropout(dropout) if dropout > 0 else nn.Identity()
         self.classifier = LinearAct(1280, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

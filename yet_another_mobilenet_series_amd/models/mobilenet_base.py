@@ -136,6 +136,22 @@ class GlobalAvgPool(nn.Module):
         return OF.global_avg_pool(x)
 
 
+class Dropout(nn.Module):
+    """Head dropout routed through the op layer (K14): HIP counter-hash
+    kernel on GPU, torch dropout on CPU.  State-dict compatible with
+    nn.Dropout (both are parameter-free)."""
+
+    def __init__(self, p: float) -> None:
+        super().__init__()
+        self.p = float(p)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return OF.dropout(x, self.p, self.training)
+
+    def extra_repr(self) -> str:
+        return f"p={self.p}"
+
+
 class LinearAct(nn.Module):
     """FC layer with optional fused activation (the V3 head's post-pool
     1x1-no-BN convs are expressed as Linear: spatial is 1x1 there)."""
@@ -151,4 +167,4 @@ class LinearAct(nn.Module):
 
 
 __all__ = ["ZooModel", "ConvBNAct", "SqueezeExcite", "InvertedResidual",
-           "GlobalAvgPool", "LinearAct", "make_divisible"]
+           "GlobalAvgPool", "LinearAct", "Dropout", "make_divisible"]

@@ -10,7 +10,7 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, LinearAct,
-                             ZooModel, make_divisible)
+                             Dropout, ZooModel, make_divisible)
 
 # (out_channels, stride) for each depthwise-separable stage
 _CFG = [(64, 1), (128, 2), (128, 1), (256, 2), (256, 1), (512, 2),
@@ -35,7 +35,7 @@ class MobileNetV1(ZooModel):
             cin = cout
         self.features = nn.Sequential(*layers)
         self.pool = GlobalAvgPool()
-        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self.dropout = Dropout(dropout) if dropout > 0 else nn.Identity()
         self.classifier = LinearAct(cin, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -9,7 +9,7 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, ZooModel, make_divisible)
+                             LinearAct, Dropout, ZooModel, make_divisible)
 
 # (expand_ratio t, out_channels c, repeats n, first stride s)
 _CFG = [(1, 16, 1, 1), (6, 24, 2, 2), (6, 32, 3, 2), (6, 64, 4, 2),
@@ -37,7 +37,7 @@ class MobileNetV2(ZooModel):
         layers.append(ConvBNAct(cin, last, 1, act="relu6", **bn_kw))
         self.features = nn.Sequential(*layers)
         self.pool = GlobalAvgPool()
-        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self.dropout = Dropout(dropout) if dropout > 0 else nn.Identity()
         self.classifier = LinearAct(last, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

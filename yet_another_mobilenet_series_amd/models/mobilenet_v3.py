@@ -14,7 +14,7 @@ import torch
 import torch.nn as nn
 
 from .mobilenet_base import (ConvBNAct, GlobalAvgPool, InvertedResidual,
-                             LinearAct, ZooModel, make_divisible)
+                             LinearAct, Dropout, ZooModel, make_divisible)
 
 # (kernel, expand_ch, out_ch, use_se, act, stride)
 _LARGE: List[Tuple[int, int, int, bool, str, int]] = [
@@ -76,7 +76,7 @@ class MobileNetV3(ZooModel):
         self.pool = GlobalAvgPool()
         # paper: "conv 1x1, NBN" head at 1x1 spatial == Linear with bias
         self.head = LinearAct(lp, head_ch, act="hswish")
-        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self.dropout = Dropout(dropout) if dropout > 0 else nn.Identity()
         self.classifier = LinearAct(head_ch, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

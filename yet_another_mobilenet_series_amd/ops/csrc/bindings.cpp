@@ -304,6 +304,22 @@ void se_wgrad(const Tensor& dsz, const Tensor& ha, const Tensor& dhz,
                  cur_stream());
 }
 
+// ---- dropout (K14) -----------------------------------------------------
+
+void dropout(const Tensor& x, Tensor& y, int64_t seed, double p,
+             bool forward) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16,
+              "dropout expects bf16 CUDA input");
+  TORCH_CHECK(x.strides() == y.strides() && x.numel() == y.numel() &&
+                  y.scalar_type() == torch::kBFloat16,
+              "dropout out mismatch");
+  TORCH_CHECK(nhwc_ok(x) || x.is_contiguous(),
+              "dropout input must be contiguous");
+  TORCH_CHECK(p >= 0.0 && p < 1.0, "dropout p must be in [0, 1)");
+  yams::dropout(x.data_ptr(), y.data_ptr(), x.numel(), uint64_t(seed),
+                float(p), forward, cur_stream());
+}
+
 // ---- stem: im2row (conv itself runs on the GEMM kernels) ---------------
 
 void im2row(const Tensor& x, Tensor& patches, int64_t k, int64_t stride,
@@ -660,6 +676,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dx"_a, "dsz"_a, "dhz"_a);
   m.def("se_wgrad", &se_wgrad, "dsz"_a, "ha"_a, "dhz"_a, "pooled"_a,
         "dw1"_a, "dw2"_a, "db1"_a, "db2"_a);
+  m.def("dropout", &dropout, "x"_a, "y"_a, "seed"_a, "p"_a,
+        "forward"_a = true);
   m.def("im2row", &im2row);
   m.def("bn_stats", &bn_stats);
   m.def("bn_finalize", &bn_finalize, "ws"_a, "splits"_a, "mean"_a,

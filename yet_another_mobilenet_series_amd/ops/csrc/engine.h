@@ -156,6 +156,11 @@ void se_wgrad(const float* dsz, const float* ha, const float* dhz,
               const float* pooled, float* dw1, float* dw2, float* db1,
               float* db2, int N, int C, int R, hipStream_t s);
 
+// --- dropout (K14): stateless counter-hash mask — backward regenerates
+// the mask from the same seed; y = x*mask/(1-p) fwd, dx = dy*mask/(1-p)
+void dropout(const void* x, void* y, int64_t total, uint64_t seed, float p,
+             bool forward, hipStream_t s);
+
 // --- loss (K11): label-smoothed softmax CE, mean over batch -------------
 // loss: fp32 scalar; dlogits fp32 [N, C]; rowloss ws fp32 [N]
 void label_smooth_ce(const void* logits, const int64_t* target,

@@ -317,7 +317,49 @@ __global__ void ce_final_kernel(const float* __restrict__ rowloss,
   if (threadIdx.x == 0) *loss = red[0] / N;
 }
 
+// ---- dropout (K14): stateless counter-based RNG ------------------------
+// Philox-style: the mask is a pure function of (seed, element index), so
+// backward regenerates it bit-exactly from the same seed — nothing is
+// stored, and runs are reproducible given the training seed (SURVEY.md
+// §5.2 determinism).  splitmix64 finalizer: passes practical uniformity
+// tests, one mul-xor chain per element.
+DEV_INLINE uint32_t yams_rand_u32(uint64_t seed, int64_t idx) {
+  uint64_t z = seed + uint64_t(idx) * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return uint32_t((z ^ (z >> 31)) >> 32);
+}
+
+template <bool FWD>
+__global__ void dropout_kernel(const ushort_t* __restrict__ x,
+                               ushort_t* __restrict__ y, int64_t total,
+                               uint64_t seed, uint32_t keep_thresh,
+                               float inv_keep) {
+  for (int64_t i = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < total; i += int64_t(gridDim.x) * blockDim.x) {
+    const bool keep = yams_rand_u32(seed, i) < keep_thresh;
+    y[i] = keep ? ushort_t(f2bf(bf2f(x[i]) * inv_keep)) : ushort_t(0);
+  }
+}
+
 }  // namespace
+
+void dropout(const void* x, void* y, int64_t total, uint64_t seed, float p,
+             bool forward, hipStream_t s) {
+  const float keep = 1.f - p;
+  const uint32_t thresh =
+      keep >= 1.f ? 0xFFFFFFFFu : uint32_t(keep * 4294967296.0);
+  const int blocks = int(std::min<int64_t>((total + 255) / 256, 4096));
+  // fwd and bwd run the SAME kernel: dx = dy * mask / keep
+  if (forward)
+    dropout_kernel<true><<<dim3(blocks), 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<ushort_t*>(y), total,
+        seed, thresh, 1.f / keep);
+  else
+    dropout_kernel<false><<<dim3(blocks), 256, 0, s>>>(
+        static_cast<const ushort_t*>(x), static_cast<ushort_t*>(y), total,
+        seed, thresh, 1.f / keep);
+}
 
 void bias_act_fwd(const void* z, const float* bias, int act, void* y,
                   int64_t M, int C, hipStream_t s) {

@@ -412,6 +412,50 @@ def squeeze_excite(x, w1, b1, w2, b2):
     return _SqueezeExciteFn.apply(x, w1, b1, w2, b2)
 
 
+_dropout_counter = [0]
+
+
+def _dropout_seed() -> int:
+    """Deterministic per-call seed: process seed x call counter.  Each
+    dropout call gets a fresh mask; a fixed torch seed reproduces the
+    whole sequence (determinism test)."""
+    _dropout_counter[0] += 1
+    return (torch.initial_seed() * 1000003 + _dropout_counter[0]) \
+        & 0x7FFFFFFFFFFFFFFF
+
+
+class _DropoutFn(torch.autograd.Function):
+    """K14: y = x*mask/(1-p); the mask is recomputed from the saved seed in
+    backward (counter-hash RNG in misc.hip) — no mask tensor stored."""
+
+    @staticmethod
+    def forward(ctx, x, p):
+        e = ext()
+        x = x.to(BF16)
+        x = (x.contiguous() if x.dim() == 2
+             else x.contiguous(memory_format=CL))
+        seed = _dropout_seed()
+        y = torch.empty_like(x)
+        e.dropout(x, y, seed, p, forward=True)
+        ctx.seed = seed
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        e = ext()
+        dy = dy.to(BF16)
+        dy = (dy.contiguous() if dy.dim() == 2
+              else dy.contiguous(memory_format=CL))
+        dx = torch.empty_like(dy)
+        e.dropout(dy, dx, ctx.seed, ctx.p, forward=False)
+        return dx, None
+
+
+def dropout(x, p):
+    return _DropoutFn.apply(x, p)
+
+
 class _LabelSmoothCEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, target, smoothing):
