@@ -607,3 +607,34 @@ def test_dropout_autograd_path():
     assert torch.equal((x.grad != 0), (y != 0).detach())
     # eval mode: identity
     assert OF.dropout(x, 0.5, training=False) is x
+
+
+def test_fused_repack_matches_python_packs(ext):
+    """After the optimizer mutates a param, refresh_packs must leave every
+    cached pack equal to a from-scratch Python repack."""
+    from yet_another_mobilenet_series_amd.ops.hip_ops import (
+        _packed, refresh_packs)
+    torch.manual_seed(2)
+    pw = torch.nn.Parameter(torch.randn(64, 32, 1, 1, device=dev()))
+    dw = torch.nn.Parameter(torch.randn(48, 1, 5, 5, device=dev()))
+    # build the caches
+    p1 = _packed(pw, "pw", lambda w: w.view(64, 32).to(BF16).contiguous(),
+                 spec="cast")
+    p2 = _packed(pw, "pw_t", lambda w: w.view(64, 32).t().to(BF16)
+                 .contiguous(), spec="t")
+    p3 = _packed(dw, "dw", lambda w: w.view(48, 25).t().to(BF16)
+                 .contiguous(), spec="t")
+    with torch.no_grad():
+        pw.mul_(1.5)
+        dw.add_(0.25)
+    refresh_packs([pw, dw])
+    torch.cuda.synchronize()
+    assert torch.equal(p1, pw.detach().view(64, 32).to(BF16))
+    assert torch.equal(p2, pw.detach().view(64, 32).t().to(BF16)
+                       .contiguous())
+    assert torch.equal(p3, dw.detach().view(48, 25).t().to(BF16)
+                       .contiguous())
+    # entries re-stamped: next _packed call is a cache HIT (same object)
+    q1 = _packed(pw, "pw", lambda w: w.view(64, 32).to(BF16).contiguous(),
+                 spec="cast")
+    assert q1 is p1

@@ -560,6 +560,38 @@ void check_f32(const Tensor& t) {
               "update tensors must be fp32 cuda dense");
 }
 
+void fused_repack(const std::vector<Tensor>& srcs,
+                  const std::vector<Tensor>& dsts,
+                  const std::vector<int64_t>& adims) {
+  const int maxc = yams::update_max_chunks();
+  const int n = int(srcs.size());
+  TORCH_CHECK(int(dsts.size()) == n && int(adims.size()) == n,
+              "fused_repack list length mismatch");
+  for (int base = 0; base < n; base += maxc) {
+    const int cnt = std::min(maxc, n - base);
+    const float* src[32];
+    void* dst[32];
+    int numel[32];
+    int adim[32];
+    for (int c = 0; c < cnt; ++c) {
+      const Tensor& s = srcs[base + c];
+      const Tensor& d = dsts[base + c];
+      TORCH_CHECK(s.scalar_type() == torch::kFloat32 &&
+                      d.scalar_type() == torch::kBFloat16 &&
+                      d.is_contiguous() && s.numel() == d.numel(),
+                  "fused_repack src/dst mismatch");
+      const int a = int(adims[base + c]);
+      TORCH_CHECK(a == 0 || (a > 0 && s.numel() % a == 0),
+                  "fused_repack bad adim");
+      src[c] = s.data_ptr<float>();
+      dst[c] = d.data_ptr();
+      numel[c] = int(s.numel());
+      adim[c] = a;
+    }
+    yams::fused_repack(src, dst, numel, adim, cnt, cur_stream());
+  }
+}
+
 void fused_sgd(const std::vector<Tensor>& params,
                const std::vector<Tensor>& grads,
                const std::vector<Tensor>& moms,
@@ -704,6 +736,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("avgpool_bwd_add", &avgpool_bwd_add);
   m.def("label_smooth_ce", &label_smooth_ce);
+  m.def("fused_repack", &fused_repack);
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_rmsprop_tf", &fused_rmsprop_tf);
   m.def("fused_ema", &fused_ema);

@@ -178,5 +178,9 @@ void fused_rmsprop_tf(const float** g, float** p, float** ms, float** mom,
                       hipStream_t s);
 void fused_ema(const float** v, float** s_, const int* numel, int count,
                float decay, hipStream_t st);
+// multi-tensor bf16 pack refresh: dst = bf16(src) flat (adim 0) or
+// dst[A,B] = bf16(src[B,A]^T) (adim = A) — one launch per ~32 packs
+void fused_repack(const float** src, void** dst, const int* numel,
+                  const int* adim, int count, hipStream_t s);
 
 }  // namespace yams

@@ -114,6 +114,42 @@ __global__ __launch_bounds__(TPB) void ema_kernel(ChunkTable t, float decay) {
   }
 }
 
+// multi-tensor bf16 weight-pack refresh: after the fused optimizer writes
+// the fp32 masters through raw pointers, ONE launch re-derives every
+// cached kernel-layout bf16 pack (flat cast, or 2D transpose-cast for the
+// dgrad/dw layouts) instead of ~95 per-tensor torch ops per step.
+struct CastTable {
+  const float* src[MAXC];
+  ushort_t* dst[MAXC];
+  int numel[MAXC];
+  int adim[MAXC];  // 0 = flat cast; else dst is [A, B] = src[B, A]^T
+  int start_block[MAXC + 1];
+  int count;
+};
+
+__global__ __launch_bounds__(TPB) void repack_kernel(CastTable t) {
+  int c = 0;
+  while (c + 1 < t.count && int(blockIdx.x) >= t.start_block[c + 1]) ++c;
+  const int nb = t.start_block[c + 1] - t.start_block[c];
+  const int lb = blockIdx.x - t.start_block[c];
+  const int n = t.numel[c];
+  const int A = t.adim[c];
+  const int B = A > 0 ? n / A : 0;
+  const float* src = t.src[c];
+  ushort_t* dst = t.dst[c];
+  for (int64_t i = (int64_t(lb) * TPB + threadIdx.x) * VEC; i < n;
+       i += int64_t(nb) * TPB * VEC) {
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      const int64_t j = i + e;
+      if (j >= n) break;
+      // weights are tiny (<1 MB total): the strided transpose read is
+      // L2-served; no LDS tiling needed
+      dst[j] = f2bf(A > 0 ? src[(j % A) * B + (j / A)] : src[j]);
+    }
+  }
+}
+
 }  // namespace
 
 int update_max_chunks() { return MAXC; }
@@ -162,6 +198,25 @@ void fused_rmsprop_tf(const float** g, float** p, float** ms, float** mom,
   const int blocks = fill_blocks(t);
   rmsprop_tf_kernel<<<dim3(blocks), TPB, 0, s>>>(t, lr, alpha, eps,
                                                  momentum);
+}
+
+void fused_repack(const float** src, void** dst, const int* numel,
+                  const int* adim, int count, hipStream_t s) {
+  CastTable t;
+  t.count = count;
+  int blocks = 0;
+  t.start_block[0] = 0;
+  for (int c = 0; c < count; ++c) {
+    t.src[c] = src[c];
+    t.dst[c] = static_cast<ushort_t*>(dst[c]);
+    t.numel[c] = numel[c];
+    t.adim[c] = adim[c];
+    const int nb = std::max(1, std::min(
+        64, int((int64_t(numel[c]) + TPB * VEC - 1) / (TPB * VEC))));
+    t.start_block[c + 1] = t.start_block[c] + nb;
+  }
+  blocks = t.start_block[count];
+  repack_kernel<<<dim3(blocks), TPB, 0, s>>>(t);
 }
 
 void fused_ema(const float** v, float** s_, const int* numel, int count,

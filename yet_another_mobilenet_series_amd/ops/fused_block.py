@@ -74,7 +74,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         if has_expand:
             ce = ew.shape[0]
             ewp = _packed(ew, "pw", lambda w: w.view(ce, cin).to(BF16)
-                          .contiguous())
+                          .contiguous(), spec="cast")
             z1 = _empty_like_nhwc(x, c=ce)
             m1 = n * h * wdt
             mean1, invstd1, ab1 = _bn_bufs(ce, dev)
@@ -103,7 +103,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         # ---- depthwise ------------------------------------------------
         cd = dww.shape[0]
         dwp = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
-                      .contiguous())
+                      .contiguous(), spec="t")
         ho = (h + 2 * pad - k) // stride + 1
         wo = (wdt + 2 * pad - k) // stride + 1
         z2 = torch.empty((n, cd, ho, wo), dtype=BF16, device=dev,
@@ -141,7 +141,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         # ---- project pw -----------------------------------------------
         cp = pw.shape[0]
         pwp = _packed(pw, "pw", lambda w: w.view(cp, cd).to(BF16)
-                      .contiguous())
+                      .contiguous(), spec="cast")
         z3 = torch.empty((n, cp, ho, wo), dtype=BF16, device=dev,
                          memory_format=CL)
         mean3, invstd3, ab3 = _bn_bufs(cp, dev)
@@ -198,7 +198,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                      bool(training), dg3, db3, dz3)
 
         pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
-                      .contiguous())
+                      .contiguous(), spec="t")
         d_in3 = torch.empty_like(z2)
         if not has_se and _EPI_ON:
             # d_in3 IS the dw-BN's backward dy: emit that BN's partial sums
@@ -244,7 +244,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
-                       .contiguous())
+                       .contiguous(), spec="t")
         dw_in = y1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
         if has_expand and stride == 1 and _EPI_ON:
@@ -274,7 +274,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                 e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
                              bool(training), dg1, db1_, dz1)
             ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
-                          .contiguous())
+                          .contiguous(), spec="t")
             d_x = _empty_like_nhwc(x)
             e.pw_conv(dz1, ewt, d_x)
             dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)

@@ -37,12 +37,22 @@ ACT_HSIGMOID = 4
 
 
 
-def _packed(param: torch.Tensor, key: str, fn):
+def _packed(param: torch.Tensor, key: str, fn, spec: Optional[str] = None):
     """Cache a packed/cast weight on the tensor, keyed by data version.
 
-    The pack ops re-run whenever the optimizer bumps the param version,
-    and ALWAYS during CUDA-graph capture (so replay re-reads the updated
-    fp32 master through the baked cast kernels)."""
+    ``spec`` records how the pack derives from the param so the fused
+    optimizers can refresh EVERY pack with one multi-tensor kernel per
+    step (:func:`refresh_packs`) instead of ~95 per-tensor torch ops:
+
+    - ``"cast"``: same element order, bf16 cast (pw/lin/SE packs — the
+      param is [A, B, 1, 1] or [A, B] and the pack its flat bf16 copy)
+    - ``"t"``: 2D transpose-cast — the pack is [A, B] bf16 from the
+      param's flat [B, A] fp32 order (dw [k*k,C], dgrad W^T packs)
+    - ``None``: custom (stem pad) — dropped on refresh, rebuilt lazily.
+
+    Packs re-run whenever the param version changes, and ALWAYS during
+    CUDA-graph capture (so replay re-reads the updated fp32 master
+    through the baked cast kernels)."""
     if param.is_cuda and torch.cuda.is_current_stream_capturing():
         return fn(param)
     cache = getattr(param, "_yams_pack", None)
@@ -57,8 +67,42 @@ def _packed(param: torch.Tensor, key: str, fn):
     if hit is not None and hit[0] == ver:
         return hit[1]
     t = fn(param)
-    cache[key] = (ver, t)
+    cache[key] = (ver, t, spec)
     return t
+
+
+def refresh_packs(params) -> None:
+    """Refresh every cached bf16 pack of ``params`` in place (one fused
+    kernel) and re-stamp the entries at the params' CURRENT versions.
+
+    Called by the fused optimizers right after they bump versions: without
+    it the next forward would rebuild ~95 packs through per-tensor torch
+    ops (measured ~2 ms/step on V3-Large)."""
+    srcs, dsts, adims = [], [], []
+    for p in params:
+        cache = getattr(p, "_yams_pack", None)
+        if not cache:
+            continue
+        ver = p._version
+        drop = []
+        for key, entry in cache.items():
+            _, t, spec = entry
+            if spec == "cast":
+                srcs.append(p)
+                dsts.append(t)
+                adims.append(0)
+            elif spec == "t":
+                srcs.append(p)
+                dsts.append(t)
+                adims.append(t.shape[0])
+            else:
+                drop.append(key)
+                continue
+            cache[key] = (ver, t, spec)
+        for key in drop:
+            del cache[key]
+    if srcs:
+        ext().fused_repack(srcs, dsts, adims)
 
 def _nhwc(x: torch.Tensor) -> torch.Tensor:
     """bf16 + channels_last view of a 4D activation tensor."""
@@ -115,10 +159,10 @@ class _ConvBNActFn(torch.autograd.Function):
         # pack weights into kernel layout (bf16)
         if kind == "pw":
             w_packed = _packed(weight, "pw", lambda w: w.view(
-                cout, cin).to(BF16).contiguous())
+                cout, cin).to(BF16).contiguous(), spec="cast")
         elif kind == "dw":
             w_packed = _packed(weight, "dw", lambda w: w.view(
-                cout, k * k).t().to(BF16).contiguous())
+                cout, k * k).t().to(BF16).contiguous(), spec="t")
         elif kind == "stem":
             # [Cout, 3, k, k] -> [Cout, k*k*3 padded to /8] matching im2row
             kp = (k * k * cin + 7) // 8 * 8
@@ -127,7 +171,7 @@ class _ConvBNActFn(torch.autograd.Function):
                 out = torch.zeros(cout, kp, dtype=BF16, device=w.device)
                 out[:, :k * k * cin] = flat.to(BF16)
                 return out
-            w_packed = _packed(weight, "stem", _pack_stem)
+            w_packed = _packed(weight, "stem", _pack_stem, spec=None)
         else:
             raise ValueError(kind)
 
@@ -190,12 +234,12 @@ class _ConvBNActFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             if kind == "pw":
                 w_t = _packed(weight, "pw_t", lambda w: w.view(
-                    cout, cin).t().to(BF16).contiguous())  # [Cin,Cout]
+                    cout, cin).t().to(BF16).contiguous(), spec="t")
                 dx = _empty_like_nhwc(x)
                 e.pw_conv(dz, w_t, dx)
             elif kind == "dw":
                 w_packed = _packed(weight, "dw", lambda w: w.view(
-                    cout, k * k).t().to(BF16).contiguous())
+                    cout, k * k).t().to(BF16).contiguous(), spec="t")
                 dx = _empty_like_nhwc(x)
                 e.dw_conv_dgrad(dz, w_packed, dx, k, s, p)
             else:
@@ -267,7 +311,7 @@ class _LinearActFn(torch.autograd.Function):
         if x2.dim() != 2:
             x2 = x2.view(x2.shape[0], -1)
         nout, nin = weight.shape
-        wt = _packed(weight, "lin", lambda w: w.to(BF16).contiguous())
+        wt = _packed(weight, "lin", lambda w: w.to(BF16).contiguous(), spec="cast")
         z = torch.empty(x2.shape[0], nout, dtype=BF16, device=x.device)
         e.pw_conv(x2.view(x2.shape[0], nin, 1, 1), wt,
                   z.view(z.shape[0], nout, 1, 1))
@@ -293,7 +337,7 @@ class _LinearActFn(torch.autograd.Function):
         dx = None
         if ctx.needs_input_grad[0]:
             w = _packed(weight, "lin_t",
-                        lambda wt_: wt_.t().to(BF16).contiguous())
+                        lambda wt_: wt_.t().to(BF16).contiguous(), spec="t")
             dx = torch.empty_like(x2)
             e.pw_conv(dz.view(-1, nout, 1, 1), w, dx.view(-1, nin, 1, 1))
         dweight = None
@@ -337,8 +381,10 @@ def global_avg_pool(x):
 def se_pack(w1, w2):
     """bf16 kernel packs: w1 [R,C], w2 [C,R] (cached per param version)."""
     r, c = w1.shape[0], w1.shape[1]
-    w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous())
-    w2m = _packed(w2, "se2", lambda w: w.view(c, r).to(BF16).contiguous())
+    w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous(),
+                  spec="cast")
+    w2m = _packed(w2, "se2", lambda w: w.view(c, r).to(BF16).contiguous(),
+                  spec="cast")
     return w1m, w2m
 
 

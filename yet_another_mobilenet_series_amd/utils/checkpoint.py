@@ -53,7 +53,7 @@ def _load_rng_state(state: Dict) -> None:
     nps = state["numpy"]
     if isinstance(nps, dict):
         np.random.set_state((nps["name"],
-                             nps["key"].numpy().astype(np.uint32),
+                             nps["key"].cpu().numpy().astype(np.uint32),
                              nps["pos"], nps["has_gauss"], nps["cached"]))
     else:
         np.random.set_state(nps)
