@@ -134,7 +134,6 @@ __global__ __launch_bounds__(TPB) void repack_kernel(CastTable t) {
   const int lb = blockIdx.x - t.start_block[c];
   const int n = t.numel[c];
   const int A = t.adim[c];
-  const int B = A > 0 ? n / A : 0;
   const float* src = t.src[c];
   ushort_t* dst = t.dst[c];
   for (int64_t i = (int64_t(lb) * TPB + threadIdx.x) * VEC; i < n;
@@ -144,8 +143,11 @@ __global__ __launch_bounds__(TPB) void repack_kernel(CastTable t) {
       const int64_t j = i + e;
       if (j >= n) break;
       // weights are tiny (<1 MB total): the strided transpose read is
-      // L2-served; no LDS tiling needed
-      dst[j] = f2bf(A > 0 ? src[(j % A) * B + (j / A)] : src[j]);
+      // L2-served; no LDS tiling needed.
+      // dst[a, b] = src[b, a]: a = j / B_cols, b = j % B_cols, where dst
+      // is [A, n/A] — so B_cols = n / A
+      const int bc = A > 0 ? n / A : 1;
+      dst[j] = f2bf(A > 0 ? src[(j % bc) * A + (j / bc)] : src[j]);
     }
   }
 }
