@@ -24,22 +24,32 @@ namespace yams {
 
 namespace {
 
-constexpr int MAX_C = 2048;  // fp32 LDS arrays sized for the zoo's C<=960
+constexpr int MAX_C = 1024;  // fp32 LDS arrays sized for the zoo's C<=960
 constexpr int MAX_R = 256;
+constexpr int BS = 1024;     // threads/block: 16 waves per CU so the
+                             // grid-N streaming phases hide HBM latency
+                             // (256-thread version measured ~4x off the
+                             // stream floor at 25% occupancy)
+
+DEV_INLINE int pow2_le(int v) {
+  int m = 1;
+  while (m * 2 <= v) m *= 2;
+  return m;
+}
 
 // lopx: LOP_NONE (x is the already-activated SE input) or LOP_BNACT
 // (x is the dw-conv output z; BN+act applied on load, the activated value
 // also streamed to `ymat` so the project conv / backward can reuse it —
 // this folds the separate bn_act_fwd pass into the SE kernel).
 template <int LOPK>
-__global__ __launch_bounds__(256) void se_fwd_kernel(
+__global__ __launch_bounds__(BS) void se_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w1,
     const float* __restrict__ b1, const ushort_t* __restrict__ w2,
     const float* __restrict__ b2, ushort_t* __restrict__ y,
     ushort_t* __restrict__ ymat, float* __restrict__ pooled_g,
     float* __restrict__ ha_g, float* __restrict__ sz_g, int HW, int C,
     int R, Lop lopx) {
-  __shared__ float red[256 * 8];
+  __shared__ float red[BS * 8];
   __shared__ float pooled_s[MAX_C];
   __shared__ float ha_s[MAX_R];
   __shared__ float s_s[MAX_C];
@@ -47,7 +57,7 @@ __global__ __launch_bounds__(256) void se_fwd_kernel(
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   const int C8 = C >> 3;
-  const int ML = 256 / C8;           // hw lanes (C8 <= 120 => ML >= 2)
+  const int ML = pow2_le(BS / C8);   // pow2 hw/r lanes (tree fold)
   const int c8 = tid % C8;
   const int mlane = tid / C8;
   const int c = c8 * 8;
@@ -82,49 +92,86 @@ __global__ __launch_bounds__(256) void se_fwd_kernel(
         *reinterpret_cast<short8_t*>(&ymat[off]) = m8;
     }
   }
+  if (mlane < ML)
 #pragma unroll
-  for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
+    for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
   __syncthreads();
+  for (int half = ML >> 1; half > 0; half >>= 1) {
+    if (mlane < half)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+    __syncthreads();
+  }
   if (tid < C8) {
     const float inv_hw = 1.f / float(HW);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      float t = 0.f;
-      for (int l = 0; l < ML; ++l) t += red[(tid + l * C8) * 8 + e];
-      t *= inv_hw;
+      const float t = red[tid * 8 + e] * inv_hw;
       pooled_s[tid * 8 + e] = t;
       pooled_g[int64_t(n) * C + tid * 8 + e] = t;
     }
   }
   __syncthreads();
 
-  // ---- phase 2: ha = relu(pooled @ w1^T + b1)  (thread = one r) --------
-  if (tid < R) {
-    float acc1 = b1[tid];
-    const ushort_t* wrow = &w1[int64_t(tid) * C];
-    for (int cc = 0; cc < C; cc += 8) {
-      short8_t w8 = *reinterpret_cast<const short8_t*>(&wrow[cc]);
+  // ---- phase 2: ha = relu(pooled @ w1^T + b1) --------------------------
+  // thread (r = tid%256, rl = tid/256): RL-way split of the C reduction
+  {
+    const int r = tid % 256;
+    const int rl = tid / 256;
+    constexpr int RL = BS / 256;
+    float acc1 = 0.f;
+    if (r < R) {
+      const ushort_t* wrow = &w1[int64_t(r) * C];
+      for (int cc = rl * 8; cc < C; cc += RL * 8) {
+        short8_t w8 = *reinterpret_cast<const short8_t*>(&wrow[cc]);
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        acc1 = fmaf(pooled_s[cc + e], bf2f(ushort_t(w8[e])), acc1);
+        for (int e = 0; e < 8; ++e)
+          acc1 = fmaf(pooled_s[cc + e], bf2f(ushort_t(w8[e])), acc1);
+      }
     }
-    acc1 = fmaxf(acc1, 0.f);
-    ha_s[tid] = acc1;
-    ha_g[int64_t(n) * R + tid] = acc1;
+    red[tid] = acc1;
+    __syncthreads();
+    if (rl == 0 && r < R) {
+#pragma unroll
+      for (int l = 1; l < RL; ++l) acc1 += red[r + l * 256];
+      acc1 = fmaxf(acc1 + b1[r], 0.f);
+      ha_s[r] = acc1;
+      ha_g[int64_t(n) * R + r] = acc1;
+    }
   }
   __syncthreads();
 
-  // ---- phase 3: s = hsigmoid(ha @ w2^T + b2)  (thread = 8 channels) ----
-  if (tid < C8) {
-#pragma unroll 1
-    for (int e = 0; e < 8; ++e) {
-      const int cc = tid * 8 + e;
-      float acc2 = b2[cc];
-      const ushort_t* wrow = &w2[int64_t(cc) * R];
-      for (int r = 0; r < R; ++r)
-        acc2 = fmaf(ha_s[r], bf2f(wrow[r]), acc2);
-      sz_g[int64_t(n) * C + cc] = acc2;
-      s_s[cc] = fminf(fmaxf(acc2 + 3.f, 0.f), 6.f) * (1.f / 6.f);
+  // ---- phase 3: s = hsigmoid(ha @ w2^T + b2) ---------------------------
+  // thread (c8, mlane): ML-way split of the R reduction, tree fold
+  {
+    float acc2[8] = {};
+    if (mlane < ML) {
+      for (int r = mlane; r < R; r += ML) {
+        const float hv = ha_s[r];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc2[e] = fmaf(hv, bf2f(w2[int64_t(c + e) * R + r]), acc2[e]);
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc2[e];
+    }
+    __syncthreads();
+    for (int half = ML >> 1; half > 0; half >>= 1) {
+      if (mlane < half)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+      __syncthreads();
+    }
+    if (tid < C8) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int cc = tid * 8 + e;
+        const float sz = red[tid * 8 + e] + b2[cc];
+        sz_g[int64_t(n) * C + cc] = sz;
+        s_s[cc] = fminf(fmaxf(sz + 3.f, 0.f), 6.f) * (1.f / 6.f);
+      }
     }
   }
   __syncthreads();
@@ -132,7 +179,7 @@ __global__ __launch_bounds__(256) void se_fwd_kernel(
   // ---- phase 4: y = x * s (flat vec8 walk, fully coalesced) ------------
   const ushort_t* src = LOPK == LOP_BNACT ? ymat : x;
   const int total8 = HW * C8;
-  for (int i = tid; i < total8; i += 256) {
+  for (int i = tid; i < total8; i += BS) {
     const int cc = (i % C8) * 8;
     const int64_t off = base + int64_t(i) * 8;
     short8_t v8 = *reinterpret_cast<const short8_t*>(&src[off]);
@@ -146,13 +193,13 @@ __global__ __launch_bounds__(256) void se_fwd_kernel(
 
 // backward per-image pass: dx (incl. the avgpool backward contribution),
 // dsz[N,C] and dhz[N,R] staged fp32 for the cross-batch wgrad kernel.
-__global__ __launch_bounds__(256) void se_bwd_kernel(
+__global__ __launch_bounds__(BS) void se_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ w1, const ushort_t* __restrict__ w2,
     const float* __restrict__ ha_g, const float* __restrict__ sz_g,
     ushort_t* __restrict__ dx, float* __restrict__ dsz_g,
     float* __restrict__ dhz_g, int HW, int C, int R) {
-  __shared__ float red[256 * 8];
+  __shared__ float red[BS * 8];
   __shared__ float dsz_s[MAX_C];
   __shared__ float dhz_s[MAX_R];
   __shared__ float s_s[MAX_C];
@@ -161,7 +208,7 @@ __global__ __launch_bounds__(256) void se_bwd_kernel(
   const int n = blockIdx.x;
   const int tid = threadIdx.x;
   const int C8 = C >> 3;
-  const int ML = 256 / C8;
+  const int ML = pow2_le(BS / C8);
   const int c8 = tid % C8;
   const int mlane = tid / C8;
   const int c = c8 * 8;
@@ -178,20 +225,25 @@ __global__ __launch_bounds__(256) void se_bwd_kernel(
       for (int e = 0; e < 8; ++e)
         acc[e] = fmaf(bf2f(ushort_t(g8[e])), bf2f(ushort_t(v8[e])), acc[e]);
     }
-  }
 #pragma unroll
-  for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
+    for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
+  }
   __syncthreads();
+  for (int half = ML >> 1; half > 0; half >>= 1) {
+    if (mlane < half)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+    __syncthreads();
+  }
   // ---- phase 2: gate backward: dsz = ds * hsig'(sz); s for phase 5 -----
   if (tid < C8) {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      float ds = 0.f;
-      for (int l = 0; l < ML; ++l) ds += red[(tid + l * C8) * 8 + e];
       const int cc = tid * 8 + e;
       const float sz = sz_g[int64_t(n) * C + cc];
-      const float dsz =
-          ds * ((sz > -3.f && sz < 3.f) ? (1.f / 6.f) : 0.f);
+      const float dsz = red[tid * 8 + e] *
+                        ((sz > -3.f && sz < 3.f) ? (1.f / 6.f) : 0.f);
       dsz_s[cc] = dsz;
       dsz_g[int64_t(n) * C + cc] = dsz;
       s_s[cc] = fminf(fmaxf(sz + 3.f, 0.f), 6.f) * (1.f / 6.f);
@@ -200,36 +252,64 @@ __global__ __launch_bounds__(256) void se_bwd_kernel(
   __syncthreads();
 
   // ---- phase 3: dhz[r] = relu'(ha) * sum_c dsz[c]*w2[c,r] --------------
-  if (tid < R) {
+  // thread (r = tid%256, rl): RL-way split of the C reduction; for a
+  // fixed c the w2[c*R + r] reads are contiguous across the r lanes
+  {
+    const int r = tid % 256;
+    const int rl = tid / 256;
+    constexpr int RL = BS / 256;
     float acc1 = 0.f;
-    for (int cc = 0; cc < C; ++cc)
-      acc1 = fmaf(dsz_s[cc], bf2f(w2[int64_t(cc) * R + tid]), acc1);
-    if (ha_g[int64_t(n) * R + tid] <= 0.f) acc1 = 0.f;
-    dhz_s[tid] = acc1;
-    dhz_g[int64_t(n) * R + tid] = acc1;
+    if (r < R) {
+      for (int cc = rl; cc < C; cc += RL)
+        acc1 = fmaf(dsz_s[cc], bf2f(w2[int64_t(cc) * R + r]), acc1);
+    }
+    red[tid] = acc1;
+    __syncthreads();
+    if (rl == 0 && r < R) {
+#pragma unroll
+      for (int l = 1; l < RL; ++l) acc1 += red[r + l * 256];
+      if (ha_g[int64_t(n) * R + r] <= 0.f) acc1 = 0.f;
+      dhz_s[r] = acc1;
+      dhz_g[int64_t(n) * R + r] = acc1;
+    }
   }
   __syncthreads();
 
   // ---- phase 4: dpooled[c]/HW = (sum_r dhz[r]*w1[r,c]) / HW ------------
-  if (tid < C8) {
+  {
     float acc2[8] = {};
-    for (int r = 0; r < R; ++r) {
-      const float dv = dhz_s[r];
-      short8_t w8 = *reinterpret_cast<const short8_t*>(
-          &w1[int64_t(r) * C + c]);
+    if (mlane < ML) {
+      for (int r = mlane; r < R; r += ML) {
+        const float dv = dhz_s[r];
+        short8_t w8 = *reinterpret_cast<const short8_t*>(
+            &w1[int64_t(r) * C + c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc2[e] = fmaf(dv, bf2f(ushort_t(w8[e])), acc2[e]);
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc2[e];
+    }
+    __syncthreads();
+    for (int half = ML >> 1; half > 0; half >>= 1) {
+      if (mlane < half)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+      __syncthreads();
+    }
+    if (tid < C8) {
+      const float inv_hw = 1.f / float(HW);
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        acc2[e] = fmaf(dv, bf2f(ushort_t(w8[e])), acc2[e]);
+        dpool_s[tid * 8 + e] = red[tid * 8 + e] * inv_hw;
     }
-    const float inv_hw = 1.f / float(HW);
-#pragma unroll
-    for (int e = 0; e < 8; ++e) dpool_s[tid * 8 + e] = acc2[e] * inv_hw;
   }
   __syncthreads();
 
   // ---- phase 5: dx = dy*s + dpooled/HW (flat vec8 walk) ----------------
   const int total8 = HW * C8;
-  for (int i = tid; i < total8; i += 256) {
+  for (int i = tid; i < total8; i += BS) {
     const int cc = (i % C8) * 8;
     const int64_t off = base + int64_t(i) * 8;
     short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
@@ -242,41 +322,62 @@ __global__ __launch_bounds__(256) void se_bwd_kernel(
   }
 }
 
-// cross-batch weight/bias grads, deterministic (serial n loop, fixed
+// cross-batch weight/bias grads, deterministic (serial n chunks, fixed
 // order; no atomics — SURVEY.md §5.2):
 //   dw1[r,c] = sum_n dhz[n,r]*pooled[n,c]   dw2[c,r] = sum_n dsz[n,c]*ha[n,r]
 //   db1[r]   = sum_n dhz[n,r]               db2[c]   = sum_n dsz[n,c]
 // grid: 16x16 tiles of dw1, then of dw2, then one block for the biases.
+// Both operand columns of a tile are staged through LDS in NT-deep chunks
+// (the untiled version re-read ~1 GB/call from L2 at the 960x240 shape —
+// measured 182 us; tiling cuts global traffic 16x).
+constexpr int NT = 64;  // n rows per LDS pass
+
 __global__ __launch_bounds__(256) void se_wgrad_kernel(
     const float* __restrict__ dsz, const float* __restrict__ ha,
     const float* __restrict__ dhz, const float* __restrict__ pooled,
     float* __restrict__ dw1, float* __restrict__ dw2,
     float* __restrict__ db1, float* __restrict__ db2, int N, int C, int R,
     int t1) {
-  const int rt = (R + 15) / 16;
+  __shared__ float al[NT][16];
+  __shared__ float bl[NT][16 + 1];
   const int bid = blockIdx.x;
   const int i = int(threadIdx.x) / 16;   // row lane
   const int j = int(threadIdx.x) % 16;   // col lane
-  if (bid < t1) {  // dw1 tile: rows = r, cols = c
-    const int r = (bid % rt) * 16 + i;
-    const int c = (bid / rt) * 16 + j;
-    if (r < R && c < C) {
-      float acc = 0.f;
-      for (int n = 0; n < N; ++n)
-        acc = fmaf(dhz[int64_t(n) * R + r], pooled[int64_t(n) * C + c],
-                   acc);
-      dw1[int64_t(r) * C + c] = acc;
+  if (bid < 2 * t1) {
+    // a = [N, Ar] operand giving tile rows, b = [N, Bc] giving tile cols
+    const float* a;
+    const float* b;
+    int Ar, Bc, r0, c0;
+    if (bid < t1) {  // dw1 tile: rows = r (of dhz), cols = c (of pooled)
+      const int rt = (R + 15) / 16;
+      a = dhz; b = pooled; Ar = R; Bc = C;
+      r0 = (bid % rt) * 16; c0 = (bid / rt) * 16;
+    } else {         // dw2 tile: rows = c (of dsz), cols = r (of ha)
+      const int ct = (C + 15) / 16;
+      const int bb = bid - t1;
+      a = dsz; b = ha; Ar = C; Bc = R;
+      r0 = (bb % ct) * 16; c0 = (bb / ct) * 16;
     }
-  } else if (bid < 2 * t1) {  // dw2 tile: rows = c, cols = r
-    const int b = bid - t1;
-    const int ct = (C + 15) / 16;
-    const int c = (b % ct) * 16 + i;
-    const int r = (b / ct) * 16 + j;
-    if (c < C && r < R) {
-      float acc = 0.f;
-      for (int n = 0; n < N; ++n)
-        acc = fmaf(dsz[int64_t(n) * C + c], ha[int64_t(n) * R + r], acc);
-      dw2[int64_t(c) * R + r] = acc;
+    float acc = 0.f;
+    for (int n0 = 0; n0 < N; n0 += NT) {
+      // cooperative stage: 256 threads = 16 n-rows x 16 cols per round
+      __syncthreads();
+      for (int nn = i; nn < NT; nn += 16) {
+        const int n = n0 + nn;
+        al[nn][j] = (n < N && r0 + j < Ar)
+                        ? a[int64_t(n) * Ar + r0 + j] : 0.f;
+        bl[nn][j] = (n < N && c0 + j < Bc)
+                        ? b[int64_t(n) * Bc + c0 + j] : 0.f;
+      }
+      __syncthreads();
+      const int hi = N - n0 < NT ? N - n0 : NT;
+      for (int nn = 0; nn < hi; ++nn)
+        acc = fmaf(al[nn][i], bl[nn][j], acc);
+    }
+    if (bid < t1) {
+      if (r0 + i < R && c0 + j < C) dw1[int64_t(r0 + i) * C + c0 + j] = acc;
+    } else {
+      if (r0 + i < C && c0 + j < R) dw2[int64_t(r0 + i) * R + c0 + j] = acc;
     }
   } else {  // biases
     for (int idx = int(threadIdx.x); idx < R + C; idx += 256) {
@@ -312,10 +413,10 @@ void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
   auto* yp = static_cast<ushort_t*>(y);
   auto* mp = static_cast<ushort_t*>(ymat);
   if (lopx.kind == LOP_NONE)
-    se_fwd_kernel<LOP_NONE><<<dim3(N), 256, 0, s>>>(
+    se_fwd_kernel<LOP_NONE><<<dim3(N), BS, 0, s>>>(
         xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
   else if (lopx.kind == LOP_BNACT && ymat != nullptr)
-    se_fwd_kernel<LOP_BNACT><<<dim3(N), 256, 0, s>>>(
+    se_fwd_kernel<LOP_BNACT><<<dim3(N), BS, 0, s>>>(
         xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
   else
     yams_launch_fail("se_fwd", lopx.kind, lopx.act, ymat != nullptr, 0);
@@ -326,7 +427,7 @@ void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
             float* dhz, int N, int HW, int C, int R, hipStream_t s) {
   if ((C & 7) || C > 960 || R > MAX_R)
     yams_launch_fail("se_bwd_shape", C, R, HW, 0);
-  se_bwd_kernel<<<dim3(N), 256, 0, s>>>(
+  se_bwd_kernel<<<dim3(N), BS, 0, s>>>(
       static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
       static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
       ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
