@@ -481,8 +481,8 @@ def test_se_fwd_fused(ext, c, r, hw):
     pooled = torch.empty(5, c, dtype=torch.float32, device=dev())
     ha = torch.empty(5, r, dtype=torch.float32, device=dev())
     sz = torch.empty(5, c, dtype=torch.float32, device=dev())
-    ext.se_fwd(x, w1.bfloat16().contiguous(), b1, w2.bfloat16().contiguous(),
-               b2, y, pooled, ha, sz)
+    ext.se_fwd(x, w1.bfloat16().t().contiguous(), b1,
+               w2.bfloat16().t().contiguous(), b2, y, pooled, ha, sz)
     # reference from the same bf16-rounded operands
     yr, pr, har, szr, _ = _se_ref(x.float(), w1.bfloat16().float(), b1,
                                   w2.bfloat16().float(), b2)
@@ -508,8 +508,9 @@ def test_se_fwd_with_bnact_lop(ext):
     pooled = torch.empty(4, c, dtype=torch.float32, device=dev())
     ha = torch.empty(4, r, dtype=torch.float32, device=dev())
     sz = torch.empty(4, c, dtype=torch.float32, device=dev())
-    ext.se_fwd(z, w1, b1, w2, b2, y, pooled, ha, sz,
-               lop_kind=1, lop_act=3, lop_coef=ab, ymat=ymat)
+    ext.se_fwd(z, w1.t().contiguous(), b1, w2.t().contiguous(), b2, y,
+               pooled, ha, sz, lop_kind=1, lop_act=3, lop_coef=ab,
+               ymat=ymat)
     zv = z.float()
     v = zv * ab[:c].view(1, c, 1, 1) + ab[c:].view(1, c, 1, 1)
     act = v * torch.clamp(v + 3, 0, 6) / 6  # hswish
@@ -539,8 +540,8 @@ def test_se_bwd_fused(ext, c, r, hw):
     pooled = torch.empty(n, c, dtype=torch.float32, device=dev())
     ha = torch.empty(n, r, dtype=torch.float32, device=dev())
     sz = torch.empty(n, c, dtype=torch.float32, device=dev())
-    ext.se_fwd(x, w1.contiguous(), b1, w2.contiguous(), b2, y, pooled, ha,
-               sz)
+    ext.se_fwd(x, w1.t().contiguous(), b1, w2.t().contiguous(), b2, y,
+               pooled, ha, sz)
     dx = torch.empty_like(x)
     dsz = torch.empty(n, c, dtype=torch.float32, device=dev())
     dhz = torch.empty(n, r, dtype=torch.float32, device=dev())

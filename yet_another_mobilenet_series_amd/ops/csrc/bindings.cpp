@@ -229,20 +229,21 @@ void dw_conv_wgrad(const Tensor& x, const Tensor& dy, Tensor& dwgt,
 
 // ---- fused squeeze-excite (K7) -----------------------------------------
 
-void se_fwd(const Tensor& x, const Tensor& w1, const Tensor& b1,
-            const Tensor& w2, const Tensor& b2, Tensor& y,
+void se_fwd(const Tensor& x, const Tensor& w1t, const Tensor& b1,
+            const Tensor& w2t, const Tensor& b2, Tensor& y,
             Tensor& pooled, Tensor& ha, Tensor& sz,
             int64_t lop_kind, int64_t lop_act,
             c10::optional<Tensor> lop_coef, c10::optional<Tensor> ymat) {
   auto xr = rows_of(x, "x");
   auto yr = rows_of(y, "y");
-  check_w(w1);
-  check_w(w2);
-  const int C = xr.C, R = int(w1.size(0));
+  check_w(w1t);
+  check_w(w2t);
+  const int C = xr.C, R = int(w1t.size(1));
   TORCH_CHECK(yr.C == C && yr.M == xr.M, "se_fwd shape mismatch");
-  TORCH_CHECK(w1.dim() == 2 && w1.size(1) == C, "w1 must be [R, C]");
-  TORCH_CHECK(w2.dim() == 2 && w2.size(0) == C && w2.size(1) == R,
-              "w2 must be [C, R]");
+  TORCH_CHECK(w1t.dim() == 2 && w1t.size(0) == C,
+              "w1t must be [C, R] (transposed fc1 weight)");
+  TORCH_CHECK(w2t.dim() == 2 && w2t.size(0) == R && w2t.size(1) == C,
+              "w2t must be [R, C] (transposed fc2 weight)");
   const int N = int(x.size(0));
   const int HW = int(xr.M / N);
   TORCH_CHECK(pooled.numel() >= int64_t(N) * C &&
@@ -255,8 +256,8 @@ void se_fwd(const Tensor& x, const Tensor& w1, const Tensor& b1,
     TORCH_CHECK(mr.C == C && mr.M == xr.M, "ymat shape mismatch");
     mp = const_cast<void*>(mr.ptr);
   }
-  yams::se_fwd(xr.ptr, w1.data_ptr(), fp32_req(b1, "b1"), w2.data_ptr(),
-               fp32_req(b2, "b2"), y.data_ptr(), mp,
+  yams::se_fwd(xr.ptr, w1t.data_ptr(), fp32_req(b1, "b1"),
+               w2t.data_ptr(), fp32_req(b2, "b2"), y.data_ptr(), mp,
                pooled.data_ptr<float>(), ha.data_ptr<float>(),
                sz.data_ptr<float>(), N, HW, C, R,
                make_lop(lop_kind, lop_act, lop_coef, c10::nullopt, C),

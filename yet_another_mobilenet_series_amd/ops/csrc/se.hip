@@ -17,6 +17,8 @@
 // Per-block LDS: 8 KB fold buffer + pooled[C] + ha[R] + s[C] fp32
 // (<= 16.6 KB of the 160 KB CU budget).
 
+#include <cstdlib>
+
 #include "common.h"
 #include "engine.h"
 
@@ -26,10 +28,9 @@ namespace {
 
 constexpr int MAX_C = 1024;  // fp32 LDS arrays sized for the zoo's C<=960
 constexpr int MAX_R = 256;
-constexpr int BS = 1024;     // threads/block: 16 waves per CU so the
-                             // grid-N streaming phases hide HBM latency
-                             // (256-thread version measured ~4x off the
-                             // stream floor at 25% occupancy)
+// threads/block is a template knob (YAMS_SE_BS env selects 256 or 1024
+// at launch): more waves hide HBM latency in the grid-N streaming phases,
+// fewer waves cut the fold/sync overhead for the small-C shapes.
 
 DEV_INLINE int pow2_le(int v) {
   int m = 1;
@@ -41,10 +42,16 @@ DEV_INLINE int pow2_le(int v) {
 // (x is the dw-conv output z; BN+act applied on load, the activated value
 // also streamed to `ymat` so the project conv / backward can reuse it —
 // this folds the separate bn_act_fwd pass into the SE kernel).
-template <int LOPK>
+// The FORWARD consumes TRANSPOSED weight packs (w1t [C,R], w2t [R,C]):
+// its FC loops then read consecutive elements across adjacent lanes
+// (one cache line per wave instruction).  With the natural layouts each
+// lane touched its own line — a measured ~64x L2 amplification that made
+// the 960x240 shape 255 us.  The backward's loop structure wants the
+// NATURAL layouts (w1 [R,C], w2 [C,R]) for exactly the same reason.
+template <int LOPK, int BS>
 __global__ __launch_bounds__(BS) void se_fwd_kernel(
-    const ushort_t* __restrict__ x, const ushort_t* __restrict__ w1,
-    const float* __restrict__ b1, const ushort_t* __restrict__ w2,
+    const ushort_t* __restrict__ x, const ushort_t* __restrict__ w1t,
+    const float* __restrict__ b1, const ushort_t* __restrict__ w2t,
     const float* __restrict__ b2, ushort_t* __restrict__ y,
     ushort_t* __restrict__ ymat, float* __restrict__ pooled_g,
     float* __restrict__ ha_g, float* __restrict__ sz_g, int HW, int C,
@@ -122,13 +129,8 @@ __global__ __launch_bounds__(BS) void se_fwd_kernel(
     constexpr int RL = BS / 256;
     float acc1 = 0.f;
     if (r < R) {
-      const ushort_t* wrow = &w1[int64_t(r) * C];
-      for (int cc = rl * 8; cc < C; cc += RL * 8) {
-        short8_t w8 = *reinterpret_cast<const short8_t*>(&wrow[cc]);
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          acc1 = fmaf(pooled_s[cc + e], bf2f(ushort_t(w8[e])), acc1);
-      }
+      for (int cc = rl; cc < C; cc += RL)
+        acc1 = fmaf(pooled_s[cc], bf2f(w1t[int64_t(cc) * R + r]), acc1);
     }
     red[tid] = acc1;
     __syncthreads();
@@ -149,9 +151,11 @@ __global__ __launch_bounds__(BS) void se_fwd_kernel(
     if (mlane < ML) {
       for (int r = mlane; r < R; r += ML) {
         const float hv = ha_s[r];
+        short8_t w8 = *reinterpret_cast<const short8_t*>(
+            &w2t[int64_t(r) * C + c]);
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          acc2[e] = fmaf(hv, bf2f(w2[int64_t(c + e) * R + r]), acc2[e]);
+          acc2[e] = fmaf(hv, bf2f(ushort_t(w8[e])), acc2[e]);
       }
 #pragma unroll
       for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc2[e];
@@ -193,6 +197,7 @@ __global__ __launch_bounds__(BS) void se_fwd_kernel(
 
 // backward per-image pass: dx (incl. the avgpool backward contribution),
 // dsz[N,C] and dhz[N,R] staged fp32 for the cross-batch wgrad kernel.
+template <int BS>
 __global__ __launch_bounds__(BS) void se_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ w1, const ushort_t* __restrict__ w2,
@@ -379,22 +384,44 @@ __global__ __launch_bounds__(256) void se_wgrad_kernel(
     } else {
       if (r0 + i < C && c0 + j < R) dw2[int64_t(r0 + i) * R + c0 + j] = acc;
     }
-  } else {  // biases
-    for (int idx = int(threadIdx.x); idx < R + C; idx += 256) {
-      float acc = 0.f;
+  } else {  // biases: 16 outputs x 16 n-lanes per block, LDS fold
+    // (the serial one-block version was the kernel's straggler: a
+    // dependent ~256-load chain per output)
+    __shared__ float redb[256];
+    const int b = bid - 2 * t1;
+    const int idx = b * 16 + int(threadIdx.x) % 16;
+    const int nl = int(threadIdx.x) / 16;
+    float acc = 0.f;
+    if (idx < R + C) {
       if (idx < R) {
-        for (int n = 0; n < N; ++n) acc += dhz[int64_t(n) * R + idx];
-        db1[idx] = acc;
+        for (int n = nl; n < N; n += 16) acc += dhz[int64_t(n) * R + idx];
       } else {
-        const int c = idx - R;
-        for (int n = 0; n < N; ++n) acc += dsz[int64_t(n) * C + c];
-        db2[c] = acc;
+        const int cc = idx - R;
+        for (int n = nl; n < N; n += 16) acc += dsz[int64_t(n) * C + cc];
       }
+    }
+    redb[threadIdx.x] = acc;
+    __syncthreads();
+    if (nl == 0 && idx < R + C) {
+      for (int l = 1; l < 16; ++l) acc += redb[threadIdx.x + l * 16];
+      if (idx < R) db1[idx] = acc;
+      else db2[idx - R] = acc;
     }
   }
 }
 
 }  // namespace
+
+// measurement knob: YAMS_SE_BS in {256, 1024} selects the block size of
+// the fused fwd/bwd kernels (A/B'd on hardware; see profiles/README.md)
+static int se_block_size() {
+  static int bs = [] {
+    const char* e = getenv("YAMS_SE_BS");
+    const int v = e ? atoi(e) : 1024;
+    return v == 256 ? 256 : 1024;
+  }();
+  return bs;
+}
 
 void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
             const float* b2, void* y, void* ymat, float* pooled,
@@ -412,14 +439,23 @@ void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
   auto* w2p = static_cast<const ushort_t*>(w2);
   auto* yp = static_cast<ushort_t*>(y);
   auto* mp = static_cast<ushort_t*>(ymat);
-  if (lopx.kind == LOP_NONE)
-    se_fwd_kernel<LOP_NONE><<<dim3(N), BS, 0, s>>>(
-        xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
-  else if (lopx.kind == LOP_BNACT && ymat != nullptr)
-    se_fwd_kernel<LOP_BNACT><<<dim3(N), BS, 0, s>>>(
-        xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
-  else
+  if (lopx.kind == LOP_NONE) {
+    if (se_block_size() == 1024)
+      se_fwd_kernel<LOP_NONE, 1024><<<dim3(N), 1024, 0, s>>>(
+          xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
+    else
+      se_fwd_kernel<LOP_NONE, 256><<<dim3(N), 256, 0, s>>>(
+          xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
+  } else if (lopx.kind == LOP_BNACT && ymat != nullptr) {
+    if (se_block_size() == 1024)
+      se_fwd_kernel<LOP_BNACT, 1024><<<dim3(N), 1024, 0, s>>>(
+          xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
+    else
+      se_fwd_kernel<LOP_BNACT, 256><<<dim3(N), 256, 0, s>>>(
+          xp, w1p, b1, w2p, b2, yp, mp, pooled, ha, sz, HW, C, R, L);
+  } else {
     yams_launch_fail("se_fwd", lopx.kind, lopx.act, ymat != nullptr, 0);
+  }
 }
 
 void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
@@ -427,17 +463,24 @@ void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
             float* dhz, int N, int HW, int C, int R, hipStream_t s) {
   if ((C & 7) || C > 960 || R > MAX_R)
     yams_launch_fail("se_bwd_shape", C, R, HW, 0);
-  se_bwd_kernel<<<dim3(N), BS, 0, s>>>(
-      static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
-      static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
-      ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
+  if (se_block_size() == 1024)
+    se_bwd_kernel<1024><<<dim3(N), 1024, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
+        static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
+        ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
+  else
+    se_bwd_kernel<256><<<dim3(N), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
+        static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
+        ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
 }
 
 void se_wgrad(const float* dsz, const float* ha, const float* dhz,
               const float* pooled, float* dw1, float* dw2, float* db1,
               float* db2, int N, int C, int R, hipStream_t s) {
   const int t1 = ((R + 15) / 16) * ((C + 15) / 16);
-  se_wgrad_kernel<<<dim3(2 * t1 + 1), 256, 0, s>>>(
+  const int bias_blocks = (R + C + 15) / 16;
+  se_wgrad_kernel<<<dim3(2 * t1 + bias_blocks), 256, 0, s>>>(
       dsz, ha, dhz, pooled, dw1, dw2, db1, db2, N, C, R, t1);
 }
 

@@ -379,13 +379,26 @@ def global_avg_pool(x):
 
 
 def se_pack(w1, w2):
-    """bf16 kernel packs: w1 [R,C], w2 [C,R] (cached per param version)."""
+    """bf16 kernel packs, natural layout: w1 [R,C], w2 [C,R] (backward)."""
     r, c = w1.shape[0], w1.shape[1]
     w1m = _packed(w1, "se", lambda w: w.view(r, c).to(BF16).contiguous(),
                   spec="cast")
     w2m = _packed(w2, "se2", lambda w: w.view(c, r).to(BF16).contiguous(),
                   spec="cast")
     return w1m, w2m
+
+
+def se_pack_t(w1, w2):
+    """Transposed packs w1t [C,R], w2t [R,C] — the forward kernel's FC
+    loops read these coalesced across lanes (se.hip)."""
+    r, c = w1.shape[0], w1.shape[1]
+    w1t = _packed(w1, "se_t",
+                  lambda w: w.view(r, c).t().to(BF16).contiguous(),
+                  spec="t")
+    w2t = _packed(w2, "se2_t",
+                  lambda w: w.view(c, r).t().to(BF16).contiguous(),
+                  spec="t")
+    return w1t, w2t
 
 
 def se_forward(e, x, w1, b1, w2, b2, lop=None, ymat=None):
@@ -397,15 +410,15 @@ def se_forward(e, x, w1, b1, w2, b2, lop=None, ymat=None):
     """
     n, c = x.shape[0], x.shape[1]
     r = w1.shape[0]
-    w1m, w2m = se_pack(w1, w2)
+    w1t, w2t = se_pack_t(w1, w2)
     y = torch.empty_like(x)
     pooled = torch.empty(n, c, dtype=torch.float32, device=x.device)
     ha = torch.empty(n, r, dtype=torch.float32, device=x.device)
     sz = torch.empty(n, c, dtype=torch.float32, device=x.device)
     if lop is None:
-        e.se_fwd(x, w1m, b1.float(), w2m, b2.float(), y, pooled, ha, sz)
+        e.se_fwd(x, w1t, b1.float(), w2t, b2.float(), y, pooled, ha, sz)
     else:
-        e.se_fwd(x, w1m, b1.float(), w2m, b2.float(), y, pooled, ha, sz,
+        e.se_fwd(x, w1t, b1.float(), w2t, b2.float(), y, pooled, ha, sz,
                  lop_kind=lop[0], lop_act=lop[1], lop_coef=lop[2],
                  ymat=ymat)
     return y, pooled, ha, sz
