@@ -125,7 +125,8 @@ DEFAULTS: Dict[str, Any] = {
     "model": {"name": "mobilenet_v3_large", "width_mult": 1.0, "num_classes": 1000,
               "dropout": 0.2, "bn_momentum": 0.1, "bn_epsilon": 1e-5},
     "data": {"dataset": "synthetic", "root": "", "image_size": 224,
-             "batch_size_per_gpu": 256, "workers": 4, "synthetic_size": 12800},
+             "batch_size_per_gpu": 256, "workers": 4, "synthetic_size": 12800,
+             "color_jitter": 0.4},
     "optim": {"name": "sgd", "lr": 0.4, "momentum": 0.9, "nesterov": True,
               "weight_decay": 4e-5, "no_wd_on_bn_bias": True,
               "rmsprop_alpha": 0.9, "rmsprop_eps": 1e-3,

@@ -6,8 +6,15 @@ Loaders:
   network ⇒ no real dataset on these boxes).
 - ``imagefolder``: ImageNet-style ``root/<split>/<class>/*.jpg`` directory
   via PIL (torchvision is not installed in this image; the transform set —
-  RandomResizedCrop(224) + hflip for train, Resize(256)+CenterCrop(224)
-  for val — is implemented here directly).
+  RandomResizedCrop(224) + hflip + color jitter for train,
+  Resize(256)+CenterCrop(224) for val — is implemented here directly).
+- ``packed``: the fast path (the reference uses an LMDB file for this,
+  SURVEY.md §2.2): one mmap'd ``<split>.pack`` file per split holding the
+  encoded images + labels back-to-back, so an epoch does zero per-sample
+  filesystem operations.  ``tools/make_packed_dataset.py`` converts an
+  imagefolder tree.  ``lmdb``: reads the reference's caffe-style LMDB
+  files directly when the ``lmdb`` module is installed (it is not in this
+  image — the class import-gates gracefully).
 
 All loaders return standard ``torch.utils.data.DataLoader`` objects with a
 ``DistributedSampler`` when the process group is initialised.
@@ -63,17 +70,77 @@ def _pil_to_tensor(img) -> torch.Tensor:
     return (t - mean) / std
 
 
-class ImageFolderDataset(Dataset):
+class _AugMixin:
+    """Inception-style train augmentation shared by the folder/packed/LMDB
+    datasets: RandomResizedCrop + hflip + color jitter."""
+
+    train: bool
+    image_size: int
+    color_jitter: float = 0.4
+
+    def _augment(self, img):
+        if self.train:
+            img = self._random_resized_crop(img)
+            if random.random() < 0.5:
+                img = img.transpose(Image.FLIP_LEFT_RIGHT)
+            if self.color_jitter > 0:
+                img = self._color_jitter(img)
+        else:
+            img = self._center_crop(img)
+        return _pil_to_tensor(img)
+
+    def _color_jitter(self, img):
+        from PIL import ImageEnhance
+        j = self.color_jitter
+        ops = [("Brightness", ImageEnhance.Brightness),
+               ("Contrast", ImageEnhance.Contrast),
+               ("Color", ImageEnhance.Color)]  # saturation
+        random.shuffle(ops)
+        for _, enh in ops:
+            factor = 1.0 + random.uniform(-j, j)
+            img = enh(img).enhance(factor)
+        return img
+
+    def _random_resized_crop(self, img):
+        w, h = img.size
+        area = w * h
+        for _ in range(10):
+            target_area = random.uniform(0.08, 1.0) * area
+            ar = random.uniform(3 / 4, 4 / 3)
+            cw = int(round((target_area * ar) ** 0.5))
+            ch = int(round((target_area / ar) ** 0.5))
+            if cw <= w and ch <= h:
+                x0 = random.randint(0, w - cw)
+                y0 = random.randint(0, h - ch)
+                img = img.crop((x0, y0, x0 + cw, y0 + ch))
+                return img.resize((self.image_size, self.image_size),
+                                  Image.BILINEAR)
+        return self._center_crop(img)
+
+    def _center_crop(self, img):
+        scale = 256 / min(img.size)
+        img = img.resize((max(1, round(img.size[0] * scale)),
+                          max(1, round(img.size[1] * scale))),
+                         Image.BILINEAR)
+        w, h = img.size
+        s = self.image_size
+        x0, y0 = (w - s) // 2, (h - s) // 2
+        return img.crop((x0, y0, x0 + s, y0 + s))
+
+
+class ImageFolderDataset(_AugMixin, Dataset):
     """Minimal ImageFolder with Inception-style train augmentation."""
 
     EXT = (".jpg", ".jpeg", ".png", ".bmp", ".webp")
 
-    def __init__(self, root: str, train: bool, image_size: int = 224) -> None:
+    def __init__(self, root: str, train: bool, image_size: int = 224,
+                 color_jitter: float = 0.4) -> None:
         if not HAS_PIL:
             raise RuntimeError("PIL is required for ImageFolderDataset")
         self.root = root
         self.train = train
         self.image_size = image_size
+        self.color_jitter = color_jitter
         classes = sorted(d for d in os.listdir(root)
                          if os.path.isdir(os.path.join(root, d)))
         self.class_to_idx = {c: i for i, c in enumerate(classes)}
@@ -90,42 +157,121 @@ class ImageFolderDataset(Dataset):
     def __len__(self) -> int:
         return len(self.samples)
 
-    def _random_resized_crop(self, img):
-        w, h = img.size
-        area = w * h
-        for _ in range(10):
-            target_area = random.uniform(0.08, 1.0) * area
-            ar = random.uniform(3 / 4, 4 / 3)
-            cw = int(round((target_area * ar) ** 0.5))
-            ch = int(round((target_area / ar) ** 0.5))
-            if cw <= w and ch <= h:
-                x0 = random.randint(0, w - cw)
-                y0 = random.randint(0, h - ch)
-                img = img.crop((x0, y0, x0 + cw, y0 + ch))
-                return img.resize((self.image_size, self.image_size),
-                                  Image.BILINEAR)
-        # fallback: center crop
-        return self._center_crop(img)
-
-    def _center_crop(self, img):
-        scale = 256 / min(img.size)
-        img = img.resize((max(1, round(img.size[0] * scale)),
-                          max(1, round(img.size[1] * scale))), Image.BILINEAR)
-        w, h = img.size
-        s = self.image_size
-        x0, y0 = (w - s) // 2, (h - s) // 2
-        return img.crop((x0, y0, x0 + s, y0 + s))
-
     def __getitem__(self, idx: int) -> Tuple[torch.Tensor, int]:
         path, label = self.samples[idx]
         img = Image.open(path).convert("RGB")
-        if self.train:
-            img = self._random_resized_crop(img)
-            if random.random() < 0.5:
-                img = img.transpose(Image.FLIP_LEFT_RIGHT)
-        else:
-            img = self._center_crop(img)
-        return _pil_to_tensor(img), label
+        return self._augment(img), label
+
+
+#: magic + version for the packed single-file format (tools/make_packed_dataset.py)
+PACK_MAGIC = b"YAMSPCK1"
+
+
+class PackedImageDataset(_AugMixin, Dataset):
+    """Single-file fast path: ``<split>.pack`` = [magic][u64 n]
+    [n x (u64 offset, u64 length, i32 label)][blob bytes...].
+
+    The index is loaded once; workers read samples with one ``mmap`` slice
+    each — zero per-sample filesystem operations, sequential-friendly
+    layout (the role LMDB plays in the reference).
+    """
+
+    def __init__(self, path: str, train: bool, image_size: int = 224,
+                 color_jitter: float = 0.4) -> None:
+        if not HAS_PIL:
+            raise RuntimeError("PIL is required for PackedImageDataset")
+        import mmap
+        import struct
+        self.path = path
+        self.train = train
+        self.image_size = image_size
+        self.color_jitter = color_jitter
+        self._mm = None  # opened lazily per worker process
+        with open(path, "rb") as f:
+            magic = f.read(8)
+            if magic != PACK_MAGIC:
+                raise RuntimeError(f"{path}: not a YAMS pack file")
+            (n,) = struct.unpack("<Q", f.read(8))
+            raw = f.read(n * 20)
+        self.index = [struct.unpack_from("<QQi", raw, i * 20)
+                      for i in range(n)]
+
+    def _mmap(self):
+        if self._mm is None:
+            import mmap
+            self._fh = open(self.path, "rb")
+            self._mm = mmap.mmap(self._fh.fileno(), 0,
+                                 access=mmap.ACCESS_READ)
+        return self._mm
+
+    def __len__(self) -> int:
+        return len(self.index)
+
+    def __getitem__(self, idx: int) -> Tuple[torch.Tensor, int]:
+        import io
+        off, length, label = self.index[idx]
+        blob = self._mmap()[off:off + length]
+        img = Image.open(io.BytesIO(blob)).convert("RGB")
+        return self._augment(img), label
+
+    def __getstate__(self):
+        d = dict(self.__dict__)
+        d["_mm"] = None  # re-open in the worker process
+        d.pop("_fh", None)
+        return d
+
+
+class ImageFolderLMDB(_AugMixin, Dataset):
+    """Reads the reference's caffe-style LMDB files (keys = decimal index,
+    values = pickled ``(encoded_bytes, label)``) when the ``lmdb`` module
+    is installed.  This image has no lmdb wheel, so construction raises a
+    clear error here; on a user box with lmdb it works as in the
+    reference (SURVEY.md §2.2 "LMDB fast path")."""
+
+    def __init__(self, path: str, train: bool, image_size: int = 224,
+                 color_jitter: float = 0.4) -> None:
+        try:
+            import lmdb  # noqa: F401
+        except ImportError as e:
+            raise RuntimeError(
+                "ImageFolderLMDB requires the 'lmdb' package; use "
+                "data.dataset=packed (tools/make_packed_dataset.py) for "
+                "the built-in single-file fast path") from e
+        import lmdb
+        self.path = path
+        self.train = train
+        self.image_size = image_size
+        self.color_jitter = color_jitter
+        self._env = None
+        env = lmdb.open(path, readonly=True, lock=False, readahead=False,
+                        meminit=False)
+        with env.begin(write=False) as txn:
+            self.length = txn.stat()["entries"]
+        env.close()
+
+    def _txn_env(self):
+        if self._env is None:
+            import lmdb
+            self._env = lmdb.open(self.path, readonly=True, lock=False,
+                                  readahead=False, meminit=False)
+        return self._env
+
+    def __len__(self) -> int:
+        return self.length
+
+    def __getitem__(self, idx: int) -> Tuple[torch.Tensor, int]:
+        import io
+        import pickle
+        with self._txn_env().begin(write=False) as txn:
+            raw = txn.get(str(idx).encode("ascii"))
+        blob, label = pickle.loads(raw)
+        img = Image.open(io.BytesIO(blob)).convert("RGB")
+        return self._augment(img), int(label)
+
+    def __getstate__(self):
+        d = dict(self.__dict__)
+        d["_env"] = None
+        return d
 
 
 class EvalDistributedSampler(torch.utils.data.Sampler):
@@ -158,10 +304,18 @@ def build_dataset(cfg, train: bool, seed: int = 0) -> Dataset:
         size = cfg.synthetic_size if train else max(cfg.synthetic_size // 10, 64)
         return SyntheticDataset(size=size, image_size=cfg.image_size,
                                 seed=seed + (0 if train else 1))
+    jitter = float(getattr(cfg, "color_jitter", 0.4))
+    split = "train" if train else "val"
     if cfg.dataset in ("imagefolder", "imagenet"):
-        split = "train" if train else "val"
         return ImageFolderDataset(os.path.join(cfg.root, split), train,
-                                  cfg.image_size)
+                                  cfg.image_size, color_jitter=jitter)
+    if cfg.dataset == "packed":
+        return PackedImageDataset(os.path.join(cfg.root, f"{split}.pack"),
+                                  train, cfg.image_size,
+                                  color_jitter=jitter)
+    if cfg.dataset == "lmdb":
+        return ImageFolderLMDB(os.path.join(cfg.root, f"{split}.lmdb"),
+                               train, cfg.image_size, color_jitter=jitter)
     raise ValueError(f"unknown dataset {cfg.dataset!r}")
 
 
