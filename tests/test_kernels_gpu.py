@@ -133,11 +133,11 @@ def test_dw_wgrad(ext, n, c, h, k, s):
     ho = (h + 2 * p - k) // s + 1
     x = randn_nhwc(n, c, h, h, seed=3)
     dy = randn_nhwc(n, c, ho, ho, seed=4)
-    dw = torch.empty(k * k, c, device=dev(), dtype=torch.float32)
+    dw = torch.empty(c, k * k, device=dev(), dtype=torch.float32)
     ext.dw_conv_wgrad(x, dy, dw, k, s, p)
     want = torch.nn.grad.conv2d_weight(
         x.float(), (c, 1, k, k), dy.float(), stride=s, padding=p, groups=c)
-    want = want.reshape(c, k * k).t()
+    want = want.reshape(c, k * k)
     assert_close(dw, want, atol=0.05 * (n * ho * ho) ** 0.5 * 0.1)
 
 

@@ -15,8 +15,8 @@
 
 namespace yams {
 
-void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
-                 hipStream_t s);
+void slab_reduce_dwt(const float* ws, float* out, int C, int KK,
+                     int splits, hipStream_t s);
 
 namespace {
 
@@ -455,23 +455,30 @@ int grid_for(int64_t total) {
 
 }  // namespace
 
-int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride) {
+int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k) {
   (void)stride;
   // measurement knob (roadmap §3): YAMS_DW_CHUNKS_DIV divides the block
   // target so the work-per-wave vs launch-count tradeoff can be swept
-  // without a rebuild.  Unset => default behavior.
+  // without a rebuild.  Unset => the per-shape table below.
   static int div_ = [] {
     const char* e = getenv("YAMS_DW_CHUNKS_DIV");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
-  // 512-block default: the chunks sweep (tools/probe_dw_sweep.py, profiles)
-  // measured div=4 off the old 2048 target 10-30% faster on most shapes —
-  // per-wave prologue + launch ramp dominate when waves do ~1 iteration
-  int64_t want = 512 / (int64_t(c_tiles) * (div_ > 0 ? div_ : 1));
+  // per-shape block target from the round-1 chunks sweep
+  // (tools/probe_dw_sweep.py; per-wave prologue + launch ramp dominate
+  // when waves run ~1 loop iteration): k5 prefers 4x fewer chunks than
+  // the 512 base, k3 at small spatial 1x (512); env div overrides.
+  int64_t target = 512;
+  if (div_ <= 0) {
+    if (k == 5) target = 256;
+  } else {
+    target = 2048 / div_;
+  }
+  int64_t want = target / c_tiles;
   int64_t maxc = P;
   int c = int(std::min(std::max<int64_t>(want, 1), maxc));
   return c < 1 ? 1 : c;
@@ -487,7 +494,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
-  const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride);
+  const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride, k);
   dim3 g(c_tiles * chunks);
   // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
   Lop LE{};
@@ -523,7 +530,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     const int C8 = C >> 2;
     const int CT8 = C8 < 256 ? C8 : 256;
     const int c_tiles = (C8 + CT8 - 1) / CT8;
-    const int chunks = dw_stats_chunks(N, H, W, C, 1);
+    const int chunks = dw_stats_chunks(N, H, W, C, 1, k);
     dim3 g(c_tiles * chunks);
     // dgrad supports only lop none (dz is materialized by bn_act_bwd);
     // with stats_ws + epi it also emits the consumer BN's backward
@@ -602,7 +609,8 @@ void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
 #undef DWW_CASE
   yams_launch_fail("dw_conv_wgrad", k, stride, lopx.kind, lopdy.kind);
 combine:
-  slab_reduce(ws, dwgt, int64_t(k) * k * C, chunks, s);
+  // transposed combine: dwgt lands as [C, k*k] (OIHW view, no copy)
+  slab_reduce_dwt(ws, dwgt, C, k * k, chunks, s);
 }
 
 void im2row(const void* x, void* patches, int N, int H, int W, int Ho,

@@ -48,6 +48,9 @@ int64_t pw_wgrad_ws_elems(int64_t M, int N1, int N2);
 // out[i] = sum_s ws[s*elems + i] (deterministic slab combine)
 void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s);
+// dw-wgrad combine with transposed store: ws [splits][KK][C] -> out [C][KK]
+void slab_reduce_dwt(const float* ws, float* out, int C, int KK,
+                     int splits, hipStream_t s);
 
 // --- depthwise conv (K3/K4) --------------------------------------------
 // w layout [k*k, C] (C fastest).  lopx transforms the input stream;
@@ -56,12 +59,13 @@ void dw_conv_fwd(const void* x, const void* w, void* y,
                  int N, int H, int W, int C, int Ho, int Wo,
                  int k, int stride, int pad, LopDesc lopx, float* stats_ws,
                  hipStream_t s);
-int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride);
+int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k);
 void dw_conv_dgrad(const void* dy, const void* w, void* dx,
                    int N, int H, int W, int C, int Ho, int Wo,
                    int k, int stride, int pad, LopDesc lopdy, LopDesc epi,
                    float* stats_ws, hipStream_t s);
-// dwgt fp32 [k*k, C]; ws fp32 [chunks, k*k, C]
+// dwgt fp32 [C, k*k] (transposed combine -> reshapes to OIHW [C,1,k,k]
+// without a copy); ws fp32 [chunks, k*k, C]
 void dw_conv_wgrad(const void* x, const void* dy, float* dwgt, float* ws,
                    int N, int H, int W, int C, int Ho, int Wo,
                    int k, int stride, int pad, LopDesc lopx, LopDesc lopdy,

@@ -485,7 +485,71 @@ __global__ __launch_bounds__(256) void slab_reduce4_kernel(
   }
 }
 
+// dw-wgrad variant: input slabs are [splits][KK][C] (C fastest) but the
+// OUTPUT is written transposed as [C][KK] so it reshapes to the OIHW
+// [C,1,k,k] param-grad layout without a separate transpose copy (15
+// eager copies/step removed).  Reads stay fully coalesced; the scattered
+// 4B stores are trivial (<100 KB).
+template <int SL>
+__global__ __launch_bounds__(256) void slab_reduce_dwt_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, int C, int KK,
+    int splits) {
+  constexpr int EL = 256 / SL;
+  __shared__ float red[256][4];
+  const int elems = C * KK;
+  const int el = int(threadIdx.x) % EL;
+  const int sl = int(threadIdx.x) / EL;
+  for (int64_t base = int64_t(blockIdx.x) * EL * 4; base < elems;
+       base += int64_t(gridDim.x) * EL * 4) {
+    const int64_t i = base + el * 4;
+    float v[4] = {};
+    if (i + 4 <= elems) {
+      for (int sp = sl; sp < splits; sp += SL) {
+        const float4 w4 = *reinterpret_cast<const float4*>(
+            &ws[int64_t(sp) * elems + i]);
+        v[0] += w4.x; v[1] += w4.y; v[2] += w4.z; v[3] += w4.w;
+      }
+    } else if (i < elems) {
+      for (int sp = sl; sp < splits; sp += SL)
+        for (int e = 0; e < 4; ++e)
+          if (i + e < elems) v[e] += ws[int64_t(sp) * elems + i + e];
+    }
+#pragma unroll
+    for (int e = 0; e < 4; ++e) red[threadIdx.x][e] = v[e];
+    __syncthreads();
+    if (sl == 0 && i < elems) {
+      for (int l = 1; l < SL; ++l)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) v[e] += red[el + l * EL][e];
+      for (int e = 0; e < 4 && i + e < elems; ++e) {
+        const int flat = int(i) + e;         // = j*C + c in slab order
+        const int j = flat / C;
+        const int cc = flat % C;
+        out[int64_t(cc) * KK + j] = v[e];
+      }
+    }
+    __syncthreads();
+  }
+}
+
 }  // namespace
+
+void slab_reduce_dwt(const float* ws, float* out, int C, int KK,
+                     int splits, hipStream_t s) {
+  const int sl = splits >= 64 ? 64 : (splits >= 16 ? 16 : 4);
+  const int el4 = (256 / sl) * 4;
+  const int64_t elems = int64_t(C) * KK;
+  const int blocks = int(std::min<int64_t>((elems + el4 - 1) / el4, 8192));
+  if (sl == 64)
+    slab_reduce_dwt_kernel<64><<<dim3(blocks), 256, 0, s>>>(ws, out, C, KK,
+                                                            splits);
+  else if (sl == 16)
+    slab_reduce_dwt_kernel<16><<<dim3(blocks), 256, 0, s>>>(ws, out, C, KK,
+                                                            splits);
+  else
+    slab_reduce_dwt_kernel<4><<<dim3(blocks), 256, 0, s>>>(ws, out, C, KK,
+                                                           splits);
+}
 
 void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
                  hipStream_t s) {
@@ -504,9 +568,17 @@ void slab_reduce(const float* ws, float* out, int64_t elems, int splits,
 }
 
 int pw_wgrad_splits(int64_t M, int N1, int N2) {
+  // measurement knob: YAMS_ATB_BLOCKS scales the split-M block target
+  // (default 1024; the split count trades slab-combine traffic against
+  // fill — swept on hardware, see profiles/)
+  static int target = [] {
+    const char* e = getenv("YAMS_ATB_BLOCKS");
+    const int v = e ? atoi(e) : 1024;
+    return v > 0 ? v : 1024;
+  }();
   const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   const int64_t max_s = (M + 127) / 128;
-  int64_t s = 1024 / tiles;
+  int64_t s = target / tiles;
   if (s < 1) s = 1;
   if (s > max_s) s = max_s;
   return int(s < 1 ? 1 : s);

@@ -111,7 +111,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         m2 = n * ho * wo
         mean2, invstd2, ab2 = _bn_bufs(cd, dev)
         if training:
-            chunks = e.dw_stats_chunks(n, ho, wo, cd, stride)
+            chunks = e.dw_stats_chunks(n, ho, wo, cd, stride, k)
             ws = torch.empty(2 * chunks * cd, dtype=F32, device=dev)
             e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad, stats_ws=ws)
             e.bn_finalize(ws, chunks, mean2, invstd2, ab2, dg, db,
@@ -249,7 +249,7 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         d_z1out = torch.empty_like(dw_in)
         if has_expand and stride == 1 and _EPI_ON:
             # d_z1out feeds the expand BN's backward: fuse its partials
-            sp1 = int(e.dw_stats_chunks(n, h, wdt, ce, 1))
+            sp1 = int(e.dw_stats_chunks(n, h, wdt, ce, 1, k))
             ws1 = torch.empty(2 * sp1 * ce, dtype=F32, device=dev)
             e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad,
                             stats_ws=ws1, bstats_act=act, bstats_coef=ab1,
@@ -257,9 +257,9 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         else:
             ws1 = sp1 = None
             e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
-        dwgt = torch.empty(k * k, cd, dtype=F32, device=dev)
+        dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
         e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
-        d_dww = dwgt.t().reshape(cd, 1, k, k)
+        d_dww = dwgt.reshape(cd, 1, k, k)
 
         # ---- expand backward --------------------------------------------
         if has_expand:

@@ -255,10 +255,10 @@ class _ConvBNActFn(torch.autograd.Function):
                 e.pw_conv_wgrad(dz, x, dwp)
                 dweight = dwp.reshape(cout, cin, 1, 1)
             elif kind == "dw":
-                dwp = torch.empty(k * k, cout, dtype=torch.float32,
+                dwp = torch.empty(cout, k * k, dtype=torch.float32,
                                   device=x.device)
                 e.dw_conv_wgrad(x, dz, dwp, k, s, p)
-                dweight = dwp.t().reshape(cout, 1, k, k)
+                dweight = dwp.reshape(cout, 1, k, k)
             else:  # stem: wgrad on the split-M GEMM over saved patches
                 patches = ctx.stem_patches
                 m = patches.shape[0]
