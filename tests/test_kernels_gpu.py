@@ -639,3 +639,16 @@ def test_fused_repack_matches_python_packs(ext):
     q1 = _packed(pw, "pw", lambda w: w.view(64, 32).to(BF16).contiguous(),
                  spec="cast")
     assert q1 is p1
+
+
+def test_pw_conv_residual_epilogue(ext):
+    """C = A @ B^T + res fused in the GEMM epilogue (backward residual)."""
+    m, k_, n = 512, 64, 32
+    a = torch.randn(m, k_, device=dev()).bfloat16().contiguous()
+    b = torch.randn(n, k_, device=dev()).bfloat16().contiguous()
+    res = torch.randn(m, n, device=dev()).bfloat16().contiguous()
+    y = torch.empty(m, n, device=dev(), dtype=BF16)
+    ext.pw_conv(a.view(m, k_, 1, 1), b, y.view(m, n, 1, 1),
+                residual=res.view(m, n, 1, 1))
+    want = (a.float() @ b.float().t() + res.float())
+    assert_close(y, want)

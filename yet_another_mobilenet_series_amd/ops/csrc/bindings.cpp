@@ -96,7 +96,8 @@ void pw_conv(const Tensor& x, const Tensor& w, Tensor& y,
              c10::optional<Tensor> lop_coef, c10::optional<Tensor> lop_z,
              c10::optional<Tensor> stats_ws,
              int64_t bstats_act, c10::optional<Tensor> bstats_coef,
-             c10::optional<Tensor> bstats_z) {
+             c10::optional<Tensor> bstats_z,
+             c10::optional<Tensor> residual) {
   auto xr = rows_of(x, "x");
   auto yr = rows_of(y, "y");
   check_w(w);
@@ -123,8 +124,15 @@ void pw_conv(const Tensor& x, const Tensor& w, Tensor& y,
     epi = make_lop(1, bstats_act, bstats_coef, bstats_z, yr.C);
     epi.z = bstats_z->data_ptr();
   }
+  const void* rp = nullptr;
+  if (residual.has_value()) {
+    auto rr = rows_of(*residual, "residual");
+    TORCH_CHECK(rr.M == yr.M && rr.C == yr.C,
+                "pw_conv residual shape mismatch");
+    rp = rr.ptr;
+  }
   yams::pw_conv(xr.ptr, w.data_ptr(), y.data_ptr(), xr.M, xr.C, yr.C, lop,
-                epi, sw, cur_stream());
+                epi, sw, rp, cur_stream());
 }
 
 int64_t pw_stats_mblocks(int64_t M) { return yams::pw_stats_mblocks(M); }
@@ -679,7 +687,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "x"_a, "w"_a, "y"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
         "lop_coef"_a = c10::nullopt, "lop_z"_a = c10::nullopt,
         "stats_ws"_a = c10::nullopt, "bstats_act"_a = 0,
-        "bstats_coef"_a = c10::nullopt, "bstats_z"_a = c10::nullopt);
+        "bstats_coef"_a = c10::nullopt, "bstats_z"_a = c10::nullopt,
+        "residual"_a = c10::nullopt);
   m.def("pw_stats_mblocks", &pw_stats_mblocks);
   m.def("pw_conv_wgrad", &pw_conv_wgrad, "dw[N1,N2] = a^T @ b (split-M)",
         "a"_a, "b"_a, "dw"_a, "lopa_kind"_a = 0, "lopa_act"_a = 0,

@@ -32,9 +32,10 @@ struct LopDesc {
 // epi: kind!=0 => stats_ws receives BN-BACKWARD partials (sum da, sum
 // da*z) of the consumer BN instead of forward sum/sumsq; epi carries the
 // z tensor, ab[2C] coefs and act (see gemm.hip EPI==2).
+// res: optional epilogue residual (y = x@w^T + res; plain-GEMM case only)
 void pw_conv(const void* x, const void* w, void* y,
              int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
-             float* stats_ws, hipStream_t s);
+             float* stats_ws, const void* res, hipStream_t s);
 int pw_stats_mblocks(int64_t M);
 
 // dw[N1,N2] = a[M,N1]^T @ b[M,N2], fp32 out, deterministic split-M slabs

@@ -45,11 +45,13 @@ inline Lop to_lop(const LopDesc& d) {
 // partial sums (sum da, sum da*z with da = dy * act'(a*z+b)) so the
 // separate bn_bwd_partial pass over dy+z never runs.  lopE carries the
 // z tensor, the ab[2C] coefs and the act for EPI==2.
-template <int EPI, int LOPK, int EACT = 0>
+// RES: epilogue residual add (C = A@B^T + res) — used by the backward's
+// d_x = pw_dgrad(dz1) + dyb so the residual path costs zero extra passes
+template <int EPI, int LOPK, int EACT = 0, bool RES = false>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
     ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA, Lop lopE,
-    float* __restrict__ stats_ws) {
+    float* __restrict__ stats_ws, const ushort_t* __restrict__ res) {
   // double-buffered register staging (guide T14: issue the next tile's
   // global loads before the current tile's MFMAs, write LDS after the
   // barrier) — the single-buffered loop parked 47% of wave cycles on
@@ -168,7 +170,11 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
-        if (row < M) C[row * N + col] = f2bf(acc[i][j][r]);
+        if (row < M) {
+          float v = acc[i][j][r];
+          if (RES) v += bf2f(res[row * N + col]);
+          C[row * N + col] = f2bf(v);
+        }
       }
     }
   }
@@ -238,7 +244,7 @@ int pw_stats_mblocks(int64_t M) { return ceil_div(M, BM); }
 
 void pw_conv(const void* x, const void* w, void* y,
              int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
-             float* stats_ws, hipStream_t s) {
+             float* stats_ws, const void* res, hipStream_t s) {
   dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   const Lop L = to_lop(lopa);
   const Lop LE = to_lop(epi);
@@ -246,16 +252,25 @@ void pw_conv(const void* x, const void* w, void* y,
   auto* ap = static_cast<const ushort_t*>(x);
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
+  auto* rp = static_cast<const ushort_t*>(res);
+  if (rp != nullptr) {  // residual epilogue: plain-GEMM case only
+    if (ep == 0 && lopa.kind == 0) {
+      pw_conv_kernel<0, 0, 0, true><<<grid, 256, 0, s>>>(
+          ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      return;
+    }
+    yams_launch_fail("pw_conv_res", ep, lopa.kind, lopa.act, epi.act);
+  }
 #define PW_CASE(EP, LK)                                                   \
   if (ep == EP && lopa.kind == LK) {                                      \
     pw_conv_kernel<EP, LK><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L,   \
-                                                LE, stats_ws);            \
+                                                LE, stats_ws, nullptr);   \
     return;                                                               \
   }
 #define PW_EPI2(AC)                                                       \
   if (ep == 2 && lopa.kind == 0 && epi.act == AC) {                       \
     pw_conv_kernel<2, 0, AC><<<grid, 256, 0, s>>>(ap, wp, yp, M, K, N, L, \
-                                                  LE, stats_ws);          \
+                                                  LE, stats_ws, nullptr); \
     return;                                                               \
   }
   PW_CASE(0, 0) PW_CASE(0, 1) PW_CASE(0, 2)

@@ -276,7 +276,9 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
                           .contiguous(), spec="t")
             d_x = _empty_like_nhwc(x)
-            e.pw_conv(dz1, ewt, d_x)
+            # residual grad fused into the dgrad epilogue (zero extra
+            # passes; the eager add cost 3 passes x 10 blocks per step)
+            e.pw_conv(dz1, ewt, d_x, residual=dyb if use_res else None)
             dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
             e.pw_conv_wgrad(dz1, x, dwp1)
             d_ew = dwp1.reshape(ce, cin, 1, 1)
@@ -284,8 +286,8 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             d_x = d_z1out
             d_ew = dg1 = db1_ = None
 
-        if use_res:
-            d_x = d_x + dyb
+        if use_res and not has_expand:
+            d_x = d_x + dyb  # rare no-expand residual block (V3-L #1)
 
         return (d_x, d_ew, dg1, db1_, None, None,
                 d_dww, dg2, db2_, None, None,
