@@ -27,6 +27,7 @@
 
 typedef unsigned short ushort_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(2))) float f32x2;  // v_pk_fma_f32 pair
 typedef __attribute__((ext_vector_type(8))) short bf16x8;   // MFMA A/B frag (4 VGPRs)
 typedef __attribute__((ext_vector_type(4))) short short4_t; // 8-byte vector load
 typedef __attribute__((ext_vector_type(8))) short short8_t; // 16-byte vector load

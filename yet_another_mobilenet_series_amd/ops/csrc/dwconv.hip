@@ -50,7 +50,14 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   __shared__ float red[EPI ? 256 * 4 : 4];
   constexpr int XW = 3 * S + K;        // x cols for 4 wo outputs
   constexpr int TH = (S == 1 && K == 3) ? 4 : 2;  // rows per thread
+  // (TH=4 for k5 measured 4x SLOWER: xr+acc registers spill at 256
+  // threads — the k5 fix is the traversal order below, not deeper tiles)
   constexpr int RR = (TH - 1) * S + K; // input rows touched
+  // k5's 6-row window thrashes L1 when quads walk along w (row reuse
+  // distance = a full row of quads ~40-80 KB); walking DOWN the column
+  // stripe (hg fastest) keeps the reuse window at XW cols x RR rows
+  // (~12 KB) — measured 1.1-1.6 TB/s -> see profiles/r02 A/B
+  constexpr bool HGFAST = (K == 5);
   const int C8 = C >> 2;
   const int CT8 = C8 < 256 ? C8 : 256;
   const int ML = 256 / CT8;
@@ -82,15 +89,36 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
       }
     }
     for (int p = lo + mlane; p < hi_; p += ML) {
-      const int w4 = p % Wo4;
-      const int rest = p / Wo4;
-      const int hg = rest % HoG;
-      const int n = rest / HoG;
+      int w4, hg, n;
+      if (HGFAST) {
+        hg = p % HoG;
+        const int rest = p / HoG;
+        w4 = rest % Wo4;
+        n = rest / Wo4;
+      } else {
+        w4 = p % Wo4;
+        const int rest = p / Wo4;
+        hg = rest % HoG;
+        n = rest / HoG;
+      }
       const int ho0 = hg * TH;
       const int wo0 = w4 * 4;
       const int wbase = wo0 * S - pad;
       const int64_t imgbase = int64_t(n) * H * W * C + c;
-      float acc[TH][4][4] = {};
+      // packed-f32 accumulators: the inner math runs on v_pk_fma_f32
+      // (2 MACs/instruction) — the scalar version measured VALU-bound
+      // (SQ_INSTS_VALU ~56% of all SIMD cycles, profiles/r02)
+      f32x2 acc[TH][4][2] = {};
+      // column clamps are row-invariant: hoist out of the r loop
+      int coff[XW];
+      bool cok[XW];
+#pragma unroll
+      for (int t = 0; t < XW; ++t) {
+        const int wi = wbase + t;
+        const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+        cok[t] = wi >= 0 && wi < W;
+        coff[t] = wic * C;
+      }
 #pragma unroll
       for (int r = 0; r < RR; ++r) {
         const int hi = ho0 * S + r - pad;
@@ -100,29 +128,34 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
         const ushort_t* zrow =
             LOPK == LOP_BNBWD ? lopx.z + imgbase + int64_t(hic) * W * C
                               : nullptr;
-        float xr[XW][4];
+        f32x2 xr[XW][2];
         // branchless: always load from a CLAMPED in-bounds address and
-        // zero by value select — a per-element branch around a load makes
-        // hipcc wait vmcnt(0) per load (serial HBM round trips)
+        // zero the raw bf16 vector by ONE select — a per-element branch
+        // around a load makes hipcc wait vmcnt(0) per load
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
-          const int wi = wbase + t;
-          const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
-          const bool ok = wi >= 0 && wi < W;
-          const int off = wic * C;
-          short4_t v = *reinterpret_cast<const short4_t*>(&xrow[off]);
-          const bool keep = ok && row_ok;
-          if (LOPK != LOP_BNBWD) {
+          short4_t v = *reinterpret_cast<const short4_t*>(&xrow[coff[t]]);
+          const bool keep = cok[t] && row_ok;
+          if (!keep) v = short4_t{0, 0, 0, 0};
+          if (LOPK == LOP_NONE) {
+#pragma unroll
+            for (int pp = 0; pp < 2; ++pp) {
+              xr[t][pp][0] = bf2f(ushort_t(v[pp * 2]));
+              xr[t][pp][1] = bf2f(ushort_t(v[pp * 2 + 1]));
+            }
+          } else if (LOPK == LOP_BNBWD) {
+            short4_t zv = *reinterpret_cast<const short4_t*>(
+                &zrow[coff[t]]);
+            if (!keep) zv = short4_t{0, 0, 0, 0};
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])), 0.f, e) : 0.f;
-          } else {
-            short4_t zv = *reinterpret_cast<const short4_t*>(&zrow[off]);
+              xr[t][e >> 1][e & 1] = lr.apply(bf2f(ushort_t(v[e])),
+                                              bf2f(ushort_t(zv[e])), e);
+          } else {  // LOP_BNACT
 #pragma unroll
             for (int e = 0; e < 4; ++e)
-              xr[t][e] = keep ? lr.apply(bf2f(ushort_t(v[e])),
-                                         bf2f(ushort_t(zv[e])), e)
-                              : 0.f;
+              xr[t][e >> 1][e & 1] =
+                  lr.apply(bf2f(ushort_t(v[e])), 0.f, e);
           }
         }
 #pragma unroll
@@ -135,12 +168,17 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
                 FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
             short4_t wv =
                 *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
+            f32x2 wf[2];
+#pragma unroll
+            for (int pp = 0; pp < 2; ++pp) {
+              wf[pp][0] = bf2f(ushort_t(wv[pp * 2]));
+              wf[pp][1] = bf2f(ushort_t(wv[pp * 2 + 1]));
+            }
 #pragma unroll
             for (int q = 0; q < 4; ++q)
 #pragma unroll
-              for (int e = 0; e < 4; ++e)
-                acc[t][q][e] = fmaf(xr[q * S + j][e],
-                                    bf2f(ushort_t(wv[e])), acc[t][q][e]);
+              for (int pp = 0; pp < 2; ++pp)
+                acc[t][q][pp] += xr[q * S + j][pp] * wf[pp];
           }
         }
       }
@@ -160,7 +198,7 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
             zq = *reinterpret_cast<const short4_t*>(&lopE.z[obase + q * C]);
 #pragma unroll
           for (int e = 0; e < 4; ++e) {
-            const ushort_t o = f2bf(acc[t][q][e]);
+            const ushort_t o = f2bf(acc[t][q][e >> 1][e & 1]);
             out[e] = short(o);
             if (EPI == 1) {
               const float v = bf2f(o);
@@ -314,10 +352,20 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
     lrx.init(lopx, c);
     lrd.init(lopdy, c);
     for (int p = lo + mlane; p < hi_; p += ML) {
-      const int w4 = p % Wo4;
-      const int rest = p / Wo4;
-      const int ho = rest % Ho;
-      const int n = rest / Ho;
+      // k5: walk down the column stripe (ho fastest) so the K-row
+      // re-reads stay in L1 (same reasoning as dw_fwd_kernel HGFAST)
+      int w4, ho, n;
+      if (K == 5) {
+        ho = p % Ho;
+        const int rest = p / Ho;
+        w4 = rest % Wo4;
+        n = rest / Wo4;
+      } else {
+        w4 = p % Wo4;
+        const int rest = p / Wo4;
+        ho = rest % Ho;
+        n = rest / Ho;
+      }
       const int wo0 = w4 * 4;
       const ushort_t* dyrow =
           dy + (int64_t(n) * Ho + ho) * int64_t(Wo) * C + c;
@@ -348,6 +396,16 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
       }
       const int wbase = wo0 * S - pad;
       const int64_t imgbase = int64_t(n) * H * W * C + c;
+      // column clamps are kernel-row invariant: hoist
+      int coff[XW];
+      bool cok[XW];
+#pragma unroll
+      for (int t = 0; t < XW; ++t) {
+        const int wi = wbase + t;
+        const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+        cok[t] = wi >= 0 && wi < W;
+        coff[t] = wic * C;
+      }
 #pragma unroll
       for (int ii = 0; ii < KR; ++ii) {
         const int i = i0 + ii;
@@ -359,10 +417,9 @@ __global__ __launch_bounds__(256) void dw_wgrad_kernel(
         float xr[XW][4];
 #pragma unroll
         for (int t = 0; t < XW; ++t) {
-          const int wi = wbase + t;
-          const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
-          const bool keep = row_ok && wi >= 0 && wi < W;
-          short4_t x4 = *reinterpret_cast<const short4_t*>(&xrow[wic * C]);
+          const bool keep = row_ok && cok[t];
+          short4_t x4 = *reinterpret_cast<const short4_t*>(
+              &xrow[coff[t]]);
           if (LX == LOP_NONE) {
 #pragma unroll
             for (int e = 0; e < 4; ++e)
