@@ -55,6 +55,21 @@ def _eval_ab(mean, invstd, gamma, beta, ab) -> None:
 #                    reductions slow the stream-optimal producer kernels
 #                    more than the removed partial pass costs.  Default off.
 _EPI_ON = os.environ.get("YAMS_BWD_EPI", "0") != "0"
+#   YAMS_WGRAD_STREAM=0  disable the side-stream weight-grad overlap
+#                    (default on: the per-block wgrad chain — two atb
+#                    GEMMs + dw wgrad + slab combines, ~0.25 ms/block of
+#                    latency-bound work — runs on a second HIP stream
+#                    under the bandwidth-bound dgrad/BN chain, joined at
+#                    the end of each block's backward)
+_WGRAD_STREAM_ON = os.environ.get("YAMS_WGRAD_STREAM", "1") != "0"
+
+_wgrad_stream = [None]
+
+
+def wgrad_stream():
+    if _wgrad_stream[0] is None:
+        _wgrad_stream[0] = torch.cuda.Stream()
+    return _wgrad_stream[0]
 
 
 class FusedInvertedResidualFn(torch.autograd.Function):
@@ -187,6 +202,21 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         cd = dww.shape[0]
         cp = pw.shape[0]
         ce = ew.shape[0] if has_expand else cin
+        # side-stream wgrad overlap: wgrads only read (dz, activations),
+        # so they run concurrently with the rest of the backward chain;
+        # a hard join before returning keeps allocator/stream semantics
+        # trivially safe (nothing outlives the block's backward)
+        overlap = (_WGRAD_STREAM_ON and
+                   not torch.cuda.is_current_stream_capturing())
+        if overlap:
+            sw = wgrad_stream()
+            cur = torch.cuda.current_stream()
+
+        def _on_wgrad_stream(ready_after=None):
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            sw.wait_event(ev)
+            return torch.cuda.stream(sw)
 
         # ---- project bn backward: dz3 materialized once ----------------
         # (computing dz as a BNBWD load-op inside both GEMMs costs more
@@ -211,12 +241,21 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         else:
             ws2 = sp2 = None
             e.pw_conv(dz3, pwt, d_in3)
-        dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
-        if has_se:
-            e.pw_conv_wgrad(dz3, y2s, dwp3)
+        if overlap:
+            with _on_wgrad_stream():
+                dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
+                if has_se:
+                    e.pw_conv_wgrad(dz3, y2s, dwp3)
+                else:
+                    e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1,
+                                    lopb_act=act, lopb_coef=ab2)
         else:
-            e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1, lopb_act=act,
-                            lopb_coef=ab2)
+            dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
+            if has_se:
+                e.pw_conv_wgrad(dz3, y2s, dwp3)
+            else:
+                e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1, lopb_act=act,
+                                lopb_coef=ab2)
         d_pw = dwp3.reshape(cp, cd, 1, 1)
 
         # ---- SE backward: fused per-image + wgrad kernels (se.hip) ------
@@ -257,8 +296,13 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         else:
             ws1 = sp1 = None
             e.dw_conv_dgrad(dz2, dwpk, d_z1out, k, stride, pad)
-        dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
-        e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
+        if overlap:
+            with _on_wgrad_stream():
+                dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
+                e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
+        else:
+            dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
+            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
         d_dww = dwgt.reshape(cd, 1, k, k)
 
         # ---- expand backward --------------------------------------------
@@ -279,8 +323,13 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             # residual grad fused into the dgrad epilogue (zero extra
             # passes; the eager add cost 3 passes x 10 blocks per step)
             e.pw_conv(dz1, ewt, d_x, residual=dyb if use_res else None)
-            dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
-            e.pw_conv_wgrad(dz1, x, dwp1)
+            if overlap:
+                with _on_wgrad_stream():
+                    dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
+                    e.pw_conv_wgrad(dz1, x, dwp1)
+            else:
+                dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
+                e.pw_conv_wgrad(dz1, x, dwp1)
             d_ew = dwp1.reshape(ce, cin, 1, 1)
         else:
             d_x = d_z1out
@@ -288,6 +337,11 @@ class FusedInvertedResidualFn(torch.autograd.Function):
 
         if use_res and not has_expand:
             d_x = d_x + dyb  # rare no-expand residual block (V3-L #1)
+
+        if overlap:
+            ev_w = torch.cuda.Event()
+            ev_w.record(sw)
+            cur.wait_event(ev_w)
 
         return (d_x, d_ew, dg1, db1_, None, None,
                 d_dww, dg2, db2_, None, None,
