@@ -594,6 +594,13 @@ int pw_wgrad_splits(int64_t M, int N1, int N2) {
   const int tiles = ceil_div(N1, tile_of(N1)) * ceil_div(N2, tile_of(N2));
   const int64_t max_s = (M + 127) / 128;
   int64_t s = target / tiles;
+  // cap the split count so the fp32 slab traffic (8*N1*N2 bytes per
+  // split, written then re-read) stays of the order of the bf16 input
+  // traffic (2*M*(N1+N2)) — uncapped, the 14^2/7^2 wgrads wrote 4-8x
+  // their input in slabs and ran at 300-800 GB/s (profiles/r02)
+  const int64_t ws_cap = std::max<int64_t>(
+      16, M * (N1 + N2) / (2LL * N1 * N2));
+  if (s > ws_cap) s = ws_cap;
   if (s < 1) s = 1;
   if (s > max_s) s = max_s;
   return int(s < 1 ? 1 : s);
