@@ -454,6 +454,17 @@ void bn_bwd_stats(const Tensor& dy, const Tensor& z, const Tensor& mean,
                    zr.M, zr.C, false, 0, cur_stream());
 }
 
+void bn_bwd_apply(const Tensor& dy, const Tensor& z, const Tensor& coef,
+                  int64_t act, Tensor& dz) {
+  auto zr = rows_of(z, "z");
+  auto gr = rows_of(dy, "dy");
+  TORCH_CHECK(zr.M == gr.M && zr.C == gr.C && zr.C % 8 == 0,
+              "bn_bwd_apply shape mismatch");
+  TORCH_CHECK(coef.numel() >= 4 * zr.C, "coef must be [4C]");
+  yams::bn_bwd_apply(gr.ptr, zr.ptr, fp32_req(coef, "coef"), int(act),
+                     dz.data_ptr(), zr.M, zr.C, cur_stream());
+}
+
 // combine conv-epilogue stats slabs [2, splits, C] into mean/invstd/ab
 void bn_finalize(const Tensor& ws, int64_t splits, Tensor& mean,
                  Tensor& invstd, Tensor& ab, const Tensor& gamma,
@@ -731,6 +742,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gamma"_a, "beta"_a, "act"_a, "training"_a, "dgamma"_a, "dbeta"_a,
         "dz"_a, "partial_ws"_a = c10::nullopt, "partial_splits"_a = 0);
   m.def("bn_bwd_stats", &bn_bwd_stats);
+  m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("bn_act_bwd_addpool", &bn_act_bwd_addpool);
   m.def("bias_act_fwd", &bias_act_fwd);
   m.def("bias_act_bwd", &bias_act_bwd);

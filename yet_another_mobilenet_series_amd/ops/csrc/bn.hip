@@ -479,6 +479,19 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
   });
 }
 
+// apply-only entry: dz from precomputed coef[4C] (used when the
+// partial/finalize ran via bn_bwd_stats and dz is materialized lazily,
+// e.g. on the side stream for wgrad consumers)
+void bn_bwd_apply(const void* dy, const void* z, const float* coef,
+                  int act, void* dz, int64_t M, int C, hipStream_t s) {
+  const int64_t total8 = M * (C >> 3);
+  ACT_SWITCH(act, {
+    bn_bwd_apply_kernel<kAct><<<elem_grid(total8), 256, 0, s>>>(
+        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(z),
+        coef, static_cast<ushort_t*>(dz), total8, C >> 3, C);
+  });
+}
+
 // bn backward where dy still needs the avg-pool grad added (SE blocks):
 // partial does add+reduce in one pass, then finalize + apply as usual.
 void bn_act_bwd_addpool(void* dy, const void* z, const void* dpooled,

@@ -108,6 +108,9 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
                 int act, bool training, float* dgamma, float* dbeta,
                 void* dz, float* ws, float* coef, int64_t M, int C,
                 bool apply, int ext_splits, hipStream_t s);
+// apply-only: dz = f(dy, z, coef[4C]) with precomputed coefs
+void bn_bwd_apply(const void* dy, const void* z, const float* coef,
+                  int act, void* dz, int64_t M, int C, hipStream_t s);
 // SE blocks: dy += avgpool-bwd(dpooled) fused into the partial pass
 void bn_act_bwd_addpool(void* dy, const void* z, const void* dpooled,
                         int HW, const float* mean, const float* invstd,

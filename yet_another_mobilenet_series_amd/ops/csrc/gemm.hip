@@ -253,9 +253,14 @@ void pw_conv(const void* x, const void* w, void* y,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   auto* rp = static_cast<const ushort_t*>(res);
-  if (rp != nullptr) {  // residual epilogue: plain-GEMM case only
+  if (rp != nullptr) {  // residual epilogue: plain or BNBWD-lop dgrad
     if (ep == 0 && lopa.kind == 0) {
       pw_conv_kernel<0, 0, 0, true><<<grid, 256, 0, s>>>(
+          ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      return;
+    }
+    if (ep == 0 && lopa.kind == 2) {
+      pw_conv_kernel<0, 2, 0, true><<<grid, 256, 0, s>>>(
           ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
       return;
     }

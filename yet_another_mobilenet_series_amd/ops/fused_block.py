@@ -62,6 +62,14 @@ _EPI_ON = os.environ.get("YAMS_BWD_EPI", "0") != "0"
 #                    under the bandwidth-bound dgrad/BN chain, joined at
 #                    the end of each block's backward)
 _WGRAD_STREAM_ON = os.environ.get("YAMS_WGRAD_STREAM", "1") != "0"
+#   YAMS_BWD_LOPDZ=1 never-materialize dz for the pw BNs: reduce-only
+#                    backward + LOP_BNBWD dz recompute in both GEMMs.
+#                    Measured NET-NEGATIVE again in round 2 (+1.6 ms
+#                    same-box A/B) even with the wgrads on the side
+#                    stream: the dual-stream staging slows the MFMA
+#                    GEMMs more than the removed apply pass costs.
+#                    Default off; kept as a measured experiment.
+_LOPDZ_ON = os.environ.get("YAMS_BWD_LOPDZ", "0") != "0"
 
 _wgrad_stream = [None]
 
@@ -218,44 +226,56 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             sw.wait_event(ev)
             return torch.cuda.stream(sw)
 
-        # ---- project bn backward: dz3 materialized once ----------------
-        # (computing dz as a BNBWD load-op inside both GEMMs costs more
-        # staging VALU/registers than the one dz write it saves)
+        # ---- project bn backward ---------------------------------------
         dg3 = torch.empty(cp, dtype=F32, device=dev)
         db3 = torch.empty_like(dg3)
-        dz3 = torch.empty_like(z3)
-        e.bn_act_bwd(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
-                     bool(training), dg3, db3, dz3)
-
         pwt = _packed(pw, "pw_t", lambda w: w.view(cp, cd).t().to(BF16)
                       .contiguous(), spec="t")
         d_in3 = torch.empty_like(z2)
-        if not has_se and _EPI_ON:
-            # d_in3 IS the dw-BN's backward dy: emit that BN's partial sums
-            # from the GEMM epilogue (EPI==2) so bn_bwd_partial never runs
-            m2 = n * ho * wo
-            sp2 = int(e.pw_stats_mblocks(m2))
-            ws2 = torch.empty(2 * sp2 * cd, dtype=F32, device=dev)
-            e.pw_conv(dz3, pwt, d_in3, stats_ws=ws2, bstats_act=act,
-                      bstats_coef=ab2, bstats_z=z2)
-        else:
+        lopdz = _LOPDZ_ON and not _EPI_ON
+        if lopdz:
+            # dz3 is NEVER materialized: reduce-only backward, then both
+            # GEMMs recompute dz on load (LOP_BNBWD) — kills the apply
+            # pass (read dy + read z + write dz) on the critical path
+            coef3 = torch.empty(4 * cp, dtype=F32, device=dev)
+            e.bn_bwd_stats(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
+                           bool(training), dg3, db3, coef3)
             ws2 = sp2 = None
-            e.pw_conv(dz3, pwt, d_in3)
-        if overlap:
-            with _on_wgrad_stream():
-                dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
-                if has_se:
-                    e.pw_conv_wgrad(dz3, y2s, dwp3)
-                else:
-                    e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1,
-                                    lopb_act=act, lopb_coef=ab2)
+            e.pw_conv(dyb, pwt, d_in3, lop_kind=2, lop_act=ACT_IDENTITY,
+                      lop_coef=coef3, lop_z=z3)
+            dz3 = dyb  # the wgrad A-stream applies the same lop
         else:
+            dz3 = torch.empty_like(z3)
+            e.bn_act_bwd(dyb, z3, mean3, invstd3, pg, pb, ACT_IDENTITY,
+                         bool(training), dg3, db3, dz3)
+            if not has_se and _EPI_ON:
+                # d_in3 IS the dw-BN's backward dy: emit that BN's partial
+                # sums from the GEMM epilogue (EPI==2)
+                m2 = n * ho * wo
+                sp2 = int(e.pw_stats_mblocks(m2))
+                ws2 = torch.empty(2 * sp2 * cd, dtype=F32, device=dev)
+                e.pw_conv(dz3, pwt, d_in3, stats_ws=ws2, bstats_act=act,
+                          bstats_coef=ab2, bstats_z=z2)
+            else:
+                ws2 = sp2 = None
+                e.pw_conv(dz3, pwt, d_in3)
+
+        def _wgrad3():
             dwp3 = torch.empty(cp, cd, dtype=F32, device=dev)
+            la = dict(lopa_kind=2, lopa_act=ACT_IDENTITY, lopa_coef=coef3,
+                      lopa_z=z3) if lopdz else {}
             if has_se:
-                e.pw_conv_wgrad(dz3, y2s, dwp3)
+                e.pw_conv_wgrad(dz3, y2s, dwp3, **la)
             else:
                 e.pw_conv_wgrad(dz3, z2, dwp3, lopb_kind=1, lopb_act=act,
-                                lopb_coef=ab2)
+                                lopb_coef=ab2, **la)
+            return dwp3
+
+        if overlap:
+            with _on_wgrad_stream():
+                dwp3 = _wgrad3()
+        else:
+            dwp3 = _wgrad3()
         d_pw = dwp3.reshape(cp, cd, 1, 1)
 
         # ---- SE backward: fused per-image + wgrad kernels (se.hip) ------
@@ -309,27 +329,40 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         if has_expand:
             dg1 = torch.empty(ce, dtype=F32, device=dev)
             db1_ = torch.empty_like(dg1)
-            dz1 = torch.empty_like(z1)
-            if ws1 is not None:
-                e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
-                             bool(training), dg1, db1_, dz1,
-                             partial_ws=ws1, partial_splits=sp1)
-            else:
-                e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
-                             bool(training), dg1, db1_, dz1)
             ewt = _packed(ew, "pw_t", lambda w: w.view(ce, cin).t().to(BF16)
                           .contiguous(), spec="t")
             d_x = _empty_like_nhwc(x)
-            # residual grad fused into the dgrad epilogue (zero extra
-            # passes; the eager add cost 3 passes x 10 blocks per step)
-            e.pw_conv(dz1, ewt, d_x, residual=dyb if use_res else None)
+            if lopdz and ws1 is None:
+                coef1 = torch.empty(4 * ce, dtype=F32, device=dev)
+                e.bn_bwd_stats(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                               bool(training), dg1, db1_, coef1)
+                # residual grad + BNBWD load-op fused into the dgrad
+                e.pw_conv(d_z1out, ewt, d_x, lop_kind=2, lop_act=act,
+                          lop_coef=coef1, lop_z=z1,
+                          residual=dyb if use_res else None)
+                dz1 = d_z1out
+                la1 = dict(lopa_kind=2, lopa_act=act, lopa_coef=coef1,
+                           lopa_z=z1)
+            else:
+                dz1 = torch.empty_like(z1)
+                if ws1 is not None:
+                    e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                                 bool(training), dg1, db1_, dz1,
+                                 partial_ws=ws1, partial_splits=sp1)
+                else:
+                    e.bn_act_bwd(d_z1out, z1, mean1, invstd1, eg, ebb, act,
+                                 bool(training), dg1, db1_, dz1)
+                # residual grad fused into the dgrad epilogue (zero extra
+                # passes; the eager add cost 3 passes x 10 blocks/step)
+                e.pw_conv(dz1, ewt, d_x, residual=dyb if use_res else None)
+                la1 = {}
             if overlap:
                 with _on_wgrad_stream():
                     dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
-                    e.pw_conv_wgrad(dz1, x, dwp1)
+                    e.pw_conv_wgrad(dz1, x, dwp1, **la1)
             else:
                 dwp1 = torch.empty(ce, cin, dtype=F32, device=dev)
-                e.pw_conv_wgrad(dz1, x, dwp1)
+                e.pw_conv_wgrad(dz1, x, dwp1, **la1)
             d_ew = dwp1.reshape(ce, cin, 1, 1)
         else:
             d_x = d_z1out
