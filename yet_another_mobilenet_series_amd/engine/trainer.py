@@ -46,6 +46,15 @@ class Trainer:
         model = model.to(self.device)
         if cfg.train.channels_last and self.device.type == "cuda":
             model = model.to(memory_format=torch.channels_last)
+        if getattr(cfg.dist, "sync_bn", False):
+            # SyncBN runs on the reference engine (exact cross-rank BN,
+            # ops/functional._SyncBNFn); the HIP fused path keeps BN
+            # local — the standard recipe at 256/GPU
+            from ..ops import backend
+            backend.set_syncbn(True)
+            backend.set_engine_mode("reference")
+            self.logger.info("sync_bn on: cross-rank BN statistics "
+                             "(reference engine)")
         reduce_dtype = (torch.bfloat16
                         if getattr(cfg.dist, "reduce_dtype", "fp32") == "bf16"
                         else None)

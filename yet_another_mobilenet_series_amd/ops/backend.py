@@ -87,3 +87,23 @@ def set_fused_blocks(on: bool) -> None:
 def ext():
     """The extension module; raises if missing (GPU path must not fall back)."""
     return load_extension(required=True)
+
+
+_SYNCBN = False
+
+
+def set_syncbn(on: bool) -> None:
+    """Cross-rank BN statistics (SURVEY.md §2.4 optional SyncBN): when on
+    and a process group is initialised, training-mode BN computes batch
+    statistics over the GLOBAL batch by all-reducing per-rank
+    {sum, sumsq} before finalising.  Off by default (at 256/GPU the local
+    statistics are the standard recipe)."""
+    global _SYNCBN
+    _SYNCBN = bool(on)
+
+
+def use_syncbn() -> bool:
+    if not _SYNCBN:
+        return False
+    import torch.distributed as dist
+    return dist.is_initialized() and dist.get_world_size() > 1

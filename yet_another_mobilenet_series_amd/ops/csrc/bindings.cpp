@@ -465,6 +465,29 @@ void bn_bwd_apply(const Tensor& dy, const Tensor& z, const Tensor& coef,
                      dz.data_ptr(), zr.M, zr.C, cur_stream());
 }
 
+// SyncBN: per-rank {sum, sumsq} [2, C] for cross-rank all-reduce
+void bn_sums(const Tensor& z, Tensor& out) {
+  auto zr = rows_of(z, "z");
+  TORCH_CHECK(zr.C % 8 == 0, "C must be a multiple of 8");
+  TORCH_CHECK(out.numel() == 2 * zr.C &&
+                  out.scalar_type() == torch::kFloat32,
+              "out must be fp32 [2C]");
+  auto ws = f32_ws(z, yams::bn_stats_ws_elems(zr.M, zr.C));
+  yams::bn_sums(zr.ptr, ws.data_ptr<float>(), out.data_ptr<float>(),
+                zr.M, zr.C, cur_stream());
+}
+
+// fold conv-epilogue slabs [2, splits, C] into [2, C] (SyncBN with
+// epilogue-produced statistics)
+void bn_fold(const Tensor& ws, int64_t splits, Tensor& out) {
+  const int C = int(out.numel() / 2);
+  TORCH_CHECK(out.scalar_type() == torch::kFloat32 &&
+                  ws.numel() >= 2 * splits * C,
+              "bn_fold shapes");
+  yams::bn_fold_sums(fp32_req(ws, "ws"), int(splits), out.data_ptr<float>(),
+                     C, cur_stream());
+}
+
 // combine conv-epilogue stats slabs [2, splits, C] into mean/invstd/ab
 void bn_finalize(const Tensor& ws, int64_t splits, Tensor& mean,
                  Tensor& invstd, Tensor& ab, const Tensor& gamma,
@@ -742,6 +765,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gamma"_a, "beta"_a, "act"_a, "training"_a, "dgamma"_a, "dbeta"_a,
         "dz"_a, "partial_ws"_a = c10::nullopt, "partial_splits"_a = 0);
   m.def("bn_bwd_stats", &bn_bwd_stats);
+  m.def("bn_sums", &bn_sums);
+  m.def("bn_fold", &bn_fold);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("bn_act_bwd_addpool", &bn_act_bwd_addpool);
   m.def("bias_act_fwd", &bias_act_fwd);

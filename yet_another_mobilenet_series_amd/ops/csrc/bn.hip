@@ -411,6 +411,27 @@ void bn_finalize(const float* ws, int splits, float* mean, float* invstd,
       splits, momentum, eps);
 }
 
+// SyncBN support: per-rank [sum, sumsq] of z reduced to ONE [2, C]
+// vector (partial slabs + parallel fold) — the caller all-reduces it
+// across ranks and finalizes with the GLOBAL element count via
+// bn_finalize(splits=1, M=M*world) (SURVEY.md §2.4 optional stat sync).
+void bn_sums(const void* z, float* ws, float* out, int64_t M, int C,
+             hipStream_t s) {
+  const RedGeom g = red_geom(M, C);
+  bn_stats_partial_kernel<<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
+      static_cast<const ushort_t*>(z), ws, M, C, g.chunks);
+  const int64_t total = 2LL * C;
+  const int blocks = int(std::min<int64_t>((total + 255) / 256, 2048));
+  bn_fold_kernel<<<dim3(blocks), 256, 0, s>>>(ws, out, g.chunks, 1, C);
+}
+
+void bn_fold_sums(const float* ws, int splits, float* out, int C,
+                  hipStream_t s) {
+  const int64_t total = 2LL * C;
+  const int blocks = int(std::min<int64_t>((total + 255) / 256, 2048));
+  bn_fold_kernel<<<dim3(blocks), 256, 0, s>>>(ws, out, splits, 1, C);
+}
+
 void bn_act_fwd(const void* z, const float* ab, int act, void* y,
                 const void* residual, int64_t M, int C, hipStream_t s) {
   const int64_t total8 = M * (C >> 3);

@@ -89,6 +89,14 @@ void bn_stats(const void* z, float* mean, float* invstd, float* ab,
 int bn_stats_chunks(int64_t M);
 int64_t bn_stats_ws_elems(int64_t M, int C);
 
+// SyncBN: one [2, C] {sum, sumsq} vector for cross-rank all-reduce;
+// ws sized bn_stats_ws_elems(M, C)
+void bn_sums(const void* z, float* ws, float* out, int64_t M, int C,
+             hipStream_t s);
+// fold epilogue slabs [2, splits, C] -> [2, C]
+void bn_fold_sums(const float* ws, int splits, float* out, int C,
+                  hipStream_t s);
+
 // finalize-only entry: combine sum/sumsq slabs ws[2, splits, C] produced
 // by a conv epilogue into mean/invstd/ab (+running update)
 void bn_finalize(const float* ws, int splits, float* mean, float* invstd,

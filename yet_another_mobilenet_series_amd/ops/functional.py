@@ -22,7 +22,7 @@ import torch
 import torch.nn.functional as F
 
 from . import hip_ops
-from .backend import use_hip
+from .backend import use_hip, use_syncbn
 
 ACT_IDENTITY = 0
 ACT_RELU = 1
@@ -94,12 +94,76 @@ def conv_bn_act(
     z = F.conv2d(x, weight, conv_bias, stride=stride, padding=padding,
                  groups=groups)
     if bn_weight is not None:
-        z = F.batch_norm(z, running_mean, running_var, bn_weight, bn_bias,
-                         training=training, momentum=momentum, eps=eps)
+        if training and use_syncbn():
+            z = _SyncBNFn.apply(z, bn_weight, bn_bias, running_mean,
+                                running_var, momentum, eps)
+        else:
+            z = F.batch_norm(z, running_mean, running_var, bn_weight,
+                             bn_bias, training=training, momentum=momentum,
+                             eps=eps)
     y = apply_act(z, act)
     if residual is not None:
         y = y + residual
     return y
+
+
+class _SyncBNFn(torch.autograd.Function):
+    """Exact cross-rank BatchNorm (SURVEY.md §2.4 SyncBN option).
+
+    Forward all-reduces per-rank {sum, sumsq, count}; backward
+    all-reduces {sum dy, sum dy*zhat} so dz matches single-process BN
+    over the concatenated global batch bit-for-math.  Runs on the
+    reference engine (``dist.sync_bn: true`` in config forces it); the
+    HIP fused path keeps BN local per GPU — the standard recipe at
+    batch 256/GPU (SURVEY.md §2.4 "keep local by default").
+    """
+
+    @staticmethod
+    def forward(ctx, z, gamma, beta, running_mean, running_var, momentum,
+                eps):
+        import torch.distributed as dist
+        n_local = z.numel() // z.shape[1]
+        sums = torch.empty(2, z.shape[1], dtype=torch.float64,
+                           device=z.device)
+        zf = z.detach().double()
+        sums[0] = zf.sum(dim=(0, 2, 3))
+        sums[1] = (zf * zf).sum(dim=(0, 2, 3))
+        dist.all_reduce(sums)
+        world = dist.get_world_size()
+        cnt = n_local * world
+        mean = (sums[0] / cnt).float()
+        var = (sums[1] / cnt).float() - mean * mean
+        var = var.clamp_min(0)
+        invstd = (var + eps).rsqrt()
+        if running_mean is not None:
+            with torch.no_grad():
+                unbiased = var * (cnt / max(cnt - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased,
+                                                    alpha=momentum)
+        sh = (1, -1, 1, 1)
+        zhat = (z - mean.view(sh)) * invstd.view(sh)
+        ctx.save_for_backward(zhat, gamma, invstd)
+        ctx.cnt = cnt
+        return zhat * gamma.view(sh) + beta.view(sh)
+
+    @staticmethod
+    def backward(ctx, dy):
+        import torch.distributed as dist
+        zhat, gamma, invstd = ctx.saved_tensors
+        sh = (1, -1, 1, 1)
+        red = torch.empty(2, dy.shape[1], dtype=torch.float64,
+                          device=dy.device)
+        red[0] = dy.double().sum(dim=(0, 2, 3))
+        red[1] = (dy.double() * zhat.double()).sum(dim=(0, 2, 3))
+        dist.all_reduce(red)
+        dbeta = red[0].float()
+        dgamma = red[1].float()
+        m = ctx.cnt
+        dz = (gamma.view(sh) * invstd.view(sh)) * (
+            dy - dbeta.view(sh) / m - zhat * dgamma.view(sh) / m)
+        return (dz.to(dy.dtype), dgamma.to(gamma.dtype),
+                dbeta.to(gamma.dtype), None, None, None, None)
 
 
 def squeeze_excite(

@@ -144,7 +144,7 @@ DEFAULTS: Dict[str, Any] = {
               "engine": "auto", "use_cuda_graph": True},
     "dist": {"backend": "auto", "bucket_mb": None, "overlap": True,
              "grad_dtype": "fp32", "reduce_dtype": "fp32",
-             "timeout_sec": 300, "watchdog_sec": 0},
+             "timeout_sec": 300, "watchdog_sec": 0, "sync_bn": False},
     "log": {"dir": "runs/default", "jsonl": True},
     "val": {"interval_epochs": 1, "batch_size_per_gpu": 256, "use_ema": True},
 }
