@@ -70,6 +70,11 @@ _WGRAD_STREAM_ON = os.environ.get("YAMS_WGRAD_STREAM", "1") != "0"
 #                    GEMMs more than the removed apply pass costs.
 #                    Default off; kept as a measured experiment.
 _LOPDZ_ON = os.environ.get("YAMS_BWD_LOPDZ", "0") != "0"
+#   YAMS_DW_LOP=1    never materialize y1: the dw forward AND the dw
+#                    wgrad read z1 with the expand BN+act as a load-op.
+#                    Round-1 measured this negative (register pressure);
+#                    re-tried after the packed-f32 dw rewrite.
+_DW_LOP_ON = os.environ.get("YAMS_DW_LOP", "0") != "0"
 
 _wgrad_stream = [None]
 
@@ -112,15 +117,24 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                 mean1.copy_(erm)
                 invstd1.copy_((erv.float() + eps).rsqrt())
                 _eval_ab(mean1, invstd1, eg, ebb, ab1)
-            # materialize y1: the dw window-load lop costs more registers
-            # than this memory-bound pass costs time (measured)
-            y1 = torch.empty_like(z1)
-            e.bn_act_fwd(z1, ab1, act, y1, None)
-            dw_in = y1
+            if _DW_LOP_ON:
+                # y1 never exists: dw fwd/wgrad apply BN+act on load
+                y1 = None
+                dw_in = z1
+                dw_lop = (1, act, ab1)
+            else:
+                # materialize y1: the dw window-load lop costs more
+                # registers than this memory-bound pass costs time
+                # (round-1 measurement; YAMS_DW_LOP re-tests it)
+                y1 = torch.empty_like(z1)
+                e.bn_act_fwd(z1, ab1, act, y1, None)
+                dw_in = y1
+                dw_lop = (0, 0, None)
         else:
             z1, mean1, invstd1, ab1 = x, None, None, None
             y1 = None
             dw_in = x
+            dw_lop = (0, 0, None)
             ce = cin
 
         # ---- depthwise ------------------------------------------------
@@ -136,11 +150,15 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         if training:
             chunks = e.dw_stats_chunks(n, ho, wo, cd, stride, k)
             ws = torch.empty(2 * chunks * cd, dtype=F32, device=dev)
-            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad, stats_ws=ws)
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
+                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
+                          lop_coef=dw_lop[2], stats_ws=ws)
             e.bn_finalize(ws, chunks, mean2, invstd2, ab2, dg, db,
                           drm, drv, momentum, eps, m2)
         else:
-            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad)
+            e.dw_conv_fwd(dw_in, dwp, z2, k, stride, pad,
+                          lop_kind=dw_lop[0], lop_act=dw_lop[1],
+                          lop_coef=dw_lop[2])
             mean2.copy_(drm)
             invstd2.copy_((drv.float() + eps).rsqrt())
             _eval_ab(mean2, invstd2, dg, db, ab2)
@@ -304,7 +322,12 @@ class FusedInvertedResidualFn(torch.autograd.Function):
                          bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)
                        .contiguous(), spec="t")
-        dw_in = y1 if has_expand else x
+        dw_lop_b = {}
+        if has_expand and y1 is None:
+            dw_in = z1  # YAMS_DW_LOP: wgrad re-applies BN+act on load
+            dw_lop_b = dict(lopx_kind=1, lopx_act=act, lopx_coef=ab1)
+        else:
+            dw_in = y1 if has_expand else x
         d_z1out = torch.empty_like(dw_in)
         if has_expand and stride == 1 and _EPI_ON:
             # d_z1out feeds the expand BN's backward: fuse its partials
@@ -319,10 +342,11 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         if overlap:
             with _on_wgrad_stream():
                 dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
-                e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
+                e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad,
+                                **dw_lop_b)
         else:
             dwgt = torch.empty(cd, k * k, dtype=F32, device=dev)
-            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad)
+            e.dw_conv_wgrad(dw_in, dz2, dwgt, k, stride, pad, **dw_lop_b)
         d_dww = dwgt.reshape(cd, 1, k, k)
 
         # ---- expand backward --------------------------------------------
