@@ -14,6 +14,7 @@
 //     (SURVEY.md §5.2).
 
 #include <algorithm>
+#include <cstdlib>
 
 #include "common.h"
 #include "engine.h"
@@ -47,7 +48,9 @@ inline Lop to_lop(const LopDesc& d) {
 // z tensor, the ab[2C] coefs and the act for EPI==2.
 // RES: epilogue residual add (C = A@B^T + res) — used by the backward's
 // d_x = pw_dgrad(dz1) + dyb so the residual path costs zero extra passes
-template <int EPI, int LOPK, int EACT = 0, bool RES = false>
+// TBM: M-tile — 128 default; 64 for small-M dgrads, where ceil(M/128)
+// blocks underfill the 256-CU chip (e.g. 7^2 layers: 98x3 blocks)
+template <int EPI, int LOPK, int EACT = 0, bool RES = false, int TBM = BM>
 __global__ __launch_bounds__(256) void pw_conv_kernel(
     const ushort_t* __restrict__ A, const ushort_t* __restrict__ Bt,
     ushort_t* __restrict__ C, int64_t M, int K, int N, Lop lopA, Lop lopE,
@@ -56,45 +59,49 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
   // global loads before the current tile's MFMAs, write LDS after the
   // barrier) — the single-buffered loop parked 47% of wave cycles on
   // staging waits (profiles/gemm_counter_collection.csv)
-  __shared__ ushort_t Alds[2][BM * LDA];
+  constexpr int FRAG_Mv = TBM / 32;   // 16x16 frags per wave (M dim)
+  constexpr int TPRA = 256 / TBM;     // staging threads per A row
+  constexpr int APT = BK / TPRA;      // A elems staged per thread
+  __shared__ ushort_t Alds[2][TBM * LDA];
   __shared__ ushort_t Blds[2][BN * LDA];
 
   const int tid = threadIdx.x;
-  const int64_t bm = int64_t(blockIdx.x) * BM;
+  const int64_t bm = int64_t(blockIdx.x) * TBM;
   const int bn = blockIdx.y * BN;
   const int wave = tid >> 6, lane = tid & 63;
-  const int wrow0 = (wave >> 1) * 64, wcol0 = (wave & 1) * 32;
+  const int wrow0 = (wave >> 1) * (TBM / 2), wcol0 = (wave & 1) * 32;
   const int fr = lane & 15, fk = (lane >> 4) * 8;
 
-  f32x4 acc[FRAG_M][FRAG_N];
+  f32x4 acc[FRAG_Mv][FRAG_N];
 #pragma unroll
-  for (int i = 0; i < FRAG_M; ++i)
+  for (int i = 0; i < FRAG_Mv; ++i)
 #pragma unroll
     for (int j = 0; j < FRAG_N; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int a_row = tid >> 1, a_seg = (tid & 1) * 16;   // 2 thr/row x 128
+  const int a_row = tid / TPRA, a_seg = (tid % TPRA) * APT;
   const int b_row = tid >> 2, b_seg = (tid & 3) * 8;    // 4 thr/row x 64
   const int64_t gm = bm + a_row;
   const int gn = bn + b_row;
 
-  ushort_t areg[16];
+  ushort_t areg[APT];
   ushort_t breg[8];
 
   auto stage_regs = [&](int k0) {
     const int gka = k0 + a_seg;
-    if (gm < M && gka + 16 <= K) {
+    if (gm < M && gka + APT <= K) {
       const int64_t off = gm * K + gka;
-      lop_vec8<LOPK>(lopA, A, off, gka, areg);
-      lop_vec8<LOPK>(lopA, A, off + 8, gka + 8, areg + 8);
+#pragma unroll
+      for (int g8 = 0; g8 < APT / 8; ++g8)
+        lop_vec8<LOPK>(lopA, A, off + g8 * 8, gka + g8 * 8, areg + g8 * 8);
     } else if (gm >= M || gka >= K) {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) areg[j] = ushort_t(0);
+      for (int j = 0; j < APT; ++j) areg[j] = ushort_t(0);
     } else {
       // partial segment: per-element conditional loads (the segment-level
       // all-OOB cases above are wave-uniform and skip loads entirely;
       // always-executed clamped loads measured slower for lop variants)
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
+      for (int j = 0; j < APT; ++j) {
         const int gkj = gka + j;
         areg[j] = (gm < M && gkj < K)
                       ? ushort_t(f2bf(lop_loadT<LOPK>(lopA, A,
@@ -121,10 +128,10 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 
   auto write_lds = [&](int buf) {
     ushort_t* da = &Alds[buf][a_row * LDA + a_seg];
-    *reinterpret_cast<short8_t*>(da) =
-        *reinterpret_cast<const short8_t*>(areg);
-    *reinterpret_cast<short8_t*>(da + 8) =
-        *reinterpret_cast<const short8_t*>(areg + 8);
+#pragma unroll
+    for (int g8 = 0; g8 < APT / 8; ++g8)
+      *reinterpret_cast<short8_t*>(da + g8 * 8) =
+          *reinterpret_cast<const short8_t*>(areg + g8 * 8);
     ushort_t* db = &Blds[buf][b_row * LDA + b_seg];
     *reinterpret_cast<short8_t*>(db) =
         *reinterpret_cast<const short8_t*>(breg);
@@ -137,9 +144,9 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
   for (int t = 0; t < nt; ++t) {
     if (t + 1 < nt) stage_regs((t + 1) * BK);  // loads fly under the MFMAs
     const int cur = t & 1;
-    bf16x8 a_frag[FRAG_M], b_frag[FRAG_N];
+    bf16x8 a_frag[FRAG_Mv], b_frag[FRAG_N];
 #pragma unroll
-    for (int i = 0; i < FRAG_M; ++i)
+    for (int i = 0; i < FRAG_Mv; ++i)
       a_frag[i] = *reinterpret_cast<const bf16x8*>(
           &Alds[cur][(wrow0 + i * 16 + fr) * LDA + fk]);
 #pragma unroll
@@ -147,7 +154,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
       b_frag[j] = *reinterpret_cast<const bf16x8*>(
           &Blds[cur][(wcol0 + j * 16 + fr) * LDA + fk]);
 #pragma unroll
-    for (int i = 0; i < FRAG_M; ++i)
+    for (int i = 0; i < FRAG_Mv; ++i)
 #pragma unroll
       for (int j = 0; j < FRAG_N; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -162,7 +169,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
   // epilogue: C/D frag layout col = lane&15, row = (lane>>4)*4 + r
   const int col_l = fr, row_l4 = (lane >> 4) * 4;
 #pragma unroll
-  for (int i = 0; i < FRAG_M; ++i) {
+  for (int i = 0; i < FRAG_Mv; ++i) {
 #pragma unroll
     for (int j = 0; j < FRAG_N; ++j) {
       const int col = bn + wcol0 + j * 16 + col_l;
@@ -193,7 +200,7 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
         eb = lopE.coef[N + col];
       }
 #pragma unroll
-      for (int i = 0; i < FRAG_M; ++i)
+      for (int i = 0; i < FRAG_Mv; ++i)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
@@ -245,7 +252,6 @@ int pw_stats_mblocks(int64_t M) { return ceil_div(M, BM); }
 void pw_conv(const void* x, const void* w, void* y,
              int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
              float* stats_ws, const void* res, hipStream_t s) {
-  dim3 grid(ceil_div(M, BM), ceil_div(N, BN));
   const Lop L = to_lop(lopa);
   const Lop LE = to_lop(epi);
   const int ep = stats_ws == nullptr ? 0 : (epi.kind == LOP_BNACT ? 2 : 1);
@@ -253,18 +259,48 @@ void pw_conv(const void* x, const void* w, void* y,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   auto* rp = static_cast<const ushort_t*>(res);
+  // small-M stats-free launches (dgrads at 7^2/14^2) use the BM=64 tile:
+  // twice the blocks on a chip that needs >>256 workgroups; 14^2 shapes
+  // measured slightly WORSE at BM=64 (more B re-staging), so only the
+  // 7^2 tier uses it (measurement
+  // knob YAMS_PW_BM64=0 disables)
+  static int bm64_on = [] {
+    const char* e = getenv("YAMS_PW_BM64");
+    return e == nullptr || atoi(e) != 0;
+  }();
+  const bool small = bm64_on && ep == 0 && M <= 16384;
+  dim3 grid(ceil_div(M, small ? 64 : BM), ceil_div(N, BN));
   if (rp != nullptr) {  // residual epilogue: plain or BNBWD-lop dgrad
     if (ep == 0 && lopa.kind == 0) {
-      pw_conv_kernel<0, 0, 0, true><<<grid, 256, 0, s>>>(
-          ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      if (small)
+        pw_conv_kernel<0, 0, 0, true, 64><<<grid, 256, 0, s>>>(
+            ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      else
+        pw_conv_kernel<0, 0, 0, true><<<grid, 256, 0, s>>>(
+            ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
       return;
     }
     if (ep == 0 && lopa.kind == 2) {
-      pw_conv_kernel<0, 2, 0, true><<<grid, 256, 0, s>>>(
-          ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      if (small)
+        pw_conv_kernel<0, 2, 0, true, 64><<<grid, 256, 0, s>>>(
+            ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
+      else
+        pw_conv_kernel<0, 2, 0, true><<<grid, 256, 0, s>>>(
+            ap, wp, yp, M, K, N, L, LE, stats_ws, rp);
       return;
     }
     yams_launch_fail("pw_conv_res", ep, lopa.kind, lopa.act, epi.act);
+  }
+  if (small) {
+#define PW64_CASE(LK)                                                     \
+    if (lopa.kind == LK) {                                                \
+      pw_conv_kernel<0, LK, 0, false, 64><<<grid, 256, 0, s>>>(           \
+          ap, wp, yp, M, K, N, L, LE, stats_ws, nullptr);                 \
+      return;                                                             \
+    }
+    PW64_CASE(0) PW64_CASE(1) PW64_CASE(2)
+#undef PW64_CASE
+    yams_launch_fail("pw_conv64", ep, lopa.kind, lopa.act, epi.act);
   }
 #define PW_CASE(EP, LK)                                                   \
   if (ep == EP && lopa.kind == LK) {                                      \
