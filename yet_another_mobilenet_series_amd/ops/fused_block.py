@@ -75,6 +75,13 @@ _LOPDZ_ON = os.environ.get("YAMS_BWD_LOPDZ", "0") != "0"
 #                    Round-1 measured this negative (register pressure);
 #                    re-tried after the packed-f32 dw rewrite.
 _DW_LOP_ON = os.environ.get("YAMS_DW_LOP", "0") != "0"
+#   YAMS_GRAPH_OVERLAP=1  keep the side-stream wgrads during hipGraph
+#                    capture (stream-capture propagates across the event
+#                    dependencies; measured: graph+overlap 14.41 vs
+#                    graph-only 14.83 vs eager+overlap 13.77 ms — graphs
+#                    stay off by default, but --graph runs capture the
+#                    overlapped step)
+_GRAPH_OVERLAP_ON = os.environ.get("YAMS_GRAPH_OVERLAP", "1") != "0"
 
 _wgrad_stream = [None]
 
@@ -232,8 +239,9 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         # so they run concurrently with the rest of the backward chain;
         # a hard join before returning keeps allocator/stream semantics
         # trivially safe (nothing outlives the block's backward)
-        overlap = (_WGRAD_STREAM_ON and
-                   not torch.cuda.is_current_stream_capturing())
+        overlap = _WGRAD_STREAM_ON and (
+            _GRAPH_OVERLAP_ON
+            or not torch.cuda.is_current_stream_capturing())
         if overlap:
             sw = wgrad_stream()
             cur = torch.cuda.current_stream()
