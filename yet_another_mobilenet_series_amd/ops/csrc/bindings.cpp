@@ -274,7 +274,9 @@ void se_fwd(const Tensor& x, const Tensor& w1t, const Tensor& b1,
 
 void se_bwd(const Tensor& dy, const Tensor& x, const Tensor& w1,
             const Tensor& w2, const Tensor& ha, const Tensor& sz,
-            Tensor& dx, Tensor& dsz, Tensor& dhz) {
+            Tensor& dx, Tensor& dsz, Tensor& dhz,
+            c10::optional<Tensor> bn_z, c10::optional<Tensor> bn_ab,
+            int64_t bn_act, c10::optional<Tensor> bn_stats_ws) {
   auto gr = rows_of(dy, "dy");
   auto xr = rows_of(x, "x");
   auto dxr = rows_of(dx, "dx");
@@ -288,10 +290,24 @@ void se_bwd(const Tensor& dy, const Tensor& x, const Tensor& w1,
   TORCH_CHECK(dsz.numel() >= int64_t(N) * C &&
                   dhz.numel() >= int64_t(N) * R,
               "se_bwd stage buffers too small");
+  const void* z2p = nullptr;
+  const float* abp = nullptr;
+  float* swp = nullptr;
+  if (bn_stats_ws.has_value()) {
+    TORCH_CHECK(bn_z.has_value() && bn_ab.has_value(),
+                "se_bwd bn stats need z and ab");
+    auto zr = rows_of(*bn_z, "bn_z");
+    TORCH_CHECK(zr.M == xr.M && zr.C == C, "se_bwd bn_z mismatch");
+    TORCH_CHECK(bn_stats_ws->numel() >= 2LL * N * C,
+                "bn_stats_ws too small");
+    z2p = zr.ptr;
+    abp = fp32_req(*bn_ab, "bn_ab");
+    swp = bn_stats_ws->data_ptr<float>();
+  }
   yams::se_bwd(gr.ptr, xr.ptr, w1.data_ptr(), w2.data_ptr(),
                fp32_req(ha, "ha"), fp32_req(sz, "sz"), dx.data_ptr(),
                dsz.data_ptr<float>(), dhz.data_ptr<float>(), N, HW, C, R,
-               cur_stream());
+               z2p, abp, int(bn_act), swp, cur_stream());
 }
 
 void se_wgrad(const Tensor& dsz, const Tensor& ha, const Tensor& dhz,
@@ -749,7 +765,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "sz"_a, "lop_kind"_a = 0, "lop_act"_a = 0,
         "lop_coef"_a = c10::nullopt, "ymat"_a = c10::nullopt);
   m.def("se_bwd", &se_bwd, "dy"_a, "x"_a, "w1"_a, "w2"_a, "ha"_a, "sz"_a,
-        "dx"_a, "dsz"_a, "dhz"_a);
+        "dx"_a, "dsz"_a, "dhz"_a, "bn_z"_a = c10::nullopt,
+        "bn_ab"_a = c10::nullopt, "bn_act"_a = 0,
+        "bn_stats_ws"_a = c10::nullopt);
   m.def("se_wgrad", &se_wgrad, "dsz"_a, "ha"_a, "dhz"_a, "pooled"_a,
         "dw1"_a, "dw2"_a, "db1"_a, "db2"_a);
   m.def("dropout", &dropout, "x"_a, "y"_a, "seed"_a, "p"_a,

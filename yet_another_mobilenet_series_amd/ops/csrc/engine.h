@@ -163,10 +163,14 @@ void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
             const float* b2, void* y, void* ymat, float* pooled,
             float* ha, float* sz, int N, int HW, int C, int R,
             LopDesc lopx, hipStream_t s);
-// dx = dy*s + avgpool_bwd(dpooled); dsz[N,C]/dhz[N,R] staged for se_wgrad
+// dx = dy*s + avgpool_bwd(dpooled); dsz[N,C]/dhz[N,R] staged for se_wgrad.
+// stats_ws != null: also emit the dw BN's backward partials [2, N, C]
+// (z2/bnab/bnact describe that BN) while writing dx — combine with
+// bn_act_bwd(partial_splits = N)
 void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
             const float* ha, const float* sz, void* dx, float* dsz,
-            float* dhz, int N, int HW, int C, int R, hipStream_t s);
+            float* dhz, int N, int HW, int C, int R, const void* z2,
+            const float* bnab, int bnact, float* stats_ws, hipStream_t s);
 // dw1[R,C], dw2[C,R], db1[R], db2[C] — deterministic serial-N reduction
 void se_wgrad(const float* dsz, const float* ha, const float* dhz,
               const float* pooled, float* dw1, float* dw2, float* db1,

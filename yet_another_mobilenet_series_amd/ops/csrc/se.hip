@@ -197,13 +197,19 @@ __global__ __launch_bounds__(BS) void se_fwd_kernel(
 
 // backward per-image pass: dx (incl. the avgpool backward contribution),
 // dsz[N,C] and dhz[N,R] staged fp32 for the cross-batch wgrad kernel.
-template <int BS>
+// BNSTATS: dx feeds the dw BN's backward — emit that BN's partial sums
+// (sum da, sum da*z2 with da = dx * act'(a*z2+b)) into stats_ws[2][N][C]
+// while WRITING dx, so the separate bn_bwd_partial pass (a full read of
+// dx + z2) never runs for SE blocks (the round-1 addpool-fusion pattern).
+template <int BS, bool BNSTATS = false>
 __global__ __launch_bounds__(BS) void se_bwd_kernel(
     const ushort_t* __restrict__ dy, const ushort_t* __restrict__ x,
     const ushort_t* __restrict__ w1, const ushort_t* __restrict__ w2,
     const float* __restrict__ ha_g, const float* __restrict__ sz_g,
     ushort_t* __restrict__ dx, float* __restrict__ dsz_g,
-    float* __restrict__ dhz_g, int HW, int C, int R) {
+    float* __restrict__ dhz_g, int HW, int C, int R,
+    const ushort_t* __restrict__ z2, const float* __restrict__ bnab,
+    int bnact, float* __restrict__ stats_ws) {
   __shared__ float red[BS * 8];
   __shared__ float dsz_s[MAX_C];
   __shared__ float dhz_s[MAX_R];
@@ -312,19 +318,83 @@ __global__ __launch_bounds__(BS) void se_bwd_kernel(
   }
   __syncthreads();
 
-  // ---- phase 5: dx = dy*s + dpooled/HW (flat vec8 walk) ----------------
-  const int total8 = HW * C8;
-  for (int i = tid; i < total8; i += BS) {
-    const int cc = (i % C8) * 8;
-    const int64_t off = base + int64_t(i) * 8;
-    short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
-    short8_t o8;
+  // ---- phase 5: dx = dy*s + dpooled/HW -----------------------------
+  if (!BNSTATS) {
+    // flat vec8 walk, fully coalesced
+    const int total8 = HW * C8;
+    for (int i = tid; i < total8; i += BS) {
+      const int cc = (i % C8) * 8;
+      const int64_t off = base + int64_t(i) * 8;
+      short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
+      short8_t o8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        o8[e] = short(f2bf(fmaf(bf2f(ushort_t(g8[e])), s_s[cc + e],
+                                dpool_s[cc + e])));
+      *reinterpret_cast<short8_t*>(&dx[off]) = o8;
+    }
+    return;
+  }
+  // BNSTATS variant: per-channel accumulators need a fixed channel per
+  // thread — strided (c8, mlane) walk like phase 1, then tree-fold
+  float s1[8] = {}, s2v[8] = {};
+  if (mlane < ML) {
+    float a[8], b[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      a[e] = bnab[c + e];
+      b[e] = bnab[C + c + e];
+    }
+    for (int hw = mlane; hw < HW; hw += ML) {
+      const int64_t off = base + int64_t(hw) * C + c;
+      short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[off]);
+      short8_t z8 = *reinterpret_cast<const short8_t*>(&z2[off]);
+      short8_t o8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const ushort_t o = f2bf(fmaf(bf2f(ushort_t(g8[e])), s_s[c8 * 8 + e],
+                                     dpool_s[c8 * 8 + e]));
+        o8[e] = short(o);
+        const float zv = bf2f(ushort_t(z8[e]));
+        const float da =
+            bf2f(o) * act_bwd_rt(fmaf(zv, a[e], b[e]), bnact);
+        s1[e] += da;
+        s2v[e] = fmaf(da, zv, s2v[e]);
+      }
+      *reinterpret_cast<short8_t*>(&dx[off]) = o8;
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) red[tid * 8 + e] = s1[e];
+  }
+  __syncthreads();
+  for (int half = ML >> 1; half > 0; half >>= 1) {
+    if (mlane < half)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+    __syncthreads();
+  }
+  if (tid < C8)
 #pragma unroll
     for (int e = 0; e < 8; ++e)
-      o8[e] = short(f2bf(fmaf(bf2f(ushort_t(g8[e])), s_s[cc + e],
-                              dpool_s[cc + e])));
-    *reinterpret_cast<short8_t*>(&dx[off]) = o8;
+      stats_ws[int64_t(n) * C + tid * 8 + e] = red[tid * 8 + e];
+  __syncthreads();
+  if (mlane < ML)
+#pragma unroll
+    for (int e = 0; e < 8; ++e) red[tid * 8 + e] = s2v[e];
+  __syncthreads();
+  for (int half = ML >> 1; half > 0; half >>= 1) {
+    if (mlane < half)
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        red[tid * 8 + e] += red[(tid + half * C8) * 8 + e];
+    __syncthreads();
   }
+  if (tid < C8)
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      stats_ws[(int64_t(gridDim.x) + n) * C + tid * 8 + e] =
+          red[tid * 8 + e];
 }
 
 // cross-batch weight/bias grads, deterministic (serial n chunks, fixed
@@ -460,19 +530,36 @@ void se_fwd(const void* x, const void* w1, const float* b1, const void* w2,
 
 void se_bwd(const void* dy, const void* x, const void* w1, const void* w2,
             const float* ha, const float* sz, void* dx, float* dsz,
-            float* dhz, int N, int HW, int C, int R, hipStream_t s) {
+            float* dhz, int N, int HW, int C, int R, const void* z2,
+            const float* bnab, int bnact, float* stats_ws,
+            hipStream_t s) {
   if ((C & 7) || C > 960 || R > MAX_R)
     yams_launch_fail("se_bwd_shape", C, R, HW, 0);
+  auto* dyp = static_cast<const ushort_t*>(dy);
+  auto* xp = static_cast<const ushort_t*>(x);
+  auto* w1p = static_cast<const ushort_t*>(w1);
+  auto* w2p = static_cast<const ushort_t*>(w2);
+  auto* dxp = static_cast<ushort_t*>(dx);
+  auto* z2p = static_cast<const ushort_t*>(z2);
+  if (stats_ws != nullptr) {
+    if (se_block_size() == 1024)
+      se_bwd_kernel<1024, true><<<dim3(N), 1024, 0, s>>>(
+          dyp, xp, w1p, w2p, ha, sz, dxp, dsz, dhz, HW, C, R, z2p, bnab,
+          bnact, stats_ws);
+    else
+      se_bwd_kernel<256, true><<<dim3(N), 256, 0, s>>>(
+          dyp, xp, w1p, w2p, ha, sz, dxp, dsz, dhz, HW, C, R, z2p, bnab,
+          bnact, stats_ws);
+    return;
+  }
   if (se_block_size() == 1024)
     se_bwd_kernel<1024><<<dim3(N), 1024, 0, s>>>(
-        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
-        static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
-        ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
+        dyp, xp, w1p, w2p, ha, sz, dxp, dsz, dhz, HW, C, R, nullptr,
+        nullptr, 0, nullptr);
   else
     se_bwd_kernel<256><<<dim3(N), 256, 0, s>>>(
-        static_cast<const ushort_t*>(dy), static_cast<const ushort_t*>(x),
-        static_cast<const ushort_t*>(w1), static_cast<const ushort_t*>(w2),
-        ha, sz, static_cast<ushort_t*>(dx), dsz, dhz, HW, C, R);
+        dyp, xp, w1p, w2p, ha, sz, dxp, dsz, dhz, HW, C, R, nullptr,
+        nullptr, 0, nullptr);
 }
 
 void se_wgrad(const float* dsz, const float* ha, const float* dhz,

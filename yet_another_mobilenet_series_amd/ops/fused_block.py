@@ -307,12 +307,17 @@ class FusedInvertedResidualFn(torch.autograd.Function):
         # ---- SE backward: fused per-image + wgrad kernels (se.hip) ------
         if has_se:
             from .hip_ops import se_backward
+            # the dw BN's backward partials ride se_bwd's dx-writing
+            # phase (one less full pass over d_y2 + z2, one less launch)
+            se_ws = torch.empty(2 * n * cd, dtype=F32, device=dev)
             d_y2, dw1p, db1se, dw2p, db2se = se_backward(
-                e, d_in3, y2, sw1, sw2, pooled, ha, sz)
+                e, d_in3, y2, sw1, sw2, pooled, ha, sz,
+                bn=(z2, ab2, act, se_ws))
             d_sw1 = dw1p.reshape_as(sw1)
             d_sw2 = dw2p.reshape_as(sw2)
         else:
             d_y2 = d_in3
+            se_ws = None
             d_sw1 = d_sw2 = db1se = db2se = None
 
         # ---- depthwise bn backward + grads (dz2 materialized) -----------
@@ -323,9 +328,12 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2,
                          partial_ws=ws2, partial_splits=sp2)
+        elif se_ws is not None:
+            # partial sums already emitted by se_bwd's dx phase
+            e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
+                         bool(training), dg2, db2_, dz2,
+                         partial_ws=se_ws, partial_splits=n)
         else:
-            # (SE blocks: the avgpool backward contribution is already in
-            # d_y2 — the fused se_bwd kernel adds it while writing dx)
             e.bn_act_bwd(d_y2, z2, mean2, invstd2, dg, db, act,
                          bool(training), dg2, db2_, dz2)
         dwpk = _packed(dww, "dw", lambda w: w.view(cd, k * k).t().to(BF16)

@@ -424,17 +424,23 @@ def se_forward(e, x, w1, b1, w2, b2, lop=None, ymat=None):
     return y, pooled, ha, sz
 
 
-def se_backward(e, dy, x, w1, w2, pooled, ha, sz):
+def se_backward(e, dy, x, w1, w2, pooled, ha, sz, bn=None):
     """Fused SE backward: one per-image kernel (dx including the pool
     gradient) + one deterministic cross-batch wgrad kernel.  All
-    SE-internal gradients stay fp32.  Returns (dx, dw1, db1, dw2, db2)."""
+    SE-internal gradients stay fp32.  ``bn=(z2, ab2, act, stats_ws)``
+    additionally emits the dw BN's backward partials while writing dx.
+    Returns (dx, dw1, db1, dw2, db2)."""
     n, c = pooled.shape
     r = ha.shape[1]
     w1m, w2m = se_pack(w1, w2)
     dx = torch.empty_like(x)
     dsz = torch.empty(n, c, dtype=torch.float32, device=x.device)
     dhz = torch.empty(n, r, dtype=torch.float32, device=x.device)
-    e.se_bwd(dy, x, w1m, w2m, ha, sz, dx, dsz, dhz)
+    if bn is not None:
+        e.se_bwd(dy, x, w1m, w2m, ha, sz, dx, dsz, dhz, bn_z=bn[0],
+                 bn_ab=bn[1], bn_act=bn[2], bn_stats_ws=bn[3])
+    else:
+        e.se_bwd(dy, x, w1m, w2m, ha, sz, dx, dsz, dhz)
     dw1 = torch.empty(r, c, dtype=torch.float32, device=x.device)
     dw2 = torch.empty(c, r, dtype=torch.float32, device=x.device)
     db1 = torch.empty(r, dtype=torch.float32, device=x.device)
