@@ -21,6 +21,16 @@ sources = [os.path.join(CSRC, f) for f in
            ("bindings.cpp", "gemm.hip", "dwconv.hip", "bn.hip", "misc.hip",
             "se.hip", "update.hip")]
 
+# Debug builds:
+#   YAMS_ASAN=1   host-side AddressSanitizer on the binding/launcher code
+#   YAMS_DEBUG=1  -g -O1 device code for AMD_SERIALIZE_KERNEL bisection
+cxx_flags = ["-O3", "-std=c++17"]
+dev_flags = ["-O3", "-std=c++17", "--offload-arch=gfx950"]
+if os.environ.get("YAMS_ASAN") == "1":
+    cxx_flags += ["-fsanitize=address", "-fno-omit-frame-pointer", "-g"]
+if os.environ.get("YAMS_DEBUG") == "1":
+    dev_flags = ["-g", "-O1", "-std=c++17", "--offload-arch=gfx950"]
+
 setup(
     name="yams_hip",
     ext_modules=[
@@ -28,8 +38,8 @@ setup(
             name="yet_another_mobilenet_series_amd.ops._yams_hip",
             sources=sources,
             extra_compile_args={
-                "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+                "cxx": cxx_flags,
+                "nvcc": dev_flags,
             },
         )
     ],
