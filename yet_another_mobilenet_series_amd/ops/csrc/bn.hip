@@ -21,6 +21,7 @@
 // All reductions deterministic (fixed-order slabs, no atomics).
 
 #include <algorithm>
+#include <cstdlib>
 
 #include "common.h"
 #include "engine.h"
@@ -29,7 +30,16 @@ namespace yams {
 
 namespace {
 
-constexpr int TARGET_BLOCKS = 1024;
+// measurement knob: YAMS_BN_BLOCKS overrides the reduction grid target
+// (1024 = round-1 sweep optimum; re-sweepable without a rebuild)
+inline int bn_target_blocks() {
+  static int t = [] {
+    const char* e = getenv("YAMS_BN_BLOCKS");
+    const int v = e ? atoi(e) : 1024;
+    return v > 0 ? v : 1024;
+  }();
+  return t;
+}
 
 struct RedGeom {
   int CT, ML, c_tiles, chunks;
@@ -42,7 +52,7 @@ RedGeom red_geom(int64_t M, int C) {
   g.ML = 256 / g.CT;
   g.c_tiles = (C8 + g.CT - 1) / g.CT;
   int64_t max_chunks = (M + 63) / 64;
-  int64_t want = TARGET_BLOCKS / g.c_tiles;
+  int64_t want = bn_target_blocks() / g.c_tiles;
   g.chunks = int(std::min<int64_t>(std::max<int64_t>(want, 1), max_chunks));
   return g;
 }
