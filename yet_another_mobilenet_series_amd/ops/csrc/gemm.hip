@@ -333,9 +333,11 @@ namespace {
 // are M-major (M is the reduction axis), so staging TRANSPOSES tiles into
 // K-major LDS rows ([n][k], k = m-chunk) and then the frag-read + MFMA
 // structure is identical to pw_conv_kernel.  The transposed scatter writes
-// use an XOR chunk swizzle (phys k-chunk = (k>>3) ^ (n&7)) so the 8
-// column-writes of one thread land in 8 distinct bank groups while frag
-// reads stay single aligned b128s.  Replaces the VALU fp32-FMA version,
+// use an XOR chunk swizzle (phys k-chunk = (k>>3) ^ ((n>>3)&7)): the
+// staging loop walks n in steps of 8, so swizzling on n&7 left every
+// lane of a write on ONE bank group (PMC: 24.8% of atb cycles in LDS
+// conflicts, profiles/r02/pmc_summary); the n>>3 bits vary across the
+// write lanes AND across the 16 consecutive n of a frag read.  Replaces the VALU fp32-FMA version,
 // which was compute-bound at ~1/16 of MFMA throughput.
 constexpr int KT = 64;        // m rows per LDS pass
 constexpr int LDK = KT + 8;   // shorts per LDS row (pad keeps b128 align)
@@ -399,7 +401,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int n = s_n + j;
-          Al[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+          Al[n * LDK +
+             ((((r >> 3) ^ ((n >> 3) & 7)) << 3) | (r & 7))] = tmp[j];
         }
       }
     }
@@ -426,7 +429,8 @@ __global__ __launch_bounds__(256) void atb_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int n = s_n + j;
-          Bl[n * LDK + ((((r >> 3) ^ (n & 7)) << 3) | (r & 7))] = tmp[j];
+          Bl[n * LDK +
+             ((((r >> 3) ^ ((n >> 3) & 7)) << 3) | (r & 7))] = tmp[j];
         }
       }
     }
@@ -438,13 +442,13 @@ __global__ __launch_bounds__(256) void atb_kernel(
 #pragma unroll
       for (int i = 0; i < F1; ++i) {
         const int n = wrow0 + i * 16 + fr;
-        const int kc = ((ks * 32 + fk) >> 3) ^ (n & 7);
+        const int kc = ((ks * 32 + fk) >> 3) ^ ((n >> 3) & 7);
         a_frag[i] = *reinterpret_cast<const bf16x8*>(&Al[n * LDK + kc * 8]);
       }
 #pragma unroll
       for (int j = 0; j < F2; ++j) {
         const int n = wcol0 + j * 16 + fr;
-        const int kc = ((ks * 32 + fk) >> 3) ^ (n & 7);
+        const int kc = ((ks * 32 + fk) >> 3) ^ ((n >> 3) & 7);
         b_frag[j] = *reinterpret_cast<const bf16x8*>(&Bl[n * LDK + kc * 8]);
       }
 #pragma unroll
