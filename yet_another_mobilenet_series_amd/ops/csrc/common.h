@@ -40,6 +40,18 @@ DEV_INLINE float bf2f(ushort_t u) {
   return v.f;
 }
 
+#ifdef __HIP_DEVICE_COMPILE__
+// native convert: hipcc lowers the __bf16 cast to v_cvt_pk_bf16_f32
+// (RNE, NaN-quieting) — ONE instruction per f32 PAIR vs the ~6-VALU
+// manual round below.  PMC showed several kernels VALU-issue-bound with
+// the store/EPI converts a large slice (profiles/r02/pmc_summary).
+DEV_INLINE ushort_t f2bf(float f) {
+  __bf16 b = static_cast<__bf16>(f);
+  union { __bf16 b; ushort_t u; } v;
+  v.b = b;
+  return v.u;
+}
+#else
 DEV_INLINE ushort_t f2bf(float f) {
   union { float f; uint32_t i; } v;
   v.f = f;
@@ -49,6 +61,7 @@ DEV_INLINE ushort_t f2bf(float f) {
   i += 0x7fffu + ((i >> 16) & 1u);
   return ushort_t(i >> 16);
 }
+#endif
 
 // ---- activations (codes shared with ops/hip_ops.py) ----
 #define ACT_IDENTITY 0
