@@ -166,21 +166,42 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
     }
   }
 
-  // epilogue: C/D frag layout col = lane&15, row = (lane>>4)*4 + r
+  // epilogue: C/D frag layout col = lane&15, row = (lane>>4)*4 + r.
+  // Stores and the EPI statistics run in ONE pass over acc (the split
+  // version converted every element to bf16 twice — measurable on the
+  // skinny-K shapes where the epilogue rivals the 1-2 mainloop tiles).
   const int col_l = fr, row_l4 = (lane >> 4) * 4;
+  float s1[FRAG_N] = {}, s2[FRAG_N] = {};
 #pragma unroll
-  for (int i = 0; i < FRAG_Mv; ++i) {
+  for (int j = 0; j < FRAG_N; ++j) {
+    const int col = bn + wcol0 + j * 16 + col_l;
+    if (col >= N) continue;
+    float ea = 0.f, eb = 0.f;
+    if (EPI == 2) {
+      ea = lopE.coef[col];
+      eb = lopE.coef[N + col];
+    }
 #pragma unroll
-    for (int j = 0; j < FRAG_N; ++j) {
-      const int col = bn + wcol0 + j * 16 + col_l;
-      if (col >= N) continue;
+    for (int i = 0; i < FRAG_Mv; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
         if (row < M) {
           float v = acc[i][j][r];
           if (RES) v += bf2f(res[row * N + col]);
-          C[row * N + col] = f2bf(v);
+          const ushort_t o = f2bf(v);
+          C[row * N + col] = o;
+          if (EPI == 1) {
+            const float vb = bf2f(o);
+            s1[j] += vb;
+            s2[j] = fmaf(vb, vb, s2[j]);
+          } else if (EPI == 2) {
+            const float vb = bf2f(o);
+            const float zv = bf2f(lopE.z[row * N + col]);
+            const float da = vb * act_bwd<EACT>(fmaf(zv, ea, eb));
+            s1[j] += da;
+            s2[j] = fmaf(da, zv, s2[j]);
+          }
         }
       }
     }
@@ -189,35 +210,6 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
   if (EPI) {
     // per-block partial column sums of the (bf16-rounded) output.
     float* red = reinterpret_cast<float*>(Alds);  // 512 floats
-    float s1[FRAG_N] = {}, s2[FRAG_N] = {};
-#pragma unroll
-    for (int j = 0; j < FRAG_N; ++j) {
-      const int col = bn + wcol0 + j * 16 + col_l;
-      float ea = 0.f, eb = 0.f;
-      if (EPI == 2) {
-        if (col >= N) continue;  // s1/s2 stay 0; red slot still written
-        ea = lopE.coef[col];
-        eb = lopE.coef[N + col];
-      }
-#pragma unroll
-      for (int i = 0; i < FRAG_Mv; ++i)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int64_t row = bm + wrow0 + i * 16 + row_l4 + r;
-          if (row < M) {
-            const float v = bf2f(f2bf(acc[i][j][r]));
-            if (EPI == 1) {
-              s1[j] += v;
-              s2[j] = fmaf(v, v, s2[j]);
-            } else {
-              const float zv = bf2f(lopE.z[row * N + col]);
-              const float da = v * act_bwd<EACT>(fmaf(zv, ea, eb));
-              s1[j] += da;
-              s2[j] = fmaf(da, zv, s2[j]);
-            }
-          }
-        }
-    }
     const int slot = (wave >> 1) * 4 + (lane >> 4);  // 0..7
     const int mblocks = gridDim.x;
     __syncthreads();
