@@ -23,8 +23,12 @@
 #include <algorithm>
 #include <cstdlib>
 
+#include <hip/hip_cooperative_groups.h>
+
 #include "common.h"
 #include "engine.h"
+
+namespace cg = cooperative_groups;
 
 namespace yams {
 
@@ -353,6 +357,148 @@ int elem_grid(int64_t total8) {
   return int(std::min<int64_t>((total8 + 255) / 256, 4096));
 }
 
+// ONE cooperative launch for the whole BN backward: partial sums ->
+// grid.sync -> finalize -> grid.sync -> apply.  Replaces three
+// dependent launches per BN (two kernel ramps x 46 BNs of latency on
+// the critical path; hipGraph replay measured unable to recover it).
+// All reductions keep the fixed slab order — the grid sync is a
+// barrier, not an atomic reduction.
+template <int ACT, bool TRAIN>
+__global__ __launch_bounds__(256) void bn_bwd_fused_kernel(
+    const ushort_t* __restrict__ dy, const ushort_t* __restrict__ z,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    float* __restrict__ ws, float* __restrict__ coef,
+    ushort_t* __restrict__ dz, int64_t M, int C, int chunks) {
+  __shared__ float red[256 * 8];
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  // ---- phase 1: per-chunk partial sums (bn_bwd_partial body) ----------
+  {
+    const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
+    const int chunk = blockIdx.x / c_tiles;
+    const int mlane = int(threadIdx.x) / CT8;
+    const int c = c8 * 8;
+    const int64_t per = (M + chunks - 1) / chunks;
+    const int64_t lo = chunk * per;
+    const int64_t hi = std::min<int64_t>(lo + per, M);
+    float s1[8] = {}, s2[8] = {};
+    if (c8 < C8 && mlane < ML && chunk < chunks) {
+      float a[8], b[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        a[e] = gamma[c + e] * invstd[c + e];
+        b[e] = beta[c + e] - mean[c + e] * a[e];
+      }
+      for (int64_t m = lo + mlane; m < hi; m += ML) {
+        short8_t z8 = *reinterpret_cast<const short8_t*>(&z[m * C + c]);
+        short8_t g8 = *reinterpret_cast<const short8_t*>(&dy[m * C + c]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float zv = bf2f(ushort_t(z8[e]));
+          const float da =
+              bf2f(ushort_t(g8[e])) * act_bwd<ACT>(fmaf(zv, a[e], b[e]));
+          s1[e] += da;
+          s2[e] = fmaf(da, zv, s2[e]);
+        }
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s1[e];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT8) && c8 < C8 && chunk < chunks) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float t = 0.f;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 8 + e];
+        ws[int64_t(chunk) * C + c + e] = t;
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int e = 0; e < 8; ++e) red[threadIdx.x * 8 + e] = s2[e];
+    __syncthreads();
+    if (threadIdx.x < unsigned(CT8) && c8 < C8 && chunk < chunks) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float t = 0.f;
+        for (int l = 0; l < ML; ++l)
+          t += red[(threadIdx.x + l * CT8) * 8 + e];
+        ws[(int64_t(chunks) + chunk) * C + c + e] = t;
+      }
+    }
+  }
+  cg::this_grid().sync();
+  // ---- phase 2: finalize (bn_bwd_finalize body on blocks 0..C/16) -----
+  if (int(blockIdx.x) < (C + 15) / 16) {
+    const int cl = int(threadIdx.x) % 16;
+    const int sl = int(threadIdx.x) / 16;
+    const int c = blockIdx.x * 16 + cl;
+    float s1 = 0.f, s2 = 0.f;
+    if (c < C) {
+#pragma unroll 4
+      for (int sp = sl; sp < chunks; sp += 16) {
+        s1 += ws[int64_t(sp) * C + c];
+        s2 += ws[int64_t(chunks + sp) * C + c];
+      }
+    }
+    red[threadIdx.x * 2] = s1;
+    red[threadIdx.x * 2 + 1] = s2;
+    __syncthreads();
+    if (sl == 0 && c < C) {
+      for (int l = 1; l < 16; ++l) {
+        s1 += red[(cl + l * 16) * 2];
+        s2 += red[(cl + l * 16) * 2 + 1];
+      }
+      const float mu = mean[c], inv = invstd[c];
+      const float dg = (s2 - mu * s1) * inv;
+      dgamma[c] = dg;
+      dbeta[c] = s1;
+      const float a = gamma[c] * inv;
+      coef[c] = a;
+      coef[C + c] = beta[c] - mu * a;
+      if (TRAIN) {
+        const float inv_m = 1.f / float(M);
+        const float k2 = a * inv * dg * inv_m;
+        coef[2 * C + c] = -a * s1 * inv_m + mu * k2;
+        coef[3 * C + c] = k2;
+      } else {
+        coef[2 * C + c] = 0.f;
+        coef[3 * C + c] = 0.f;
+      }
+    }
+  }
+  cg::this_grid().sync();
+  // ---- phase 3: apply (grid-stride; bn_bwd_apply body) ----------------
+  {
+    const int64_t total8 = M * C8;
+    for (int64_t idx = int64_t(blockIdx.x) * 256 + threadIdx.x;
+         idx < total8; idx += int64_t(gridDim.x) * 256) {
+      const int c = int(idx % C8) * 8;
+      const int64_t off = idx * 8;
+      short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
+      short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
+      short8_t out;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float a = coef[c + e];
+        const float b = coef[C + c + e];
+        const float k0 = coef[2 * C + c + e];
+        const float k2 = coef[3 * C + c + e];
+        const float zf = bf2f(ushort_t(zv[e]));
+        const float da =
+            bf2f(ushort_t(gv[e])) * act_bwd<ACT>(fmaf(zf, a, b));
+        out[e] = short(f2bf(fmaf(a, da, k0) - k2 * zf));
+      }
+      *reinterpret_cast<short8_t*>(&dz[off]) = out;
+    }
+  }
+}
+
 }  // namespace
 
 int bn_stats_chunks(int64_t M) { return red_geom(M, 256).chunks; }
@@ -492,6 +638,34 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
   auto* zp = static_cast<const ushort_t*>(z);
   auto* dzp = static_cast<ushort_t*>(dz);
   const int64_t total8 = M * (C >> 3);
+  // fused cooperative path: one launch instead of partial+finalize+apply.
+  // Measured NEGATIVE and unstable (grid.sync cost ~18.3 vs 12.5 ms/step
+  // and a crash in the small-shape tests) — default OFF, kept only as a
+  // documented experiment (YAMS_BN_COOP=1 enables).
+  static int coop_on = [] {
+    const char* e = getenv("YAMS_BN_COOP");
+    return e != nullptr && atoi(e) != 0;
+  }();
+  if (coop_on && ext_splits <= 0 && apply && training) {
+    bool launched = false;
+    ACT_SWITCH(act, {
+      void* kernel = reinterpret_cast<void*>(
+          &bn_bwd_fused_kernel<kAct, true>);
+      int chunks = g.chunks;
+      void* args[] = {&gp, &zp,
+                      const_cast<float**>(&mean),
+                      const_cast<float**>(&invstd),
+                      const_cast<float**>(&gamma),
+                      const_cast<float**>(&beta),
+                      &dgamma, &dbeta, &ws, &coef, &dzp, &M,
+                      const_cast<int*>(&C), &chunks};
+      const hipError_t err = hipLaunchCooperativeKernel(
+          kernel, dim3(g.c_tiles * g.chunks), dim3(256), args, 0, s);
+      launched = err == hipSuccess;
+      if (!launched) (void)hipGetLastError();  // clear, fall back
+    });
+    if (launched) return;
+  }
   ACT_SWITCH(act, {
     if (ext_splits <= 0)
       bn_bwd_partial_kernel<kAct><<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
