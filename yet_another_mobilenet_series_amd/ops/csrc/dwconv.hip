@@ -472,61 +472,37 @@ __global__ __launch_bounds__(256) void im2row_kernel(
     const ushort_t* __restrict__ x, ushort_t* __restrict__ patches,
     int N, int H, int W, int Ho, int Wo, int k, int stride, int pad,
     int KP, int64_t total) {
-  // one block per output row (n, ho): the k input rows are staged into
-  // LDS with coalesced vec loads, then each thread packs patches from
-  // LDS.  (The thread-per-patch version issued 27 scalar global loads
-  // per patch — load-issue bound at 220 us for the stem.)
-  constexpr int Cin = 3;
-  __shared__ ushort_t rows[3 * (3 * 256 + 8)];  // k<=3 rows x W*3 (W<=256)
-  const int rw = W * Cin;
-  for (int blk = blockIdx.x; blk < N * Ho; blk += gridDim.x) {
-    const int ho = blk % Ho;
-    const int n = blk / Ho;
-    // stage k clamped input rows
+  const int Cin = 3;
+  for (int64_t m = int64_t(blockIdx.x) * blockDim.x + threadIdx.x;
+       m < total; m += int64_t(gridDim.x) * blockDim.x) {
+    const int wo = int(m % Wo);
+    int64_t rest = m / Wo;
+    const int ho = int(rest % Ho);
+    const int n = int(rest / Ho);
+    ushort_t buf[32];
+    int t = 0;
     for (int i = 0; i < k; ++i) {
       const int hi = ho * stride + i - pad;
       const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
-      const bool row_ok = hi >= 0 && hi < H;
-      const ushort_t* src = x + (int64_t(n) * H + hic) * rw;
-      ushort_t* dst = &rows[i * (rw + 8)];
-      for (int t = int(threadIdx.x); t * 4 < rw; t += 256) {
-        short4_t v = (t * 4 + 4 <= rw)
-                         ? *reinterpret_cast<const short4_t*>(&src[t * 4])
-                         : short4_t{0, 0, 0, 0};
-        if (t * 4 + 4 > rw)
-          for (int e = 0; e < 4 && t * 4 + e < rw; ++e) v[e] = src[t * 4 + e];
-        if (!row_ok) v = short4_t{0, 0, 0, 0};
-        *reinterpret_cast<short4_t*>(&dst[t * 4]) = v;
-      }
-    }
-    __syncthreads();
-    // pack patches for this output row
-    for (int wo = int(threadIdx.x); wo < Wo; wo += 256) {
-      ushort_t buf[32];
-      int t = 0;
-      for (int i = 0; i < k; ++i) {
-        const ushort_t* rrow = &rows[i * (rw + 8)];
-        for (int j = 0; j < k; ++j) {
-          const int wi = wo * stride + j - pad;
-          const bool ok = wi >= 0 && wi < W;
-          const int wic = ok ? wi : 0;
+      for (int j = 0; j < k; ++j) {
+        const int wi = wo * stride + j - pad;
+        const int wic = wi < 0 ? 0 : (wi >= W ? W - 1 : wi);
+        const bool ok = hi >= 0 && hi < H && wi >= 0 && wi < W;
 #pragma unroll
-          for (int ci = 0; ci < Cin; ++ci) {
-            const ushort_t v = rrow[wic * Cin + ci];
-            buf[t++] = ok ? v : ushort_t(0);
-          }
+        for (int ci = 0; ci < 3; ++ci) {
+          const ushort_t v = x[idx4(n, hic, wic, ci, H, W, Cin)];
+          buf[t++] = ok ? v : ushort_t(0);
         }
       }
-      for (; t < KP; ++t) buf[t] = ushort_t(0);
-      ushort_t* dst = &patches[(int64_t(blk) * Wo + wo) * KP];
-#pragma unroll
-      for (int v = 0; v < 32; v += 8) {
-        if (v >= KP) break;
-        *reinterpret_cast<short8_t*>(&dst[v]) =
-            *reinterpret_cast<const short8_t*>(&buf[v]);
-      }
     }
-    __syncthreads();
+    for (; t < KP; ++t) buf[t] = ushort_t(0);
+    ushort_t* dst = &patches[m * KP];
+#pragma unroll
+    for (int v = 0; v < 32; v += 8) {
+      if (v >= KP) break;
+      *reinterpret_cast<short8_t*>(&dst[v]) =
+          *reinterpret_cast<const short8_t*>(&buf[v]);
+    }
   }
 }
 
@@ -696,10 +672,8 @@ combine:
 
 void im2row(const void* x, void* patches, int N, int H, int W, int Ho,
             int Wo, int k, int stride, int pad, int KP, hipStream_t s) {
-  if (k > 3 || W > 256) yams_launch_fail("im2row", k, W, stride, 0);
   const int64_t total = int64_t(N) * Ho * Wo;
-  const int blocks = int(std::min<int64_t>(int64_t(N) * Ho, MAX_BLOCKS));
-  im2row_kernel<<<dim3(blocks), 256, 0, s>>>(
+  im2row_kernel<<<grid_for((total + 3) / 4), 256, 0, s>>>(
       static_cast<const ushort_t*>(x), static_cast<ushort_t*>(patches),
       N, H, W, Ho, Wo, k, stride, pad, KP, total);
 }
