@@ -246,7 +246,9 @@ class FusedInvertedResidualFn(torch.autograd.Function):
             sw = wgrad_stream()
             cur = torch.cuda.current_stream()
 
-        def _on_wgrad_stream(ready_after=None):
+        def _on_wgrad_stream():
+            # make the side stream wait for everything main has queued,
+            # then return the context that routes the wgrads onto it
             ev = torch.cuda.Event()
             ev.record(cur)
             sw.wait_event(ev)
