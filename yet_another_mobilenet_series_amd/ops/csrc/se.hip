@@ -129,8 +129,19 @@ __global__ __launch_bounds__(BS) void se_fwd_kernel(
     constexpr int RL = BS / 256;
     float acc1 = 0.f;
     if (r < R) {
-      for (int cc = rl; cc < C; cc += RL)
-        acc1 = fmaf(pooled_s[cc], bf2f(w1t[int64_t(cc) * R + r]), acc1);
+      // 4 independent accumulators break the fma dependency chain (the
+      // phase measured ~85% wave-wait in PMC)
+      float a4[4] = {};
+      int cc = rl;
+      for (; cc + 3 * RL < C; cc += 4 * RL) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          a4[u] = fmaf(pooled_s[cc + u * RL],
+                       bf2f(w1t[int64_t(cc + u * RL) * R + r]), a4[u]);
+      }
+      for (; cc < C; cc += RL)
+        a4[0] = fmaf(pooled_s[cc], bf2f(w1t[int64_t(cc) * R + r]), a4[0]);
+      acc1 = (a4[0] + a4[1]) + (a4[2] + a4[3]);
     }
     red[tid] = acc1;
     __syncthreads();
@@ -271,8 +282,17 @@ __global__ __launch_bounds__(BS) void se_bwd_kernel(
     constexpr int RL = BS / 256;
     float acc1 = 0.f;
     if (r < R) {
-      for (int cc = rl; cc < C; cc += RL)
-        acc1 = fmaf(dsz_s[cc], bf2f(w2[int64_t(cc) * R + r]), acc1);
+      float a4[4] = {};
+      int cc = rl;
+      for (; cc + 3 * RL < C; cc += 4 * RL) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          a4[u] = fmaf(dsz_s[cc + u * RL],
+                       bf2f(w2[int64_t(cc + u * RL) * R + r]), a4[u]);
+      }
+      for (; cc < C; cc += RL)
+        a4[0] = fmaf(dsz_s[cc], bf2f(w2[int64_t(cc) * R + r]), a4[0]);
+      acc1 = (a4[0] + a4[1]) + (a4[2] + a4[3]);
     }
     red[tid] = acc1;
     __syncthreads();
