@@ -453,7 +453,7 @@ __global__ __launch_bounds__(256) void se_wgrad_kernel(
       a = dsz; b = ha; Ar = C; Bc = R;
       r0 = (bb % ct) * 16; c0 = (bb / ct) * 16;
     }
-    float acc = 0.f;
+    float a4[4] = {};
     for (int n0 = 0; n0 < N; n0 += NT) {
       // cooperative stage: 256 threads = 16 n-rows x 16 cols per round
       __syncthreads();
@@ -466,9 +466,14 @@ __global__ __launch_bounds__(256) void se_wgrad_kernel(
       }
       __syncthreads();
       const int hi = N - n0 < NT ? N - n0 : NT;
-      for (int nn = 0; nn < hi; ++nn)
-        acc = fmaf(al[nn][i], bl[nn][j], acc);
+      int nn = 0;
+      for (; nn + 4 <= hi; nn += 4)
+#pragma unroll
+        for (int u = 0; u < 4; ++u)
+          a4[u] = fmaf(al[nn + u][i], bl[nn + u][j], a4[u]);
+      for (; nn < hi; ++nn) a4[0] = fmaf(al[nn][i], bl[nn][j], a4[0]);
     }
+    const float acc = (a4[0] + a4[1]) + (a4[2] + a4[3]);
     if (bid < t1) {
       if (r0 + i < R && c0 + j < C) dw1[int64_t(r0 + i) * C + c0 + j] = acc;
     } else {
