@@ -269,7 +269,7 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
     const int c = c8 * 8;
     LopRegT<LOPK, ACT, 8> lr;
     lr.init(lopdy, c);
-    float acc[8] = {};
+    f32x2 acc[4] = {};  // packed pairs: taps accumulate on v_pk_fma_f32
     const int ip0 = (hi + pad) & 1;
     const int jp0 = (wi + pad) & 1;
 #pragma unroll
@@ -288,27 +288,36 @@ __global__ __launch_bounds__(256) void dw_dgrad_s2_kernel(
         const int woc = wo < 0 ? 0 : (wo >= Wo ? Wo - 1 : wo);
         const int64_t off = idx4(n, hoc, woc, c, Ho, Wo, C);
         short8_t gv = *reinterpret_cast<const short8_t*>(&dy[off]);
+        if (!ok) gv = short8_t{0, 0, 0, 0, 0, 0, 0, 0};
         short8_t wv = *reinterpret_cast<const short8_t*>(
             &w[(i * K + j) * C + c]);
         if (LOPK == LOP_NONE) {
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(ok ? bf2f(ushort_t(gv[e])) : 0.f,
-                          bf2f(ushort_t(wv[e])), acc[e]);
+          for (int pp = 0; pp < 4; ++pp) {
+            f32x2 g2 = {bf2f(ushort_t(gv[pp * 2])),
+                        bf2f(ushort_t(gv[pp * 2 + 1]))};
+            f32x2 w2v = {bf2f(ushort_t(wv[pp * 2])),
+                         bf2f(ushort_t(wv[pp * 2 + 1]))};
+            acc[pp] += g2 * w2v;
+          }
         } else {
+          // (gv pre-zeroing is NOT a valid mask for the BNBWD lop — its
+          // affine terms are nonzero at 0 input — so mask the applied
+          // value explicitly)
           short8_t zv = *reinterpret_cast<const short8_t*>(&lopdy.z[off]);
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            acc[e] = fmaf(ok ? lr.apply(bf2f(ushort_t(gv[e])),
-                                        bf2f(ushort_t(zv[e])), e)
-                             : 0.f,
-                          bf2f(ushort_t(wv[e])), acc[e]);
+            acc[e >> 1][e & 1] =
+                fmaf(ok ? lr.apply(bf2f(ushort_t(gv[e])),
+                                   bf2f(ushort_t(zv[e])), e)
+                        : 0.f,
+                     bf2f(ushort_t(wv[e])), acc[e >> 1][e & 1]);
         }
       }
     }
     short8_t out;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e]));
+    for (int e = 0; e < 8; ++e) out[e] = short(f2bf(acc[e >> 1][e & 1]));
     *reinterpret_cast<short8_t*>(&dx[idx4(n, hi, wi, c, H, W, C)]) = out;
   }
 }
