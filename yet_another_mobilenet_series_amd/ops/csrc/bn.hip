@@ -215,7 +215,8 @@ __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
   const int64_t per = (M + chunks - 1) / chunks;
   const int64_t lo = chunk * per;
   const int64_t hi = std::min<int64_t>(lo + per, M);
-  float s1[8] = {}, s2[8] = {};
+  float s1[8] __attribute__((aligned(8))) = {};
+  float s2[8] __attribute__((aligned(8))) = {};
   if (c8 < C8 && mlane < ML) {
     float a[8], b[8];
 #pragma unroll
@@ -236,13 +237,19 @@ __global__ __launch_bounds__(256) void bn_bwd_partial_kernel(
                              bf2f(ushort_t(p8[e])) * inv_hw));
         *reinterpret_cast<short8_t*>(&dy[m * C + c]) = g8;
       }
+      // packed accumulation: the two per-element fmas run as
+      // v_pk_fma_f32 (the act' factor stays scalar — it has compares)
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const float zv = bf2f(ushort_t(z8[e]));
-        const float da =
-            bf2f(ushort_t(g8[e])) * act_bwd<ACT>(fmaf(zv, a[e], b[e]));
-        s1[e] += da;
-        s2[e] = fmaf(da, zv, s2[e]);
+      for (int pp = 0; pp < 4; ++pp) {
+        f32x2 zv = {bf2f(ushort_t(z8[pp * 2])),
+                    bf2f(ushort_t(z8[pp * 2 + 1]))};
+        f32x2 da = {
+            bf2f(ushort_t(g8[pp * 2])) *
+                act_bwd<ACT>(fmaf(zv[0], a[pp * 2], b[pp * 2])),
+            bf2f(ushort_t(g8[pp * 2 + 1])) *
+                act_bwd<ACT>(fmaf(zv[1], a[pp * 2 + 1], b[pp * 2 + 1]))};
+        *reinterpret_cast<f32x2*>(&s1[pp * 2]) += da;
+        *reinterpret_cast<f32x2*>(&s2[pp * 2]) += da * zv;
       }
     }
   }
