@@ -10,6 +10,8 @@ from __future__ import annotations
 import datetime
 import functools
 import os
+import threading
+import time
 from typing import Callable, Optional
 
 import torch
@@ -114,7 +116,6 @@ class Watchdog:
 
     def __init__(self, timeout_sec: float, name: str = "step",
                  logger=None) -> None:
-        import threading
         self.timeout_sec = float(timeout_sec)
         self.name = name
         self.logger = logger
@@ -126,8 +127,6 @@ class Watchdog:
     def start(self) -> "Watchdog":
         if self.timeout_sec <= 0 or self._thread is not None:
             return self
-        import threading
-        import time
         self._last = time.monotonic()
         self._stop.clear()
 
@@ -145,7 +144,7 @@ class Watchdog:
                     print(msg, flush=True)
                     os._exit(self.EXIT_CODE)
 
-        self._thread = __import__("threading").Thread(
+        self._thread = threading.Thread(
             target=_watch, name="yams-watchdog", daemon=True)
         self._thread.start()
         return self
@@ -153,7 +152,6 @@ class Watchdog:
     def kick(self) -> None:
         if self._thread is None:
             return
-        import time
         with self._lock:
             self._last = time.monotonic()
 
