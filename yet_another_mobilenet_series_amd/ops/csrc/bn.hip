@@ -128,10 +128,20 @@ __global__ __launch_bounds__(256) void bn_stats_finalize_kernel(
   const int c = blockIdx.x * 16 + cl;
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
-    for (int sp = sl; sp < splits; sp += 16) {
+    float s1b = 0.f, s2b = 0.f;  // second chain: ILP on the wait-bound
+    int sp = sl;                 // serial slab walk
+    for (; sp + 16 < splits; sp += 32) {
+      s1 += ws[int64_t(sp) * C + c];
+      s2 += ws[int64_t(splits + sp) * C + c];
+      s1b += ws[int64_t(sp + 16) * C + c];
+      s2b += ws[int64_t(splits + sp + 16) * C + c];
+    }
+    for (; sp < splits; sp += 16) {
       s1 += ws[int64_t(sp) * C + c];
       s2 += ws[int64_t(splits + sp) * C + c];
     }
+    s1 += s1b;
+    s2 += s2b;
   }
   red[threadIdx.x] = s1;
   __syncthreads();
@@ -292,11 +302,20 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
   const int c = blockIdx.x * 16 + cl;
   float s1 = 0.f, s2 = 0.f;
   if (c < C) {
-#pragma unroll 4
-    for (int sp = sl; sp < splits; sp += 16) {
+    float s1b = 0.f, s2b = 0.f;
+    int sp = sl;
+    for (; sp + 16 < splits; sp += 32) {
+      s1 += ws[int64_t(sp) * C + c];
+      s2 += ws[int64_t(splits + sp) * C + c];
+      s1b += ws[int64_t(sp + 16) * C + c];
+      s2b += ws[int64_t(splits + sp + 16) * C + c];
+    }
+    for (; sp < splits; sp += 16) {
       s1 += ws[int64_t(sp) * C + c];
       s2 += ws[int64_t(splits + sp) * C + c];
     }
+    s1 += s1b;
+    s2 += s2b;
   }
   red[threadIdx.x][0] = s1;
   red[threadIdx.x][1] = s2;
