@@ -251,16 +251,21 @@ void pw_conv(const void* x, const void* w, void* y,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   auto* rp = static_cast<const ushort_t*>(res);
-  // small-M stats-free launches (dgrads at 7^2/14^2) use the BM=64 tile:
-  // twice the blocks on a chip that needs >>256 workgroups; 14^2 shapes
-  // measured slightly WORSE at BM=64 (more B re-staging), so only the
-  // 7^2 tier uses it (measurement
-  // knob YAMS_PW_BM64=0 disables)
+  // stats-free launches (dgrads) use the BM=64 tile: twice the blocks
+  // and 12 waves/CU on a chip that needs >>256 workgroups
+  // (YAMS_PW_BM64=0 disables, YAMS_PW_BM64_MAX sweeps the threshold)
   static int bm64_on = [] {
     const char* e = getenv("YAMS_PW_BM64");
     return e == nullptr || atoi(e) != 0;
   }();
-  const bool small = bm64_on && ep == 0 && M <= 16384;
+  static int64_t bm64_max = [] {  // sweep knob for the small-M threshold
+    // same-lease sweep: BM64 for ALL stats-free (dgrad) launches won
+    // (11.77 vs 11.89 at 16k) — 2x blocks/12 waves beats the extra
+    // B-tile re-staging even at the 112^2 shapes
+    const char* e = getenv("YAMS_PW_BM64_MAX");
+    return e ? atoll(e) : (1LL << 40);
+  }();
+  const bool small = bm64_on && ep == 0 && M <= bm64_max;
   dim3 grid(ceil_div(M, small ? 64 : BM), ceil_div(N, BN));
   if (rp != nullptr) {  // residual epilogue: plain or BNBWD-lop dgrad
     if (ep == 0 && lopa.kind == 0) {
