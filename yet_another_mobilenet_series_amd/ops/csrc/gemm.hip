@@ -239,7 +239,22 @@ __global__ __launch_bounds__(256) void pw_conv_kernel(
 
 }  // namespace
 
-int pw_stats_mblocks(int64_t M) { return ceil_div(M, BM); }
+// BM policy shared by dispatch and the stats-ws sizing: BM=64 doubles
+// the block count (12 waves/CU) — measured faster for the dgrads and
+// the forward alike; YAMS_PW_BM64{,_MAX} sweep it.
+static int pw_bm_for(int64_t M) {
+  static int on = [] {
+    const char* e = getenv("YAMS_PW_BM64");
+    return e == nullptr || atoi(e) != 0;
+  }();
+  static int64_t mx = [] {
+    const char* e = getenv("YAMS_PW_BM64_MAX");
+    return e ? atoll(e) : (1LL << 40);
+  }();
+  return (on && M <= mx) ? 64 : BM;
+}
+
+int pw_stats_mblocks(int64_t M) { return ceil_div(M, pw_bm_for(M)); }
 
 void pw_conv(const void* x, const void* w, void* y,
              int64_t M, int K, int N, LopDesc lopa, LopDesc epi,
@@ -254,18 +269,9 @@ void pw_conv(const void* x, const void* w, void* y,
   // stats-free launches (dgrads) use the BM=64 tile: twice the blocks
   // and 12 waves/CU on a chip that needs >>256 workgroups
   // (YAMS_PW_BM64=0 disables, YAMS_PW_BM64_MAX sweeps the threshold)
-  static int bm64_on = [] {
-    const char* e = getenv("YAMS_PW_BM64");
-    return e == nullptr || atoi(e) != 0;
-  }();
-  static int64_t bm64_max = [] {  // sweep knob for the small-M threshold
-    // same-lease sweep: BM64 for ALL stats-free (dgrad) launches won
-    // (11.77 vs 11.89 at 16k) — 2x blocks/12 waves beats the extra
-    // B-tile re-staging even at the 112^2 shapes
-    const char* e = getenv("YAMS_PW_BM64_MAX");
-    return e ? atoll(e) : (1LL << 40);
-  }();
-  const bool small = bm64_on && ep == 0 && M <= bm64_max;
+  // the BM policy must match pw_stats_mblocks (the Python side sizes
+  // stats workspaces with it), so EVERY ep follows it
+  const bool small = pw_bm_for(M) == 64;
   dim3 grid(ceil_div(M, small ? 64 : BM), ceil_div(N, BN));
   if (rp != nullptr) {  // residual epilogue: plain or BNBWD-lop dgrad
     if (ep == 0 && lopa.kind == 0) {
@@ -289,14 +295,23 @@ void pw_conv(const void* x, const void* w, void* y,
     yams_launch_fail("pw_conv_res", ep, lopa.kind, lopa.act, epi.act);
   }
   if (small) {
-#define PW64_CASE(LK)                                                     \
-    if (lopa.kind == LK) {                                                \
-      pw_conv_kernel<0, LK, 0, false, 64><<<grid, 256, 0, s>>>(           \
+#define PW64_CASE(EP, LK)                                                 \
+    if (ep == EP && lopa.kind == LK) {                                    \
+      pw_conv_kernel<EP, LK, 0, false, 64><<<grid, 256, 0, s>>>(          \
           ap, wp, yp, M, K, N, L, LE, stats_ws, nullptr);                 \
       return;                                                             \
     }
-    PW64_CASE(0) PW64_CASE(1) PW64_CASE(2)
+    PW64_CASE(0, 0) PW64_CASE(0, 1) PW64_CASE(0, 2)
+    PW64_CASE(1, 0) PW64_CASE(1, 1) PW64_CASE(1, 2)
 #undef PW64_CASE
+#define PW64_EPI2(AC)                                                     \
+    if (ep == 2 && lopa.kind == 0 && epi.act == AC) {                     \
+      pw_conv_kernel<2, 0, AC, false, 64><<<grid, 256, 0, s>>>(           \
+          ap, wp, yp, M, K, N, L, LE, stats_ws, nullptr);                 \
+      return;                                                             \
+    }
+    PW64_EPI2(0) PW64_EPI2(1) PW64_EPI2(2) PW64_EPI2(3)
+#undef PW64_EPI2
     yams_launch_fail("pw_conv64", ep, lopa.kind, lopa.act, epi.act);
   }
 #define PW_CASE(EP, LK)                                                   \
