@@ -203,3 +203,38 @@ def test_gpu_refuses_silent_fallback(tmp_path):
     from yet_another_mobilenet_series_amd.ops import backend
     assert backend.extension_available(), (
         "HIP extension must be built and shipped with the tree")
+
+
+@pytest.mark.parametrize("name,width", [("mobilenet_v3_large", 1.0),
+                                        ("mobilenet_v2", 0.75),
+                                        ("mobilenet_v3_small", 0.5)])
+def test_model_eval_parity(name, width):
+    """EVAL-mode forward (BN running stats + the eval lop paths) vs the
+    fp32 reference — covers inference/validation numerics, including
+    non-1.0 width multipliers."""
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    model = models.build_model(name, dropout=0.0, width_mult=width).to(dev)
+    model = model.to(memory_format=torch.channels_last)
+    # give running stats a non-trivial state
+    model.train()
+    with torch.no_grad():
+        for _ in range(2):
+            model(torch.randn(8, 3, 224, 224, device=dev).bfloat16()
+                  .contiguous(memory_format=torch.channels_last))
+    model.eval()
+    x = torch.randn(4, 3, 224, 224, device=dev)
+    with torch.no_grad():
+        lo_hip = model(x.bfloat16()
+                       .contiguous(memory_format=torch.channels_last))
+        from yet_another_mobilenet_series_amd.ops.backend import (
+            set_engine_mode)
+        set_engine_mode("reference")
+        try:
+            lo_ref = model(x)
+        finally:
+            set_engine_mode("hip")
+    assert torch.isfinite(lo_hip).all()
+    err = ((lo_hip.float() - lo_ref.float()).norm() /
+           lo_ref.float().norm().clamp(min=1e-6)).item()
+    assert err < 0.1, err
