@@ -170,29 +170,41 @@ __global__ __launch_bounds__(256) void bn_stats_finalize_kernel(
 }
 
 // --- forward normalize+act(+residual): flat vec8, 2 coef arrays ---------
+// (c8, mlane) layout: the per-channel {a, b} pairs load once per thread
+// (same restructure as bn_bwd_apply_kernel; the flat version re-read
+// 4 float4 of coefficients per 8 elements)
 template <int ACT, bool RES>
 __global__ __launch_bounds__(256) void bn_act_fwd_kernel(
     const ushort_t* __restrict__ z, const float* __restrict__ ab,
     ushort_t* __restrict__ y, const ushort_t* __restrict__ res,
-    int64_t total8, int C8, int C) {
-  const int tot = int(total8);  // < 2^31 for every supported shape
-  for (int idx = int(blockIdx.x) * int(blockDim.x) + int(threadIdx.x);
-       idx < tot; idx += int(gridDim.x) * int(blockDim.x)) {
-    const int c = (idx % C8) * 8;
-    const int64_t off = idx * 8;
+    int64_t M, int C, int chunks) {
+  const int C8 = C >> 3;
+  const int CT8 = C8 < 256 ? C8 : 256;
+  const int ML = 256 / CT8;
+  const int c_tiles = (C8 + CT8 - 1) / CT8;
+  const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
+  const int chunk = blockIdx.x / c_tiles;
+  const int mlane = int(threadIdx.x) / CT8;
+  const int c = c8 * 8;
+  if (c8 >= C8 || mlane >= ML) return;
+  const int64_t per = (M + chunks - 1) / chunks;
+  const int64_t lo = chunk * per;
+  const int64_t hi = std::min<int64_t>(lo + per, M);
+  float a[8], b[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    a[e] = ab[c + e];
+    b[e] = ab[C + c + e];
+  }
+  for (int64_t m = lo + mlane; m < hi; m += ML) {
+    const int64_t off = m * C + c;
     short8_t zv = *reinterpret_cast<const short8_t*>(&z[off]);
-    float4_t a0 = *reinterpret_cast<const float4_t*>(&ab[c]);
-    float4_t a1 = *reinterpret_cast<const float4_t*>(&ab[c + 4]);
-    float4_t b0 = *reinterpret_cast<const float4_t*>(&ab[C + c]);
-    float4_t b1 = *reinterpret_cast<const float4_t*>(&ab[C + c + 4]);
     short8_t rv;
     if (RES) rv = *reinterpret_cast<const short8_t*>(&res[off]);
     short8_t out;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
-      const float a = e < 4 ? a0[e] : a1[e - 4];
-      const float b = e < 4 ? b0[e] : b1[e - 4];
-      float v = act_fwd<ACT>(fmaf(bf2f(ushort_t(zv[e])), a, b));
+      float v = act_fwd<ACT>(fmaf(bf2f(ushort_t(zv[e])), a[e], b[e]));
       if (RES) v += bf2f(ushort_t(rv[e]));
       out[e] = short(f2bf(v));
     }
@@ -626,17 +638,19 @@ void bn_fold_sums(const float* ws, int splits, float* out, int C,
 
 void bn_act_fwd(const void* z, const float* ab, int act, void* y,
                 const void* residual, int64_t M, int C, hipStream_t s) {
-  const int64_t total8 = M * (C >> 3);
+  const RedGeom g = red_geom(M, C);
   auto* zp = static_cast<const ushort_t*>(z);
   auto* yp = static_cast<ushort_t*>(y);
   auto* rp = static_cast<const ushort_t*>(residual);
   ACT_SWITCH(act, {
     if (rp != nullptr)
-      bn_act_fwd_kernel<kAct, true><<<elem_grid(total8), 256, 0, s>>>(
-          zp, ab, yp, rp, total8, C >> 3, C);
+      bn_act_fwd_kernel<kAct, true>
+          <<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(zp, ab, yp, rp, M,
+                                                      C, g.chunks);
     else
-      bn_act_fwd_kernel<kAct, false><<<elem_grid(total8), 256, 0, s>>>(
-          zp, ab, yp, nullptr, total8, C >> 3, C);
+      bn_act_fwd_kernel<kAct, false>
+          <<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(zp, ab, yp, nullptr,
+                                                      M, C, g.chunks);
   });
 }
 
