@@ -174,8 +174,13 @@ def test_rccl_world1_smoke():
     from yet_another_mobilenet_series_amd.parallel.ddp import (
         DataParallelEngine)
 
-    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29541")
+    import socket
+    sock = socket.socket()
+    sock.bind(("127.0.0.1", 0))
+    port = sock.getsockname()[1]
+    sock.close()
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
     assert torch.distributed.is_nccl_available()
     dist.init_process_group("nccl", rank=0, world_size=1)
     try:
