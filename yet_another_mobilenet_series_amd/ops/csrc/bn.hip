@@ -400,10 +400,6 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
   }
 }
 
-int elem_grid(int64_t total8) {
-  return int(std::min<int64_t>((total8 + 255) / 256, 4096));
-}
-
 // ONE cooperative launch for the whole BN backward: partial sums ->
 // grid.sync -> finalize -> grid.sync -> apply.  Replaces three
 // dependent launches per BN (two kernel ramps x 46 BNs of latency on
@@ -686,7 +682,6 @@ void bn_act_bwd(const void* dy, const void* z, const float* mean,
   auto* gp = static_cast<const ushort_t*>(dy);
   auto* zp = static_cast<const ushort_t*>(z);
   auto* dzp = static_cast<ushort_t*>(dz);
-  const int64_t total8 = M * (C >> 3);
   // fused cooperative path: one launch instead of partial+finalize+apply.
   // Measured NEGATIVE and unstable (grid.sync cost ~18.3 vs 12.5 ms/step
   // and a crash in the small-shape tests) — default OFF, kept only as a
@@ -760,7 +755,6 @@ void bn_act_bwd_addpool(void* dy, const void* z, const void* dpooled,
   auto* zp = static_cast<const ushort_t*>(z);
   auto* pp = static_cast<const ushort_t*>(dpooled);
   auto* dzp = static_cast<ushort_t*>(dz);
-  const int64_t total8 = M * (C >> 3);
   ACT_SWITCH(act, {
     bn_bwd_partial_kernel<kAct, true>
         <<<dim3(g.c_tiles * g.chunks), 256, 0, s>>>(
