@@ -107,3 +107,36 @@ def test_trainer_grad_accum_two(tmp_path):
     changed = sum((a != b).any().item()
                   for a, b in zip(before, t.model.parameters()))
     assert changed > 0
+
+
+def test_trainer_with_packed_dataset(tmp_path):
+    """End-to-end Trainer over the packed single-file fast path."""
+    import sys
+
+    import numpy as np
+
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    try:
+        from PIL import Image
+    except ImportError:
+        import pytest
+        pytest.skip("PIL required")
+    from tools.make_packed_dataset import pack_folder
+
+    rng = np.random.default_rng(0)
+    for split in ("train", "val"):
+        for c in ("a", "b"):
+            d = tmp_path / "imgs" / split / c
+            d.mkdir(parents=True)
+            for i in range(3):
+                arr = rng.integers(0, 255, (72, 72, 3), dtype=np.uint8)
+                Image.fromarray(arr).save(d / f"{i}.jpg")
+        pack_folder(str(tmp_path / "imgs" / split),
+                    str(tmp_path / f"{split}.pack"))
+    cfg = _smoke_cfg(tmp_path, **{
+        "data.dataset": "packed", "data.root": str(tmp_path),
+        "data.image_size": 64, "data.batch_size_per_gpu": 2,
+        "train.epochs": 1, "ema.enabled": False})
+    t = Trainer(cfg)
+    t.fit()
+    assert t.global_step == 3  # 6 train samples / batch 2
