@@ -652,3 +652,29 @@ def test_pw_conv_residual_epilogue(ext):
                 residual=res.view(m, n, 1, 1))
     want = (a.float() @ b.float().t() + res.float())
     assert_close(y, want)
+
+
+def test_guard_bands_model_step(ext):
+    """OOB sweep: a full fused V3-Small fwd+bwd+optimizer
+    step with EVERY Python-level output allocation wrapped in canary
+    guard bands — any HIP kernel writing outside its buffers trips it."""
+    from yet_another_mobilenet_series_amd.models import build_model
+    from yet_another_mobilenet_series_amd.ops.guard import guard_allocations
+    from yet_another_mobilenet_series_amd.utils.optim import FusedSGD
+
+    torch.manual_seed(0)
+    model = build_model("mobilenet_v3_small").to(dev()) \
+        .to(memory_format=CL).train()
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9,
+                   weight_decay=1e-5)
+    x = torch.randn(32, 3, 224, 224, device=dev()) \
+        .to(BF16).contiguous(memory_format=CL)
+    tgt = torch.randint(0, 1000, (32,), device=dev())
+    with guard_allocations() as chk:
+        for _ in range(2):
+            opt.zero_grad(set_to_none=True)
+            loss = F.cross_entropy(model(x).float(), tgt)
+            loss.backward()
+            opt.step()
+        chk()
+    assert torch.isfinite(loss.detach()).item()
