@@ -140,3 +140,19 @@ def test_trainer_with_packed_dataset(tmp_path):
     t = Trainer(cfg)
     t.fit()
     assert t.global_step == 3  # 6 train samples / batch 2
+
+
+def test_trainer_step_interval_checkpoints(tmp_path):
+    """SURVEY §5.3 every-N-step checkpoints: a mid-epoch save exists even
+    when the per-epoch save never fires, and records the CURRENT epoch as
+    the resume point (the interrupted epoch replays from its start)."""
+    cfg = _smoke_cfg(tmp_path, **{
+        "train.save_interval_steps": 1, "train.save_interval_epochs": 1000,
+        "val.interval_epochs": 1000})
+    t = Trainer(cfg)
+    t.fit()
+    path = os.path.join(cfg.log.dir, "checkpoint.pt")
+    assert os.path.exists(path)
+    state = torch.load(path, map_location="cpu", weights_only=True)
+    assert state["epoch"] == 0          # replay epoch 0 on resume
+    assert state["global_step"] == t.global_step > 0

@@ -184,6 +184,9 @@ class Trainer:
                                         for n, v in ms.items()})
             self.global_step += 1
             self.watchdog.kick()
+            sis = int(getattr(cfg.train, "save_interval_steps", 0) or 0)
+            if sis > 0 and self.global_step % sis == 0:
+                self.save(epoch, 0.0, mid_epoch=True)
             if prof is not None:
                 prof.step()
                 if it >= int(getattr(cfg.train, "trace_steps", 3)) + 2:
@@ -244,13 +247,20 @@ class Trainer:
                              ema=bool(use_ema))
         return top1
 
-    def save(self, epoch: int, top1: float) -> None:
+    def save(self, epoch: int, top1: float, mid_epoch: bool = False) -> None:
+        """Write checkpoint.pt (rank 0, atomic).  ``mid_epoch`` saves record
+        the CURRENT epoch as the resume point (SURVEY.md §5.3 every-N-step
+        checkpoints bound lost work on long epochs; resume replays the
+        interrupted epoch from its start with the saved weights)."""
         if self.rank != 0:
             return
         meta = {"best_top1": self.best_top1, "config": self.cfg.to_dict()}
         path = os.path.join(self.cfg.log.dir, "checkpoint.pt")
         ckpt.save_checkpoint(path, self.model, self.optimizer, self.ema,
-                             self.lr_sched, epoch + 1, self.global_step, meta)
+                             self.lr_sched, epoch + (0 if mid_epoch else 1),
+                             self.global_step, meta)
+        if mid_epoch:
+            return
         if top1 > self.best_top1:
             self.best_top1 = top1
             ckpt.save_checkpoint(os.path.join(self.cfg.log.dir, "best.pt"),

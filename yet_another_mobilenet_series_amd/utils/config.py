@@ -138,7 +138,8 @@ DEFAULTS: Dict[str, Any] = {
     "train": {"epochs": 150, "seed": 42, "log_interval": 50,
               "grad_accum_steps": 1, "dtype": "bf16",
               "resume": "", "pretrained": "",
-              "save_interval_epochs": 1, "channels_last": True,
+              "save_interval_epochs": 1, "save_interval_steps": 0,
+              "channels_last": True,
               "deterministic": False, "perf_breakdown": False,
               "trace": False, "trace_steps": 3,
               "engine": "auto", "use_cuda_graph": True},
