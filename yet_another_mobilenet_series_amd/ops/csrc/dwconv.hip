@@ -42,7 +42,8 @@ inline Lop to_lop(const LopDesc& d) {
 // FLIP dgrad output dy feeds a BN backward, and the epilogue accumulates
 // (sum da, sum da*z) with da = dy * act'(a*z+b) so no separate
 // bn_bwd_partial pass reads dy+z again.  lopE carries z/ab/act for EPI==2.
-template <int K, int S, bool FLIP, int EPI, int LOPK, int ACT>
+template <int K, int S, bool FLIP, int EPI, int LOPK, int ACT,
+          bool WL = false>
 __global__ __launch_bounds__(256) void dw_fwd_kernel(
     const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
     ushort_t* __restrict__ y, int N, int H, int W, int C, int Ho, int Wo,
@@ -59,13 +60,37 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
   // (~12 KB) — measured 1.1-1.6 TB/s -> see profiles/r02 A/B
   constexpr bool HGFAST = (K == 5);
   const int C8 = C >> 2;
-  const int CT8 = C8 < 256 ? C8 : 256;
+  // WL (k5): the register-cached weight decode costs 256 VGPR + 148
+  // AGPR per thread = ONE wave/SIMD (objdump count, docs/ROADMAP §3b);
+  // staging the decoded f32 filter bank in LDS instead caps the
+  // register demand (LDS is never register-promoted) at the price of a
+  // ds_read_b128 per tap use.  CT8 is capped at 64 so the bank fits.
+  const int CT8 = WL ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
   const int ML = 256 / CT8;
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int c8 = (blockIdx.x % c_tiles) * CT8 + int(threadIdx.x) % CT8;
   const int chunk = blockIdx.x / c_tiles;
   const int mlane = int(threadIdx.x) / CT8;
   const int c = c8 * 4;
+  __shared__ float wlds[WL ? 64 * K * K * 4 : 4];
+  if (WL) {
+    const int cbase = (blockIdx.x % c_tiles) * CT8;
+    for (int idx = int(threadIdx.x); idx < CT8 * K * K; idx += 256) {
+      const int c8l = idx / (K * K);
+      const int tap = idx % (K * K);
+      const int cg = cbase + c8l;
+      if (cg < C8) {
+        // bake the FLIP into the bank so the inner loop indexes i*K+j
+        const int tsrc = FLIP ? K * K - 1 - tap : tap;
+        const short4_t wv =
+            *reinterpret_cast<const short4_t*>(&w[tsrc * C + cg * 4]);
+#pragma unroll
+        for (int e = 0; e < 4; ++e)
+          wlds[(c8l * K * K + tap) * 4 + e] = bf2f(ushort_t(wv[e]));
+      }
+    }
+    __syncthreads();
+  }
 
   const int Wo4 = (Wo + 3) / 4;
   const int HoG = (Ho + TH - 1) / TH;
@@ -119,8 +144,14 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
         cok[t] = wi >= 0 && wi < W;
         coff[t] = wic * C;
       }
+      int rr = RR;
+      // WL: opaque bound keeps the row loop ROLLED — full unrolling is
+      // what drives the load hoisting + weight caching that exhausts
+      // the register file; with 2+ waves/SIMD the waves hide the row
+      // load latency instead of one wave's software pipeline.
+      if (WL) asm volatile("" : "+r"(rr));
 #pragma unroll
-      for (int r = 0; r < RR; ++r) {
+      for (int r = 0; r < rr; ++r) {
         const int hi = ho0 * S + r - pad;
         const int hic = hi < 0 ? 0 : (hi >= H ? H - 1 : hi);
         const bool row_ok = hi >= 0 && hi < H;
@@ -164,15 +195,24 @@ __global__ __launch_bounds__(256) void dw_fwd_kernel(
           if (i < 0 || i >= K) continue;
 #pragma unroll
           for (int j = 0; j < K; ++j) {
-            const int tap =
-                FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
-            short4_t wv =
-                *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
             f32x2 wf[2];
+            if (WL) {
+              const float4 wf4 = *reinterpret_cast<const float4*>(
+                  &wlds[((int(threadIdx.x) % CT8) * K * K + i * K + j) * 4]);
+              wf[0][0] = wf4.x;
+              wf[0][1] = wf4.y;
+              wf[1][0] = wf4.z;
+              wf[1][1] = wf4.w;
+            } else {
+              const int tap =
+                  FLIP ? (K - 1 - i) * K + (K - 1 - j) : i * K + j;
+              short4_t wv =
+                  *reinterpret_cast<const short4_t*>(&w[tap * C + c]);
 #pragma unroll
-            for (int pp = 0; pp < 2; ++pp) {
-              wf[pp][0] = bf2f(ushort_t(wv[pp * 2]));
-              wf[pp][1] = bf2f(ushort_t(wv[pp * 2 + 1]));
+              for (int pp = 0; pp < 2; ++pp) {
+                wf[pp][0] = bf2f(ushort_t(wv[pp * 2]));
+                wf[pp][1] = bf2f(ushort_t(wv[pp * 2 + 1]));
+              }
             }
 #pragma unroll
             for (int q = 0; q < 4; ++q)
@@ -550,6 +590,16 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k) {
   return c < 1 ? 1 : c;
 }
 
+// YAMS_DW_K5LDS=0 disables the LDS filter-bank k5 variant (see the WL
+// template comment in dw_fwd_kernel — default on, A/B in profiles/r02)
+static bool dw_k5_wlds() {
+  static const bool on = [] {
+    const char* e = getenv("YAMS_DW_K5LDS");
+    return !(e && e[0] == '0');
+  }();
+  return on;
+}
+
 void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
                  int C, int Ho, int Wo, int k, int stride, int pad,
                  LopDesc lopx, float* stats_ws, hipStream_t s) {
@@ -557,25 +607,30 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   const Lop L = to_lop(lopx);
+  const bool wl = (k == 5) && dw_k5_wlds();
   const int C8 = C >> 2;
-  const int CT8 = C8 < 256 ? C8 : 256;
+  const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride, k);
   dim3 g(c_tiles * chunks);
   // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
   Lop LE{};
-#define DWF_CASE(KK, SS, ST, LK, AC)                                       \
-  if (k == KK && stride == SS && (stats_ws != nullptr) == (ST != 0) &&     \
+#define DWF_CASE(KK, SS, ST, LK, AC, WLW)                                  \
+  if (k == KK && stride == SS && wl == WLW &&                              \
+      (stats_ws != nullptr) == (ST != 0) &&                                \
       lopx.kind == LK && (LK == 0 || lopx.act == AC)) {                    \
-    dw_fwd_kernel<KK, SS, false, ST, LK, AC><<<g, 256, 0, s>>>(            \
+    dw_fwd_kernel<KK, SS, false, ST, LK, AC, WLW><<<g, 256, 0, s>>>(       \
         xp, wp, yp, N, H, W, C, Ho, Wo, pad, L, LE, stats_ws, chunks);     \
     return;                                                                \
   }
-#define DWF_ST(KK, SS, ST)                                                 \
-  DWF_CASE(KK, SS, ST, 0, 0) DWF_CASE(KK, SS, ST, 1, 1)                    \
-  DWF_CASE(KK, SS, ST, 1, 2) DWF_CASE(KK, SS, ST, 1, 3)
-#define DWF_KS(KK, SS) DWF_ST(KK, SS, false) DWF_ST(KK, SS, true)
-  DWF_KS(3, 1) DWF_KS(3, 2) DWF_KS(5, 1) DWF_KS(5, 2)
+#define DWF_ST(KK, SS, ST, WLW)                                            \
+  DWF_CASE(KK, SS, ST, 0, 0, WLW) DWF_CASE(KK, SS, ST, 1, 1, WLW)          \
+  DWF_CASE(KK, SS, ST, 1, 2, WLW) DWF_CASE(KK, SS, ST, 1, 3, WLW)
+#define DWF_KS(KK, SS, WLW)                                                \
+  DWF_ST(KK, SS, false, WLW) DWF_ST(KK, SS, true, WLW)
+  DWF_KS(3, 1, false) DWF_KS(3, 2, false)
+  DWF_KS(5, 1, false) DWF_KS(5, 2, false)
+  DWF_KS(5, 1, true) DWF_KS(5, 2, true)
 #undef DWF_KS
 #undef DWF_ST
 #undef DWF_CASE
@@ -593,29 +648,31 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
   const Lop LE = to_lop(epi);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
+    const bool wl = (k == 5) && dw_k5_wlds();
     const int C8 = C >> 2;
-    const int CT8 = C8 < 256 ? C8 : 256;
+    const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
     const int c_tiles = (C8 + CT8 - 1) / CT8;
     const int chunks = dw_stats_chunks(N, H, W, C, 1, k);
     dim3 g(c_tiles * chunks);
     // dgrad supports only lop none (dz is materialized by bn_act_bwd);
     // with stats_ws + epi it also emits the consumer BN's backward
     // partial sums (EPI==2) so bn_bwd_partial is skipped
-#define DWD_EPI(KK, AC)                                                    \
-    if (stats_ws != nullptr && k == KK && lopdy.kind == 0 &&               \
-        epi.act == AC) {                                                   \
-      dw_fwd_kernel<KK, 1, true, 2, 0, AC><<<g, 256, 0, s>>>(              \
+#define DWD_EPI(KK, AC, WLW)                                               \
+    if (stats_ws != nullptr && k == KK && wl == WLW &&                     \
+        lopdy.kind == 0 && epi.act == AC) {                                \
+      dw_fwd_kernel<KK, 1, true, 2, 0, AC, WLW><<<g, 256, 0, s>>>(         \
           gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, stats_ws, chunks);   \
       return;                                                              \
     }
-#define DWD_CASE(KK)                                                       \
-    DWD_EPI(KK, 0) DWD_EPI(KK, 1) DWD_EPI(KK, 2) DWD_EPI(KK, 3)            \
-    if (k == KK && lopdy.kind == 0) {                                      \
-      dw_fwd_kernel<KK, 1, true, 0, 0, 0><<<g, 256, 0, s>>>(               \
+#define DWD_CASE(KK, WLW)                                                  \
+    DWD_EPI(KK, 0, WLW) DWD_EPI(KK, 1, WLW)                                \
+    DWD_EPI(KK, 2, WLW) DWD_EPI(KK, 3, WLW)                                \
+    if (k == KK && wl == WLW && lopdy.kind == 0) {                         \
+      dw_fwd_kernel<KK, 1, true, 0, 0, 0, WLW><<<g, 256, 0, s>>>(          \
           gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, nullptr, chunks);    \
       return;                                                              \
     }
-    DWD_CASE(3) DWD_CASE(5)
+    DWD_CASE(3, false) DWD_CASE(5, false) DWD_CASE(5, true)
 #undef DWD_CASE
 #undef DWD_EPI
     yams_launch_fail("dw_conv_dgrad_s1", k, lopdy.kind, lopdy.act, epi.act);
