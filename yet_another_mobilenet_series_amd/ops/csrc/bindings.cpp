@@ -179,7 +179,8 @@ void dw_conv_fwd(const Tensor& x, const Tensor& w, Tensor& y, int64_t k,
 int64_t dw_stats_chunks(int64_t N, int64_t Ho, int64_t Wo, int64_t C,
                         int64_t stride, int64_t k) {
   return yams::dw_stats_chunks(int(N), int(Ho), int(Wo), int(C),
-                               int(stride), int(k));
+                               int(stride), int(k),
+                               k == 5 && yams::dw_k5_wlds());
 }
 
 void dw_conv_dgrad(const Tensor& dy, const Tensor& w, Tensor& dx, int64_t k,

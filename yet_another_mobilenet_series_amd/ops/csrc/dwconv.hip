@@ -561,7 +561,8 @@ int grid_for(int64_t total) {
 
 }  // namespace
 
-int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k) {
+int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k,
+                    bool wl) {
   (void)stride;
   // measurement knob (roadmap §3): YAMS_DW_CHUNKS_DIV divides the block
   // target so the work-per-wave vs launch-count tradeoff can be swept
@@ -571,16 +572,20 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k) {
     return e ? atoi(e) : 0;
   }();
   const int C8 = C >> 2;
-  const int CT8 = C8 < 256 ? C8 : 256;
+  const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
   const int c_tiles = (C8 + CT8 - 1) / CT8;
   const int64_t P = int64_t(N) * ((Ho + 1) / 2) * ((Wo + 3) / 4);
   // per-shape block target from the round-1 chunks sweep
   // (tools/probe_dw_sweep.py; per-wave prologue + launch ramp dominate
   // when waves run ~1 loop iteration): k5 prefers 4x fewer chunks than
   // the 512 base, k3 at small spatial 1x (512); env div overrides.
+  // The WL k5 variant runs at ~134 VGPR (3 blocks/CU) instead of the
+  // register-saturated 1 block/CU the 256 target was tuned for, so its
+  // grid target is 3x — without this the extra occupancy sits idle
+  // (measured: the c_tiles==1 shapes did not move under WL alone).
   int64_t target = 512;
   if (div_ <= 0) {
-    if (k == 5) target = 256;
+    if (k == 5) target = wl ? 768 : 256;
   } else {
     target = 2048 / div_;
   }
@@ -592,7 +597,7 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k) {
 
 // YAMS_DW_K5LDS=0 disables the LDS filter-bank k5 variant (see the WL
 // template comment in dw_fwd_kernel — default on, A/B in profiles/r02)
-static bool dw_k5_wlds() {
+bool dw_k5_wlds() {
   static const bool on = [] {
     const char* e = getenv("YAMS_DW_K5LDS");
     return !(e && e[0] == '0');
@@ -611,7 +616,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   const int C8 = C >> 2;
   const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
   const int c_tiles = (C8 + CT8 - 1) / CT8;
-  const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride, k);
+  const int chunks = dw_stats_chunks(N, Ho, Wo, C, stride, k, wl);
   dim3 g(c_tiles * chunks);
   // fwd lop kinds: none, or bnact with act in {relu, relu6, hswish}
   Lop LE{};
@@ -652,7 +657,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
     const int C8 = C >> 2;
     const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
     const int c_tiles = (C8 + CT8 - 1) / CT8;
-    const int chunks = dw_stats_chunks(N, H, W, C, 1, k);
+    const int chunks = dw_stats_chunks(N, H, W, C, 1, k, wl);
     dim3 g(c_tiles * chunks);
     // dgrad supports only lop none (dz is materialized by bn_act_bwd);
     // with stats_ws + epi it also emits the consumer BN's backward

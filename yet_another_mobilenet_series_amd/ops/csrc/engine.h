@@ -60,7 +60,9 @@ void dw_conv_fwd(const void* x, const void* w, void* y,
                  int N, int H, int W, int C, int Ho, int Wo,
                  int k, int stride, int pad, LopDesc lopx, float* stats_ws,
                  hipStream_t s);
-int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k);
+int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k,
+                    bool wl);
+bool dw_k5_wlds();  // YAMS_DW_K5LDS gate (LDS filter-bank k5 variant)
 void dw_conv_dgrad(const void* dy, const void* w, void* dx,
                    int N, int H, int W, int C, int Ho, int Wo,
                    int k, int stride, int pad, LopDesc lopdy, LopDesc epi,
