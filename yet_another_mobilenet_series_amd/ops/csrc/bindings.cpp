@@ -180,7 +180,8 @@ int64_t dw_stats_chunks(int64_t N, int64_t Ho, int64_t Wo, int64_t C,
                         int64_t stride, int64_t k) {
   return yams::dw_stats_chunks(int(N), int(Ho), int(Wo), int(C),
                                int(stride), int(k),
-                               k == 5 && yams::dw_k5_wlds());
+                               k == 5 ? yams::dw_k5_wlds()
+                                      : k == 3 && yams::dw_k3_wlds());
 }
 
 void dw_conv_dgrad(const Tensor& dy, const Tensor& w, Tensor& dx, int64_t k,

@@ -586,6 +586,7 @@ int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k,
   int64_t target = 512;
   if (div_ <= 0) {
     if (k == 5) target = wl ? 768 : 256;
+    if (k == 3 && wl) target = 1024;
   } else {
     target = 2048 / div_;
   }
@@ -605,6 +606,16 @@ bool dw_k5_wlds() {
   return on;
 }
 
+// YAMS_DW_K3LDS=1 extends the LDS filter bank to k3 (9.2 KB bank; k3
+// sits at ~200-256 VGPR = 2 waves/SIMD — default off pending the A/B)
+bool dw_k3_wlds() {
+  static const bool on = [] {
+    const char* e = getenv("YAMS_DW_K3LDS");
+    return e && e[0] == '1';
+  }();
+  return on;
+}
+
 void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
                  int C, int Ho, int Wo, int k, int stride, int pad,
                  LopDesc lopx, float* stats_ws, hipStream_t s) {
@@ -612,7 +623,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   auto* wp = static_cast<const ushort_t*>(w);
   auto* yp = static_cast<ushort_t*>(y);
   const Lop L = to_lop(lopx);
-  const bool wl = (k == 5) && dw_k5_wlds();
+  const bool wl = k == 5 ? dw_k5_wlds() : (k == 3 && dw_k3_wlds());
   const int C8 = C >> 2;
   const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
   const int c_tiles = (C8 + CT8 - 1) / CT8;
@@ -635,6 +646,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y, int N, int H, int W,
   DWF_ST(KK, SS, false, WLW) DWF_ST(KK, SS, true, WLW)
   DWF_KS(3, 1, false) DWF_KS(3, 2, false)
   DWF_KS(5, 1, false) DWF_KS(5, 2, false)
+  DWF_KS(3, 1, true) DWF_KS(3, 2, true)
   DWF_KS(5, 1, true) DWF_KS(5, 2, true)
 #undef DWF_KS
 #undef DWF_ST
@@ -653,7 +665,7 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
   const Lop LE = to_lop(epi);
   if (stride == 1) {
     // dgrad == correlation of dy with the flipped filter at same pad
-    const bool wl = (k == 5) && dw_k5_wlds();
+    const bool wl = k == 5 ? dw_k5_wlds() : (k == 3 && dw_k3_wlds());
     const int C8 = C >> 2;
     const int CT8 = wl ? (C8 < 64 ? C8 : 64) : (C8 < 256 ? C8 : 256);
     const int c_tiles = (C8 + CT8 - 1) / CT8;
@@ -677,7 +689,8 @@ void dw_conv_dgrad(const void* dy, const void* w, void* dx, int N, int H,
           gp, wp, xp, N, Ho, Wo, C, H, W, pad, L, LE, nullptr, chunks);    \
       return;                                                              \
     }
-    DWD_CASE(3, false) DWD_CASE(5, false) DWD_CASE(5, true)
+    DWD_CASE(3, false) DWD_CASE(3, true)
+    DWD_CASE(5, false) DWD_CASE(5, true)
 #undef DWD_CASE
 #undef DWD_EPI
     yams_launch_fail("dw_conv_dgrad_s1", k, lopdy.kind, lopdy.act, epi.act);

@@ -63,6 +63,7 @@ void dw_conv_fwd(const void* x, const void* w, void* y,
 int dw_stats_chunks(int N, int Ho, int Wo, int C, int stride, int k,
                     bool wl);
 bool dw_k5_wlds();  // YAMS_DW_K5LDS gate (LDS filter-bank k5 variant)
+bool dw_k3_wlds();  // YAMS_DW_K3LDS gate (default off)
 void dw_conv_dgrad(const void* dy, const void* w, void* dx,
                    int N, int H, int W, int C, int Ho, int Wo,
                    int k, int stride, int pad, LopDesc lopdy, LopDesc epi,
