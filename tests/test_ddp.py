@@ -310,3 +310,45 @@ def test_dp_world8_gloo(tmp_path):
                 p -= 0.1 * p.grad
     for a, b in zip(got, model.parameters()):
         assert torch.allclose(a, b.detach(), rtol=1e-4, atol=1e-5)
+
+
+def _broadcast_worker(rank, world, tmpfile, out_path):
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpfile}", rank=rank, world_size=world)
+    try:
+        from yet_another_mobilenet_series_amd.parallel.distributed import (
+            broadcast_module)
+        torch.manual_seed(123 + rank)  # divergent init AND buffers
+        model = nn.Sequential(nn.Linear(8, 16), nn.BatchNorm1d(16),
+                              nn.ReLU(), nn.Linear(16, 4))
+        model(torch.randn(4 + rank, 8))  # divergent BN running stats
+        for b in model.buffers():
+            b.add_(rank)
+        broadcast_module(model, src=0)
+        state = {k: v.clone() for k, v in model.state_dict().items()}
+        torch.save(state, f"{out_path}.{rank}")
+        dist.barrier()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_broadcast_module_syncs_params_and_buffers(tmp_path):
+    """broadcast_module makes every rank bit-identical to rank 0 —
+    params AND buffers (BN running stats, counters)."""
+    for attempt in range(3):
+        ctx = mp.get_context("spawn")
+        tmpfile = str(tmp_path / f"bcast_init_{attempt}")
+        out = str(tmp_path / "bcast_state")
+        ps = [ctx.Process(target=_broadcast_worker,
+                          args=(r, 2, tmpfile, out)) for r in range(2)]
+        [p.start() for p in ps]
+        [p.join(120) for p in ps]
+        if all(p.exitcode == 0 for p in ps):
+            break
+    assert all(p.exitcode == 0 for p in ps)
+    s0 = torch.load(out + ".0", weights_only=True)
+    s1 = torch.load(out + ".1", weights_only=True)
+    assert set(s0) == set(s1)
+    for k in s0:
+        assert torch.equal(s0[k], s1[k]), k
