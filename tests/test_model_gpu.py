@@ -238,3 +238,29 @@ def test_model_eval_parity(name, width):
     err = ((lo_hip.float() - lo_ref.float()).norm() /
            lo_ref.float().norm().clamp(min=1e-6)).item()
     assert err < 0.1, err
+
+
+def test_eval_forward_graph_capture_matches_eager():
+    """The serving path: eval-mode forward captured in a hipGraph replays
+    bit-identically to eager (static shapes, no stats writes) — guards
+    the tools/bench_infer.py --graph numbers."""
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = models.build_model("mobilenet_v3_large").to(dev) \
+        .to(memory_format=torch.channels_last).eval()
+    x = torch.randn(8, 3, 224, 224, device=dev).bfloat16() \
+        .contiguous(memory_format=torch.channels_last)
+    with torch.no_grad():
+        want = model(x).float().clone()
+        s0 = torch.cuda.Stream()
+        s0.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s0):
+            for _ in range(3):
+                model(x)
+        torch.cuda.current_stream().wait_stream(s0)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            got_buf = model(x)
+        g.replay()
+        torch.cuda.synchronize()
+    assert torch.equal(got_buf.float(), want)
