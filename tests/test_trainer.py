@@ -184,3 +184,18 @@ def test_train_cli_torchrun_two_ranks(tmp_path):
     metrics = [json.loads(l) for l in open(logdir / "metrics.jsonl")]
     vals = [m for m in metrics if m["kind"] == "val"]
     assert vals, "no val record from the master rank"
+
+
+def test_trainer_resume_auto(tmp_path):
+    """train.resume=auto: fresh start when no checkpoint exists, then a
+    relaunch with the same log.dir resumes from this run's checkpoint —
+    the watchdog/torchrun crash-loop restart story."""
+    cfg = _smoke_cfg(tmp_path, **{"train.resume": "auto"})
+    ta = Trainer(cfg)
+    assert ta.start_epoch == 0       # no checkpoint yet -> fresh
+    ta.fit()
+    cfg2 = _smoke_cfg(tmp_path, **{"train.resume": "auto",
+                                   "train.epochs": 2})
+    tb = Trainer(cfg2)
+    assert tb.start_epoch == 1       # picked up epoch-1 checkpoint
+    assert tb.global_step == ta.global_step

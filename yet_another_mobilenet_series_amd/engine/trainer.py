@@ -90,15 +90,21 @@ class Trainer:
         if cfg.train.pretrained:
             ckpt.load_pretrained(cfg.train.pretrained, model)
             self.logger.info("loaded pretrained %s", cfg.train.pretrained)
-        if cfg.train.resume and os.path.exists(cfg.train.resume):
+        resume = cfg.train.resume
+        if resume == "auto":
+            # crash-loop restarts: watchdog exits non-zero -> torchrun
+            # relaunches -> pick up this run's own latest checkpoint
+            # (fresh start when none exists yet)
+            resume = os.path.join(cfg.log.dir, "checkpoint.pt")
+        if resume and os.path.exists(resume):
             state = ckpt.load_checkpoint(
-                cfg.train.resume, model, self.optimizer, self.ema,
+                resume, model, self.optimizer, self.ema,
                 self.lr_sched, map_location=str(self.device))
             self.start_epoch = state.get("epoch", 0)
             self.global_step = state.get("global_step", 0)
             self.best_top1 = state.get("meta", {}).get("best_top1", 0.0)
             self.logger.info("resumed from %s at epoch %d",
-                             cfg.train.resume, self.start_epoch)
+                             resume, self.start_epoch)
 
     # ------------------------------------------------------------------
 
